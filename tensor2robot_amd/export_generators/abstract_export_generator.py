@@ -1,0 +1,179 @@
+"""Servable export: the SavedModel-production analog.
+
+Reference: `export_generators/abstract_export_generator.py:38-142` and
+`export_generators/default_export_generator.py:33-133`.
+
+An export is a timestamped directory:
+
+    <export_root>/<timestamp>/
+        servable.pt            torchscript trace (or pickled module fallback)
+        metadata.json          input/output key order, receiver mode, format
+        assets.extra/t2r_assets.pbtxt   feature/label specs + global_step
+
+The serving signature matches the reference's two receiver families:
+  * numpy receiver: feed {flat_key: ndarray} for the required in-specs
+    (preprocessing INCLUDED in the exported graph unless
+    export_raw_receivers).
+  * tf_example receiver: feed serialized tf.Example bytes; the predictor
+    runs the spec-derived parser before the servable (the parse stage lives
+    host-side, as in the reference's serving input receiver).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import time
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from tensor2robot_amd import gin
+from tensor2robot_amd.specs import tensorspec_utils as tsu
+from tensor2robot_amd.utils import modes as run_modes
+
+SERVABLE_NAME = "servable.pt"
+METADATA_NAME = "metadata.json"
+
+
+class ServableWrapper(torch.nn.Module):
+  """Flat-tensor-tuple adapter around a model's PREDICT path."""
+
+  def __init__(self, model, in_keys: List[str], out_keys: List[str],
+               apply_preprocessor: bool):
+    super().__init__()
+    self._model = model
+    self.network = model.network  # registers params for tracing
+    self.in_keys = in_keys
+    self.out_keys = out_keys
+    self.apply_preprocessor = apply_preprocessor
+
+  def forward(self, *tensors: torch.Tensor) -> Tuple[torch.Tensor, ...]:
+    features = tsu.TensorSpecStruct()
+    for key, t in zip(self.in_keys, tensors):
+      features[key] = t
+    if self.apply_preprocessor:
+      features, _ = self._model.preprocessor.preprocess(
+          features, None, run_modes.PREDICT)
+    ops = self._model.model_fn(features, None, run_modes.PREDICT)
+    return tuple(ops.predictions[k] for k in self.out_keys)
+
+
+class AbstractExportGenerator:
+  """Produces servable exports from a model (reference :38)."""
+
+  def __init__(self, export_raw_receivers: bool = False):
+    # export_raw_receivers: skip preprocessing in the exported graph so a
+    # client can preprocess on its side (reference :42-47).
+    self._export_raw_receivers = export_raw_receivers
+    self._in_feature_spec = None
+    self._in_label_spec = None
+    self._out_feature_spec = None
+    self._out_label_spec = None
+
+  def set_specification_from_model(self, model,
+                                   mode=run_modes.PREDICT):
+    preprocessor = model.preprocessor
+    self._in_feature_spec = preprocessor.get_in_feature_specification(mode)
+    self._in_label_spec = preprocessor.get_in_label_specification(mode)
+    self._out_feature_spec = preprocessor.get_out_feature_specification(mode)
+    self._out_label_spec = preprocessor.get_out_label_specification(mode)
+
+  def serving_input_spec(self) -> tsu.TensorSpecStruct:
+    spec = self._out_feature_spec if self._export_raw_receivers else \
+        self._in_feature_spec
+    return tsu.filter_required_flat_tensor_spec(spec)
+
+  def _build_wrapper(self, model) -> Tuple[ServableWrapper, Dict]:
+    input_spec = self.serving_input_spec()
+    in_keys = list(input_spec.keys())
+    example = tsu.make_random_tensors(input_spec, batch_size=1,
+                                      device=model.device, seed=0)
+    example_tensors = tuple(example[k] for k in in_keys)
+    probe = ServableWrapper(model, in_keys, [],
+                            apply_preprocessor=not self._export_raw_receivers)
+    # Discover output keys with one eager run.
+    features = tsu.TensorSpecStruct()
+    for key, t in zip(in_keys, example_tensors):
+      features[key] = t
+    if probe.apply_preprocessor:
+      features, _ = model.preprocessor.preprocess(features, None,
+                                                  run_modes.PREDICT)
+    with torch.no_grad():
+      ops = model.model_fn(features, None, run_modes.PREDICT)
+    out_keys = sorted(ops.predictions.keys())
+    wrapper = ServableWrapper(
+        model, in_keys, out_keys,
+        apply_preprocessor=not self._export_raw_receivers)
+    return wrapper, {"example_tensors": example_tensors}
+
+  def export(self, model, export_root: str, global_step: int = 0,
+             receiver_mode: str = "numpy") -> str:
+    """Writes a timestamped servable export; returns its directory."""
+    was_training = model.network.training
+    model.network.eval()
+    try:
+      wrapper, aux = self._build_wrapper(model)
+      fmt = "torchscript"
+      try:
+        with torch.no_grad():
+          scripted = torch.jit.trace(wrapper, aux["example_tensors"],
+                                     strict=False, check_trace=False)
+      except Exception:
+        scripted = None
+        fmt = "pickled_module"
+      version = int(time.time())
+      os.makedirs(export_root, exist_ok=True)
+      while os.path.exists(os.path.join(export_root, str(version))):
+        version += 1
+      tmp_dir = os.path.join(export_root, f"temp-{version}")
+      final_dir = os.path.join(export_root, str(version))
+      os.makedirs(tmp_dir, exist_ok=True)
+      servable_path = os.path.join(tmp_dir, SERVABLE_NAME)
+      if fmt == "torchscript":
+        scripted.save(servable_path)
+      else:
+        torch.save(wrapper, servable_path)
+      metadata = {
+          "format": fmt,
+          "receiver_mode": receiver_mode,
+          "in_keys": wrapper.in_keys,
+          "out_keys": wrapper.out_keys,
+          "apply_preprocessor": wrapper.apply_preprocessor,
+          "global_step": int(global_step),
+      }
+      with open(os.path.join(tmp_dir, METADATA_NAME), "w") as f:
+        json.dump(metadata, f, indent=2)
+      assets_dir = os.path.join(tmp_dir, tsu.EXTRA_ASSETS_DIRECTORY)
+      os.makedirs(assets_dir, exist_ok=True)
+      assets = tsu.T2RAssets(self.serving_input_spec(),
+                             self._in_label_spec, global_step=global_step)
+      tsu.write_t2r_assets_to_file(
+          assets, os.path.join(assets_dir, tsu.T2R_ASSETS_FILENAME))
+      os.replace(tmp_dir, final_dir)
+      return final_dir
+    finally:
+      if was_training:
+        model.network.train()
+
+  def create_warmup_requests_numpy(self, batch_sizes: List[int],
+                                   export_dir: str) -> str:
+    """TFRecord of zero-filled request Examples for serving warmup.
+
+    Reference :109-142 (PredictionLog warmup records; here plain Examples
+    whose features are the zero-filled required inputs).
+    """
+    from tensor2robot_amd.data import example as example_codec
+    from tensor2robot_amd.data import tfrecord
+    input_spec = self.serving_input_spec()
+    os.makedirs(export_dir, exist_ok=True)
+    path = os.path.join(export_dir, "warmup_requests.tfrecord")
+    with tfrecord.TFRecordWriter(path) as writer:
+      for bs in batch_sizes:
+        feed = tsu.make_constant_numpy(input_spec, 0.0, batch_size=bs)
+        features = {}
+        for key, arr in feed.items():
+          name = input_spec[key].name or key
+          features[name] = arr
+        writer.write(example_codec.encode_example(features))
+    return path

@@ -637,8 +637,15 @@ def _value_shape_dtype(value):
 def assert_equal_spec_or_tensor(expected_spec, actual, ignore_batch=False):
   e_shape, e_dtype = _value_shape_dtype(expected_spec)
   a_shape, a_dtype = _value_shape_dtype(actual)
-  e_shape = maybe_ignore_batch(e_shape, ignore_batch)
-  a_shape = maybe_ignore_batch(a_shape, ignore_batch)
+  if ignore_batch:
+    # Specs are usually batchless while tensors carry a leading batch dim:
+    # strip the batch from whichever side has it.
+    if len(a_shape) == len(e_shape) + 1:
+      a_shape = a_shape[1:]
+    elif len(e_shape) == len(a_shape) + 1:
+      e_shape = e_shape[1:]
+    elif len(e_shape) == len(a_shape) and e_shape:
+      e_shape, a_shape = e_shape[1:], a_shape[1:]
   if e_dtype != a_dtype:
     raise ValueError(
         f"dtype mismatch: expected {e_dtype}, got {a_dtype}")

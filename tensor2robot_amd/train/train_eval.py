@@ -1,0 +1,476 @@
+"""Native train/eval driver: the Estimator + train_and_evaluate replacement.
+
+Reference: `utils/train_eval.py:424-613` (train_eval_model),
+`:296-386` (create_default_exporters), `:390-420` (predict_from_model).
+
+The TF1 graph/session machinery collapses into an explicit step loop:
+  * lazy build: first batch -> network construction -> optimizer creation.
+  * bf16 autocast on the HIP device; f32 master weights.
+  * data parallelism: one process per GPU, bucketed RCCL all-reduce
+    overlapped with backward (tensor2robot_amd/parallel/ddp.py); chief
+    (rank 0) owns checkpoints, eval, export and summaries.
+  * hooks observe the loop through TrainContext (train/hooks).
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+import time
+from typing import Callable, Dict, List, Optional
+
+import torch
+
+from tensor2robot_amd import gin
+from tensor2robot_amd.data import pipeline
+from tensor2robot_amd.models import abstract_model as abstract_model_mod
+from tensor2robot_amd.specs import tensorspec_utils as tsu
+from tensor2robot_amd.train import checkpointing
+from tensor2robot_amd.train import hooks as hooks_mod
+from tensor2robot_amd.utils import modes as run_modes
+from tensor2robot_amd.utils import summaries as summaries_mod
+
+_log = logging.getLogger(__name__)
+
+TRAIN = run_modes.TRAIN
+EVAL = run_modes.EVAL
+PREDICT = run_modes.PREDICT
+
+
+def _dist_info():
+  if torch.distributed.is_available() and \
+      torch.distributed.is_initialized():
+    return torch.distributed.get_rank(), torch.distributed.get_world_size()
+  return 0, 1
+
+
+def resolve_device(model) -> torch.device:
+  if model.device_type == "gpu" and torch.cuda.is_available():
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    return torch.device(f"cuda:{local_rank}")
+  return torch.device("cpu")
+
+
+class TrainContext:
+  """What hooks and export callbacks see of the running loop."""
+
+  def __init__(self, trainer):
+    self._trainer = trainer
+
+  @property
+  def model(self):
+    return self._trainer.model
+
+  @property
+  def network(self):
+    return self._trainer.network
+
+  @property
+  def optimizer(self):
+    return self._trainer.optimizer
+
+  @property
+  def global_step(self):
+    return self._trainer.global_step
+
+  @property
+  def model_dir(self):
+    return self._trainer.model_dir
+
+  def save_checkpoint(self) -> str:
+    return self._trainer.save_checkpoint()
+
+
+class Trainer:
+  """Owns the step loop for one model + device (+ DP group)."""
+
+  def __init__(self, model, model_dir: str,
+               hooks: Optional[List[hooks_mod.TrainHook]] = None,
+               max_to_keep: int = 5,
+               keep_checkpoint_every_n_hours: Optional[float] = None,
+               log_every_n_steps: int = 100):
+    self.model = model
+    self.model_dir = model_dir
+    self.hooks = hooks or []
+    self.global_step = 0
+    self.optimizer = None
+    self.ema = None
+    self._dp_engine = None
+    self._log_every = log_every_n_steps
+    self.rank, self.world_size = _dist_info()
+    self.is_chief = self.rank == 0
+    self.device = resolve_device(model)
+    model.to_device(self.device)
+    self._autocast_enabled = (
+        self.device.type == "cuda" and
+        model.compute_dtype in (torch.bfloat16, torch.float16))
+    self.checkpointer = checkpointing.Checkpointer(
+        model_dir, max_to_keep=max_to_keep,
+        keep_checkpoint_every_n_hours=keep_checkpoint_every_n_hours) \
+        if self.is_chief and model_dir else None
+    self.summary_writer = summaries_mod.SummaryWriter(model_dir) \
+        if self.is_chief and model_dir else None
+    self.context = TrainContext(self)
+
+  # -- lifecycle -----------------------------------------------------------
+  @property
+  def network(self):
+    return self.model.network
+
+  def _ensure_built(self):
+    if self.optimizer is not None:
+      return
+    _ = self.model.network  # lazy construction + device placement
+    if self.world_size > 1:
+      from tensor2robot_amd.parallel import ddp
+      self._dp_engine = ddp.DataParallelEngine(self.model.network)
+    self.optimizer = self.model.create_optimizer()
+    self.ema = self.model.create_ema()
+    if self.model_dir:
+      latest = checkpointing.latest_checkpoint(self.model_dir)
+      if latest is not None:
+        self.global_step = checkpointing.Checkpointer.restore(
+            latest, self.network, self.optimizer, self.ema)
+        _log.info("Resumed from %s at step %d", latest, self.global_step)
+
+  def save_checkpoint(self) -> str:
+    if self.checkpointer is None:
+      return ""
+    path = self.checkpointer.save(self.global_step, self.network,
+                                  self.optimizer, self.ema)
+    for hook in self.hooks:
+      hook.after_checkpoint(self.context, path)
+    return path
+
+  def _autocast(self):
+    return torch.autocast(device_type="cuda",
+                          dtype=self.model.compute_dtype,
+                          enabled=self._autocast_enabled)
+
+  # -- train ---------------------------------------------------------------
+  def train(self, input_fn, max_steps: int,
+            save_checkpoint_steps: Optional[int] = None) -> Dict[str, float]:
+    self._ensure_built()
+    self.network.train()
+    for hook in self.hooks:
+      if self.is_chief or hook.every_rank:
+        hook.begin(self.context)
+    iterator = iter(input_fn())
+    last_log = time.time()
+    last_loss = float("nan")
+    while self.global_step < max_steps:
+      features, labels = next(iterator)
+      features = pipeline.move_struct_to_device(features, self.device)
+      labels = pipeline.move_struct_to_device(labels, self.device)
+      for hook in self.hooks:
+        if self.is_chief or hook.every_rank:
+          hook.before_step(self.context)
+      self.optimizer.zero_grad(set_to_none=True)
+      with self._autocast():
+        ops = self.model.model_fn(features, labels, TRAIN)
+      loss = ops.loss
+      if self._dp_engine is not None:
+        self._dp_engine.backward(loss)
+      else:
+        loss.backward()
+      self.optimizer.step(self.global_step)
+      if self.ema is not None:
+        self.ema.update()
+      self.global_step += 1
+      last_loss = float(loss.detach().float().cpu())
+      if self.is_chief and self.summary_writer is not None and \
+          ops.scalar_summaries:
+        self.summary_writer.add_scalars(ops.scalar_summaries,
+                                        self.global_step)
+      for hook in self.hooks:
+        if self.is_chief or hook.every_rank:
+          hook.after_step(self.context, ops)
+      if save_checkpoint_steps and \
+          self.global_step % save_checkpoint_steps == 0:
+        self.save_checkpoint()
+      if self.is_chief and self.global_step % self._log_every == 0:
+        now = time.time()
+        _log.info("step=%d loss=%.5f steps/s=%.2f", self.global_step,
+                  last_loss, self._log_every / max(now - last_log, 1e-9))
+        last_log = now
+    if self.checkpointer is not None:
+      self.save_checkpoint()
+    for hook in self.hooks:
+      if self.is_chief or hook.every_rank:
+        hook.end(self.context)
+    if self.summary_writer is not None:
+      self.summary_writer.flush()
+    return {"loss": last_loss, "global_step": self.global_step}
+
+  # -- eval ----------------------------------------------------------------
+  def evaluate(self, input_fn, eval_steps: int,
+               eval_name: str = "") -> Dict[str, float]:
+    self._ensure_built()
+    self.network.eval()
+    if self.ema is not None:
+      self.ema.swap_in()
+    totals: Dict[str, float] = {}
+    count = 0
+    try:
+      iterator = iter(input_fn())
+      with torch.no_grad():
+        for _ in range(eval_steps):
+          try:
+            features, labels = next(iterator)
+          except StopIteration:
+            break
+          features = pipeline.move_struct_to_device(features, self.device)
+          labels = pipeline.move_struct_to_device(labels, self.device)
+          with self._autocast():
+            ops = self.model.model_fn(features, labels, EVAL)
+          batch_metrics = dict(ops.metrics or {})
+          if ops.loss is not None:
+            batch_metrics.setdefault("loss", ops.loss)
+          for key, value in batch_metrics.items():
+            if isinstance(value, torch.Tensor):
+              value = float(value.detach().float().cpu())
+            totals[key] = totals.get(key, 0.0) + value
+          count += 1
+    finally:
+      if self.ema is not None:
+        self.ema.swap_out()
+      self.network.train()
+    metrics = {k: v / max(count, 1) for k, v in totals.items()}
+    metrics["global_step"] = self.global_step
+    if self.summary_writer is not None:
+      prefix = f"eval_{eval_name}/" if eval_name else "eval/"
+      self.summary_writer.add_scalars(
+          {prefix + k: v for k, v in metrics.items()}, self.global_step)
+      self.summary_writer.flush()
+    return metrics
+
+  def predict(self, input_fn, yield_single_examples: bool = True):
+    self._ensure_built()
+    self.network.eval()
+    with torch.no_grad():
+      for features, _ in input_fn():
+        features = pipeline.move_struct_to_device(features, self.device)
+        with self._autocast():
+          ops = self.model.model_fn(features, None, PREDICT)
+        preds = {k: v.detach().float().cpu().numpy()
+                 for k, v in ops.predictions.items()}
+        if yield_single_examples:
+          batch = next(iter(preds.values())).shape[0]
+          for i in range(batch):
+            yield {k: v[i] for k, v in preds.items()}
+        else:
+          yield preds
+
+
+# ---------------------------------------------------------------------------
+# Exporters (Best/Latest; reference train_eval.py:207-386)
+# ---------------------------------------------------------------------------
+
+
+@gin.configurable
+def create_valid_result_smaller(result_key: str = "loss"):
+  """BestExporter compare fn: smaller metric is better (reference :207)."""
+
+  def compare(best: Dict[str, float], current: Dict[str, float]) -> bool:
+    if result_key not in current:
+      return False
+    if best is None or result_key not in best:
+      return True
+    return current[result_key] < best[result_key]
+
+  return compare
+
+
+@gin.configurable
+def create_valid_result_larger(result_key: str = "loss"):
+
+  def compare(best, current) -> bool:
+    if result_key not in current:
+      return False
+    if best is None or result_key not in best:
+      return True
+    return current[result_key] > best[result_key]
+
+  return compare
+
+
+class Exporter:
+
+  def __init__(self, name: str, export_generator=None):
+    from tensor2robot_amd.export_generators import default_export_generator
+    self.name = name
+    self.export_generator = export_generator or \
+        default_export_generator.DefaultExportGenerator()
+
+  def export(self, trainer: Trainer, eval_result: Dict[str, float],
+             export_root: str) -> Optional[str]:
+    raise NotImplementedError
+
+
+class LatestExporter(Exporter):
+
+  def __init__(self, name: str = "latest_exporter_numpy",
+               export_generator=None, exports_to_keep: int = 5):
+    super().__init__(name, export_generator)
+    self._gc = None
+    self._keep = exports_to_keep
+
+  def export(self, trainer, eval_result, export_root):
+    out_dir = os.path.join(export_root, self.name)
+    self.export_generator.set_specification_from_model(trainer.model)
+    if trainer.ema is not None:
+      trainer.ema.swap_in()
+    try:
+      path = self.export_generator.export(trainer.model, out_dir,
+                                          global_step=trainer.global_step)
+    finally:
+      if trainer.ema is not None:
+        trainer.ema.swap_out()
+    hooks_mod._DirectoryVersionGC(out_dir, self._keep).collect()
+    return path
+
+
+class BestExporter(LatestExporter):
+
+  def __init__(self, name: str = "best_exporter_numpy",
+               compare_fn=None, export_generator=None,
+               exports_to_keep: int = 1):
+    super().__init__(name, export_generator, exports_to_keep)
+    self._compare_fn = compare_fn or create_valid_result_smaller()
+    self._best: Optional[Dict[str, float]] = None
+
+  def export(self, trainer, eval_result, export_root):
+    if not self._compare_fn(self._best, eval_result):
+      return None
+    self._best = dict(eval_result)
+    return super().export(trainer, eval_result, export_root)
+
+
+@gin.configurable
+def create_default_exporters(export_generator=None,
+                             compare_fn=create_valid_result_smaller,
+                             exports_to_keep: int = 5):
+  """Best + Latest exporters (reference :296-386)."""
+  return [
+      BestExporter(name="best_exporter_numpy", compare_fn=compare_fn(),
+                   export_generator=export_generator),
+      LatestExporter(name="latest_exporter_numpy",
+                     export_generator=export_generator,
+                     exports_to_keep=exports_to_keep),
+  ]
+
+
+# ---------------------------------------------------------------------------
+# train_eval_model: THE entry point (reference :424-613)
+# ---------------------------------------------------------------------------
+
+
+@gin.configurable
+def train_eval_model(t2r_model=None,
+                     input_generator_train=None,
+                     input_generator_eval=None,
+                     max_train_steps: int = 1000,
+                     eval_steps: int = 100,
+                     model_dir: str = "/tmp/t2r_model_dir",
+                     eval_throttle_secs: float = 0.0,
+                     eval_every_n_steps: Optional[int] = None,
+                     save_checkpoint_steps: Optional[int] = None,
+                     create_exporters_fn=None,
+                     train_hook_builders: Optional[List] = None,
+                     chief_train_hook_builders: Optional[List] = None,
+                     eval_hook_builders: Optional[List] = None,
+                     multi_eval_name: Optional[str] = None,
+                     use_continuous_eval: bool = True,
+                     log_every_n_steps: int = 100) -> Dict[str, float]:
+  """Trains and/or evaluates; dispatch mirrors the reference.
+
+  train+eval | train-only (no eval generator) | eval-only (no train
+  generator; continuous polling loop over new checkpoints when
+  use_continuous_eval).
+  """
+  if t2r_model is None:
+    raise ValueError("t2r_model is required")
+  rank, world_size = _dist_info()
+
+  # Hook builders -> hooks (train + chief-only; reference :516-528).
+  trainer = Trainer(t2r_model, model_dir,
+                    log_every_n_steps=log_every_n_steps)
+  hook_list: List[hooks_mod.TrainHook] = [
+      hooks_mod.GinConfigLoggerHook()]
+  for builder in (train_hook_builders or []):
+    hook_list.extend(builder.create_hooks(t2r_model, trainer))
+  if trainer.is_chief:
+    for builder in (chief_train_hook_builders or []):
+      hook_list.extend(builder.create_hooks(t2r_model, trainer))
+  trainer.hooks = hook_list
+
+  exporters = create_exporters_fn() if create_exporters_fn else []
+
+  def run_eval(eval_trainer) -> Dict[str, float]:
+    if input_generator_eval is None:
+      return {}
+    input_generator_eval.set_specification_from_model(t2r_model, EVAL)
+    eval_input_fn = input_generator_eval.create_dataset_input_fn(EVAL)
+    result = eval_trainer.evaluate(eval_input_fn, eval_steps,
+                                   eval_name=multi_eval_name or "")
+    export_root = os.path.join(model_dir, "export")
+    for exporter in exporters:
+      exporter.export(eval_trainer, result, export_root)
+    return result
+
+  # --- eval-only mode (reference :585-611) ---
+  if input_generator_train is None:
+    if input_generator_eval is None:
+      raise ValueError("Need at least one input generator")
+    if not use_continuous_eval:
+      return run_eval(trainer)
+    last_seen = None
+    result: Dict[str, float] = {}
+    while True:
+      ckpt = checkpointing.wait_for_checkpoint(model_dir, last_seen,
+                                               timeout=30.0)
+      if ckpt is None:
+        break
+      last_seen = ckpt
+      trainer._ensure_built()
+      trainer.global_step = checkpointing.Checkpointer.restore(
+          ckpt, trainer.network, trainer.optimizer, trainer.ema)
+      result = run_eval(trainer)
+      if trainer.global_step >= max_train_steps:
+        break
+    return result
+
+  # --- train (+eval) mode ---
+  input_generator_train.set_specification_from_model(t2r_model, TRAIN)
+  train_input_fn = input_generator_train.create_dataset_input_fn(
+      TRAIN, pin_memory=trainer.device.type == "cuda")
+
+  if input_generator_eval is None or not trainer.is_chief:
+    result = trainer.train(train_input_fn, max_train_steps,
+                           save_checkpoint_steps)
+    return result
+
+  # Interleaved train/eval on the chief (train_and_evaluate semantics).
+  eval_interval = eval_every_n_steps or max_train_steps
+  result: Dict[str, float] = {}
+  while trainer.global_step < max_train_steps:
+    target = min(trainer.global_step + eval_interval, max_train_steps)
+    result = trainer.train(train_input_fn, target, save_checkpoint_steps)
+    eval_result = run_eval(trainer)
+    result.update({f"eval_{k}": v for k, v in eval_result.items()})
+    if eval_throttle_secs:
+      time.sleep(eval_throttle_secs)
+  return result
+
+
+@gin.configurable
+def predict_from_model(t2r_model=None, input_generator=None,
+                       model_dir: str = "", yield_single_examples: bool =
+                       True):
+  """Offline batch predict (reference :390-420)."""
+  if t2r_model is None or input_generator is None:
+    raise ValueError("model and input generator required")
+  trainer = Trainer(t2r_model, model_dir)
+  input_generator.set_specification_from_model(t2r_model, PREDICT)
+  input_fn = input_generator.create_dataset_input_fn(PREDICT)
+  return trainer.predict(input_fn, yield_single_examples)
