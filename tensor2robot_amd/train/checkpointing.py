@@ -87,7 +87,7 @@ class Checkpointer:
     try:
       payload = {
           "global_step": step,
-          "model_state": {k: v.detach().cpu()
+          "model_state": {k: v.detach().cpu().clone()
                           for k, v in network.state_dict().items()},
       }
     finally:
@@ -98,12 +98,14 @@ class Checkpointer:
     if ema is not None:
       payload["ema_state"] = {
           "decay": ema.decay,
-          "shadow": {k: v.detach().cpu() for k, v in ema.shadow.items()},
+          "shadow": {k: v.detach().cpu().clone()
+                     for k, v in ema.shadow.items()},
       }
       # The live (non-averaged) weights ride along so RESUME is exact even
       # though the canonical model_state holds averaged params.
       payload["raw_model_state"] = {
-          k: v.detach().cpu() for k, v in network.state_dict().items()}
+          k: v.detach().cpu().clone()
+          for k, v in network.state_dict().items()}
     if extra:
       payload["extra"] = extra
     path = checkpoint_path(self._model_dir, step)
