@@ -1,0 +1,191 @@
+"""Flagship benchmark: QT-Opt grasping critic training throughput.
+
+Measures images/sec (whole job) for the BASELINE.json headline metric:
+QT-Opt critic train, 472x472 input, bs=32/GPU, bf16, on 1..8 MI355X.
+
+Each timed step is the full training step: on-GPU preprocessing of the raw
+synthetic 512x640 uint8 batch (convert + crop + photometric distortion),
+Grasping44 forward, sigmoid log-loss, backward, bucketed RCCL gradient
+all-reduce (N>1), momentum optimizer step and EMA update — nothing skipped.
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from tensor2robot_amd.models import optimizers as optimizers_mod
+from tensor2robot_amd.research.qtopt import t2r_models
+from tensor2robot_amd.specs import tensorspec_utils as tsu
+from tensor2robot_amd.utils import modes as run_modes
+
+
+def parse_args():
+  p = argparse.ArgumentParser()
+  p.add_argument("--gpus", type=int, default=1)
+  p.add_argument("--steps", type=int, default=30)
+  p.add_argument("--warmup", type=int, default=10)
+  p.add_argument("--batch-size", type=int, default=32)
+  p.add_argument("--no-ema", action="store_true")
+  p.add_argument("--no-preprocess", action="store_true",
+                 help="feed pre-cropped 472x472 f32 (ablation only)")
+  return p.parse_args()
+
+
+def make_synthetic_pool(batch_size, device, n_batches=4, seed=0):
+  """Fixed pool of raw uint8 batches, resident on the GPU."""
+  g = torch.Generator(device="cpu").manual_seed(seed)
+  pool = []
+  for _ in range(n_batches):
+    images = torch.randint(0, 256,
+                           (batch_size, t2r_models.RAW_HEIGHT,
+                            t2r_models.RAW_WIDTH, 3),
+                           generator=g, dtype=torch.uint8)
+    action = torch.rand((batch_size, t2r_models.ACTION_DIM), generator=g)
+    labels = (torch.rand((batch_size, 1), generator=g) > 0.5).float()
+    pool.append((images.to(device), action.to(device), labels.to(device)))
+  return pool
+
+
+def build_features(model, images, action, mode, preprocess=True):
+  features = tsu.TensorSpecStruct()
+  if preprocess:
+    features["state/image"] = images
+    offset = 0
+    for name, size in t2r_models.ACTION_COMPONENTS:
+      features["action/" + name] = action[:, offset: offset + size]
+      offset += size
+    features, _ = model.preprocessor.preprocess(features, None, mode)
+  else:
+    features["state/image"] = images
+    offset = 0
+    for name, size in t2r_models.ACTION_COMPONENTS:
+      features["action/" + name] = action[:, offset: offset + size]
+      offset += size
+  return features
+
+
+def main():
+  args = parse_args()
+  world_size = int(os.environ.get("WORLD_SIZE", "1"))
+  rank = int(os.environ.get("RANK", "0"))
+  local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+  distributed = world_size > 1
+
+  use_cuda = torch.cuda.is_available()
+  if distributed:
+    backend = "nccl" if use_cuda else "gloo"
+    torch.distributed.init_process_group(backend=backend)
+  device = torch.device(f"cuda:{local_rank}") if use_cuda else \
+      torch.device("cpu")
+  if use_cuda:
+    torch.cuda.set_device(device)
+
+  model = t2r_models.GraspingModel(
+      device_type="gpu" if use_cuda else "cpu",
+      compute_dtype="bfloat16" if use_cuda else "float32",
+      use_avg_model_params=not args.no_ema)
+  model.to_device(device)
+  network = model.network
+  network.to(memory_format=torch.channels_last)
+
+  dp_engine = None
+  if distributed:
+    from tensor2robot_amd.parallel import ddp
+    dp_engine = ddp.DataParallelEngine(network)
+
+  optimizer = model.create_optimizer()
+  ema = model.create_ema()
+
+  pool = make_synthetic_pool(args.batch_size, device, seed=1234 + rank)
+  autocast = torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                            enabled=use_cuda)
+
+  def step(i, global_step):
+    images, action, labels_t = pool[i % len(pool)]
+    optimizer.zero_grad(set_to_none=True)
+    with autocast:
+      features = build_features(model, images, action, run_modes.TRAIN,
+                                preprocess=not args.no_preprocess)
+      image = features["state/image"].permute(0, 3, 1, 2).contiguous(
+          memory_format=torch.channels_last)
+      actions = model.pack_action_vector(features)
+      q = network(image, actions)
+    loss = torch.nn.functional.binary_cross_entropy(
+        torch.clamp(q.float(), 1e-7, 1 - 1e-7),
+        labels_t.reshape(q.shape))
+    if dp_engine is not None:
+      dp_engine.backward(loss)
+    else:
+      loss.backward()
+    optimizer.step(global_step)
+    if ema is not None:
+      ema.update()
+    return loss
+
+  def barrier_sync():
+    if distributed:
+      torch.distributed.barrier()
+    if use_cuda:
+      torch.cuda.synchronize()
+
+  for i in range(args.warmup):
+    step(i, i)
+  barrier_sync()
+  t0 = time.perf_counter()
+  for i in range(args.steps):
+    step(i, args.warmup + i)
+  barrier_sync()
+  elapsed = time.perf_counter() - t0
+
+  # Max over ranks.
+  if distributed:
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if use_cuda else "cpu")
+    torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+  n_gpus = world_size
+  total_images = args.batch_size * args.steps * n_gpus
+  images_per_sec = total_images / elapsed
+  ms_per_step = elapsed / args.steps * 1000.0
+
+  if rank == 0:
+    result = {
+        "metric": "images/sec (whole node) QT-Opt critic train, "
+                  "472x472 bs=32/GPU",
+        "value": round(images_per_sec, 2),
+        "unit": "images/sec",
+        "n_gpus": n_gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(ms_per_step, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16" if use_cuda else "float32",
+        "data": "synthetic",
+        "config": {
+            "model": "qtopt_grasping44_critic",
+            "global_batch": args.batch_size * n_gpus,
+            "input": "512x640 uint8 -> 472x472 crop (on-GPU preprocess)"
+                     if not args.no_preprocess else "472x472 f32",
+            "ema": not args.no_ema,
+            "parallelism": f"dp{n_gpus}",
+        },
+    }
+    print(json.dumps(result))
+
+  if distributed:
+    torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+  main()

@@ -1,0 +1,250 @@
+"""Image crop + photometric distortion library (torch ops).
+
+Reference: `preprocessors/image_transformations.py` (RandomCropImages :25,
+CenterCropImages :62, CustomCropImages :104, ApplyPhotometricImageDistortions
+:176/:268/:365, ApplyRandomFlips :387, ApplyDepthImageDistortions :403).
+
+All functions operate on float images in [0, 1] with shape [N, H, W, C]
+(NHWC, matching the serialized layout; the model layer converts to the
+device-preferred memory format).  On GPU these dispatch to the fused HIP
+kernels in tensor2robot_amd/ops when available.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional, Sequence, Tuple
+
+import torch
+
+from tensor2robot_amd import gin
+
+
+def _check_nhwc(image: torch.Tensor):
+  if image.dim() != 4:
+    raise ValueError(f"Expected [N,H,W,C] image batch, got {image.shape}")
+
+
+def random_crop_offsets(input_shape, target_shape,
+                        generator: Optional[torch.Generator] = None
+                        ) -> Tuple[int, int]:
+  ih, iw = input_shape[:2]
+  th, tw = target_shape[:2]
+  if th > ih or tw > iw:
+    raise ValueError(f"Crop {target_shape} larger than input {input_shape}")
+  oy = int(torch.randint(0, ih - th + 1, (1,), generator=generator))
+  ox = int(torch.randint(0, iw - tw + 1, (1,), generator=generator))
+  return oy, ox
+
+
+@gin.configurable
+def RandomCropImages(images: Sequence[torch.Tensor], input_shape,
+                     target_shape,
+                     generator: Optional[torch.Generator] = None
+                     ) -> List[torch.Tensor]:
+  """Crops every image in the list with ONE shared random offset (:25-60)."""
+  oy, ox = random_crop_offsets(input_shape, target_shape, generator)
+  th, tw = target_shape[:2]
+  out = []
+  for img in images:
+    _check_nhwc(img)
+    out.append(img[:, oy: oy + th, ox: ox + tw, :])
+  return out
+
+
+@gin.configurable
+def CenterCropImages(images: Sequence[torch.Tensor], input_shape,
+                     target_shape) -> List[torch.Tensor]:
+  """Center crop of every image in the list (:62-102)."""
+  ih, iw = input_shape[:2]
+  th, tw = target_shape[:2]
+  oy, ox = (ih - th) // 2, (iw - tw) // 2
+  out = []
+  for img in images:
+    _check_nhwc(img)
+    out.append(img[:, oy: oy + th, ox: ox + tw, :])
+  return out
+
+
+@gin.configurable
+def CustomCropImages(images: Sequence[torch.Tensor],
+                     crop_centers: torch.Tensor,
+                     target_shape) -> List[torch.Tensor]:
+  """Per-sample crops at given centers, clamped to bounds (:104-174)."""
+  th, tw = target_shape[:2]
+  out = []
+  for img in images:
+    _check_nhwc(img)
+    n, ih, iw, c = img.shape
+    crops = []
+    for i in range(n):
+      cy = int(crop_centers[i, 0])
+      cx = int(crop_centers[i, 1])
+      oy = min(max(cy - th // 2, 0), ih - th)
+      ox = min(max(cx - tw // 2, 0), iw - tw)
+      crops.append(img[i, oy: oy + th, ox: ox + tw, :])
+    out.append(torch.stack(crops))
+  return out
+
+
+def _rand(n, lo, hi, device, generator=None):
+  return torch.rand(n, device=device, generator=generator) * (hi - lo) + lo
+
+
+def _adjust_hue(image: torch.Tensor, delta: torch.Tensor) -> torch.Tensor:
+  """Hue rotation per batch element; delta in turns [-0.5, 0.5]. NHWC RGB."""
+  r, g, b = image.unbind(-1)
+  maxc = torch.maximum(torch.maximum(r, g), b)
+  minc = torch.minimum(torch.minimum(r, g), b)
+  v = maxc
+  diff = maxc - minc
+  s = torch.where(maxc > 0, diff / torch.clamp(maxc, min=1e-8),
+                  torch.zeros_like(maxc))
+  diff_safe = torch.clamp(diff, min=1e-8)
+  rc = (maxc - r) / diff_safe
+  gc = (maxc - g) / diff_safe
+  bc = (maxc - b) / diff_safe
+  h = torch.where(r == maxc, bc - gc,
+                  torch.where(g == maxc, 2.0 + rc - bc, 4.0 + gc - rc))
+  h = (h / 6.0) % 1.0
+  h = torch.where(diff > 0, h, torch.zeros_like(h))
+  h = (h + delta.view(-1, 1, 1)) % 1.0
+  i = torch.floor(h * 6.0)
+  f = h * 6.0 - i
+  p = v * (1.0 - s)
+  q = v * (1.0 - s * f)
+  t = v * (1.0 - s * (1.0 - f))
+  i = i.long() % 6
+  r2 = torch.where(i == 0, v, torch.where(i == 1, q, torch.where(
+      i == 2, p, torch.where(i == 3, p, torch.where(i == 4, t, v)))))
+  g2 = torch.where(i == 0, t, torch.where(i == 1, v, torch.where(
+      i == 2, v, torch.where(i == 3, q, torch.where(i == 4, p, p)))))
+  b2 = torch.where(i == 0, p, torch.where(i == 1, p, torch.where(
+      i == 2, t, torch.where(i == 3, v, torch.where(i == 4, v, q)))))
+  return torch.stack([r2, g2, b2], dim=-1)
+
+
+@gin.configurable
+def ApplyPhotometricImageDistortions(
+    images: Sequence[torch.Tensor],
+    random_brightness: bool = False,
+    max_delta_brightness: float = 0.125,
+    random_saturation: bool = False,
+    lower_saturation: float = 0.5,
+    upper_saturation: float = 1.5,
+    random_hue: bool = False,
+    max_delta_hue: float = 0.2,
+    random_contrast: bool = False,
+    lower_contrast: float = 0.5,
+    upper_contrast: float = 1.5,
+    random_noise_levels: float = 0.0,
+    random_noise_apply_probability: float = 0.5,
+    generator: Optional[torch.Generator] = None) -> List[torch.Tensor]:
+  """Per-image-independent photometric distortions, clipped to [0,1].
+
+  Reference :176-363 (the Parallel variant is the default here: each batch
+  element draws independent parameters — one fused pass).
+  """
+  out = []
+  for img in images:
+    _check_nhwc(img)
+    n = img.shape[0]
+    device = img.device
+    x = img
+    if random_brightness:
+      delta = _rand(n, -max_delta_brightness, max_delta_brightness, device,
+                    generator).view(-1, 1, 1, 1)
+      x = x + delta
+    if random_saturation:
+      factor = _rand(n, lower_saturation, upper_saturation, device,
+                     generator).view(-1, 1, 1, 1)
+      gray = x.mean(dim=-1, keepdim=True)
+      x = gray + (x - gray) * factor
+    if random_hue:
+      delta = _rand(n, -max_delta_hue, max_delta_hue, device, generator)
+      x = _adjust_hue(torch.clamp(x, 0.0, 1.0), delta)
+    if random_contrast:
+      factor = _rand(n, lower_contrast, upper_contrast, device,
+                     generator).view(-1, 1, 1, 1)
+      mean = x.mean(dim=(1, 2), keepdim=True)
+      x = (x - mean) * factor + mean
+    if random_noise_levels:
+      sigma = _rand(n, 0.0, random_noise_levels, device,
+                    generator).view(-1, 1, 1, 1)
+      apply = (_rand(n, 0.0, 1.0, device, generator) <
+               random_noise_apply_probability).float().view(-1, 1, 1, 1)
+      noise = torch.randn(x.shape, device=device, generator=generator) \
+          * sigma * apply
+      x = x + noise
+    out.append(torch.clamp(x, 0.0, 1.0))
+  return out
+
+
+@gin.configurable
+def ApplyPhotometricImageDistortionsCheap(
+    images: Sequence[torch.Tensor],
+    max_delta_brightness: float = 32.0 / 255.0,
+    lower_contrast: float = 0.5, upper_contrast: float = 1.5,
+    generator: Optional[torch.Generator] = None) -> List[torch.Tensor]:
+  """Brightness + contrast only (reference :365-385)."""
+  return ApplyPhotometricImageDistortions(
+      images, random_brightness=True,
+      max_delta_brightness=max_delta_brightness, random_contrast=True,
+      lower_contrast=lower_contrast, upper_contrast=upper_contrast,
+      generator=generator)
+
+
+@gin.configurable
+def ApplyRandomFlips(images: Sequence[torch.Tensor],
+                     flip_probability: float = 0.5,
+                     generator: Optional[torch.Generator] = None
+                     ) -> List[torch.Tensor]:
+  """One shared horizontal-flip decision across the image list (:387-401)."""
+  flip = bool(torch.rand(1, generator=generator) < flip_probability)
+  return [torch.flip(img, dims=[2]) if flip else img for img in images]
+
+
+@gin.configurable
+def ApplyDepthImageDistortions(depth_images: Sequence[torch.Tensor],
+                               random_noise_level: float = 0.05,
+                               random_noise_apply_probability: float = 0.5,
+                               scaling_noise: bool = True,
+                               gamma_shape: float = 1000.0,
+                               gamma_scale_inverse: float = 1000.0,
+                               min_depth_allowed: float = 0.25,
+                               max_depth_allowed: float = 2.5,
+                               generator: Optional[torch.Generator] = None
+                               ) -> List[torch.Tensor]:
+  """Gamma-scaled multiplicative depth noise, clipped (reference :403-459)."""
+  out = []
+  for img in depth_images:
+    _check_nhwc(img)
+    n = img.shape[0]
+    device = img.device
+    x = img
+    if random_noise_level:
+      sigma = _rand(n, 0.0, random_noise_level, device,
+                    generator).view(-1, 1, 1, 1)
+      apply = (_rand(n, 0.0, 1.0, device, generator) <
+               random_noise_apply_probability).float().view(-1, 1, 1, 1)
+      x = x + torch.randn(x.shape, device=device,
+                          generator=generator) * sigma * apply
+    if scaling_noise:
+      gamma = torch.distributions.Gamma(
+          gamma_shape, gamma_scale_inverse).sample((n,)).to(device)
+      x = x * gamma.view(-1, 1, 1, 1)
+    out.append(torch.clamp(x, min_depth_allowed, max_depth_allowed))
+  return out
+
+
+@gin.configurable
+def mixup(images: torch.Tensor, labels: torch.Tensor, alpha: float = 0.2,
+          generator: Optional[torch.Generator] = None):
+  """Beta-mixing of batch elements (BC-Z/vrgripper mixup)."""
+  n = images.shape[0]
+  lam = torch.distributions.Beta(alpha, alpha).sample((n,)).to(images.device)
+  perm = torch.randperm(n, device=images.device, generator=generator)
+  lam_img = lam.view(-1, *([1] * (images.dim() - 1)))
+  mixed_images = lam_img * images + (1 - lam_img) * images[perm]
+  lam_lab = lam.view(-1, *([1] * (labels.dim() - 1)))
+  mixed_labels = lam_lab * labels + (1 - lam_lab) * labels[perm]
+  return mixed_images, mixed_labels
