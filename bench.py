@@ -117,10 +117,9 @@ def main():
       image = features["state/image"].permute(0, 3, 1, 2).contiguous(
           memory_format=torch.channels_last)
       actions = model.pack_action_vector(features)
-      q = network(image, actions)
-    loss = torch.nn.functional.binary_cross_entropy(
-        torch.clamp(q.float(), 1e-7, 1 - 1e-7),
-        labels_t.reshape(q.shape))
+      logit = network(image, actions)
+    loss = torch.nn.functional.binary_cross_entropy_with_logits(
+        logit.float(), labels_t.reshape(logit.shape))
     if dp_engine is not None:
       dp_engine.backward(loss)
     else:

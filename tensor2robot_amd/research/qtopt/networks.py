@@ -103,7 +103,9 @@ class Grasping44(nn.Module):
               ) -> torch.Tensor:
     """image: [N,3,H,W]; action: [N,A] or [N,S,A] (CEM megabatch).
 
-    Returns sigmoid Q of shape [N] (or [N,S] for the megabatch).
+    Returns Q LOGITS of shape [N] (or [N,S] for the megabatch); apply
+    sigmoid for probabilities (kept out of the module so training can use
+    the numerically-stable fused logits loss).
     """
     tile_batch = action.dim() == 3
     action_samples = action.shape[1] if tile_batch else 1
@@ -116,11 +118,10 @@ class Grasping44(nn.Module):
     context = self.embed_action(action)
     net = emb + context[:, :, None, None]
     logits = self.head(net)
-    q = torch.sigmoid(logits)
     if tile_batch:
       if self.num_classes > 1:
-        return q.reshape(-1, action_samples, self.num_classes)
-      return q.reshape(-1, action_samples)
+        return logits.reshape(-1, action_samples, self.num_classes)
+      return logits.reshape(-1, action_samples)
     if self.num_classes == 1:
-      return q.squeeze(-1)
-    return q
+      return logits.squeeze(-1)
+    return logits

@@ -128,17 +128,16 @@ class GraspingModel(classification_model.CriticModel):
       image = image.permute(0, 3, 1, 2)  # NHWC (wire) -> NCHW tower input
     image = image.contiguous(memory_format=torch.channels_last)
     action = self.pack_action_vector(features)
-    q = self.network(image, action)
-    return {"q_predicted": q}
+    logit = self.network(image, action)
+    return {"q_predicted": torch.sigmoid(logit), "logit": logit}
 
   def model_train_fn(self, features, labels, inference_outputs, mode,
                      params=None):
     """Sigmoid log loss on grasp success (reference :229-241)."""
-    q = inference_outputs["q_predicted"]
-    target = labels.grasp_success.reshape(q.shape).to(q.dtype)
-    loss = torch.nn.functional.binary_cross_entropy(
-        torch.clamp(q.float(), 1e-7, 1.0 - 1e-7), target.float())
-    return loss
+    logit = inference_outputs["logit"]
+    target = labels.grasp_success.reshape(logit.shape)
+    return torch.nn.functional.binary_cross_entropy_with_logits(
+        logit.float(), target.float())
 
   def model_eval_fn(self, features, labels, inference_outputs, train_loss,
                     train_outputs, mode, params=None):
