@@ -80,6 +80,7 @@ def main():
   distributed = world_size > 1
 
   use_cuda = torch.cuda.is_available()
+  torch.backends.cudnn.benchmark = True  # let MIOpen autotune conv algos
   if distributed:
     backend = "nccl" if use_cuda else "gloo"
     torch.distributed.init_process_group(backend=backend)

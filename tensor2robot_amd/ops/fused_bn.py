@@ -1,0 +1,115 @@
+"""FusedBatchNormReLU: drop-in BN(+ReLU) module backed by HIP kernels.
+
+On CUDA (ROCm) bf16 channels-last tensors this runs the hand-written CDNA4
+kernels (ops/hip/fused_bn_relu.hip): 2 forward + 3 backward kernels instead
+of MIOpen's BN chain + separate ReLU (see profiles/ for the baseline cost).
+On CPU it runs the plain torch reference (also used by numerics tests).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+from torch import nn
+
+from tensor2robot_amd import ops as ops_mod
+
+
+class _FusedBNReLUFunction(torch.autograd.Function):
+
+  @staticmethod
+  def forward(ctx, x_flat, gamma, beta, running_mean, running_var, eps,
+              momentum, fuse_relu):
+    ext = ops_mod.require_hip()
+    y, mean, invstd = ext.fused_bn_relu_forward(
+        x_flat, gamma, beta, running_mean, running_var, eps, momentum,
+        fuse_relu)
+    ctx.save_for_backward(x_flat, y, gamma, mean, invstd)
+    ctx.fuse_relu = fuse_relu
+    return y
+
+  @staticmethod
+  def backward(ctx, dy):
+    ext = ops_mod.require_hip()
+    x_flat, y, gamma, mean, invstd = ctx.saved_tensors
+    dx, dgamma, dbeta = ext.fused_bn_relu_backward(
+        dy.contiguous(), x_flat, y, gamma, mean, invstd, ctx.fuse_relu)
+    return dx, dgamma, dbeta, None, None, None, None, None
+
+
+def _flat_nhwc(x: torch.Tensor):
+  """[N,C,H,W] channels_last or [N,C] -> flat [M,C] view + restore info."""
+  if x.dim() == 2:
+    return x.contiguous(), None
+  if x.dim() == 4:
+    if not x.is_contiguous(memory_format=torch.channels_last):
+      x = x.contiguous(memory_format=torch.channels_last)
+    n, c, h, w = x.shape
+    flat = x.permute(0, 2, 3, 1).reshape(n * h * w, c)
+    return flat, (n, c, h, w)
+  raise ValueError(f"FusedBatchNormReLU supports 2D/4D, got {x.dim()}D")
+
+
+def _unflat(y_flat: torch.Tensor, shape_info):
+  if shape_info is None:
+    return y_flat
+  n, c, h, w = shape_info
+  return y_flat.reshape(n, h, w, c).permute(0, 3, 1, 2).contiguous(
+      memory_format=torch.channels_last)
+
+
+class FusedBatchNormReLU(nn.Module):
+  """BatchNorm2d/1d + optional ReLU; HIP-fused on GPU bf16."""
+
+  def __init__(self, num_features: int, eps: float = 1e-3,
+               momentum: float = 0.003, fuse_relu: bool = True):
+    super().__init__()
+    self.num_features = num_features
+    self.eps = eps
+    self.momentum = momentum
+    self.fuse_relu = fuse_relu
+    self.weight = nn.Parameter(torch.ones(num_features))
+    self.bias = nn.Parameter(torch.zeros(num_features))
+    self.register_buffer("running_mean", torch.zeros(num_features))
+    self.register_buffer("running_var", torch.ones(num_features))
+    self.register_buffer("num_batches_tracked",
+                         torch.tensor(0, dtype=torch.long))
+
+  def _use_hip(self, x: torch.Tensor) -> bool:
+    return (x.is_cuda and x.dtype == torch.bfloat16 and
+            self.num_features % 8 == 0 and self.num_features <= 2048)
+
+  def forward(self, x: torch.Tensor) -> torch.Tensor:
+    if self._use_hip(x):
+      flat, shape_info = _flat_nhwc(x)
+      if self.training:
+        y = _FusedBNReLUFunction.apply(
+            flat, self.weight, self.bias,
+            self.running_mean, self.running_var, self.eps, self.momentum,
+            self.fuse_relu)
+        self.num_batches_tracked += 1
+      else:
+        invstd = torch.rsqrt(self.running_var + self.eps)
+        scale = (self.weight * invstd).float()
+        shift = (self.bias - self.running_mean * self.weight * invstd
+                 ).float()
+        y = ops_mod.require_hip().bn_inference_apply(
+            flat, scale.contiguous(), shift.contiguous(), self.fuse_relu)
+      return _unflat(y, shape_info)
+    # Torch reference path (CPU / non-bf16): identical math.
+    if x.dim() == 4:
+      y = torch.nn.functional.batch_norm(
+          x, self.running_mean, self.running_var, self.weight, self.bias,
+          self.training, self.momentum, self.eps)
+    else:
+      y = torch.nn.functional.batch_norm(
+          x, self.running_mean, self.running_var, self.weight, self.bias,
+          self.training, self.momentum, self.eps)
+    if self.fuse_relu:
+      y = torch.relu(y)
+    return y
+
+  def extra_repr(self):
+    return (f"{self.num_features}, eps={self.eps}, "
+            f"momentum={self.momentum}, fuse_relu={self.fuse_relu}")

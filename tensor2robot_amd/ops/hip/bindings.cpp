@@ -1,0 +1,33 @@
+// pybind11 bindings for the hand-written CDNA4 HIP kernels.
+
+#include <torch/extension.h>
+
+std::vector<at::Tensor> fused_bn_relu_forward(
+    at::Tensor x, at::Tensor gamma, at::Tensor beta,
+    c10::optional<at::Tensor> running_mean,
+    c10::optional<at::Tensor> running_var, double eps, double momentum,
+    bool fuse_relu);
+
+at::Tensor bn_inference_apply(at::Tensor x, at::Tensor scale,
+                              at::Tensor shift, bool fuse_relu);
+
+std::vector<at::Tensor> fused_bn_relu_backward(
+    at::Tensor dy, at::Tensor x, at::Tensor y, at::Tensor gamma,
+    at::Tensor mean, at::Tensor invstd, bool fused_relu);
+
+at::Tensor fused_preprocess(at::Tensor raw, int64_t oy, int64_t ox,
+                            int64_t th, int64_t tw,
+                            c10::optional<at::Tensor> delta_b,
+                            c10::optional<at::Tensor> f_sat,
+                            c10::optional<at::Tensor> f_con, bool out_bf16);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_bn_relu_forward", &fused_bn_relu_forward,
+        "Fused training BN(+ReLU) forward (NHWC bf16)");
+  m.def("bn_inference_apply", &bn_inference_apply,
+        "BN inference apply (+ReLU) (NHWC bf16)");
+  m.def("fused_bn_relu_backward", &fused_bn_relu_backward,
+        "Fused BN(+ReLU) backward (NHWC bf16)");
+  m.def("fused_preprocess", &fused_preprocess,
+        "Fused crop+convert+photometric distortion (uint8 NHWC -> f32/bf16)");
+}

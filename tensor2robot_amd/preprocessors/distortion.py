@@ -59,6 +59,17 @@ def preprocess_image(image: torch.Tensor, mode: str,
   if is_sequence or image.dim() == 5:
     leading = image.shape[:2]
     image = image.reshape(-1, *image.shape[2:])
+  crop = crop_size or target_size
+  if image.is_cuda and image.dtype == torch.uint8 and image.dim() == 4 \
+      and image.shape[-1] == 3:
+    # Fused HIP path: crop + convert + photometric distortion in one
+    # kernel pair (tensor2robot_amd/ops/hip/preprocess.hip).
+    from tensor2robot_amd.ops import preprocess as fused
+    image = fused.fused_preprocess_image(image, mode, tuple(crop),
+                                         generator=generator)
+    if leading is not None:
+      image = image.reshape(*leading, *image.shape[1:])
+    return image
   if image.dtype == torch.uint8:
     image = image.to(torch.float32) / 255.0
   crop_size = crop_size or target_size

@@ -1,0 +1,133 @@
+"""GPU numerics tests: HIP kernels vs plain-torch fp32 references."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                                  reason="needs MI355X")
+
+
+def _bn_reference(x32, gamma, beta, eps, relu=True):
+  mean = x32.mean(dim=0)
+  var = x32.var(dim=0, unbiased=False)
+  xhat = (x32 - mean) / torch.sqrt(var + eps)
+  y = xhat * gamma + beta
+  return torch.relu(y) if relu else y
+
+
+@requires_gpu
+def test_fused_bn_relu_forward_matches_fp32_reference():
+  from tensor2robot_amd.ops import _t2r_hip
+  torch.manual_seed(0)
+  M, C = 10000, 64
+  x = (torch.randn(M, C, device="cuda") * 2 + 0.5).to(torch.bfloat16)
+  gamma = torch.rand(C, device="cuda") + 0.5
+  beta = torch.randn(C, device="cuda")
+  y, mean, invstd = _t2r_hip.fused_bn_relu_forward(
+      x, gamma, beta, None, None, 1e-3, 0.003, True)
+  ref = _bn_reference(x.float(), gamma, beta, 1e-3)
+  assert torch.allclose(y.float(), ref, atol=3e-2, rtol=3e-2)
+  ref_mean = x.float().mean(dim=0)
+  assert torch.allclose(mean, ref_mean, atol=1e-3, rtol=1e-3)
+
+
+@requires_gpu
+def test_fused_bn_relu_backward_matches_autograd():
+  from tensor2robot_amd.ops import fused_bn
+  torch.manual_seed(1)
+  M, C = 4096, 64
+  x32 = torch.randn(M, C, device="cuda", requires_grad=True)
+  gamma32 = (torch.rand(C, device="cuda") + 0.5).requires_grad_(True)
+  beta32 = torch.randn(C, device="cuda").requires_grad_(True)
+  ref = _bn_reference(x32, gamma32, beta32, 1e-3)
+  dy = torch.randn_like(ref)
+  ref.backward(dy)
+
+  x_bf = x32.detach().to(torch.bfloat16).requires_grad_(True)
+  gamma = gamma32.detach().clone().requires_grad_(True)
+  beta = beta32.detach().clone().requires_grad_(True)
+  y = fused_bn._FusedBNReLUFunction.apply(
+      x_bf, gamma, beta, None, None, 1e-3, 0.003, True)
+  y.backward(dy.to(torch.bfloat16))
+  assert torch.allclose(y.float(), ref, atol=3e-2, rtol=3e-2)
+  # Gradients: bf16 inputs => loose tolerances, but structure must match.
+  assert torch.allclose(gamma.grad, gamma32.grad, atol=0.5, rtol=3e-2)
+  assert torch.allclose(beta.grad, beta32.grad, atol=0.5, rtol=3e-2)
+  rel = (x_bf.grad.float() - x32.grad).abs().max() / \
+      x32.grad.abs().max().clamp(min=1e-6)
+  assert rel < 0.1, f"dx relative error {rel}"
+
+
+@requires_gpu
+def test_fused_bn_module_4d_channels_last():
+  from tensor2robot_amd.ops import fused_bn
+  torch.manual_seed(2)
+  m = fused_bn.FusedBatchNormReLU(64).cuda()
+  x = torch.randn(8, 64, 27, 27, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+  y = m(x)
+  assert y.shape == x.shape
+  y.sum().backward()
+  assert x.grad is not None and m.weight.grad is not None
+  # Running stats updated.
+  assert not torch.allclose(m.running_mean, torch.zeros(64, device="cuda"))
+  # Eval path.
+  m.eval()
+  with torch.no_grad():
+    y2 = m(x)
+  assert y2.shape == x.shape
+  assert (y2.float() >= 0).all()
+
+
+@requires_gpu
+def test_fused_preprocess_matches_torch_reference():
+  from tensor2robot_amd.ops import _t2r_hip
+  torch.manual_seed(3)
+  n, H, W, th, tw = 3, 64, 80, 48, 48
+  raw = torch.randint(0, 256, (n, H, W, 3), dtype=torch.uint8,
+                      device="cuda")
+  delta_b = (torch.rand(n, device="cuda") * 2 - 1) * 0.125
+  f_sat = torch.rand(n, device="cuda") + 0.5
+  f_con = torch.rand(n, device="cuda") + 0.5
+  oy, ox = 5, 7
+  out = _t2r_hip.fused_preprocess(raw, oy, ox, th, tw, delta_b, f_sat,
+                                  f_con, False)
+  # Torch reference: crop -> /255 -> brightness -> saturation -> contrast.
+  x = raw[:, oy: oy + th, ox: ox + tw, :].float() / 255.0
+  x = x + delta_b.view(-1, 1, 1, 1)
+  gray = x.mean(dim=-1, keepdim=True)
+  x = gray + (x - gray) * f_sat.view(-1, 1, 1, 1)
+  mean = x.mean(dim=(1, 2), keepdim=True)
+  x = (x - mean) * f_con.view(-1, 1, 1, 1) + mean
+  ref = torch.clamp(x, 0.0, 1.0)
+  assert torch.allclose(out, ref, atol=2e-3), \
+      f"max diff {(out-ref).abs().max()}"
+
+
+@requires_gpu
+def test_fused_preprocess_center_crop_no_distort():
+  from tensor2robot_amd.ops import _t2r_hip
+  raw = torch.randint(0, 256, (2, 32, 32, 3), dtype=torch.uint8,
+                      device="cuda")
+  out = _t2r_hip.fused_preprocess(raw, 4, 4, 24, 24, None, None, None,
+                                  False)
+  ref = raw[:, 4:28, 4:28, :].float() / 255.0
+  assert torch.allclose(out, ref, atol=1e-6)
+
+
+@requires_gpu
+def test_grasping44_uses_hip_bn():
+  """The flagship network must run the HIP BN path on GPU (fail loudly)."""
+  from tensor2robot_amd.ops import fused_bn, hip_available
+  assert hip_available(), "HIP extension must be importable on a GPU box"
+  from tensor2robot_amd.research.qtopt import networks
+  net = networks.Grasping44().cuda().to(memory_format=torch.channels_last)
+  x = torch.randn(2, 3, 472, 472, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last)
+  a = torch.rand(2, 10, device="cuda")
+  with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+    out = net(x, a)
+  assert out.shape == (2,)
