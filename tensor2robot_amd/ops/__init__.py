@@ -10,11 +10,13 @@ from __future__ import annotations
 
 import torch
 
+import importlib
+
 _t2r_hip = None
 _load_error = None
 
 try:
-  from tensor2robot_amd.ops import _t2r_hip  # type: ignore  # noqa: F401
+  _t2r_hip = importlib.import_module("tensor2robot_amd.ops._t2r_hip")
 except ImportError as e:  # pragma: no cover
   _load_error = e
 
