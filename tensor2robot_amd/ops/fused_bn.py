@@ -22,20 +22,20 @@ class _FusedBNReLUFunction(torch.autograd.Function):
   def forward(ctx, x_flat, gamma, beta, running_mean, running_var, eps,
               momentum, fuse_relu):
     ext = ops_mod.require_hip()
-    y, mean, invstd = ext.fused_bn_relu_forward(
+    y, stats = ext.fused_bn_relu_forward(
         x_flat, gamma, beta, running_mean, running_var, eps, momentum,
         fuse_relu)
-    ctx.save_for_backward(x_flat, y, gamma, mean, invstd)
+    ctx.save_for_backward(x_flat, gamma, beta, stats)
     ctx.fuse_relu = fuse_relu
     return y
 
   @staticmethod
   def backward(ctx, dy):
     ext = ops_mod.require_hip()
-    x_flat, y, gamma, mean, invstd = ctx.saved_tensors
-    dx, dgamma, dbeta = ext.fused_bn_relu_backward(
-        dy.contiguous(), x_flat, y, gamma, mean, invstd, ctx.fuse_relu)
-    return dx, dgamma, dbeta, None, None, None, None, None
+    x_flat, gamma, beta, stats = ctx.saved_tensors
+    dx, grads = ext.fused_bn_relu_backward(
+        dy.contiguous(), x_flat, gamma, beta, stats, ctx.fuse_relu)
+    return dx, grads[1], grads[0], None, None, None, None, None
 
 
 def _flat_nhwc(x: torch.Tensor):

@@ -12,8 +12,8 @@ at::Tensor bn_inference_apply(at::Tensor x, at::Tensor scale,
                               at::Tensor shift, bool fuse_relu);
 
 std::vector<at::Tensor> fused_bn_relu_backward(
-    at::Tensor dy, at::Tensor x, at::Tensor y, at::Tensor gamma,
-    at::Tensor mean, at::Tensor invstd, bool fused_relu);
+    at::Tensor dy, at::Tensor x, at::Tensor gamma, at::Tensor beta,
+    at::Tensor stats, bool fused_relu);
 
 at::Tensor fused_preprocess(at::Tensor raw, int64_t oy, int64_t ox,
                             int64_t th, int64_t tw,

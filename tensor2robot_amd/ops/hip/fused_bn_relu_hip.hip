@@ -1,21 +1,24 @@
 // Fused training BatchNorm + ReLU for NHWC bf16 tensors (gfx950 / CDNA4).
 //
 // Replaces the MIOpen BN kernel chain (MeanVariance, FinalMeanVariance,
-// Norm) + separate ReLU clamp + their backward counterparts with 2 forward
-// and 2 backward kernels.  The tensor is viewed as a flat [M, C] matrix
-// (NHWC channels-last: C contiguous, M = N*H*W), so every wave issues
-// 16-byte (bf16x8) loads that are perfectly coalesced.
+// Norm) + separate ReLU clamp + their backward counterparts with 3 forward
+// and 3 backward kernels, fewer HBM passes and no f32 atomics:
+//
+//   fwd: stats (per-WG partial slabs) -> finalize (reduce + scale/shift +
+//        running stats) -> apply (normalize + ReLU, one pass)
+//   bwd: reduce (partial dbeta/dgamma slabs; ReLU mask RECOMPUTED from x,
+//        so only x and dy are read - y is never saved) -> coeffs ->
+//        dx = k1*g + k2*x + k3 (one pass)
 //
 // Design notes (cdna_hip_programming.md):
-//  * wave64; 256-thread workgroups; each thread owns 8 consecutive channels
-//    (one uint4 load) => C must be a multiple of 8 (python falls back to
-//    torch otherwise).
-//  * grid is oversubscribed (>> 256 CUs) with a grid-stride loop; LDS
-//    tree-reduce inside the workgroup, one float atomicAdd per channel per
-//    workgroup to the global partial buffers (few thousand atomics total).
-//  * stats/params/accumulators are fp32; data is bf16.
-//  * ReLU is fused into the normalize pass; backward masks with y > 0 so no
-//    separate mask tensor is stored.
+//  * wave64; 256-thread workgroups; each thread owns 8 consecutive
+//    channels (one uint4 = bf16x8 load) => C % 8 == 0.
+//  * grid-stride loops over the flat [M, C] view (channels-last memory);
+//    a wave's 64 lanes cover 8 rows x 128 B contiguous = 1 KiB/instr.
+//  * partial reductions go to per-WG slabs (no atomic contention; the
+//    earlier atomicAdd design serialized 4096 WGs on 64 addresses and was
+//    2-4x slower than MIOpen on large feature maps).
+//  * stats/params/accumulators fp32; data bf16.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
@@ -23,8 +26,6 @@
 #include <ATen/hip/HIPContext.h>
 
 #include <cstdint>
-
-#define T2R_CHECK(cond, msg) TORCH_CHECK(cond, msg)
 
 typedef __hip_bfloat16 bf16_t;
 
@@ -40,25 +41,33 @@ union Vec8 {
   bf16_t v[8];
 };
 
+// Workgroups for the streaming kernels: enough to fill 256 CUs several
+// times over, small enough that the partial-slab reduction stays cheap.
+static int pick_grid(long M, int rows, int cap = 1024) {
+  long wgs = (M + rows - 1) / rows;
+  if (wgs > cap) wgs = cap;
+  if (wgs < 1) wgs = 1;
+  return (int)wgs;
+}
+
 // ---------------------------------------------------------------------------
-// Forward stats: partial per-channel sum / sum-of-squares.
+// Forward stats: per-WG partial sum / sumsq slabs [n_wgs, 2, C].
 // ---------------------------------------------------------------------------
 
 extern "C" __global__ void __launch_bounds__(256)
-bn_stats_kernel(const bf16_t* __restrict__ x, float* __restrict__ psum,
-                float* __restrict__ psq, long M, int C) {
-  const int tpr = C >> 3;              // threads per row
-  const int rows = 256 / tpr;          // rows handled per wg iteration
-  const int rg = threadIdx.x / tpr;    // row group within wg
+bn_stats_kernel(const bf16_t* __restrict__ x, float* __restrict__ partial,
+                long M, int C) {
+  const int tpr = C >> 3;
+  const int rows = 256 / tpr;
+  const int rg = threadIdx.x / tpr;
   const int cbase = (threadIdx.x % tpr) << 3;
-  extern __shared__ float lds[];       // [2][rows][C]
+  extern __shared__ float lds[];  // [2][rows][C]
   float* s_sum = lds;
   float* s_sq = lds + (long)rows * C;
 
   float sum[8], sq[8];
 #pragma unroll
   for (int i = 0; i < 8; ++i) { sum[i] = 0.f; sq[i] = 0.f; }
-
   for (long r = (long)blockIdx.x * rows + rg; r < M;
        r += (long)gridDim.x * rows) {
     Vec8 vec;
@@ -76,25 +85,24 @@ bn_stats_kernel(const bf16_t* __restrict__ x, float* __restrict__ psum,
     s_sq[(long)rg * C + cbase + i] = sq[i];
   }
   __syncthreads();
-  // Threads 0..C-1 reduce over row groups (C <= 256 assumed; python layer
-  // enforces C <= 2048 by splitting, in practice C is 64..512).
+  float* out = partial + (long)blockIdx.x * 2 * C;
   for (int c = threadIdx.x; c < C; c += 256) {
     float a = 0.f, b = 0.f;
     for (int g = 0; g < rows; ++g) {
       a += s_sum[(long)g * C + c];
       b += s_sq[(long)g * C + c];
     }
-    atomicAdd(&psum[c], a);
-    atomicAdd(&psq[c], b);
+    out[c] = a;
+    out[C + c] = b;
   }
 }
 
 // ---------------------------------------------------------------------------
-// Finalize: mean/invstd + scale/shift + running-stat update (1 workgroup).
+// Finalize: reduce slabs; mean/invstd + scale/shift + running stats.
 // ---------------------------------------------------------------------------
 
 extern "C" __global__ void bn_finalize_kernel(
-    const float* __restrict__ psum, const float* __restrict__ psq,
+    const float* __restrict__ partial, int n_wgs,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ mean_out, float* __restrict__ invstd_out,
     float* __restrict__ scale_out, float* __restrict__ shift_out,
@@ -102,8 +110,13 @@ extern "C" __global__ void bn_finalize_kernel(
     long M, int C, float eps, float momentum) {
   for (int c = threadIdx.x + blockIdx.x * blockDim.x; c < C;
        c += blockDim.x * gridDim.x) {
-    float mean = psum[c] / (float)M;
-    float var = fmaxf(psq[c] / (float)M - mean * mean, 0.f);
+    float s = 0.f, q = 0.f;
+    for (int w = 0; w < n_wgs; ++w) {
+      s += partial[(long)w * 2 * C + c];
+      q += partial[(long)w * 2 * C + C + c];
+    }
+    float mean = s / (float)M;
+    float var = fmaxf(q / (float)M - mean * mean, 0.f);
     float invstd = rsqrtf(var + eps);
     float sc = gamma[c] * invstd;
     mean_out[c] = mean;
@@ -111,7 +124,6 @@ extern "C" __global__ void bn_finalize_kernel(
     scale_out[c] = sc;
     shift_out[c] = beta[c] - mean * sc;
     if (running_mean != nullptr) {
-      // torch semantics: running stats use unbiased variance.
       float unbiased = (M > 1) ? var * (float)M / (float)(M - 1) : var;
       running_mean[c] += momentum * (mean - running_mean[c]);
       running_var[c] += momentum * (unbiased - running_var[c]);
@@ -153,17 +165,18 @@ bn_apply_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
 }
 
 // ---------------------------------------------------------------------------
-// Backward reduction: dbeta = sum g, dgamma = sum g*xhat, g = dy * (y>0).
+// Backward reduction: partial dbeta/dgamma slabs.  g = dy * relu_mask,
+// relu_mask recomputed as (xhat*gamma + beta > 0) - no y stream.
 // ---------------------------------------------------------------------------
 
 extern "C" __global__ void __launch_bounds__(256)
 bn_bwd_reduce_kernel(const bf16_t* __restrict__ x,
                      const bf16_t* __restrict__ dy,
-                     const bf16_t* __restrict__ y,
                      const float* __restrict__ mean,
                      const float* __restrict__ invstd,
-                     float* __restrict__ pdbeta,
-                     float* __restrict__ pdgamma, long M, int C,
+                     const float* __restrict__ gamma,
+                     const float* __restrict__ beta,
+                     float* __restrict__ partial, long M, int C,
                      int fused_relu) {
   const int tpr = C >> 3;
   const int rows = 256 / tpr;
@@ -172,27 +185,26 @@ bn_bwd_reduce_kernel(const bf16_t* __restrict__ x,
   extern __shared__ float lds[];
   float* s_db = lds;
   float* s_dg = lds + (long)rows * C;
-  float mu[8], is[8], db[8], dg[8];
+  float mu[8], is[8], ga[8], be[8], db[8], dg[8];
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
     mu[i] = mean[cbase + i];
     is[i] = invstd[cbase + i];
+    ga[i] = gamma[cbase + i];
+    be[i] = beta[cbase + i];
     db[i] = 0.f;
     dg[i] = 0.f;
   }
   for (long r = (long)blockIdx.x * rows + rg; r < M;
        r += (long)gridDim.x * rows) {
-    Vec8 vx, vdy, vy;
+    Vec8 vx, vdy;
     vx.raw = *reinterpret_cast<const uint4*>(x + r * C + cbase);
     vdy.raw = *reinterpret_cast<const uint4*>(dy + r * C + cbase);
-    if (fused_relu) {
-      vy.raw = *reinterpret_cast<const uint4*>(y + r * C + cbase);
-    }
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
-      float g = bf2f(vdy.v[i]);
-      if (fused_relu && bf2f(vy.v[i]) <= 0.f) g = 0.f;
       float xhat = (bf2f(vx.v[i]) - mu[i]) * is[i];
+      float g = bf2f(vdy.v[i]);
+      if (fused_relu && (xhat * ga[i] + be[i]) <= 0.f) g = 0.f;
       db[i] += g;
       dg[i] += g * xhat;
     }
@@ -203,41 +215,54 @@ bn_bwd_reduce_kernel(const bf16_t* __restrict__ x,
     s_dg[(long)rg * C + cbase + i] = dg[i];
   }
   __syncthreads();
+  float* out = partial + (long)blockIdx.x * 2 * C;
   for (int c = threadIdx.x; c < C; c += 256) {
     float a = 0.f, b = 0.f;
     for (int g = 0; g < rows; ++g) {
       a += s_db[(long)g * C + c];
       b += s_dg[(long)g * C + c];
     }
-    atomicAdd(&pdbeta[c], a);
-    atomicAdd(&pdgamma[c], b);
+    out[c] = a;
+    out[C + c] = b;
   }
 }
 
 // ---------------------------------------------------------------------------
-// Backward dx: dx = k1*g + k2*x + k3 with per-channel coefficients.
+// Coefficients: reduce slabs; dgamma/dbeta out; k1/k2/k3 for the dx pass.
 //   dx = gamma*invstd * (g - dbeta/M - xhat * dgamma/M)
+//      = k1*g + k2*x + k3
 // ---------------------------------------------------------------------------
 
 extern "C" __global__ void bn_bwd_coeffs_kernel(
+    const float* __restrict__ partial, int n_wgs,
     const float* __restrict__ gamma, const float* __restrict__ mean,
-    const float* __restrict__ invstd, const float* __restrict__ dbeta,
-    const float* __restrict__ dgamma, float* __restrict__ k1,
+    const float* __restrict__ invstd, float* __restrict__ dbeta_out,
+    float* __restrict__ dgamma_out, float* __restrict__ k1,
     float* __restrict__ k2, float* __restrict__ k3, long M, int C) {
   for (int c = threadIdx.x + blockIdx.x * blockDim.x; c < C;
        c += blockDim.x * gridDim.x) {
+    float db = 0.f, dg = 0.f;
+    for (int w = 0; w < n_wgs; ++w) {
+      db += partial[(long)w * 2 * C + c];
+      dg += partial[(long)w * 2 * C + C + c];
+    }
+    dbeta_out[c] = db;
+    dgamma_out[c] = dg;
     float gs = gamma[c] * invstd[c];
-    float t = dgamma[c] / (float)M * invstd[c];
+    float t = dg / (float)M * invstd[c];
     k1[c] = gs;
     k2[c] = -gs * t;
-    k3[c] = gs * (mean[c] * t - dbeta[c] / (float)M);
+    k3[c] = gs * (mean[c] * t - db / (float)M);
   }
 }
 
 extern "C" __global__ void __launch_bounds__(256)
 bn_bwd_dx_kernel(const bf16_t* __restrict__ x,
                  const bf16_t* __restrict__ dy,
-                 const bf16_t* __restrict__ y,
+                 const float* __restrict__ mean,
+                 const float* __restrict__ invstd,
+                 const float* __restrict__ gamma,
+                 const float* __restrict__ beta,
                  const float* __restrict__ k1, const float* __restrict__ k2,
                  const float* __restrict__ k3, bf16_t* __restrict__ dx,
                  long M, int C, int fused_relu) {
@@ -245,26 +270,30 @@ bn_bwd_dx_kernel(const bf16_t* __restrict__ x,
   const int rows = 256 / tpr;
   const int rg = threadIdx.x / tpr;
   const int cbase = (threadIdx.x % tpr) << 3;
-  float a[8], b[8], c3[8];
+  float a[8], b[8], c3[8], mu[8], is[8], ga[8], be[8];
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
     a[i] = k1[cbase + i];
     b[i] = k2[cbase + i];
     c3[i] = k3[cbase + i];
+    mu[i] = mean[cbase + i];
+    is[i] = invstd[cbase + i];
+    ga[i] = gamma[cbase + i];
+    be[i] = beta[cbase + i];
   }
   for (long r = (long)blockIdx.x * rows + rg; r < M;
        r += (long)gridDim.x * rows) {
-    Vec8 vx, vdy, vy, out;
+    Vec8 vx, vdy, out;
     vx.raw = *reinterpret_cast<const uint4*>(x + r * C + cbase);
     vdy.raw = *reinterpret_cast<const uint4*>(dy + r * C + cbase);
-    if (fused_relu) {
-      vy.raw = *reinterpret_cast<const uint4*>(y + r * C + cbase);
-    }
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
-      float g = bf2f(vdy.v[i]);
-      if (fused_relu && bf2f(vy.v[i]) <= 0.f) g = 0.f;
       float xv = bf2f(vx.v[i]);
+      float g = bf2f(vdy.v[i]);
+      if (fused_relu) {
+        float xhat = (xv - mu[i]) * is[i];
+        if (xhat * ga[i] + be[i] <= 0.f) g = 0.f;
+      }
       out.v[i] = f2bf(a[i] * g + b[i] * xv + c3[i]);
     }
     *reinterpret_cast<uint4*>(dx + r * C + cbase) = out.raw;
@@ -281,13 +310,6 @@ static void check_flat(const at::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_contiguous(), name, " must be a contiguous [M,C] view");
 }
 
-static int pick_grid(long M, int rows) {
-  long wgs = (M + rows - 1) / rows;
-  if (wgs > 4096) wgs = 4096;  // oversubscribe 256 CUs, bounded atomics
-  if (wgs < 1) wgs = 1;
-  return (int)wgs;
-}
-
 std::vector<at::Tensor> fused_bn_relu_forward(
     at::Tensor x, at::Tensor gamma, at::Tensor beta,
     c10::optional<at::Tensor> running_mean,
@@ -298,42 +320,37 @@ std::vector<at::Tensor> fused_bn_relu_forward(
   const int C = (int)x.size(1);
   TORCH_CHECK(C % 8 == 0 && C <= 2048, "C must be a multiple of 8, <=2048");
   auto opts = x.options().dtype(at::kFloat);
-  auto psum = at::zeros({C}, opts);
-  auto psq = at::zeros({C}, opts);
-  auto mean = at::empty({C}, opts);
-  auto invstd = at::empty({C}, opts);
-  auto scale = at::empty({C}, opts);
-  auto shift = at::empty({C}, opts);
-  auto y = at::empty_like(x);
-
   const int tpr = C / 8;
   const int rows = 256 / tpr;
   const int grid = pick_grid(M, rows);
+  auto partial = at::empty({(long)grid * 2 * C}, opts);
+  auto stats = at::empty({4, C}, opts);  // mean, invstd, scale, shift
+  auto y = at::empty_like(x);
   const size_t lds_bytes = 2l * rows * C * sizeof(float);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
 
   hipLaunchKernelGGL(bn_stats_kernel, dim3(grid), dim3(256), lds_bytes,
                      stream.stream(),
                      reinterpret_cast<const bf16_t*>(x.data_ptr()),
-                     psum.data_ptr<float>(), psq.data_ptr<float>(), M, C);
+                     partial.data_ptr<float>(), M, C);
   float* rm = running_mean.has_value()
                   ? running_mean->data_ptr<float>() : nullptr;
   float* rv = running_var.has_value()
                   ? running_var->data_ptr<float>() : nullptr;
-  hipLaunchKernelGGL(bn_finalize_kernel, dim3(1), dim3(256), 0,
-                     stream.stream(), psum.data_ptr<float>(),
-                     psq.data_ptr<float>(), gamma.data_ptr<float>(),
-                     beta.data_ptr<float>(), mean.data_ptr<float>(),
-                     invstd.data_ptr<float>(), scale.data_ptr<float>(),
-                     shift.data_ptr<float>(), rm, rv, M, C, (float)eps,
+  float* stats_ptr = stats.data_ptr<float>();
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+                     0, stream.stream(), partial.data_ptr<float>(), grid,
+                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                     stats_ptr, stats_ptr + C, stats_ptr + 2 * C,
+                     stats_ptr + 3 * C, rm, rv, M, C, (float)eps,
                      (float)momentum);
   hipLaunchKernelGGL(bn_apply_kernel, dim3(grid), dim3(256), 0,
                      stream.stream(),
                      reinterpret_cast<const bf16_t*>(x.data_ptr()),
                      reinterpret_cast<bf16_t*>(y.data_ptr()),
-                     scale.data_ptr<float>(), shift.data_ptr<float>(), M, C,
+                     stats_ptr + 2 * C, stats_ptr + 3 * C, M, C,
                      fuse_relu ? 1 : 0);
-  return {y, mean, invstd};
+  return {y, stats};
 }
 
 at::Tensor bn_inference_apply(at::Tensor x, at::Tensor scale,
@@ -356,46 +373,45 @@ at::Tensor bn_inference_apply(at::Tensor x, at::Tensor scale,
 }
 
 std::vector<at::Tensor> fused_bn_relu_backward(
-    at::Tensor dy, at::Tensor x, at::Tensor y, at::Tensor gamma,
-    at::Tensor mean, at::Tensor invstd, bool fused_relu) {
+    at::Tensor dy, at::Tensor x, at::Tensor gamma, at::Tensor beta,
+    at::Tensor stats, bool fused_relu) {
   check_flat(x, "x");
   check_flat(dy, "dy");
   const long M = x.size(0);
   const int C = (int)x.size(1);
   auto opts = x.options().dtype(at::kFloat);
-  auto dbeta = at::zeros({C}, opts);
-  auto dgamma = at::zeros({C}, opts);
-  auto k1 = at::empty({C}, opts);
-  auto k2 = at::empty({C}, opts);
-  auto k3 = at::empty({C}, opts);
-  auto dx = at::empty_like(x);
   const int tpr = C / 8;
   const int rows = 256 / tpr;
   const int grid = pick_grid(M, rows);
+  auto partial = at::empty({(long)grid * 2 * C}, opts);
+  auto grads = at::empty({2, C}, opts);   // dbeta, dgamma
+  auto coeffs = at::empty({3, C}, opts);  // k1, k2, k3
+  auto dx = at::empty_like(x);
   const size_t lds_bytes = 2l * rows * C * sizeof(float);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  float* stats_ptr = stats.data_ptr<float>();  // mean, invstd, scale, shift
+  float* grads_ptr = grads.data_ptr<float>();
+  float* coeffs_ptr = coeffs.data_ptr<float>();
   hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(grid), dim3(256), lds_bytes,
                      stream.stream(),
                      reinterpret_cast<const bf16_t*>(x.data_ptr()),
                      reinterpret_cast<const bf16_t*>(dy.data_ptr()),
-                     reinterpret_cast<const bf16_t*>(y.data_ptr()),
-                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                     dbeta.data_ptr<float>(), dgamma.data_ptr<float>(), M, C,
-                     fused_relu ? 1 : 0);
-  hipLaunchKernelGGL(bn_bwd_coeffs_kernel, dim3(1), dim3(256), 0,
-                     stream.stream(), gamma.data_ptr<float>(),
-                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                     dbeta.data_ptr<float>(), dgamma.data_ptr<float>(),
-                     k1.data_ptr<float>(), k2.data_ptr<float>(),
-                     k3.data_ptr<float>(), M, C);
+                     stats_ptr, stats_ptr + C, gamma.data_ptr<float>(),
+                     beta.data_ptr<float>(), partial.data_ptr<float>(), M,
+                     C, fused_relu ? 1 : 0);
+  hipLaunchKernelGGL(bn_bwd_coeffs_kernel, dim3((C + 255) / 256), dim3(256),
+                     0, stream.stream(), partial.data_ptr<float>(), grid,
+                     gamma.data_ptr<float>(), stats_ptr, stats_ptr + C,
+                     grads_ptr, grads_ptr + C, coeffs_ptr, coeffs_ptr + C,
+                     coeffs_ptr + 2 * C, M, C);
   hipLaunchKernelGGL(bn_bwd_dx_kernel, dim3(grid), dim3(256), 0,
                      stream.stream(),
                      reinterpret_cast<const bf16_t*>(x.data_ptr()),
                      reinterpret_cast<const bf16_t*>(dy.data_ptr()),
-                     reinterpret_cast<const bf16_t*>(y.data_ptr()),
-                     k1.data_ptr<float>(), k2.data_ptr<float>(),
-                     k3.data_ptr<float>(),
+                     stats_ptr, stats_ptr + C, gamma.data_ptr<float>(),
+                     beta.data_ptr<float>(), coeffs_ptr, coeffs_ptr + C,
+                     coeffs_ptr + 2 * C,
                      reinterpret_cast<bf16_t*>(dx.data_ptr()), M, C,
                      fused_relu ? 1 : 0);
-  return {dx, dgamma, dbeta};
+  return {dx, grads};
 }
