@@ -101,20 +101,35 @@ bn_stats_kernel(const bf16_t* __restrict__ x, float* __restrict__ partial,
 // Finalize: reduce slabs; mean/invstd + scale/shift + running stats.
 // ---------------------------------------------------------------------------
 
-extern "C" __global__ void bn_finalize_kernel(
+// One block per channel; 256 threads stride the WG slabs, LDS tree-reduce.
+extern "C" __global__ void __launch_bounds__(256)
+bn_finalize_kernel(
     const float* __restrict__ partial, int n_wgs,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ mean_out, float* __restrict__ invstd_out,
     float* __restrict__ scale_out, float* __restrict__ shift_out,
     float* __restrict__ running_mean, float* __restrict__ running_var,
     long M, int C, float eps, float momentum) {
-  for (int c = threadIdx.x + blockIdx.x * blockDim.x; c < C;
-       c += blockDim.x * gridDim.x) {
-    float s = 0.f, q = 0.f;
-    for (int w = 0; w < n_wgs; ++w) {
-      s += partial[(long)w * 2 * C + c];
-      q += partial[(long)w * 2 * C + C + c];
+  const int c = blockIdx.x;
+  float s = 0.f, q = 0.f;
+  for (int w = threadIdx.x; w < n_wgs; w += 256) {
+    s += partial[(long)w * 2 * C + c];
+    q += partial[(long)w * 2 * C + C + c];
+  }
+  __shared__ float ls[256], lq[256];
+  ls[threadIdx.x] = s;
+  lq[threadIdx.x] = q;
+  __syncthreads();
+  for (int st = 128; st > 0; st >>= 1) {
+    if (threadIdx.x < st) {
+      ls[threadIdx.x] += ls[threadIdx.x + st];
+      lq[threadIdx.x] += lq[threadIdx.x + st];
     }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    s = ls[0];
+    q = lq[0];
     float mean = s / (float)M;
     float var = fmaxf(q / (float)M - mean * mean, 0.f);
     float invstd = rsqrtf(var + eps);
@@ -233,19 +248,33 @@ bn_bwd_reduce_kernel(const bf16_t* __restrict__ x,
 //      = k1*g + k2*x + k3
 // ---------------------------------------------------------------------------
 
-extern "C" __global__ void bn_bwd_coeffs_kernel(
+extern "C" __global__ void __launch_bounds__(256)
+bn_bwd_coeffs_kernel(
     const float* __restrict__ partial, int n_wgs,
     const float* __restrict__ gamma, const float* __restrict__ mean,
     const float* __restrict__ invstd, float* __restrict__ dbeta_out,
     float* __restrict__ dgamma_out, float* __restrict__ k1,
     float* __restrict__ k2, float* __restrict__ k3, long M, int C) {
-  for (int c = threadIdx.x + blockIdx.x * blockDim.x; c < C;
-       c += blockDim.x * gridDim.x) {
-    float db = 0.f, dg = 0.f;
-    for (int w = 0; w < n_wgs; ++w) {
-      db += partial[(long)w * 2 * C + c];
-      dg += partial[(long)w * 2 * C + C + c];
+  const int c = blockIdx.x;
+  float db = 0.f, dg = 0.f;
+  for (int w = threadIdx.x; w < n_wgs; w += 256) {
+    db += partial[(long)w * 2 * C + c];
+    dg += partial[(long)w * 2 * C + C + c];
+  }
+  __shared__ float ls[256], lq[256];
+  ls[threadIdx.x] = db;
+  lq[threadIdx.x] = dg;
+  __syncthreads();
+  for (int st = 128; st > 0; st >>= 1) {
+    if (threadIdx.x < st) {
+      ls[threadIdx.x] += ls[threadIdx.x + st];
+      lq[threadIdx.x] += lq[threadIdx.x + st];
     }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    db = ls[0];
+    dg = lq[0];
     dbeta_out[c] = db;
     dgamma_out[c] = dg;
     float gs = gamma[c] * invstd[c];
@@ -338,7 +367,7 @@ std::vector<at::Tensor> fused_bn_relu_forward(
   float* rv = running_var.has_value()
                   ? running_var->data_ptr<float>() : nullptr;
   float* stats_ptr = stats.data_ptr<float>();
-  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(256),
                      0, stream.stream(), partial.data_ptr<float>(), grid,
                      gamma.data_ptr<float>(), beta.data_ptr<float>(),
                      stats_ptr, stats_ptr + C, stats_ptr + 2 * C,
@@ -399,7 +428,7 @@ std::vector<at::Tensor> fused_bn_relu_backward(
                      stats_ptr, stats_ptr + C, gamma.data_ptr<float>(),
                      beta.data_ptr<float>(), partial.data_ptr<float>(), M,
                      C, fused_relu ? 1 : 0);
-  hipLaunchKernelGGL(bn_bwd_coeffs_kernel, dim3((C + 255) / 256), dim3(256),
+  hipLaunchKernelGGL(bn_bwd_coeffs_kernel, dim3(C), dim3(256),
                      0, stream.stream(), partial.data_ptr<float>(), grid,
                      gamma.data_ptr<float>(), stats_ptr, stats_ptr + C,
                      grads_ptr, grads_ptr + C, coeffs_ptr, coeffs_ptr + C,

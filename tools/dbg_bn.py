@@ -9,7 +9,7 @@ def _bn_reference(x32, gamma, beta, eps, relu=True):
   return torch.relu(y) if relu else y
 torch.manual_seed(1)
 M, C = 4096, 64
-x32 = torch.randn(M, C, device="cuda", requires_grad=True)
+x32 = torch.randn(M, C, device="cuda").to(torch.bfloat16).float().requires_grad_(True)
 gamma32 = (torch.rand(C, device="cuda") + 0.5).requires_grad_(True)
 beta32 = torch.randn(C, device="cuda").requires_grad_(True)
 ref = _bn_reference(x32, gamma32, beta32, 1e-3)
