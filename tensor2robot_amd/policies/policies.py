@@ -1,0 +1,279 @@
+"""Policies: action selection on numpy I/O over predictors.
+
+Reference: `policies/policies.py:33-360` — Policy.SelectAction /
+sample_action / reset / restore; CEM over a critic Q, direct regression,
+exploration-noise and per-episode-switch variants; LSTM-CEM hidden-state
+caching.
+"""
+
+from __future__ import annotations
+
+import abc
+from typing import Dict, Optional
+
+import numpy as np
+
+from tensor2robot_amd import gin
+from tensor2robot_amd.utils import cross_entropy
+
+
+class Policy(abc.ABC):
+  """Base policy (reference :33-102)."""
+
+  def __init__(self, predictor=None):
+    self._predictor = predictor
+
+  @property
+  def predictor(self):
+    return self._predictor
+
+  @abc.abstractmethod
+  def SelectAction(self, state, context=None, timestep: int = 0
+                   ) -> np.ndarray:
+    pass
+
+  def sample_action(self, obs, explore_prob: float = 0.0):
+    """run_env adapter (reference :83-102): returns (action, debug)."""
+    del explore_prob
+    action = self.SelectAction(obs, None, 0)
+    return action, {}
+
+  def reset(self):
+    pass
+
+  def restore(self) -> bool:
+    if self._predictor is not None:
+      return self._predictor.restore()
+    return True
+
+  def init_randomly(self):
+    if self._predictor is not None:
+      self._predictor.init_randomly()
+
+  @property
+  def global_step(self) -> int:
+    if self._predictor is not None:
+      return self._predictor.global_step
+    return -1
+
+
+@gin.configurable
+class CEMPolicy(Policy):
+  """CEM argmax over the critic's Q (reference :106-186).
+
+  Evaluates all CEM samples in ONE megabatch-tiled critic call
+  (`action/...` fed as [1, num_samples, action_dim]; the Grasping44 network
+  tiles the image embedding, reference networks.py:515-521).
+  """
+
+  def __init__(self, predictor=None, action_size: int = 10,
+               cem_iterations: int = 3, cem_samples: int = 64,
+               num_elites: int = 6, state_key: str = "state/image",
+               q_key: str = "q_predicted",
+               action_bounds=None, seed: Optional[int] = None):
+    super().__init__(predictor)
+    self._action_size = action_size
+    self._cem = cross_entropy.CrossEntropyMethod(
+        num_samples=cem_samples, num_elites=num_elites,
+        num_iterations=cem_iterations, seed=seed)
+    self._state_key = state_key
+    self._q_key = q_key
+    self._action_bounds = action_bounds
+
+  def pack_fn(self, state, action_samples: np.ndarray
+              ) -> Dict[str, np.ndarray]:
+    """Builds the tiled feed: state batch=1, actions [1, S, d]."""
+    feed: Dict[str, np.ndarray] = {}
+    if isinstance(state, dict):
+      for k, v in state.items():
+        feed[k] = np.asarray(v)[None]
+    else:
+      feed[self._state_key] = np.asarray(state)[None]
+    feed.update(self._split_action(action_samples[None]))
+    return feed
+
+  def _split_action(self, action: np.ndarray) -> Dict[str, np.ndarray]:
+    """Overridable action-vector -> component-features mapping."""
+    return {"action": action.astype(np.float32)}
+
+  def objective_fn(self, state):
+    def objective(samples: np.ndarray) -> np.ndarray:
+      feed = self.pack_fn(state, samples)
+      out = self._predictor.predict(feed)
+      q = np.asarray(out[self._q_key]).reshape(-1)
+      return q
+    return objective
+
+  def get_cem_action(self, state) -> np.ndarray:
+    mean = np.zeros(self._action_size, np.float32)
+    std = np.full(self._action_size, 0.5, np.float32)
+    best, _, _, _ = self._cem.run(self.objective_fn(state), mean, std,
+                                  bounds=self._action_bounds)
+    return best
+
+  def SelectAction(self, state, context=None, timestep: int = 0):
+    return self.get_cem_action(state)
+
+
+@gin.configurable
+class LSTMCEMPolicy(CEMPolicy):
+  """CEM with critic LSTM hidden-state carry (reference :188-220)."""
+
+  def __init__(self, hidden_state_key: str = "lstm_hidden_state",
+               **kwargs):
+    super().__init__(**kwargs)
+    self._hidden_state_key = hidden_state_key
+    self._hidden = None
+    self._best_hidden = None
+
+  def reset(self):
+    self._hidden = None
+    self._best_hidden = None
+
+  def pack_fn(self, state, action_samples):
+    feed = super().pack_fn(state, action_samples)
+    if self._hidden is not None:
+      feed[self._hidden_state_key] = self._hidden
+    return feed
+
+  def SelectAction(self, state, context=None, timestep: int = 0):
+    best_idx = {"i": 0}
+    base_objective = self.objective_fn(state)
+
+    def objective(samples):
+      feed = self.pack_fn(state, samples)
+      out = self._predictor.predict(feed)
+      q = np.asarray(out[self._q_key]).reshape(-1)
+      best_idx["i"] = int(np.argmax(q))
+      if self._hidden_state_key in out:
+        # Keep hidden state of the best sample (reference :199-218).
+        h = out[self._hidden_state_key]
+        self._best_hidden = h.reshape(
+            (-1,) + h.shape[2:])[best_idx["i"]][None] \
+            if h.ndim > 2 else h
+      return q
+
+    mean = np.zeros(self._action_size, np.float32)
+    std = np.full(self._action_size, 0.5, np.float32)
+    best, _, _, _ = self._cem.run(objective, mean, std,
+                                  bounds=self._action_bounds)
+    self._hidden = self._best_hidden
+    return best
+
+
+@gin.configurable
+class RegressionPolicy(Policy):
+  """Direct inference_output action (reference :222-238)."""
+
+  def __init__(self, predictor=None, state_key: str = "state",
+               action_key: str = "inference_output"):
+    super().__init__(predictor)
+    self._state_key = state_key
+    self._action_key = action_key
+
+  def _build_feed(self, state) -> Dict[str, np.ndarray]:
+    if isinstance(state, dict):
+      return {k: np.asarray(v)[None] for k, v in state.items()}
+    return {self._state_key: np.asarray(state)[None]}
+
+  def SelectAction(self, state, context=None, timestep: int = 0):
+    out = self._predictor.predict(self._build_feed(state))
+    return np.asarray(out[self._action_key])[0]
+
+
+@gin.configurable
+class SequentialRegressionPolicy(RegressionPolicy):
+  """Keeps a running context of past states (reference :240-256)."""
+
+  def __init__(self, **kwargs):
+    super().__init__(**kwargs)
+    self._history = []
+
+  def reset(self):
+    self._history = []
+
+  def SelectAction(self, state, context=None, timestep: int = 0):
+    self._history.append(np.asarray(state))
+    return super().SelectAction(state, context, timestep)
+
+
+@gin.configurable
+class OUExploreRegressionPolicy(RegressionPolicy):
+  """Ornstein-Uhlenbeck action noise (reference :258-293)."""
+
+  def __init__(self, theta: float = 0.15, sigma: float = 0.2,
+               seed: Optional[int] = None, **kwargs):
+    super().__init__(**kwargs)
+    self._theta = theta
+    self._sigma = sigma
+    self._rng = np.random.RandomState(seed)
+    self._noise = None
+
+  def reset(self):
+    self._noise = None
+
+  def SelectAction(self, state, context=None, timestep: int = 0):
+    action = super().SelectAction(state, context, timestep)
+    if self._noise is None:
+      self._noise = np.zeros_like(action)
+    self._noise = self._noise - self._theta * self._noise + \
+        self._sigma * self._rng.randn(*action.shape)
+    return action + self._noise
+
+
+@gin.configurable
+class ScheduledExplorationRegressionPolicy(RegressionPolicy):
+  """Linear-schedule gaussian noise by global_step (reference :295-322)."""
+
+  def __init__(self, initial_sigma: float = 0.5, final_sigma: float = 0.05,
+               decay_steps: int = 100000, seed: Optional[int] = None,
+               **kwargs):
+    super().__init__(**kwargs)
+    self._initial_sigma = initial_sigma
+    self._final_sigma = final_sigma
+    self._decay_steps = decay_steps
+    self._rng = np.random.RandomState(seed)
+
+  def _sigma(self) -> float:
+    step = max(self.global_step, 0)
+    frac = min(step / float(self._decay_steps), 1.0)
+    return self._initial_sigma + frac * (self._final_sigma -
+                                         self._initial_sigma)
+
+  def SelectAction(self, state, context=None, timestep: int = 0):
+    action = super().SelectAction(state, context, timestep)
+    return action + self._sigma() * self._rng.randn(*action.shape)
+
+
+@gin.configurable
+class PerEpisodeSwitchPolicy(Policy):
+  """Chooses explore vs greedy policy per episode (reference :324-360)."""
+
+  def __init__(self, explore_policy_class=None, greedy_policy_class=None,
+               explore_prob: float = 0.1, seed: Optional[int] = None):
+    super().__init__(None)
+    self._explore_policy = explore_policy_class()
+    self._greedy_policy = greedy_policy_class()
+    self._explore_prob = explore_prob
+    self._rng = np.random.RandomState(seed)
+    self._current = self._greedy_policy
+
+  def reset(self):
+    self._current = self._explore_policy if \
+        self._rng.rand() < self._explore_prob else self._greedy_policy
+    self._current.reset()
+
+  def restore(self):
+    ok = self._explore_policy.restore()
+    return self._greedy_policy.restore() and ok
+
+  def init_randomly(self):
+    self._explore_policy.init_randomly()
+    self._greedy_policy.init_randomly()
+
+  @property
+  def global_step(self):
+    return self._greedy_policy.global_step
+
+  def SelectAction(self, state, context=None, timestep: int = 0):
+    return self._current.SelectAction(state, context, timestep)

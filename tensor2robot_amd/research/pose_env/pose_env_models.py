@@ -1,0 +1,184 @@
+"""Pose-env models: regression + continuous-MC critic (CPU smoke workload).
+
+Reference: `research/pose_env/pose_env_models.py` —
+PoseEnvRegressionModel :231 (image -> pose), PoseEnvContinuousMCModel :92
+(critic + CEM), preprocessor decoding 64x64 encoded images (:48-77).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Optional
+
+import numpy as np
+import torch
+from torch import nn
+
+from tensor2robot_amd import gin
+from tensor2robot_amd.models import abstract_model
+from tensor2robot_amd.models import classification_model
+from tensor2robot_amd.preprocessors import abstract_preprocessor
+from tensor2robot_amd.research.pose_env import pose_env
+from tensor2robot_amd.specs import tensorspec_utils as tsu
+
+TSPEC = tsu.ExtendedTensorSpec
+IMG = pose_env.IMAGE_SIZE
+
+
+class _PoseEnvPreprocessor(
+    abstract_preprocessor.SpecTransformationPreprocessor):
+  """Encoded 64x64 image -> float [0,1] (reference :48-77)."""
+
+  def _transform_in_feature_specification(self, flat_spec, mode):
+    out = tsu.TensorSpecStruct()
+    for key, spec in flat_spec.items():
+      if "image" in key:
+        out[key] = TSPEC((IMG, IMG, 3), torch.uint8, name=spec.name,
+                         data_format=spec.data_format or "PNG",
+                         dataset_key=spec.dataset_key)
+      else:
+        out[key] = spec
+    return out
+
+  def _preprocess_fn(self, features, labels, mode):
+    for key in list(features.keys()):
+      if "image" in key and features[key].dtype == torch.uint8:
+        features[key] = features[key].to(torch.float32) / 255.0
+    return features, labels
+
+
+class _SmallConvNet(nn.Module):
+
+  def __init__(self, out_dim: int, extra_in: int = 0):
+    super().__init__()
+    self.conv = nn.Sequential(
+        nn.Conv2d(3, 16, 3, stride=2, padding=1), nn.ReLU(),
+        nn.Conv2d(16, 16, 3, stride=2, padding=1), nn.ReLU(),
+        nn.Conv2d(16, 16, 3, stride=2, padding=1), nn.ReLU(),
+    )
+    self.head = nn.Sequential(
+        nn.Linear(16 * 8 * 8 + extra_in, 64), nn.ReLU(),
+        nn.Linear(64, out_dim),
+    )
+
+  def forward(self, image, extra=None):
+    h = self.conv(image).flatten(1)
+    if extra is not None:
+      h = torch.cat([h, extra], dim=1)
+    return self.head(h)
+
+
+@gin.configurable
+class PoseEnvRegressionModel(abstract_model.AbstractT2RModel):
+  """Image -> 2D pose regression (reference :231-300)."""
+
+  def __init__(self, device_type: str = "cpu", **kwargs):
+    kwargs.setdefault("compute_dtype", "float32"
+                      if device_type == "cpu" else "bfloat16")
+    kwargs.setdefault("preprocessor_cls", _PoseEnvPreprocessor)
+    super().__init__(device_type=device_type, **kwargs)
+
+  def get_feature_specification(self, mode):
+    spec = tsu.TensorSpecStruct()
+    spec["state/image"] = TSPEC((IMG, IMG, 3), torch.float32,
+                                name="state/image", data_format="PNG")
+    return spec
+
+  def get_label_specification(self, mode):
+    spec = tsu.TensorSpecStruct()
+    spec["pose"] = TSPEC((2,), torch.float32, name="pose")
+    return spec
+
+  def create_network(self):
+    return _SmallConvNet(out_dim=2)
+
+  def inference_network_fn(self, features, labels, mode, params=None):
+    image = features["state/image"]
+    if image.shape[-1] == 3:
+      image = image.permute(0, 3, 1, 2).contiguous()
+    pose = self.network(image)
+    return {"inference_output": pose}
+
+  def model_train_fn(self, features, labels, inference_outputs, mode,
+                     params=None):
+    return torch.nn.functional.mse_loss(
+        inference_outputs["inference_output"], labels.pose)
+
+  def model_eval_fn(self, features, labels, inference_outputs, train_loss,
+                    train_outputs, mode, params=None):
+    err = (inference_outputs["inference_output"] -
+           labels.pose).norm(dim=-1).mean()
+    return {"mean_pose_error": err}
+
+  def create_export_outputs_fn(self, features, inference_outputs, mode,
+                               params=None):
+    return {"inference_output": inference_outputs["inference_output"]}
+
+
+@gin.configurable
+class PoseEnvContinuousMCModel(classification_model.CriticModel):
+  """Critic Q(image, pose-action) for CEM serving (reference :92-230)."""
+
+  def __init__(self, device_type: str = "cpu", **kwargs):
+    kwargs.setdefault("compute_dtype", "float32"
+                      if device_type == "cpu" else "bfloat16")
+    kwargs.setdefault("preprocessor_cls", _PoseEnvPreprocessor)
+    super().__init__(device_type=device_type, **kwargs)
+
+  def get_state_specification(self):
+    spec = tsu.TensorSpecStruct()
+    spec["image"] = TSPEC((IMG, IMG, 3), torch.float32,
+                          name="state/image", data_format="PNG")
+    return spec
+
+  def get_action_specification(self):
+    spec = tsu.TensorSpecStruct()
+    spec["pose"] = TSPEC((2,), torch.float32, name="action/pose")
+    return spec
+
+  def create_network(self):
+    return _SmallConvNet(out_dim=1, extra_in=2)
+
+  def q_func(self, features, mode, params=None):
+    image = features["state/image"]
+    if image.shape[-1] == 3:
+      image = image.permute(0, 3, 1, 2).contiguous()
+    action = features["action/pose"]
+    tile_batch = action.dim() == 3
+    if tile_batch:
+      samples = action.shape[1]
+      action = action.reshape(-1, action.shape[-1])
+      image = image.repeat_interleave(samples, dim=0)
+    logit = self.network(image, action)
+    q = torch.sigmoid(logit)
+    if tile_batch:
+      q = q.reshape(-1, samples)
+      logit = logit.reshape(-1, samples)
+    return {"q_predicted": q, "logit": logit}
+
+  def model_train_fn(self, features, labels, inference_outputs, mode,
+                     params=None):
+    logit = inference_outputs["logit"]
+    target = labels.reward.reshape(logit.shape)
+    return torch.nn.functional.binary_cross_entropy_with_logits(
+        logit.float(), target.float())
+
+
+def generate_test_tfrecord(path: str, num_records: int = 20,
+                           seed: int = 0) -> str:
+  """Builds the pose_env test fixture (PNG-encoded synthetic episodes)."""
+  from tensor2robot_amd.data import example as example_codec
+  from tensor2robot_amd.data import image_codec
+  from tensor2robot_amd.data import tfrecord
+  env = pose_env.PoseToyEnv(seed=seed)
+  with tfrecord.TFRecordWriter(path) as writer:
+    for _ in range(num_records):
+      obs = env.reset()
+      png = image_codec.encode_png(obs)
+      features = {
+          "state/image": [png],
+          "pose": env.target_pose.astype(np.float32),
+          "reward": np.asarray([1.0], np.float32),
+          "action/pose": env.target_pose.astype(np.float32),
+      }
+      writer.write(example_codec.encode_example(features))
+  return path

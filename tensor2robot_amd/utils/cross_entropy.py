@@ -1,0 +1,73 @@
+"""Cross-entropy method (CEM) optimizer for action selection.
+
+Reference: `utils/cross_entropy.py:30-...` — iterative sample -> evaluate ->
+elite-refit loop over Gaussian action distributions; supports dict or array
+sample batches and early termination.
+"""
+
+from __future__ import annotations
+
+from typing import Callable, Optional, Tuple
+
+import numpy as np
+
+from tensor2robot_amd import gin
+
+
+@gin.configurable
+class CrossEntropyMethod:
+
+  def __init__(self, num_samples: int = 64, num_elites: int = 6,
+               num_iterations: int = 3,
+               early_termination_value: Optional[float] = None,
+               seed: Optional[int] = None):
+    self.num_samples = num_samples
+    self.num_elites = num_elites
+    self.num_iterations = num_iterations
+    self.early_termination_value = early_termination_value
+    self._rng = np.random.RandomState(seed)
+
+  def run(self, objective_fn: Callable[[np.ndarray], np.ndarray],
+          initial_mean: np.ndarray, initial_std: np.ndarray,
+          bounds: Optional[Tuple[np.ndarray, np.ndarray]] = None
+          ) -> Tuple[np.ndarray, float, np.ndarray, np.ndarray]:
+    """Maximizes objective_fn over actions.
+
+    objective_fn: [num_samples, action_dim] -> [num_samples] scores.
+    Returns (best_action, best_score, final_mean, final_std).
+    """
+    mean = np.asarray(initial_mean, np.float32).copy()
+    std = np.asarray(initial_std, np.float32).copy()
+    best_action, best_score = mean.copy(), -np.inf
+    for _ in range(self.num_iterations):
+      samples = self._rng.normal(
+          mean[None, :], std[None, :],
+          size=(self.num_samples, mean.shape[0])).astype(np.float32)
+      if bounds is not None:
+        samples = np.clip(samples, bounds[0], bounds[1])
+      scores = np.asarray(objective_fn(samples)).reshape(-1)
+      order = np.argsort(-scores)
+      elites = samples[order[: self.num_elites]]
+      if scores[order[0]] > best_score:
+        best_score = float(scores[order[0]])
+        best_action = samples[order[0]].copy()
+      mean = elites.mean(axis=0)
+      std = elites.std(axis=0) + 1e-6
+      if self.early_termination_value is not None and \
+          best_score >= self.early_termination_value:
+        break
+    return best_action, best_score, mean, std
+
+
+@gin.configurable
+def cross_entropy_optimize(objective_fn, action_size: int,
+                           num_samples: int = 64, num_elites: int = 6,
+                           num_iterations: int = 3,
+                           initial_std: float = 0.5,
+                           seed: Optional[int] = None):
+  """Functional convenience wrapper."""
+  cem = CrossEntropyMethod(num_samples=num_samples, num_elites=num_elites,
+                           num_iterations=num_iterations, seed=seed)
+  mean = np.zeros(action_size, np.float32)
+  std = np.full(action_size, initial_std, np.float32)
+  return cem.run(objective_fn, mean, std)

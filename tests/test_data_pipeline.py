@@ -1,0 +1,233 @@
+"""Data layer tests: wire codec, TFRecord IO, spec-driven parser, pipeline."""
+
+import numpy as np
+import pytest
+import torch
+
+from tensor2robot_amd.data import example as example_codec
+from tensor2robot_amd.data import image_codec
+from tensor2robot_amd.data import input_generators
+from tensor2robot_amd.data import parser as parser_mod
+from tensor2robot_amd.data import pipeline
+from tensor2robot_amd.data import tfrecord
+from tensor2robot_amd.specs import tensorspec_utils as tsu
+
+T = tsu.ExtendedTensorSpec
+
+
+def test_example_roundtrip():
+  features = {
+      "floats": np.array([1.5, -2.25, 3.0], np.float32),
+      "ints": np.array([1, -7, 1 << 40], np.int64),
+      "bytes": [b"hello", b"world"],
+  }
+  data = example_codec.encode_example(features)
+  decoded = example_codec.decode_example(data)
+  np.testing.assert_allclose(decoded["floats"], features["floats"])
+  np.testing.assert_array_equal(decoded["ints"], features["ints"])
+  assert decoded["bytes"] == features["bytes"]
+
+
+def test_sequence_example_roundtrip():
+  context = {"task": np.array([3], np.int64)}
+  feature_lists = {
+      "obs": [np.array([1.0, 2.0], np.float32),
+              np.array([3.0, 4.0], np.float32)],
+      "act": [np.array([0.5], np.float32)] * 3,
+  }
+  data = example_codec.encode_sequence_example(context, feature_lists)
+  ctx, fl = example_codec.decode_sequence_example(data)
+  np.testing.assert_array_equal(ctx["task"], [3])
+  assert len(fl["obs"]) == 2 and len(fl["act"]) == 3
+  np.testing.assert_allclose(fl["obs"][1], [3.0, 4.0])
+
+
+def test_tfrecord_roundtrip(tmp_path):
+  path = str(tmp_path / "test.tfrecord")
+  records = [b"first", b"second" * 100, b""]
+  with tfrecord.TFRecordWriter(path) as w:
+    for r in records:
+      w.write(r)
+  out = list(tfrecord.read_records(path, verify_crc=True))
+  assert out == records
+
+
+def test_crc32c_known_value():
+  # CRC32C of "123456789" is 0xE3069283 (RFC 3720 test vector).
+  assert tfrecord.crc32c(b"123456789") == 0xE3069283
+
+
+def test_png_roundtrip():
+  rng = np.random.RandomState(0)
+  img = rng.randint(0, 256, (16, 20, 3), np.uint8)
+  data = image_codec.encode_png(img)
+  out = image_codec.decode_png(data)
+  np.testing.assert_array_equal(out, img)
+
+
+def _write_records(tmp_path, records, name="data.tfrecord"):
+  path = str(tmp_path / name)
+  with tfrecord.TFRecordWriter(path) as w:
+    for r in records:
+      w.write(r)
+  return path
+
+
+def test_parser_fixed_and_image(tmp_path):
+  rng = np.random.RandomState(1)
+  imgs = [rng.randint(0, 256, (8, 8, 3), np.uint8) for _ in range(4)]
+  records = []
+  for i, img in enumerate(imgs):
+    records.append(example_codec.encode_example({
+        "img": [image_codec.encode_png(img)],
+        "vec": np.full(3, i, np.float32),
+        "label": np.array([i % 2], np.float32),
+    }))
+  feature_spec = tsu.TensorSpecStruct()
+  feature_spec["image"] = T((8, 8, 3), torch.uint8, name="img",
+                            data_format="PNG")
+  feature_spec["vec"] = T((3,), torch.float32, name="vec")
+  label_spec = tsu.TensorSpecStruct()
+  label_spec["label"] = T((1,), torch.float32, name="label")
+  parse = parser_mod.create_parse_example_fn(feature_spec, label_spec)
+  features, labels = parse(records)
+  assert features["image"].shape == (4, 8, 8, 3)
+  np.testing.assert_array_equal(features["image"][2].numpy(), imgs[2])
+  assert features["vec"].shape == (4, 3)
+  assert labels["label"].shape == (4, 1)
+
+
+def test_parser_empty_image_is_zero():
+  rec = example_codec.encode_example({"img": [b""]})
+  spec = tsu.TensorSpecStruct()
+  spec["image"] = T((8, 8, 3), torch.uint8, name="img", data_format="PNG")
+  parse = parser_mod.create_parse_example_fn(spec)
+  features, _ = parse([rec])
+  assert features["image"].eq(0).all()
+
+
+def test_parser_varlen_pad_clip():
+  records = [
+      example_codec.encode_example(
+          {"v": np.array([1.0, 2.0], np.float32)}),
+      example_codec.encode_example(
+          {"v": np.arange(10, dtype=np.float32)}),
+  ]
+  spec = tsu.TensorSpecStruct()
+  spec["v"] = T((4,), torch.float32, name="v", varlen_default_value=9.0)
+  parse = parser_mod.create_parse_example_fn(spec)
+  features, _ = parse(records)
+  np.testing.assert_allclose(features["v"][0].numpy(), [1, 2, 9, 9])
+  np.testing.assert_allclose(features["v"][1].numpy(), [0, 1, 2, 3])
+
+
+def test_parser_sequence_with_length():
+  records = []
+  for t_len in (2, 4):
+    records.append(example_codec.encode_sequence_example(
+        {"ctx": np.array([1.0], np.float32)},
+        {"obs": [np.array([float(t), 0.0], np.float32)
+                 for t in range(t_len)]}))
+  spec = tsu.TensorSpecStruct()
+  spec["ctx"] = T((1,), torch.float32, name="ctx")
+  spec["obs"] = T((2,), torch.float32, name="obs", is_sequence=True)
+  parse = parser_mod.create_parse_example_fn(spec)
+  features, _ = parse(records)
+  assert features["obs"].shape == (2, 4, 2)  # padded to max len
+  np.testing.assert_array_equal(features["obs_length"].numpy(), [2, 4])
+  assert features["obs"][0, 2].eq(0).all()  # padding
+
+
+def test_parser_bf16_spec_parsed_as_f32():
+  rec = example_codec.encode_example(
+      {"x": np.array([1.0, 2.0], np.float32)})
+  spec = tsu.TensorSpecStruct()
+  spec["x"] = T((2,), torch.bfloat16, name="x")
+  parse = parser_mod.create_parse_example_fn(spec)
+  features, _ = parse([rec])
+  assert features["x"].dtype == torch.bfloat16
+
+
+def test_parser_multi_dataset():
+  rec1 = example_codec.encode_example({"a": np.array([1.0], np.float32)})
+  rec2 = example_codec.encode_example({"b": np.array([2.0], np.float32)})
+  spec = tsu.TensorSpecStruct()
+  spec["a"] = T((1,), torch.float32, name="a", dataset_key="d1")
+  spec["b"] = T((1,), torch.float32, name="b", dataset_key="d2")
+  parse = parser_mod.create_parse_example_fn(spec)
+  features, _ = parse({"d1": [rec1], "d2": [rec2]})
+  assert float(features["a"][0]) == 1.0
+  assert float(features["b"][0]) == 2.0
+
+
+def test_parser_optional_missing_ok():
+  rec = example_codec.encode_example({"x": np.array([1.0], np.float32)})
+  spec = tsu.TensorSpecStruct()
+  spec["x"] = T((1,), torch.float32, name="x")
+  spec["opt"] = T((1,), torch.float32, name="opt", is_optional=True)
+  parse = parser_mod.create_parse_example_fn(spec)
+  features, _ = parse([rec])
+  assert "opt" not in features
+  # Missing REQUIRED raises.
+  spec2 = tsu.TensorSpecStruct()
+  spec2["missing"] = T((1,), torch.float32, name="nope")
+  parse2 = parser_mod.create_parse_example_fn(spec2)
+  with pytest.raises(ValueError):
+    parse2([rec])
+
+
+def test_record_batch_iterator(tmp_path):
+  records = [example_codec.encode_example(
+      {"x": np.array([float(i)], np.float32)}) for i in range(10)]
+  path = _write_records(tmp_path, records)
+  it = pipeline.RecordBatchIterator({"": [path]}, batch_size=4,
+                                    shuffle=False, repeat=False)
+  batches = list(it)
+  assert len(batches) == 2  # drop_remainder
+  assert len(batches[0][""]) == 4
+
+
+def test_record_input_generator_end_to_end(tmp_path):
+  records = [example_codec.encode_example({
+      "x": np.array([float(i), 0.0], np.float32),
+      "y": np.array([1.0], np.float32),
+  }) for i in range(8)]
+  path = _write_records(tmp_path, records)
+  gen = input_generators.DefaultRecordInputGenerator(
+      file_patterns=path, batch_size=2, seed=0)
+  feature_spec = tsu.TensorSpecStruct()
+  feature_spec["x"] = T((2,), torch.float32, name="x")
+  label_spec = tsu.TensorSpecStruct()
+  label_spec["y"] = T((1,), torch.float32, name="y")
+  gen.set_feature_specifications(feature_spec)
+  gen.set_label_specifications(label_spec)
+  it = gen._iterate("eval")
+  features, labels = next(it)
+  assert features["x"].shape == (2, 2)
+  assert labels["y"].shape == (2, 1)
+
+
+def test_weighted_generator(tmp_path):
+  p1 = _write_records(tmp_path, [example_codec.encode_example(
+      {"x": np.array([0.0], np.float32)})] * 5, "a.tfrecord")
+  p2 = _write_records(tmp_path, [example_codec.encode_example(
+      {"x": np.array([1.0], np.float32)})] * 5, "b.tfrecord")
+  gen = input_generators.WeightedRecordInputGenerator(
+      file_patterns=[p1, p2], weights=[0.9, 0.1], batch_size=16, seed=1)
+  spec = tsu.TensorSpecStruct()
+  spec["x"] = T((1,), torch.float32, name="x")
+  gen.set_feature_specifications(spec)
+  gen.set_label_specifications(tsu.TensorSpecStruct())
+  features, _ = next(gen._iterate("train"))
+  frac_zero = float((features["x"] == 0).float().mean())
+  assert frac_zero > 0.5  # heavily weighted toward dataset a
+
+
+def test_prefetch_iterator_propagates_errors():
+  def bad_source():
+    yield 1
+    raise RuntimeError("boom")
+
+  it = pipeline.PrefetchIterator(bad_source, depth=2)
+  with pytest.raises(RuntimeError):
+    list(it)

@@ -1,0 +1,235 @@
+"""Predictor / policy / collect-eval-loop tests.
+
+Mirrors the reference's 'distributed tests without a cluster' pattern
+(SURVEY §4.8): train -> export -> predictor restore -> policy -> env loop,
+all in-process through the filesystem contract.
+"""
+
+import os
+import threading
+import time
+
+import numpy as np
+import pytest
+import torch
+
+from tensor2robot_amd.data import input_generators
+from tensor2robot_amd.models import optimizers
+from tensor2robot_amd.policies import policies
+from tensor2robot_amd.predictors import checkpoint_predictor
+from tensor2robot_amd.predictors import ensemble_predictor
+from tensor2robot_amd.predictors import exported_savedmodel_predictor as esp
+from tensor2robot_amd.research.pose_env import pose_env
+from tensor2robot_amd.research.pose_env import pose_env_models
+from tensor2robot_amd.research.dql_grasping_lib import run_env as run_env_mod
+from tensor2robot_amd.train import train_eval
+from tensor2robot_amd.utils import continuous_collect_eval
+from tensor2robot_amd.utils import cross_entropy
+from tensor2robot_amd.utils import mocks
+from tensor2robot_amd.utils import writer as writer_mod
+from tensor2robot_amd.data import example as example_codec
+
+
+def _train_and_export(tmp_path, steps=50):
+  model = mocks.MockT2RModel(
+      device_type="cpu",
+      create_optimizer_fn=lambda: optimizers.create_adam_optimizer(5e-2))
+  train_gen = mocks.MockInputGenerator(batch_size=16)
+  eval_gen = mocks.MockInputGenerator(batch_size=16, seed=5)
+  train_eval.train_eval_model(
+      t2r_model=model, input_generator_train=train_gen,
+      input_generator_eval=eval_gen, max_train_steps=steps, eval_steps=4,
+      model_dir=str(tmp_path),
+      create_exporters_fn=train_eval.create_default_exporters)
+  export_dir = os.path.join(tmp_path, "export", "latest_exporter_numpy")
+  return model, export_dir
+
+
+def test_exported_predictor_roundtrip(tmp_path):
+  model, export_dir = _train_and_export(tmp_path)
+  predictor = esp.ExportedSavedModelPredictor(export_dir, timeout=5)
+  assert predictor.is_loaded
+  assert predictor.global_step == 50
+  spec = predictor.get_feature_specification()
+  assert "measured_position" in spec
+  out = predictor.predict(
+      {"measured_position": np.ones((3, 3), np.float32)})
+  assert out["prediction"].shape == (3, 1)
+  # Predictions are sign-consistent with the learned rule (sum > 0 -> 1).
+  pos = predictor.predict(
+      {"measured_position": np.full((2, 3), 0.9, np.float32)})
+  neg = predictor.predict(
+      {"measured_position": np.full((2, 3), -0.9, np.float32)})
+  assert pos["prediction"].mean() > 0.5
+  assert neg["prediction"].mean() < 0.5
+
+
+def test_exported_predictor_times_out_gracefully(tmp_path):
+  predictor = esp.ExportedSavedModelPredictor(
+      str(tmp_path / "nothing"), timeout=0.5,
+      restore_model_option=esp.RestoreOptions.DO_NOT_RESTORE)
+  assert not predictor.restore()
+  assert not predictor.is_loaded
+
+
+def test_exported_predictor_async_restore(tmp_path):
+  model, export_dir = _train_and_export(tmp_path, steps=10)
+  predictor = esp.ExportedSavedModelPredictor(
+      export_dir, timeout=10,
+      restore_model_option=esp.RestoreOptions.RESTORE_ASYNCHRONOUSLY)
+  assert predictor.is_loaded  # join guard waits for the thread
+
+
+def test_checkpoint_predictor(tmp_path):
+  model, _ = _train_and_export(tmp_path, steps=20)
+  model2 = mocks.MockT2RModel(device_type="cpu")
+  predictor = checkpoint_predictor.CheckpointPredictor(
+      t2r_model=model2, checkpoint_dir=str(tmp_path), timeout=5)
+  assert predictor.restore()
+  assert predictor.global_step == 20
+  out = predictor.predict(
+      {"measured_position": np.ones((2, 3), np.float32)})
+  assert "prediction" in out
+
+
+def test_checkpoint_predictor_init_randomly():
+  model = mocks.MockT2RModel(device_type="cpu")
+  predictor = checkpoint_predictor.CheckpointPredictor(
+      t2r_model=model, checkpoint_dir=None)
+  predictor.init_randomly()
+  out = predictor.predict(
+      {"measured_position": np.zeros((1, 3), np.float32)})
+  assert out["logit"].shape == (1, 1)
+
+
+def test_ensemble_predictor(tmp_path):
+  _, export_dir = _train_and_export(tmp_path, steps=10)
+  predictor = ensemble_predictor.EnsembleExportedSavedModelPredictor(
+      export_dirs=f"{export_dir},{export_dir}", ensemble_size=2,
+      timeout=5, seed=0)
+  assert predictor.restore()
+  out = predictor.predict(
+      {"measured_position": np.zeros((2, 3), np.float32)})
+  assert out["prediction"].shape == (2, 1)
+
+
+def test_cem_optimizer_finds_maximum():
+  cem = cross_entropy.CrossEntropyMethod(num_samples=128, num_elites=12,
+                                         num_iterations=10, seed=0)
+  target = np.array([0.3, -0.7], np.float32)
+
+  def objective(samples):
+    return -np.sum((samples - target) ** 2, axis=1)
+
+  best, score, mean, std = cem.run(objective, np.zeros(2, np.float32),
+                                   np.ones(2, np.float32))
+  assert np.linalg.norm(best - target) < 0.15
+
+
+def test_regression_policy_env_loop(tmp_path):
+  """Full predictor -> policy -> env -> replay-writer loop on pose_env."""
+  model = pose_env_models.PoseEnvRegressionModel(
+      create_optimizer_fn=lambda: optimizers.create_adam_optimizer(1e-3))
+  predictor = checkpoint_predictor.CheckpointPredictor(
+      t2r_model=model, checkpoint_dir=None)
+  predictor.init_randomly()
+  policy = policies.RegressionPolicy(predictor=predictor,
+                                     state_key="state/image")
+  env = pose_env.PoseToyEnv(seed=1)
+  replay_writer = writer_mod.TFRecordReplayWriter()
+
+  def episode_to_transitions(episode_data):
+    out = []
+    for t in episode_data:
+      out.append(example_codec.encode_example({
+          "state/image": [b""],
+          "action/pose": np.asarray(t.action, np.float32),
+          "reward": np.asarray([t.reward], np.float32),
+      }))
+    return out
+
+  rewards = run_env_mod.run_env(
+      env, policy=policy, replay_writer=replay_writer,
+      episode_to_transitions_fn=episode_to_transitions,
+      root_dir=str(tmp_path), num_episodes=2, max_episode_steps=3)
+  assert len(rewards) == 2
+  collect_dir = os.path.join(tmp_path, "policy_collect")
+  files = os.listdir(collect_dir)
+  assert any(f.endswith(".tfrecord") for f in files)
+
+
+def test_cem_policy_with_critic():
+  # action_batch_size expands the serving action spec to [S, d] (CEM tiling).
+  model = pose_env_models.PoseEnvContinuousMCModel(action_batch_size=16)
+  predictor = checkpoint_predictor.CheckpointPredictor(
+      t2r_model=model, checkpoint_dir=None)
+  predictor.init_randomly()
+
+  class PoseCEMPolicy(policies.CEMPolicy):
+
+    def _split_action(self, action):
+      return {"action/pose": action.astype(np.float32)}
+
+  policy = PoseCEMPolicy(predictor=predictor, action_size=2,
+                         cem_samples=16, cem_iterations=2,
+                         state_key="state/image", seed=0)
+  state = pose_env.PoseToyEnv(seed=3).reset()  # uint8 render (env contract)
+  action = policy.SelectAction(state)
+  assert action.shape == (2,)
+
+
+def test_collect_eval_loop(tmp_path):
+  """Actor loop polls exports produced by a trainer (filesystem contract)."""
+  _, export_dir = _train_and_export(tmp_path, steps=10)
+
+  calls = []
+
+  class _StubPolicy:
+
+    def restore(self):
+      return True
+
+    def init_randomly(self):
+      pass
+
+    @property
+    def global_step(self):
+      return 10 + len(calls)
+
+    def reset(self):
+      pass
+
+    def sample_action(self, obs, explore_prob=0.0):
+      return np.zeros(2, np.float32), {}
+
+  def run_agent_fn(env, policy=None, global_step=0, root_dir="",
+                   num_episodes=1, tag=""):
+    calls.append((tag, global_step))
+
+  result = continuous_collect_eval.collect_eval_loop(
+      collect_env=object(), eval_env=object(),
+      policy_class=_StubPolicy, run_agent_fn=run_agent_fn,
+      root_dir=str(tmp_path), max_steps=10, poll_sleep_secs=0.01,
+      max_loops=5)
+  assert result >= 10
+  assert ("collect", 10) in calls and ("eval", 10) in calls
+
+
+def test_exploration_policies():
+  model = pose_env_models.PoseEnvRegressionModel()
+  predictor = checkpoint_predictor.CheckpointPredictor(
+      t2r_model=model, checkpoint_dir=None)
+  predictor.init_randomly()
+  state = pose_env.PoseToyEnv(seed=4).reset()
+  base = policies.RegressionPolicy(predictor=predictor,
+                                   state_key="state/image")
+  a0 = base.SelectAction(state)
+  ou = policies.OUExploreRegressionPolicy(
+      predictor=predictor, state_key="state/image", seed=0)
+  ou.reset()
+  a1 = ou.SelectAction(state)
+  assert a1.shape == a0.shape and not np.allclose(a0, a1)
+  sched = policies.ScheduledExplorationRegressionPolicy(
+      predictor=predictor, state_key="state/image", seed=0)
+  a2 = sched.SelectAction(state)
+  assert a2.shape == a0.shape
