@@ -1,0 +1,264 @@
+"""Task-Embedded Control (TEC) embedding networks and contrastive losses.
+
+Reference `layers/tec.py`: embed_fullstate :30 (FC stack + layer-norm),
+embed_condition_images :61 (Berkeley-Net embedding + optional FC/1x1
+stack), reduce_temporal_embeddings :114 (temporal conv / mean reduce ->
+FC), compute_embedding_contrastive_loss :173 (modes default /
+both_directions / reverse_direction / cross_entropy / triplet
+:212-258), cosine triplet-semihard machinery :260-383.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Sequence, Tuple
+
+import torch
+from torch import nn
+import torch.nn.functional as F
+
+from tensor2robot_amd import gin
+from tensor2robot_amd.layers import vision_layers
+
+
+class EmbedFullstate(nn.Module):
+  """FC stack (relu + layer norm) -> linear embed (reference :30-55)."""
+
+  def __init__(self, in_dim: int, embed_size: int,
+               fc_layers: Sequence[int] = (100,)):
+    super().__init__()
+    layers = []
+    d = in_dim
+    for width in fc_layers:
+      layers += [nn.Linear(d, width), nn.LayerNorm(width), nn.ReLU()]
+      d = width
+    self.stack = nn.Sequential(*layers)
+    self.head = nn.Linear(d, embed_size)
+
+  def forward(self, fullstate: torch.Tensor) -> torch.Tensor:
+    return self.head(self.stack(fullstate))
+
+
+@gin.configurable
+class EmbedConditionImages(nn.Module):
+  """Berkeley-Net image embedding + optional FC stack (reference :61-109).
+
+  With use_spatial_softmax the torso yields [N, 64] feature points and
+  fc_layers (if given) are fully-connected; without, the torso yields a
+  spatial map and fc_layers become 1x1 convs.
+  """
+
+  def __init__(self, in_channels: int = 3,
+               fc_layers: Optional[Sequence[int]] = None,
+               use_spatial_softmax: bool = True):
+    super().__init__()
+    self.use_spatial_softmax = use_spatial_softmax
+    self.torso = vision_layers.ImagesToFeaturesNet(
+        in_channels=in_channels, use_spatial_softmax=use_spatial_softmax)
+    self.stack = None
+    self.head = None
+    if fc_layers is not None:
+      d = 64 if use_spatial_softmax else 32
+      layers = []
+      for width in fc_layers[:-1]:
+        if use_spatial_softmax:
+          layers += [nn.Linear(d, width), nn.LayerNorm(width), nn.ReLU()]
+        else:
+          layers += [nn.Conv2d(d, width, 1), nn.GroupNorm(1, width),
+                     nn.ReLU()]
+        d = width
+      self.stack = nn.Sequential(*layers)
+      self.head = (nn.Linear(d, fc_layers[-1]) if use_spatial_softmax
+                   else nn.Conv2d(d, fc_layers[-1], 1))
+    self.out_dim = (fc_layers[-1] if fc_layers is not None
+                    else (64 if use_spatial_softmax else 32))
+
+  def forward(self, condition_image: torch.Tensor) -> torch.Tensor:
+    if condition_image.dim() != 4:
+      raise ValueError(
+          f"Image has unexpected shape {tuple(condition_image.shape)}")
+    embedding, _ = self.torso(condition_image)
+    if self.stack is not None:
+      embedding = self.head(self.stack(embedding))
+    return embedding
+
+
+@gin.configurable
+class ReduceTemporalEmbeddings(nn.Module):
+  """Reduce [N, T, F] over time (reference :114-170).
+
+  combine_mode: 'temporal_conv' (conv1d k=10 no-bias + layer norm, then
+  flatten), 'temporal_conv_avg_after' (then mean over time), anything
+  else = plain time-mean.  Followed by an FC stack and linear head.
+  A rank-5 input [N, T, H, W, F] is spatially mean-pooled first.
+  """
+
+  def __init__(self, in_dim: int, output_size: int, time_dim: int,
+               conv1d_layers: Optional[Sequence[int]] = (64,),
+               fc_hidden_layers: Sequence[int] = (100,),
+               combine_mode: str = "temporal_conv"):
+    super().__init__()
+    self.combine_mode = combine_mode
+    self.convs = None
+    d, t = in_dim, time_dim
+    if "temporal_conv" in combine_mode and conv1d_layers is not None:
+      convs = []
+      for width in conv1d_layers:
+        convs.append(nn.Conv1d(d, width, 10, bias=False))
+        convs.append(nn.GroupNorm(1, width))  # layer norm over (C, T)
+        d = width
+        t = t - 9  # VALID conv shrinks time
+      self.convs = nn.ModuleList(convs)
+    if "temporal_conv" not in self.combine_mode:
+      flat_dim = in_dim
+    elif combine_mode == "temporal_conv_avg_after":
+      flat_dim = d
+    else:
+      if t <= 0:
+        raise ValueError(
+            f"time_dim {time_dim} too short for conv1d stack")
+      flat_dim = d * t
+    layers = []
+    d = flat_dim
+    for width in fc_hidden_layers:
+      layers += [nn.Linear(d, width), nn.LayerNorm(width), nn.ReLU()]
+      d = width
+    self.stack = nn.Sequential(*layers)
+    self.head = nn.Linear(d, output_size)
+
+  def forward(self, temporal_embedding: torch.Tensor) -> torch.Tensor:
+    if temporal_embedding.dim() == 5:
+      temporal_embedding = temporal_embedding.mean(dim=(2, 3))
+    if temporal_embedding.dim() != 3:
+      raise ValueError("Temporal embedding has unexpected shape "
+                       f"{tuple(temporal_embedding.shape)}")
+    x = temporal_embedding
+    if "temporal_conv" not in self.combine_mode:
+      x = x.mean(dim=1)
+    else:
+      if self.convs is not None:
+        y = x.transpose(1, 2)  # [N, F, T]
+        for m in self.convs:
+          y = F.relu(m(y)) if isinstance(m, nn.Conv1d) else m(y)
+        x = y.transpose(1, 2)
+      if self.combine_mode == "temporal_conv_avg_after":
+        x = x.mean(dim=1)
+      else:
+        x = x.flatten(1)
+    return self.head(self.stack(x))
+
+
+def contrastive_loss(labels: torch.Tensor, anchor: torch.Tensor,
+                     embeddings: torch.Tensor,
+                     margin: float = 1.0) -> torch.Tensor:
+  """TF-slim metric_learning.contrastive_loss equivalent.
+
+  labels [T] bool/float, anchor [1, K], embeddings [T, K]:
+  mean(y * d^2 + (1-y) * max(margin - d, 0)^2), d = euclidean distance.
+  """
+  y = labels.float()
+  d = torch.sqrt(((anchor - embeddings) ** 2).sum(dim=1) + 1e-12)
+  return (y * d ** 2 + (1.0 - y) * F.relu(margin - d) ** 2).mean()
+
+
+def masked_maximum(data: torch.Tensor, mask: torch.Tensor,
+                   dim: int = 1) -> torch.Tensor:
+  """Max over masked elements, stable under empty masks (reference :260)."""
+  axis_min = data.min(dim, keepdim=True).values
+  return ((data - axis_min) * mask).max(dim, keepdim=True).values + axis_min
+
+
+def masked_minimum(data: torch.Tensor, mask: torch.Tensor,
+                   dim: int = 1) -> torch.Tensor:
+  """Min over masked elements (reference :279)."""
+  axis_max = data.max(dim, keepdim=True).values
+  return ((data - axis_max) * mask).min(dim, keepdim=True).values + axis_max
+
+
+def cosine_pairwise_distance(feature: torch.Tensor) -> torch.Tensor:
+  """1 - cosine similarity matrix, zeroed diagonal (reference :298)."""
+  sim = feature @ feature.t()
+  dist = 1.0 - sim
+  return dist * (1.0 - torch.eye(feature.shape[0], device=feature.device,
+                                 dtype=dist.dtype))
+
+
+def cosine_triplet_semihard_loss(labels: torch.Tensor,
+                                 embeddings: torch.Tensor,
+                                 margin: float = 1.0) -> torch.Tensor:
+  """Triplet semi-hard loss with cosine distance (reference :322-383)."""
+  labels = labels.reshape(-1, 1)
+  batch_size = labels.shape[0]
+  pdist = cosine_pairwise_distance(embeddings)
+  adjacency = labels == labels.t()
+  adjacency_not = ~adjacency
+
+  pdist_tile = pdist.repeat(batch_size, 1)
+  mask = adjacency_not.repeat(batch_size, 1) & (
+      pdist_tile > pdist.t().reshape(-1, 1))
+  mask_final = (mask.float().sum(1, keepdim=True) > 0.0).reshape(
+      batch_size, batch_size).t()
+
+  adjacency_not_f = adjacency_not.float()
+  mask_f = mask.float()
+
+  negatives_outside = masked_minimum(pdist_tile, mask_f).reshape(
+      batch_size, batch_size).t()
+  negatives_inside = masked_maximum(pdist, adjacency_not_f).expand(
+      -1, batch_size)
+  semi_hard_negatives = torch.where(mask_final, negatives_outside,
+                                    negatives_inside)
+  loss_mat = margin + pdist - semi_hard_negatives
+
+  mask_positives = adjacency.float() - torch.eye(
+      batch_size, device=embeddings.device)
+  num_positives = mask_positives.sum().clamp(min=1.0)
+  return (loss_mat * mask_positives).clamp(min=0.0).sum() / num_positives
+
+
+@gin.configurable
+def compute_embedding_contrastive_loss(
+    inf_embedding: torch.Tensor, con_embedding: torch.Tensor,
+    positives: Optional[torch.Tensor] = None,
+    contrastive_loss_mode: str = "both_directions") -> torch.Tensor:
+  """Contrastive loss between inference and condition embeddings.
+
+  Reference :173-258.  Embeddings are [num_tasks, num_episodes, K],
+  expected L2-normalized; task 0 is the anchor unless `positives` gives
+  explicit labels.
+  """
+  if inf_embedding.dim() != 3 or con_embedding.dim() != 3:
+    raise ValueError("Embeddings must be rank 3")
+  avg_inf = inf_embedding.mean(dim=1)
+  avg_con = con_embedding.mean(dim=1)
+  anchor = avg_inf[0:1]
+  if positives is not None and contrastive_loss_mode != "triplet":
+    labels = positives
+  else:
+    labels = torch.arange(avg_con.shape[0],
+                          device=avg_con.device) == 0
+  if contrastive_loss_mode == "default":
+    return contrastive_loss(labels, anchor, avg_con)
+  if contrastive_loss_mode == "both_directions":
+    anchor_cond = avg_con[0:1]
+    return (contrastive_loss(labels, anchor, avg_con) +
+            contrastive_loss(labels, anchor_cond, avg_inf))
+  if contrastive_loss_mode == "reverse_direction":
+    anchor_cond = avg_con[0:1]
+    return contrastive_loss(labels, anchor_cond, avg_inf)
+  if contrastive_loss_mode == "cross_entropy":
+    temperature = 2.0
+    anchor_cond = avg_con[0:1]
+    y = labels.float()
+    loss1 = F.binary_cross_entropy_with_logits(
+        temperature * (anchor * avg_con).sum(dim=1), y)
+    loss2 = F.binary_cross_entropy_with_logits(
+        temperature * (anchor_cond * avg_inf).sum(dim=1), y)
+    return loss1 + loss2
+  if contrastive_loss_mode == "triplet":
+    if positives is None:
+      positives = torch.arange(avg_inf.shape[0], device=avg_inf.device)
+    labels = positives.repeat(2)
+    embeds = torch.cat([avg_inf, avg_con], dim=0)
+    return cosine_triplet_semihard_loss(labels, embeds, margin=1.0)
+  raise ValueError("Did not understand contrastive_loss_mode "
+                   f"{contrastive_loss_mode!r}")

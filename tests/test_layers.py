@@ -1,0 +1,303 @@
+"""CPU unit tests for the NN layer library (SURVEY §2.5 parity).
+
+Mirrors the reference's layer tests (resnet_test, spatial_softmax,
+mdn_test, snail_test, tec) with torch-native checks.
+"""
+
+import math
+
+import pytest
+import torch
+
+from tensor2robot_amd.layers import bcz_networks
+from tensor2robot_amd.layers import mdn
+from tensor2robot_amd.layers import resnet
+from tensor2robot_amd.layers import snail
+from tensor2robot_amd.layers import spatial_softmax
+from tensor2robot_amd.layers import tec
+from tensor2robot_amd.layers import vision_layers
+
+
+# ---------------------------------------------------------------- resnet
+@pytest.mark.parametrize("size,version", [(18, 1), (18, 2), (50, 2)])
+def test_resnet_shapes_and_endpoints(size, version):
+  net = resnet.ResNet(resnet_size=size, num_classes=10, version=version)
+  x = torch.randn(2, 3, 64, 64)
+  out, endpoints = net(x)
+  assert out.shape == (2, 10)
+  for key in ["initial_conv", "initial_max_pool", "block_layer1",
+              "block_layer2", "block_layer3", "block_layer4",
+              "pre_final_pool", "final_reduce_mean", "final_dense"]:
+    assert key in endpoints, key
+  expansion = 4 if size >= 50 else 1
+  assert endpoints["pre_final_pool"].shape[1] == 512 * expansion
+  # 64 input: /2 stem, /2 pool, /2 x3 stages -> 2x2 spatial.
+  assert endpoints["pre_final_pool"].shape[-2:] == (2, 2)
+
+
+def test_resnet_film_changes_output():
+  torch.manual_seed(0)
+  net = resnet.ResNet(resnet_size=18, num_classes=4, version=2).eval()
+  gen = resnet.LinearFiLMGenerator(embedding_dim=6, resnet=net)
+  x = torch.randn(2, 3, 32, 32)
+  emb = torch.randn(2, 6)
+  gbs = gen(emb)
+  assert len(gbs) == sum(net.blocks_per_layer)
+  for gb, width in zip(gbs, net.film_channels):
+    assert gb.shape == (2, width)
+  out_film, _ = net(x, film_gamma_betas=gbs)
+  out_plain, _ = net(x)
+  assert not torch.allclose(out_film, out_plain)
+  # Zero embedding -> zero gamma/beta bias-free? (linear has bias, so just
+  # check determinism instead)
+  out_film2, _ = net(x, film_gamma_betas=gen(emb))
+  torch.testing.assert_close(out_film, out_film2)
+
+
+def test_resnet_film_disabled_layers():
+  net = resnet.ResNet(resnet_size=18, num_classes=0)
+  gen = resnet.LinearFiLMGenerator(
+      embedding_dim=4, resnet=net,
+      enabled_block_layers=[True, False, True, False])
+  gbs = gen(torch.randn(3, 4))
+  assert len(gbs) == 8  # 2+2+2+2 blocks
+  assert gbs[2] is None and gbs[3] is None  # block_layer2 disabled
+  assert gbs[6] is None and gbs[7] is None  # block_layer4 disabled
+  out, _ = net(torch.randn(3, 3, 32, 32), film_gamma_betas=gbs)
+  assert out.shape == (3, 512)
+
+
+def test_resnet_warm_start_skips_head(tmp_path):
+  net = resnet.ResNet(resnet_size=18, num_classes=5)
+  path = str(tmp_path / "ckpt.pt")
+  torch.save(net.state_dict(), path)
+  net2 = resnet.ResNet(resnet_size=18, num_classes=7)  # different head
+  init_fn = resnet.resnet_init_from_checkpoint_fn(path, skip_dense=True)
+  report = init_fn(net2)
+  assert any(k.startswith("final_dense") for k in report["missing"])
+  torch.testing.assert_close(net2.initial_conv.weight,
+                             net.initial_conv.weight)
+
+
+# -------------------------------------------------------- spatial softmax
+def test_spatial_softmax_peak_location():
+  # A single hot pixel must give back its own normalized coordinates.
+  n, c, h, w = 1, 2, 5, 7
+  feat = torch.zeros(n, c, h, w)
+  feat[0, 0, 0, 0] = 50.0     # top-left -> (-1, -1)
+  feat[0, 1, 4, 6] = 50.0     # bottom-right -> (+1, +1)
+  points, softmax = spatial_softmax.SpatialSoftmax()(feat)
+  assert points.shape == (1, 4)
+  assert softmax.shape == (1, 2, 5, 7)
+  # Layout: [x0, y0, x1, y1] (channel-major (x, y) pairs, reference
+  # spatial_softmax.py:83-86 reshape semantics).
+  torch.testing.assert_close(points[0],
+                             torch.tensor([-1.0, -1.0, 1.0, 1.0]),
+                             atol=1e-3, rtol=0)
+
+
+def test_spatial_softmax_uniform_center():
+  feat = torch.zeros(2, 3, 9, 9)
+  points, _ = spatial_softmax.SpatialSoftmax()(feat)
+  torch.testing.assert_close(points, torch.zeros(2, 6), atol=1e-5, rtol=0)
+
+
+# ------------------------------------------------------------------ mdn
+def test_mdn_distribution_log_prob_matches_manual():
+  torch.manual_seed(0)
+  a, s = 3, 2
+  params = torch.randn(4, a + 2 * a * s)
+  dist = mdn.get_mixture_distribution(params, a, s)
+  x = torch.randn(4, s)
+  lp = dist.log_prob(x)
+  # Manual: logsumexp over components of log w_i + N(x; mu_i, sigma_i).
+  logits = params[:, :a]
+  mus = params[:, a:a + a * s].reshape(4, a, s)
+  sigmas = torch.nn.functional.softplus(
+      params[:, a + a * s:]).reshape(4, a, s) + mdn.SIGMA_EPS
+  comp = -0.5 * (((x[:, None] - mus) / sigmas) ** 2
+                 + 2 * sigmas.log() + math.log(2 * math.pi)).sum(-1)
+  expected = torch.logsumexp(torch.log_softmax(logits, -1) + comp, -1)
+  torch.testing.assert_close(lp, expected, rtol=1e-5, atol=1e-5)
+
+
+def test_mdn_head_and_decoder():
+  torch.manual_seed(0)
+  dec = mdn.MDNDecoder(in_dim=8, action_size=3, num_mixture=4)
+  x = torch.randn(5, 8)
+  action = dec(x)
+  assert action.shape == (5, 3)
+  loss = dec.loss(torch.randn(5, 3))
+  assert loss.dim() == 0 and torch.isfinite(loss)
+  loss.backward()
+  assert dec.head.fc.weight.grad is not None
+  # Unconditioned sigmas are free parameters.
+  assert dec.head.sigma_param.grad is not None
+
+
+def test_mdn_approximate_mode_picks_top_component():
+  logits = torch.tensor([[10.0, -10.0]])
+  mus = torch.tensor([[[1.0, 2.0], [5.0, 6.0]]])
+  sigmas = torch.ones(1, 2, 2)
+  dist = mdn.GaussianMixture(logits, mus, sigmas)
+  torch.testing.assert_close(dist.approximate_mode(),
+                             torch.tensor([[1.0, 2.0]]))
+
+
+# ---------------------------------------------------------------- snail
+def test_causal_conv_is_causal():
+  torch.manual_seed(0)
+  net = snail.CausalConv(4, 8, kernel_size=2, dilation=2)
+  x = torch.randn(2, 10, 4)
+  y1 = net(x)
+  x2 = x.clone()
+  x2[:, 7:] += 100.0  # perturb the future
+  y2 = net(x2)
+  torch.testing.assert_close(y1[:, :7], y2[:, :7])
+  assert y1.shape == (2, 10, 8)
+
+
+def test_attention_block_is_causal():
+  torch.manual_seed(0)
+  net = snail.AttentionBlock(4, key_size=8, value_size=8)
+  x = torch.randn(1, 6, 4)
+  y1 = net(x)
+  x2 = x.clone()
+  x2[:, 4:] += 10.0
+  y2 = net(x2)
+  torch.testing.assert_close(y1[:, :4], y2[:, :4])
+  assert y1.shape == (1, 6, 12)
+
+
+def test_tc_block_receptive_field():
+  net = snail.TCBlock(3, sequence_length=8, filters=5)
+  # ceil(log2(8)) = 3 dense blocks, each adds `filters` channels.
+  assert net.out_dim == 3 + 3 * 5
+  y = net(torch.randn(2, 8, 3))
+  assert y.shape == (2, 8, 18)
+
+
+def test_snail_net_runs():
+  net = snail.SNAILNet(in_dim=6, sequence_length=8, out_dim=4)
+  y = net(torch.randn(2, 8, 6))
+  assert y.shape == (2, 8, 4)
+
+
+# -------------------------------------------------------- vision layers
+def test_images_to_features_shapes():
+  net = vision_layers.ImagesToFeaturesNet(num_output_maps=16)
+  points, extra = net(torch.rand(2, 3, 64, 64))
+  assert points.shape == (2, 32)
+  assert "softmax" in extra
+
+
+def test_images_to_features_film():
+  torch.manual_seed(0)
+  net = vision_layers.ImagesToFeaturesNet().eval()
+  film = vision_layers.FiLMParams(embedding_dim=5)
+  x = torch.rand(2, 3, 64, 64)
+  p_plain, _ = net(x)
+  p_film, _ = net(x, film_params=film(torch.randn(2, 5)))
+  assert p_plain.shape == p_film.shape
+  assert not torch.allclose(p_plain, p_film)
+  with pytest.raises(ValueError):
+    net(x, film_params=torch.zeros(2, 7))
+
+
+def test_images_to_features_no_softmax_returns_map():
+  net = vision_layers.ImagesToFeaturesNet(use_spatial_softmax=False,
+                                          num_output_maps=8)
+  fmap, extra = net(torch.rand(1, 3, 64, 64))
+  assert fmap.dim() == 4 and fmap.shape[1] == 8
+  assert extra == {}
+
+
+def test_high_res_torso():
+  net = vision_layers.ImagesToFeaturesNetHighRes(num_blocks=3)
+  points, extra = net(torch.rand(2, 3, 128, 128))
+  assert points.shape == (2, 64)
+
+
+def test_pose_net_aux_and_bias_transform():
+  net = vision_layers.ImageFeaturesToPoseNet(
+      feature_dim=64, num_outputs=7, aux_input_dim=5, aux_output_dim=3,
+      bias_transform_size=10)
+  pose, aux = net(torch.randn(4, 64), aux_input=torch.randn(4, 5))
+  assert pose.shape == (4, 7)
+  assert aux.shape == (4, 3)
+  assert net.bias_transform.requires_grad
+
+
+# --------------------------------------------------------- bcz networks
+def test_multihead_mlp_shapes_and_stop_gradient():
+  torch.manual_seed(0)
+  net = bcz_networks.MultiHeadMLP(in_dim=16, action_sizes=[3, 4],
+                                  num_waypoints=5, fc_layers=[32])
+  net.train()
+  x = torch.randn(2, 16, requires_grad=True)
+  outs = net(x)
+  assert [tuple(o.shape) for o in outs] == [(2, 5, 3), (2, 5, 4)]
+  # Gradient of the future waypoints must not reach the torso input.
+  outs[0][:, 1:].sum().backward(retain_graph=True)
+  assert x.grad is None or torch.all(x.grad == 0)
+  outs[0][:, :1].sum().backward()
+  assert x.grad is not None and torch.any(x.grad != 0)
+
+
+def test_conv_lstm_and_snail_policy():
+  torso = bcz_networks.SpatialSoftmaxTorso(aux_dim=2)
+  net = bcz_networks.ConvLSTMNet(torso, lstm_num_units=16, output_size=7)
+  image = torch.rand(2, 4, 3, 64, 64)
+  aux = torch.rand(2, 4, 2)
+  out, eps = net(image, aux)
+  assert out.shape == (2, 4, 7)
+  assert "feature_points" in eps
+
+  torso2 = bcz_networks.SpatialSoftmaxTorso(aux_dim=0)
+  policy = bcz_networks.SNAILPolicyNet(
+      torso2, output_size=7, condition_sequence_length=2,
+      inference_sequence_length=2)
+  out2, _ = policy(image)
+  assert out2.shape == (2, 4, 7)
+
+
+# ------------------------------------------------------------------ tec
+def test_contrastive_loss_manual():
+  labels = torch.tensor([1.0, 0.0])
+  anchor = torch.zeros(1, 2)
+  emb = torch.tensor([[3.0, 4.0], [0.3, 0.4]])  # d = 5, 0.5
+  loss = tec.contrastive_loss(labels, anchor, emb, margin=1.0)
+  expected = (25.0 + 0.25) / 2  # y*d^2 + (1-y)*max(1-d,0)^2
+  assert abs(loss.item() - expected) < 1e-4
+
+
+@pytest.mark.parametrize("mode", ["default", "both_directions",
+                                  "reverse_direction", "cross_entropy",
+                                  "triplet"])
+def test_embedding_contrastive_loss_modes(mode):
+  torch.manual_seed(0)
+  inf = torch.nn.functional.normalize(torch.randn(4, 2, 8), dim=-1)
+  con = torch.nn.functional.normalize(torch.randn(4, 3, 8), dim=-1)
+  loss = tec.compute_embedding_contrastive_loss(
+      inf, con, contrastive_loss_mode=mode)
+  assert loss.dim() == 0 and torch.isfinite(loss)
+
+
+def test_reduce_temporal_embeddings_modes():
+  x = torch.randn(2, 20, 16)
+  for mode, _ in [("temporal_conv", None), ("temporal_conv_avg_after", None),
+                  ("mean", None)]:
+    net = tec.ReduceTemporalEmbeddings(16, 10, time_dim=20,
+                                       combine_mode=mode)
+    assert net(x).shape == (2, 10)
+  # Rank-5 input is spatially pooled.
+  net = tec.ReduceTemporalEmbeddings(16, 10, time_dim=20,
+                                     combine_mode="mean")
+  assert net(torch.randn(2, 20, 4, 4, 16)).shape == (2, 10)
+
+
+def test_embed_modules():
+  full = tec.EmbedFullstate(in_dim=9, embed_size=12)
+  assert full(torch.randn(3, 9)).shape == (3, 12)
+  img = tec.EmbedConditionImages(fc_layers=(32, 16))
+  assert img(torch.rand(2, 3, 64, 64)).shape == (2, 16)
