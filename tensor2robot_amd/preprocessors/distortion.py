@@ -60,8 +60,9 @@ def preprocess_image(image: torch.Tensor, mode: str,
     leading = image.shape[:2]
     image = image.reshape(-1, *image.shape[2:])
   crop = crop_size or target_size
-  if image.is_cuda and image.dtype == torch.uint8 and image.dim() == 4 \
-      and image.shape[-1] == 3:
+  needs_resize = tuple(crop) != tuple(target_size)
+  if not needs_resize and image.is_cuda and image.dtype == torch.uint8 \
+      and image.dim() == 4 and image.shape[-1] == 3:
     # Fused HIP path: crop + convert + photometric distortion in one
     # kernel pair (tensor2robot_amd/ops/hip/preprocess.hip).
     from tensor2robot_amd.ops import preprocess as fused
@@ -72,9 +73,12 @@ def preprocess_image(image: torch.Tensor, mode: str,
     return image
   if image.dtype == torch.uint8:
     image = image.to(torch.float32) / 255.0
-  crop_size = crop_size or target_size
-  image = crop_image(image, mode, crop_size[0], crop_size[1],
-                     generator=generator)
+  image = crop_image(image, mode, crop[0], crop[1], generator=generator)
+  if needs_resize:
+    # Reference :96-97 resize_images (bilinear) after the crop.
+    image = torch.nn.functional.interpolate(
+        image.permute(0, 3, 1, 2), size=tuple(target_size),
+        mode="bilinear", align_corners=False).permute(0, 2, 3, 1)
   image = maybe_distort_image_batch(image, mode, generator=generator)
   if leading is not None:
     image = image.reshape(*leading, *image.shape[1:])
