@@ -1,0 +1,151 @@
+"""Grasp2Vec tests (reference research/grasp2vec/losses_test.py shape)."""
+
+import numpy as np
+import pytest
+import torch
+
+from tensor2robot_amd.data import input_generators
+from tensor2robot_amd.models import optimizers
+from tensor2robot_amd.research.grasp2vec import grasp2vec_model
+from tensor2robot_amd.research.grasp2vec import losses
+from tensor2robot_amd.research.grasp2vec import visualization
+from tensor2robot_amd.specs import tensorspec_utils as tsu
+from tensor2robot_amd.train import train_eval
+from tensor2robot_amd.utils import modes as run_modes
+
+
+def test_npairs_loss_prefers_aligned_pairs():
+  torch.manual_seed(0)
+  labels = torch.arange(4)
+  aligned = torch.eye(4) * 3.0
+  loss_aligned = losses.npairs_loss(labels, aligned, aligned)
+  shuffled = aligned[[1, 2, 3, 0]]
+  loss_shuffled = losses.npairs_loss(labels, aligned, shuffled)
+  assert loss_aligned < loss_shuffled
+
+
+def test_npairs_loss_equal_labels_share_probability():
+  labels = torch.tensor([0, 0, 1])
+  emb = torch.randn(3, 5)
+  loss = losses.npairs_loss(labels, emb, emb)
+  assert torch.isfinite(loss)
+
+
+def test_arithmetic_losses_zero_when_consistent():
+  pre = torch.tensor([[2.0, 1.0], [4.0, 0.0]])
+  goal = torch.tensor([[1.0, 1.0], [3.0, -1.0]])
+  post = pre - goal
+  mask = torch.ones(2)
+  assert float(losses.L2ArithmeticLoss(pre, goal, post, mask)) < 1e-10
+  assert float(losses.CosineArithmeticLoss(pre, goal, post, mask)) < 1e-6
+  # Empty mask -> zero loss, no NaN.
+  assert float(losses.L2ArithmeticLoss(pre, goal, post,
+                                       torch.zeros(2))) == 0.0
+
+
+def test_triplet_loss_runs_and_semihard_sane():
+  torch.manual_seed(0)
+  pre, goal, post = (torch.randn(6, 8) for _ in range(3))
+  loss, pairs, labels = losses.TripletLoss(pre, goal, post)
+  assert torch.isfinite(loss)
+  assert pairs.shape == (12, 8)
+  assert labels.shape == (12,)
+  # Perfectly separated clusters within margin -> small loss.
+  sep = torch.eye(4).repeat(2, 1) * 100.0
+  lbl = torch.arange(4).repeat(2)
+  small = losses.triplet_semihard_loss(lbl, torch.nn.functional.normalize(
+      sep, dim=1), margin=0.1)
+  assert float(small) < 0.2
+
+
+def test_keypoint_accuracy():
+  keypoints = torch.tensor([[0.5, -0.5], [-0.5, 0.5]])
+  labels = torch.tensor([0, 3])
+  acc, loss = losses.KeypointAccuracy(keypoints, labels)
+  assert float(acc) == 1.0
+  assert torch.isfinite(loss)
+
+
+def test_softmax_response_detects_presence():
+  torch.manual_seed(0)
+  goal = torch.zeros(1, 4)
+  goal[0, 2] = 1.0
+  scene_with = torch.zeros(1, 4, 3, 3)
+  scene_with[0, 2, 1, 1] = 5.0
+  scene_without = torch.zeros(1, 4, 3, 3)
+  heat_with, _ = losses.get_softmax_response(goal, scene_with)
+  heat_without, _ = losses.get_softmax_response(goal, scene_without)
+  assert heat_with > heat_without
+
+
+def test_tyloss_sign():
+  goal = torch.tensor([[1.0, 0.0]])
+  pre = torch.zeros(1, 2, 2, 2)
+  pre[0, 0] = 1.0   # object present pregrasp
+  post = torch.zeros(1, 2, 2, 2)
+  post[0, 1] = 1.0  # absent postgrasp
+  assert float(losses.TYloss(pre, post, goal)) < 0
+
+
+def test_match_norms_loss_gradient_only_on_paired():
+  anchor = torch.randn(3, 4, requires_grad=True)
+  paired = torch.randn(3, 4, requires_grad=True)
+  loss = losses.MatchNormsLoss(anchor, paired)
+  loss.backward()
+  assert anchor.grad is None or torch.all(anchor.grad == 0)
+  assert paired.grad is not None
+
+
+def test_npairs_multilabel_runs():
+  torch.manual_seed(0)
+  pre, goal, post = (torch.randn(4, 6) for _ in range(3))
+  success = torch.tensor([1.0, 0.0, 1.0, 1.0])
+  loss = losses.NPairsLossMultilabel(pre, goal, post, success)
+  assert torch.isfinite(loss)
+
+
+def test_preprocessor_crop_and_flip():
+  model = grasp2vec_model.Grasp2VecModel(
+      scene_size=(472, 472), goal_size=(472, 472), device_type="cpu",
+      compute_dtype="float32")
+  prep = model.preprocessor
+  in_spec = prep.get_in_feature_specification(run_modes.TRAIN)
+  assert tuple(in_spec["pregrasp_image"].shape) == (512, 640, 3)
+  features = tsu.TensorSpecStruct()
+  for key in ("pregrasp_image", "postgrasp_image", "goal_image"):
+    features[key] = torch.randint(0, 256, (2, 512, 640, 3),
+                                  dtype=torch.uint8)
+  f, _ = prep._preprocess_fn(features, None, run_modes.EVAL)
+  for key in ("pregrasp_image", "postgrasp_image", "goal_image"):
+    assert f[key].shape == (2, 472, 472, 3)
+    assert f[key].dtype == torch.float32
+
+
+def test_grasp2vec_small_train_smoke(tmp_path):
+  model = grasp2vec_model.Grasp2VecModel(
+      scene_size=(64, 64), goal_size=(64, 64), resnet_size=18,
+      device_type="cpu", compute_dtype="float32",
+      preprocessor_cls=None,
+      create_optimizer_fn=lambda: optimizers.create_adam_optimizer(1e-3))
+  # Bypass the 512x640 preprocessor: feed model-spec-shaped data.
+  from tensor2robot_amd.preprocessors import abstract_preprocessor
+  model._preprocessor = abstract_preprocessor.NoOpPreprocessor(
+      model_feature_specification_fn=model.get_feature_specification,
+      model_label_specification_fn=model.get_label_specification)
+  gen = input_generators.DefaultRandomInputGenerator(batch_size=2, seed=7)
+  result = train_eval.train_eval_model(
+      t2r_model=model, input_generator_train=gen,
+      input_generator_eval=None, max_train_steps=2,
+      model_dir=str(tmp_path))
+  assert result["global_step"] == 2
+  assert np.isfinite(result["loss"])
+
+
+def test_visualization_heatmap():
+  goal = torch.randn(2, 8)
+  spatial = torch.randn(2, 8, 4, 4)
+  heat = visualization.compute_heatmap(goal, spatial)
+  assert heat.shape == (2, 1, 4, 4)
+  assert float(heat.min()) >= 0.0 and float(heat.max()) <= 1.0
+  points, _ = visualization.heatmap_keypoints(heat)
+  assert points.shape == (2, 2)
