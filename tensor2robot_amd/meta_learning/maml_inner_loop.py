@@ -1,0 +1,167 @@
+"""MAML inner-loop gradient descent without modifying base models.
+
+Reference `meta_learning/maml_inner_loop.py:27-327`: the TF version
+intercepts variable reads with a custom getter and substitutes
+functional `theta - alpha * grad(L)` tensors (:106-187).  The torch-native
+equivalent swaps a module's parameters for plain (graph-connected)
+tensors for the duration of a forward pass — same trick, torch idiom —
+so any base model runs under adapted fast weights with zero changes.
+
+`use_second_order=True` keeps the gradient graph (create_graph) so the
+outer loss backpropagates through the inner update; False detaches the
+gradient term (first-order MAML, reference :184-185).  `learn_inner_lr`
+gives every parameter its own learned inner learning rate (:82-94);
+`var_scope` restricts which parameters the inner loop adapts (:174-177).
+"""
+
+from __future__ import annotations
+
+import contextlib
+from typing import Callable, Dict, List, Optional, Tuple
+
+import torch
+from torch import nn
+
+from tensor2robot_amd import gin
+
+
+@contextlib.contextmanager
+def swap_parameters(module: nn.Module, fast: Dict[str, torch.Tensor]):
+  """Temporarily replace module parameters with plain tensors.
+
+  Forward passes inside the context read `fast[name]` wherever the
+  module would read its registered parameter `name` — the torch
+  equivalent of the reference's custom-getter interception.
+  """
+  saved = []
+  try:
+    for name, tensor in fast.items():
+      mod = module
+      parts = name.split(".")
+      for p in parts[:-1]:
+        mod = getattr(mod, p)
+      leaf = parts[-1]
+      saved.append((mod, leaf, mod._parameters[leaf]))
+      mod._parameters[leaf] = tensor
+    yield
+  finally:
+    for mod, leaf, original in reversed(saved):
+      mod._parameters[leaf] = original
+
+
+@gin.configurable
+class MAMLInnerLoopGradientDescent:
+  """Functional theta - alpha*grad inner loop (reference :27-327)."""
+
+  def __init__(self, learning_rate: float = 0.001,
+               use_second_order: bool = True,
+               var_scope: Optional[str] = None,
+               learn_inner_lr: bool = False,
+               inner_lr_params: Optional[nn.ParameterDict] = None):
+    self._learning_rate = learning_rate
+    self._use_second_order = use_second_order
+    self._var_scope = var_scope
+    self._learn_inner_lr = learn_inner_lr
+    self._inner_lr_params = inner_lr_params
+
+  @staticmethod
+  def lr_key(param_name: str) -> str:
+    return param_name.replace(".", "_") + "_inner_lr"
+
+  def create_inner_lr_params(self, network: nn.Module) -> nn.ParameterDict:
+    """Per-parameter learned LRs, trained by the outer loop (ref :82-94)."""
+    lrs = nn.ParameterDict()
+    for name, _ in network.named_parameters():
+      if self._adapts(name):
+        lrs[self.lr_key(name)] = nn.Parameter(
+            torch.tensor(float(self._learning_rate)))
+    self._inner_lr_params = lrs
+    return lrs
+
+  def _adapts(self, name: str) -> bool:
+    return self._var_scope is None or name.startswith(self._var_scope)
+
+  def _lr(self, name: str, device) -> torch.Tensor:
+    if self._learn_inner_lr:
+      if self._inner_lr_params is None:
+        raise RuntimeError("learn_inner_lr requires create_inner_lr_params "
+                           "before inner_loop")
+      return self._inner_lr_params[self.lr_key(name)]
+    return torch.tensor(self._learning_rate, device=device)
+
+  @staticmethod
+  def _extract_train_loss(train_fn_result):
+    if isinstance(train_fn_result, torch.Tensor):
+      return train_fn_result
+    if isinstance(train_fn_result, tuple):
+      return train_fn_result[0]
+    raise ValueError("model_train_fn should return loss or "
+                     "(loss, train_outputs)")
+
+  def _apply_gradients(self, loss: torch.Tensor,
+                       fast: Dict[str, torch.Tensor]
+                       ) -> Dict[str, torch.Tensor]:
+    names = [n for n in fast if self._adapts(n)]
+    grads = torch.autograd.grad(
+        loss, [fast[n] for n in names],
+        create_graph=self._use_second_order, allow_unused=True)
+    updated = dict(fast)
+    for name, grad in zip(names, grads):
+      if grad is None:
+        continue
+      if not self._use_second_order:
+        grad = grad.detach()  # first-order MAML (reference :184-185)
+      updated[name] = fast[name] - self._lr(name, grad.device) * grad
+    return updated
+
+  def inner_loop(self, inputs_list, inference_network_fn: Callable,
+                 model_train_fn: Callable, network: nn.Module,
+                 mode=None, params=None
+                 ) -> Tuple[List, List, List[torch.Tensor]]:
+    """len(inputs_list)-1 adaptation steps, then val forwards (ref :212-327).
+
+    inputs_list: [(cond_f, cond_l), ..., (inference_f, inference_l)].
+    Returns ([unconditioned, conditioned] val outputs, inner_outputs,
+    inner_losses).
+    """
+    val_features, val_labels = inputs_list[-1]
+    params = dict(params or {})
+    params["is_inner_loop"] = True
+
+    fast = {name: p for name, p in network.named_parameters()}
+    inner_outputs, inner_losses = [], []
+    for train_features, train_labels in inputs_list[:-1]:
+      with swap_parameters(network, fast):
+        outputs = inference_network_fn(features=train_features,
+                                       labels=train_labels, mode=mode,
+                                       params=params)
+      inner_outputs.append(outputs)
+      loss = self._extract_train_loss(model_train_fn(
+          features=train_features, labels=train_labels,
+          inference_outputs=outputs, mode=mode, params=params))
+      inner_losses.append(loss)
+      fast = self._apply_gradients(loss, fast)
+
+    # Monitor adaptation: final forward on the last condition step.
+    final_features, final_labels = inputs_list[-2]
+    with swap_parameters(network, fast):
+      final_outputs = inference_network_fn(features=final_features,
+                                           labels=final_labels, mode=mode,
+                                           params=params)
+    inner_outputs.append(final_outputs)
+    inner_losses.append(self._extract_train_loss(model_train_fn(
+        features=final_features, labels=final_labels,
+        inference_outputs=final_outputs, mode=mode, params=params)))
+
+    with swap_parameters(network, fast):
+      params_cond = dict(params)
+      params_cond["is_inner_loop"] = False
+      conditioned = inference_network_fn(features=val_features,
+                                         labels=val_labels, mode=mode,
+                                         params=params_cond)
+    # Unconditioned val forward under the ORIGINAL weights (reference
+    # :321-324) — insight into what the adaptation changed.
+    unconditioned = inference_network_fn(features=val_features,
+                                         labels=val_labels, mode=mode,
+                                         params=params)
+    return [unconditioned, conditioned], inner_outputs, inner_losses

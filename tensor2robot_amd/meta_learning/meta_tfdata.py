@@ -1,0 +1,84 @@
+"""Batch-reshaping utilities for meta-learning data.
+
+Reference `meta_learning/meta_tfdata.py`: TrainValPair :27,
+flatten_batch_examples :174 / unflatten_batch_examples :201
+([tasks, samples, ...] <-> [tasks*samples, ...]), multi_batch_apply :261
+(merge N batch dims -> f -> unmerge), split_train_val :130.
+"""
+
+from __future__ import annotations
+
+import collections
+from typing import Callable
+
+import torch
+
+from tensor2robot_amd.specs import tensorspec_utils as tsu
+
+TrainValPair = collections.namedtuple("TrainValPair", ["train", "val"])
+
+
+def _map_struct(fn, struct):
+  if isinstance(struct, torch.Tensor):
+    return fn(struct)
+  if isinstance(struct, dict) or isinstance(struct, tsu.TensorSpecStruct):
+    out = tsu.TensorSpecStruct()
+    for k, v in tsu.flatten_spec_structure(struct).items():
+      out[k] = fn(v) if isinstance(v, torch.Tensor) else v
+    return out
+  if isinstance(struct, (list, tuple)):
+    return type(struct)(_map_struct(fn, s) for s in struct)
+  return struct
+
+
+def flatten_batch_examples(struct):
+  """[tasks, samples, ...] -> [tasks*samples, ...] (reference :174)."""
+  return _map_struct(lambda t: t.reshape(-1, *t.shape[2:]), struct)
+
+
+def unflatten_batch_examples(struct, samples_per_task: int):
+  """[tasks*samples, ...] -> [tasks, samples, ...] (reference :201)."""
+  return _map_struct(
+      lambda t: t.reshape(-1, samples_per_task, *t.shape[1:]), struct)
+
+
+def multi_batch_apply(fn: Callable, num_batch_dims: int, *args, **kwargs):
+  """Merge the first num_batch_dims dims, apply fn, unmerge (reference :261).
+
+  All tensor arguments must share the leading batch dims.
+  """
+  batch_shape = None
+
+  def find_shape(struct):
+    nonlocal batch_shape
+    if isinstance(struct, torch.Tensor) and batch_shape is None:
+      batch_shape = struct.shape[:num_batch_dims]
+    elif isinstance(struct, (dict, tsu.TensorSpecStruct)):
+      for v in tsu.flatten_spec_structure(struct).values():
+        find_shape(v)
+    elif isinstance(struct, (list, tuple)):
+      for v in struct:
+        find_shape(v)
+
+  for a in args:
+    find_shape(a)
+  if batch_shape is None:
+    raise ValueError("multi_batch_apply found no tensors")
+
+  def merge(t):
+    return t.reshape(-1, *t.shape[num_batch_dims:])
+
+  def unmerge(t):
+    return t.reshape(*batch_shape, *t.shape[1:])
+
+  merged_args = [_map_struct(merge, a) for a in args]
+  merged_kwargs = {k: _map_struct(merge, v) for k, v in kwargs.items()}
+  result = fn(*merged_args, **merged_kwargs)
+  return _map_struct(unmerge, result)
+
+
+def split_train_val(struct, num_train_samples: int) -> TrainValPair:
+  """Split [tasks, train+val, ...] into a TrainValPair (reference :130)."""
+  train = _map_struct(lambda t: t[:, :num_train_samples], struct)
+  val = _map_struct(lambda t: t[:, num_train_samples:], struct)
+  return TrainValPair(train, val)
