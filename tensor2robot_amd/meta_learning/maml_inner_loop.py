@@ -131,16 +131,19 @@ class MAMLInnerLoopGradientDescent:
     fast = {name: p for name, p in network.named_parameters()}
     inner_outputs, inner_losses = [], []
     for train_features, train_labels in inputs_list[:-1]:
-      with swap_parameters(network, fast):
+      # The adaptation needs autograd even under inference no_grad —
+      # MAML serving adapts at SelectAction time (reference meta
+      # policies), so grad mode is forced on for the inner steps.
+      with torch.enable_grad(), swap_parameters(network, fast):
         outputs = inference_network_fn(features=train_features,
                                        labels=train_labels, mode=mode,
                                        params=params)
+        loss = self._extract_train_loss(model_train_fn(
+            features=train_features, labels=train_labels,
+            inference_outputs=outputs, mode=mode, params=params))
+        fast = self._apply_gradients(loss, fast)
       inner_outputs.append(outputs)
-      loss = self._extract_train_loss(model_train_fn(
-          features=train_features, labels=train_labels,
-          inference_outputs=outputs, mode=mode, params=params))
       inner_losses.append(loss)
-      fast = self._apply_gradients(loss, fast)
 
     # Monitor adaptation: final forward on the last condition step.
     final_features, final_labels = inputs_list[-2]
