@@ -1,0 +1,161 @@
+"""GPU tests for the layer library and model families (bf16, fused BN).
+
+Each numerics test compares the HIP/bf16 path against a plain-torch fp32
+reference of the same op/net.
+"""
+
+import copy
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                                  reason="needs MI355X")
+
+
+@requires_gpu
+def test_resnet18_gpu_bf16_matches_cpu_fp32():
+  from tensor2robot_amd.layers import resnet
+  torch.manual_seed(0)
+  net_cpu = resnet.ResNet(resnet_size=18, num_classes=8).eval()
+  net_gpu = copy.deepcopy(net_cpu).cuda().to(
+      memory_format=torch.channels_last).eval()
+  x = torch.randn(4, 3, 64, 64)
+  with torch.no_grad():
+    out_cpu, _ = net_cpu(x)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+      out_gpu, _ = net_gpu(x.cuda().contiguous(
+          memory_format=torch.channels_last))
+  assert torch.allclose(out_cpu, out_gpu.float().cpu(), atol=0.1,
+                        rtol=0.1), \
+      float((out_cpu - out_gpu.float().cpu()).abs().max())
+
+
+@requires_gpu
+def test_resnet_train_step_fused_bn_runs():
+  from tensor2robot_amd.layers import resnet
+  torch.manual_seed(0)
+  net = resnet.ResNet(resnet_size=18, num_classes=4).cuda().to(
+      memory_format=torch.channels_last).train()
+  opt = torch.optim.SGD(net.parameters(), lr=1e-2)
+  x = torch.randn(8, 3, 64, 64, device="cuda").contiguous(
+      memory_format=torch.channels_last)
+  with torch.autocast("cuda", dtype=torch.bfloat16):
+    out, _ = net(x)
+    loss = out.float().pow(2).mean()
+  loss.backward()
+  opt.step()
+  for p in net.parameters():
+    assert torch.isfinite(p).all()
+  # BN running stats moved.
+  bn = net.block_layers[0].blocks[0].bn1
+  assert not torch.allclose(bn.running_mean,
+                            torch.zeros_like(bn.running_mean))
+
+
+@requires_gpu
+def test_spatial_softmax_gpu_matches_cpu():
+  from tensor2robot_amd.layers import spatial_softmax
+  torch.manual_seed(0)
+  feat = torch.randn(3, 8, 14, 14)
+  p_cpu, _ = spatial_softmax.SpatialSoftmax()(feat)
+  p_gpu, _ = spatial_softmax.SpatialSoftmax()(feat.cuda())
+  assert torch.allclose(p_cpu, p_gpu.cpu(), atol=1e-5, rtol=1e-4)
+
+
+@requires_gpu
+def test_bcz_train_step_gpu_bf16():
+  import functools
+  from tensor2robot_amd.models import optimizers
+  from tensor2robot_amd.research.bcz import model as bcz_model
+  from tensor2robot_amd.specs import tensorspec_utils as tsu
+  from tensor2robot_amd.utils import modes as run_modes
+  torch.manual_seed(0)
+  model = bcz_model.BCZModel(
+      image_size=(100, 100), input_size=(512, 640),
+      preprocessor_cls=functools.partial(
+          bcz_model.BCZPreprocessor, image_size=(100, 100),
+          input_size=(512, 640), crop_size=(472, 472),
+          mock_subtask=True),
+      create_optimizer_fn=lambda: optimizers.create_adam_optimizer(1e-4),
+      device_type="gpu", compute_dtype="bfloat16")
+  model.to_device(torch.device("cuda:0"))
+  model.network.to(memory_format=torch.channels_last)
+  g = torch.Generator().manual_seed(0)
+  features = tsu.TensorSpecStruct()
+  features["image"] = torch.randint(0, 256, (4, 512, 640, 3),
+                                    dtype=torch.uint8,
+                                    generator=g).cuda()
+  features["subtask_id"] = torch.zeros(4, 1, dtype=torch.int64).cuda()
+  for name, size, _, _ in model._action_components:
+    features["present/" + name] = torch.rand(4, size,
+                                             generator=g).cuda()
+  labels = tsu.TensorSpecStruct()
+  labels["future/xyz_residual"] = torch.randn(4, 1, 3,
+                                              generator=g).cuda()
+  labels["future/quaternion"] = torch.randn(4, 1, 4, generator=g).cuda()
+  labels["future/target_close"] = (torch.rand(4, 1, 1, generator=g)
+                                   > 0.5).float().cuda()
+  with torch.autocast("cuda", dtype=torch.bfloat16):
+    f, l = model.preprocessor.preprocess(features, labels,
+                                         run_modes.TRAIN)
+    ops = model.model_fn(f, l, run_modes.TRAIN)
+  ops.loss.backward()
+  torch.cuda.synchronize()
+  assert torch.isfinite(ops.loss)
+
+
+@requires_gpu
+def test_grasp2vec_step_gpu():
+  from tensor2robot_amd.research.grasp2vec import grasp2vec_model
+  from tensor2robot_amd.specs import tensorspec_utils as tsu
+  from tensor2robot_amd.utils import modes as run_modes
+  torch.manual_seed(0)
+  model = grasp2vec_model.Grasp2VecModel(
+      scene_size=(128, 128), goal_size=(128, 128), resnet_size=18,
+      device_type="gpu", compute_dtype="bfloat16")
+  model.to_device(torch.device("cuda:0"))
+  model.network.to(memory_format=torch.channels_last)
+  features = tsu.TensorSpecStruct()
+  for key in ("pregrasp_image", "postgrasp_image", "goal_image"):
+    features[key] = torch.rand(4, 128, 128, 3, device="cuda")
+  with torch.autocast("cuda", dtype=torch.bfloat16):
+    ops = model.model_fn(features, None, run_modes.TRAIN)
+  ops.loss.backward()
+  torch.cuda.synchronize()
+  assert torch.isfinite(ops.loss)
+
+
+@requires_gpu
+def test_maml_inner_loop_gpu():
+  from tensor2robot_amd.meta_learning import maml_inner_loop
+  from tensor2robot_amd.specs import tensorspec_utils as tsu
+
+  class _Base:
+    def __init__(self):
+      torch.manual_seed(0)
+      self.network = torch.nn.Linear(4, 2).cuda()
+
+    def inference_network_fn(self, features, labels, mode, params=None):
+      return {"prediction": self.network(features["x"])}
+
+    def model_train_fn(self, features, labels, inference_outputs, mode,
+                       params=None):
+      return torch.nn.functional.mse_loss(
+          inference_outputs["prediction"], labels["y"])
+
+  base = _Base()
+  loop = maml_inner_loop.MAMLInnerLoopGradientDescent(learning_rate=0.05)
+  f = tsu.TensorSpecStruct()
+  f["x"] = torch.randn(8, 4, device="cuda")
+  l = tsu.TensorSpecStruct()
+  l["y"] = torch.randn(8, 2, device="cuda")
+  outputs, _, inner_losses = loop.inner_loop(
+      [(f, l), (f, l), (f, l)], base.inference_network_fn,
+      base.model_train_fn, base.network, mode="train")
+  assert float(inner_losses[-1]) < float(inner_losses[0])
+  outputs[1]["prediction"].pow(2).mean().backward()
+  assert base.network.weight.grad is not None
