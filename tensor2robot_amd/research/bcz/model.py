@@ -43,6 +43,7 @@ GRIPPER_CLOSE_FRACTION_TO_OPEN_GRIPPER = 0.4  # reference :58
 MIN_GRIPPER_CLOSE = 0.2  # reference :59
 
 
+@gin.constants_from_enum
 class ConditionMode(enum.Enum):
   ONEHOT_TASKID = 1
   LANGUAGE_EMBEDDING = 2

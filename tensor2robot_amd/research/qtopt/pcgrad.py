@@ -1,0 +1,124 @@
+"""PCGrad: per-task gradient projection (Gradient Surgery, arXiv:2001.06782).
+
+Reference `research/qtopt/pcgrad.py:29-...`: wraps a base optimizer;
+compute_gradients takes a LIST of per-task losses, shuffles it (:112),
+computes per-task flattened gradients over the PCGrad variable subset
+(allowlist/denylist wildcard filtering :73-87), projects away conflicting
+components g_i -= min(g_i.g_k / ||g_k||^2, 0) * g_k (:179-205), sums the
+projected task gradients, and trains filtered-out variables on the plain
+summed loss.
+
+Torch design: PCGrad wraps any torch.optim.Optimizer (or this repo's
+ScheduledOptimizer); call pcgrad_backward(losses) instead of
+loss.backward(), then step() as usual.
+"""
+
+from __future__ import annotations
+
+import fnmatch
+import random
+from typing import Iterable, List, Optional, Sequence, Tuple
+
+import torch
+
+from tensor2robot_amd import gin
+
+
+def _filter_params(named_params, allowlist, denylist):
+  allowlist = list(allowlist) if allowlist is not None else ["*"]
+  denylist = list(denylist) if denylist is not None else []
+  accepts, rejects = [], []
+  for name, p in named_params:
+    if any(fnmatch.fnmatchcase(name, w) for w in allowlist) and not any(
+        fnmatch.fnmatchcase(name, w) for w in denylist):
+      accepts.append((name, p))
+    else:
+      rejects.append((name, p))
+  return accepts, rejects
+
+
+def project_conflicting(grads_task: torch.Tensor) -> torch.Tensor:
+  """[T, D] per-task flat grads -> summed PCGrad gradient [D]."""
+  num_tasks = grads_task.shape[0]
+  projected = []
+  for i in range(num_tasks):
+    g = grads_task[i].clone()
+    for k in range(num_tasks):
+      gk = grads_task[k]
+      inner = (g * gk).sum()
+      denom = (gk * gk).sum() + 1e-5
+      g = g - torch.clamp(inner / denom, max=0.0) * gk
+    projected.append(g)
+  return torch.stack(projected).sum(dim=0)
+
+
+@gin.configurable
+class PCGrad:
+  """Optimizer wrapper applying PCGrad to a list of task losses."""
+
+  def __init__(self, optimizer_to_wrap, model: Optional[
+      torch.nn.Module] = None,
+               allowlist: Optional[Iterable[str]] = None,
+               denylist: Optional[Iterable[str]] = None,
+               seed: Optional[int] = None):
+    self._optimizer = optimizer_to_wrap
+    self._model = model
+    self._allowlist = allowlist
+    self._denylist = denylist
+    self._rng = random.Random(seed)
+
+  def __getattr__(self, name):
+    return getattr(self._optimizer, name)
+
+  def pcgrad_backward(self, losses: Sequence[torch.Tensor],
+                      named_parameters=None) -> None:
+    """Sets .grad on all parameters from the projected task losses."""
+    if not isinstance(losses, (list, tuple)):
+      raise TypeError(f"losses must be a list, got {type(losses)}")
+    losses = list(losses)
+    self._rng.shuffle(losses)
+    if named_parameters is None:
+      named_parameters = list(self._model.named_parameters())
+    named_parameters = [(n, p) for n, p in named_parameters
+                        if p.requires_grad]
+    pcgrad_nps, other_nps = _filter_params(named_parameters,
+                                           self._allowlist,
+                                           self._denylist)
+    pcgrad_params = [p for _, p in pcgrad_nps]
+    other_params = [p for _, p in other_nps]
+
+    other_grads = None
+    if other_params:
+      other_grads = torch.autograd.grad(
+          sum(losses), other_params, retain_graph=bool(pcgrad_params),
+          allow_unused=True)
+
+    if pcgrad_params:
+      task_flat = []
+      for i, loss in enumerate(losses):
+        grads = torch.autograd.grad(
+            loss, pcgrad_params, retain_graph=i < len(losses) - 1,
+            allow_unused=True)
+        flat = torch.cat([
+            (g if g is not None else torch.zeros_like(p)).reshape(-1)
+            for g, p in zip(grads, pcgrad_params)])
+        task_flat.append(flat)
+      summed = project_conflicting(torch.stack(task_flat))
+      offset = 0
+      for p in pcgrad_params:
+        n = p.numel()
+        g = summed[offset: offset + n].reshape(p.shape)
+        p.grad = g if p.grad is None else p.grad + g
+        offset += n
+
+    if other_grads is not None:
+      for p, g in zip(other_params, other_grads):
+        if g is None:
+          continue
+        p.grad = g if p.grad is None else p.grad + g
+
+  def step(self, *args, **kwargs):
+    return self._optimizer.step(*args, **kwargs)
+
+  def zero_grad(self, *args, **kwargs):
+    return self._optimizer.zero_grad(*args, **kwargs)

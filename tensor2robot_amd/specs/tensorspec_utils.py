@@ -755,10 +755,10 @@ def copy_tensorspec(spec_structure, batch_size=None,
 def _concrete_shape(spec, batch_size=None, sequence_length=None):
   shape = list(spec.shape)
   if batch_size is not None:
-    if shape and shape[0] is None:
-      shape[0] = batch_size
-    else:
-      shape = [batch_size] + shape
+    # Reference :817-925: the batch dim is always PREPENDED; a leading
+    # None (e.g. the meta-learning samples dim from batch_size=-1 specs)
+    # stays as an inner dim and concretizes to 1 below.
+    shape = [batch_size] + shape
   if spec.is_sequence:
     # A sequence spec's data carries an episode/time dim after batch.
     seq = sequence_length if sequence_length is not None else 1
