@@ -18,6 +18,50 @@ from tensor2robot_amd.specs import tensorspec_utils as tsu
 TrainValPair = collections.namedtuple("TrainValPair", ["train", "val"])
 
 
+def parallel_read(file_patterns, parse_fn, shuffle_filenames: bool = True,
+                  num_train_samples_per_task: int = 4,
+                  num_val_samples_per_task: int = 4,
+                  shuffle_buffer_size: int = 50, filter_fn=None,
+                  interleave_cycle_length=None, mode: str = "train",
+                  seed=None):
+  """ONE FILE == ONE TASK reading (reference meta_tfdata.py:32-126).
+
+  Yields per-task batches: parse_fn applied to
+  num_train + num_val consecutive (shuffled in train) records of each
+  task file, interleaving tasks round-robin; task order reshuffles per
+  epoch.  parse_fn: List[bytes] -> parsed batch struct.
+  """
+  import random as _random
+  from tensor2robot_amd.data import tfrecord as tfrecord_mod
+
+  _, filenames = tfrecord_mod.get_data_format_and_filenames(file_patterns)
+  rng = _random.Random(seed)
+  samples_per_task = (num_train_samples_per_task
+                      + num_val_samples_per_task)
+
+  def task_stream(path):
+    """Infinite per-file batch stream (shuffled within a buffer)."""
+    while True:
+      records = list(tfrecord_mod.read_records(path))
+      if mode == "train":
+        rng.shuffle(records)
+      for i in range(0, len(records) - samples_per_task + 1,
+                     samples_per_task):
+        batch = records[i: i + samples_per_task]
+        parsed = parse_fn(batch)
+        if filter_fn is not None and not filter_fn(parsed):
+          continue
+        yield parsed
+
+  order = list(filenames)
+  streams = {f: task_stream(f) for f in filenames}
+  while True:
+    if shuffle_filenames:
+      rng.shuffle(order)
+    for f in order:
+      yield next(streams[f])
+
+
 def _map_struct(fn, struct):
   if isinstance(struct, torch.Tensor):
     return fn(struct)
