@@ -20,7 +20,13 @@ hip_sources = [
     os.path.join(HIP_DIR, "bindings.cpp"),
     os.path.join(HIP_DIR, "fused_bn_relu.hip"),
     os.path.join(HIP_DIR, "preprocess.hip"),
+    os.path.join(HIP_DIR, "maxpool.hip"),
 ]
+
+import pybind11
+from setuptools import Extension
+
+NATIVE_DIR = os.path.join(ROOT, "tensor2robot_amd", "data", "native")
 
 ext_modules = [
     cpp_extension.CUDAExtension(
@@ -30,6 +36,15 @@ ext_modules = [
             "cxx": ["-O3", "-std=c++17"],
             "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
         },
+    ),
+    # CPU-only data codecs (baseline JPEG): plain pybind11, no torch dep.
+    Extension(
+        name="tensor2robot_amd.ops._t2r_native",
+        sources=[os.path.join(NATIVE_DIR, "jpeg_codec.cpp"),
+                 os.path.join(NATIVE_DIR, "native_bindings.cpp")],
+        include_dirs=[pybind11.get_include()],
+        extra_compile_args=["-O3", "-std=c++17"],
+        language="c++",
     ),
 ]
 

@@ -1,0 +1,631 @@
+// Baseline sequential JPEG codec (encode + decode), dependency-free C++.
+//
+// Backs tensor2robot_amd/data/image_codec.{encode,decode}_jpeg — the
+// CPU data-pipeline equivalent of the reference's tf.image.decode_image /
+// tf.io.encode_jpeg path (`utils/tfdata.py:426-484,546-627`).  Supports
+// baseline DCT (SOF0), 8-bit, grayscale or YCbCr with 1x1/2x1/1x2/2x2
+// luma sampling, restart markers; encoder emits 4:4:4 YCbCr (or
+// grayscale) with the Annex-K quantization/huffman tables.
+
+#include <cmath>
+#include <cstdint>
+#include <cstring>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+namespace t2r_jpeg {
+
+// ---------------------------------------------------------------------------
+// Shared tables
+// ---------------------------------------------------------------------------
+
+static const uint8_t kZigzag[64] = {
+    0,  1,  8,  16, 9,  2,  3,  10, 17, 24, 32, 25, 18, 11, 4,  5,
+    12, 19, 26, 33, 40, 48, 41, 34, 27, 20, 13, 6,  7,  14, 21, 28,
+    35, 42, 49, 56, 57, 50, 43, 36, 29, 22, 15, 23, 30, 37, 44, 51,
+    58, 59, 52, 45, 38, 31, 39, 46, 53, 60, 61, 54, 47, 55, 62, 63};
+
+// Annex K luminance / chrominance quantization tables (natural order).
+static const int kLumaQ[64] = {
+    16, 11, 10, 16, 24,  40,  51,  61,  12, 12, 14, 19, 26,  58,  60,  55,
+    14, 13, 16, 24, 40,  57,  69,  56,  14, 17, 22, 29, 51,  87,  80,  62,
+    18, 22, 37, 56, 68,  109, 103, 77,  24, 35, 55, 64, 81,  104, 113, 92,
+    49, 64, 78, 87, 103, 121, 120, 101, 72, 92, 95, 98, 112, 100, 103, 99};
+static const int kChromaQ[64] = {
+    17, 18, 24, 47, 99, 99, 99, 99, 18, 21, 26, 66, 99, 99, 99, 99,
+    24, 26, 56, 99, 99, 99, 99, 99, 47, 66, 99, 99, 99, 99, 99, 99,
+    99, 99, 99, 99, 99, 99, 99, 99, 99, 99, 99, 99, 99, 99, 99, 99,
+    99, 99, 99, 99, 99, 99, 99, 99, 99, 99, 99, 99, 99, 99, 99, 99};
+
+// Annex K huffman specs: {bits[1..16], values}.
+static const uint8_t kDcLumaBits[17] = {0, 0, 1, 5, 1, 1, 1, 1, 1,
+                                        1, 0, 0, 0, 0, 0, 0, 0};
+static const uint8_t kDcLumaVals[12] = {0, 1, 2, 3,  4,  5,
+                                        6, 7, 8, 9, 10, 11};
+static const uint8_t kDcChromaBits[17] = {0, 0, 3, 1, 1, 1, 1, 1, 1,
+                                          1, 1, 1, 0, 0, 0, 0, 0};
+static const uint8_t kDcChromaVals[12] = {0, 1, 2, 3,  4,  5,
+                                          6, 7, 8, 9, 10, 11};
+static const uint8_t kAcLumaBits[17] = {0, 0, 2, 1, 3, 3, 2, 4, 3,
+                                        5, 5, 4, 4, 0, 0, 1, 0x7d};
+static const uint8_t kAcLumaVals[162] = {
+    0x01, 0x02, 0x03, 0x00, 0x04, 0x11, 0x05, 0x12, 0x21, 0x31, 0x41, 0x06,
+    0x13, 0x51, 0x61, 0x07, 0x22, 0x71, 0x14, 0x32, 0x81, 0x91, 0xa1, 0x08,
+    0x23, 0x42, 0xb1, 0xc1, 0x15, 0x52, 0xd1, 0xf0, 0x24, 0x33, 0x62, 0x72,
+    0x82, 0x09, 0x0a, 0x16, 0x17, 0x18, 0x19, 0x1a, 0x25, 0x26, 0x27, 0x28,
+    0x29, 0x2a, 0x34, 0x35, 0x36, 0x37, 0x38, 0x39, 0x3a, 0x43, 0x44, 0x45,
+    0x46, 0x47, 0x48, 0x49, 0x4a, 0x53, 0x54, 0x55, 0x56, 0x57, 0x58, 0x59,
+    0x5a, 0x63, 0x64, 0x65, 0x66, 0x67, 0x68, 0x69, 0x6a, 0x73, 0x74, 0x75,
+    0x76, 0x77, 0x78, 0x79, 0x7a, 0x83, 0x84, 0x85, 0x86, 0x87, 0x88, 0x89,
+    0x8a, 0x92, 0x93, 0x94, 0x95, 0x96, 0x97, 0x98, 0x99, 0x9a, 0xa2, 0xa3,
+    0xa4, 0xa5, 0xa6, 0xa7, 0xa8, 0xa9, 0xaa, 0xb2, 0xb3, 0xb4, 0xb5, 0xb6,
+    0xb7, 0xb8, 0xb9, 0xba, 0xc2, 0xc3, 0xc4, 0xc5, 0xc6, 0xc7, 0xc8, 0xc9,
+    0xca, 0xd2, 0xd3, 0xd4, 0xd5, 0xd6, 0xd7, 0xd8, 0xd9, 0xda, 0xe1, 0xe2,
+    0xe3, 0xe4, 0xe5, 0xe6, 0xe7, 0xe8, 0xe9, 0xea, 0xf1, 0xf2, 0xf3, 0xf4,
+    0xf5, 0xf6, 0xf7, 0xf8, 0xf9, 0xfa};
+static const uint8_t kAcChromaBits[17] = {0, 0, 2, 1, 2, 4, 4, 3, 4,
+                                          7, 5, 4, 4, 0, 1, 2, 0x77};
+static const uint8_t kAcChromaVals[162] = {
+    0x00, 0x01, 0x02, 0x03, 0x11, 0x04, 0x05, 0x21, 0x31, 0x06, 0x12, 0x41,
+    0x51, 0x07, 0x61, 0x71, 0x13, 0x22, 0x32, 0x81, 0x08, 0x14, 0x42, 0x91,
+    0xa1, 0xb1, 0xc1, 0x09, 0x23, 0x33, 0x52, 0xf0, 0x15, 0x62, 0x72, 0xd1,
+    0x0a, 0x16, 0x24, 0x34, 0xe1, 0x25, 0xf1, 0x17, 0x18, 0x19, 0x1a, 0x26,
+    0x27, 0x28, 0x29, 0x2a, 0x35, 0x36, 0x37, 0x38, 0x39, 0x3a, 0x43, 0x44,
+    0x45, 0x46, 0x47, 0x48, 0x49, 0x4a, 0x53, 0x54, 0x55, 0x56, 0x57, 0x58,
+    0x59, 0x5a, 0x63, 0x64, 0x65, 0x66, 0x67, 0x68, 0x69, 0x6a, 0x73, 0x74,
+    0x75, 0x76, 0x77, 0x78, 0x79, 0x7a, 0x82, 0x83, 0x84, 0x85, 0x86, 0x87,
+    0x88, 0x89, 0x8a, 0x92, 0x93, 0x94, 0x95, 0x96, 0x97, 0x98, 0x99, 0x9a,
+    0xa2, 0xa3, 0xa4, 0xa5, 0xa6, 0xa7, 0xa8, 0xa9, 0xaa, 0xb2, 0xb3, 0xb4,
+    0xb5, 0xb6, 0xb7, 0xb8, 0xb9, 0xba, 0xc2, 0xc3, 0xc4, 0xc5, 0xc6, 0xc7,
+    0xc8, 0xc9, 0xca, 0xd2, 0xd3, 0xd4, 0xd5, 0xd6, 0xd7, 0xd8, 0xd9, 0xda,
+    0xe2, 0xe3, 0xe4, 0xe5, 0xe6, 0xe7, 0xe8, 0xe9, 0xea, 0xf2, 0xf3, 0xf4,
+    0xf5, 0xf6, 0xf7, 0xf8, 0xf9, 0xfa};
+
+// ---------------------------------------------------------------------------
+// DCT
+// ---------------------------------------------------------------------------
+
+static void fdct8x8(const float in[64], float out[64]) {
+  static float c[8][8];
+  static bool init = false;
+  if (!init) {
+    for (int u = 0; u < 8; ++u)
+      for (int x = 0; x < 8; ++x)
+        c[u][x] = std::cos((2 * x + 1) * u * M_PI / 16.0);
+    init = true;
+  }
+  float tmp[64];
+  for (int y = 0; y < 8; ++y)           // rows
+    for (int u = 0; u < 8; ++u) {
+      float s = 0;
+      for (int x = 0; x < 8; ++x) s += in[y * 8 + x] * c[u][x];
+      tmp[y * 8 + u] = s;
+    }
+  for (int u = 0; u < 8; ++u)           // cols
+    for (int v = 0; v < 8; ++v) {
+      float s = 0;
+      for (int y = 0; y < 8; ++y) s += tmp[y * 8 + u] * c[v][y];
+      float cu = (u == 0) ? 0.70710678f : 1.0f;
+      float cv = (v == 0) ? 0.70710678f : 1.0f;
+      out[v * 8 + u] = 0.25f * cu * cv * s;
+    }
+}
+
+static void idct8x8(const float in[64], float out[64]) {
+  static float c[8][8];
+  static bool init = false;
+  if (!init) {
+    for (int u = 0; u < 8; ++u)
+      for (int x = 0; x < 8; ++x)
+        c[u][x] = std::cos((2 * x + 1) * u * M_PI / 16.0);
+    init = true;
+  }
+  float tmp[64];
+  for (int v = 0; v < 8; ++v)           // cols
+    for (int y = 0; y < 8; ++y) {
+      float s = 0;
+      for (int u = 0; u < 8; ++u) {
+        float cu = (u == 0) ? 0.70710678f : 1.0f;
+        s += cu * in[u * 8 + v] * c[u][y];
+      }
+      tmp[y * 8 + v] = s;
+    }
+  for (int y = 0; y < 8; ++y)           // rows
+    for (int x = 0; x < 8; ++x) {
+      float s = 0;
+      for (int u = 0; u < 8; ++u) {
+        float cu = (u == 0) ? 0.70710678f : 1.0f;
+        s += cu * tmp[y * 8 + u] * c[u][x];
+      }
+      out[y * 8 + x] = 0.25f * s;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Encoder
+// ---------------------------------------------------------------------------
+
+struct BitWriter {
+  std::vector<uint8_t>& out;
+  uint32_t acc = 0;
+  int nbits = 0;
+  explicit BitWriter(std::vector<uint8_t>& o) : out(o) {}
+  void put(uint32_t bits, int n) {
+    acc = (acc << n) | (bits & ((1u << n) - 1));
+    nbits += n;
+    while (nbits >= 8) {
+      uint8_t b = (acc >> (nbits - 8)) & 0xff;
+      out.push_back(b);
+      if (b == 0xff) out.push_back(0x00);  // byte stuffing
+      nbits -= 8;
+    }
+  }
+  void flush() {
+    if (nbits > 0) put(0x7f, 8 - nbits);  // pad with 1s
+  }
+};
+
+struct HuffEncTable {
+  uint16_t code[256];
+  uint8_t size[256];
+};
+
+static void build_enc_table(const uint8_t* bits, const uint8_t* vals,
+                            HuffEncTable& t) {
+  std::memset(t.size, 0, sizeof(t.size));
+  int code = 0, k = 0;
+  for (int len = 1; len <= 16; ++len) {
+    for (int i = 0; i < bits[len]; ++i) {
+      t.code[vals[k]] = code;
+      t.size[vals[k]] = len;
+      ++code;
+      ++k;
+    }
+    code <<= 1;
+  }
+}
+
+static int bit_length(int v) {
+  int n = 0;
+  while (v) {
+    ++n;
+    v >>= 1;
+  }
+  return n;
+}
+
+static void emit_marker(std::vector<uint8_t>& out, uint8_t m) {
+  out.push_back(0xff);
+  out.push_back(m);
+}
+
+static void emit_u16(std::vector<uint8_t>& out, uint16_t v) {
+  out.push_back(v >> 8);
+  out.push_back(v & 0xff);
+}
+
+std::vector<uint8_t> encode(const uint8_t* rgb, int h, int w, int channels,
+                            int quality) {
+  if (channels != 1 && channels != 3)
+    throw std::runtime_error("JPEG encode: channels must be 1 or 3");
+  if (quality < 1) quality = 1;
+  if (quality > 100) quality = 100;
+  int scale = quality < 50 ? 5000 / quality : 200 - 2 * quality;
+  int qluma[64], qchroma[64];
+  for (int i = 0; i < 64; ++i) {
+    int ql = (kLumaQ[i] * scale + 50) / 100;
+    int qc = (kChromaQ[i] * scale + 50) / 100;
+    qluma[i] = ql < 1 ? 1 : (ql > 255 ? 255 : ql);
+    qchroma[i] = qc < 1 ? 1 : (qc > 255 ? 255 : qc);
+  }
+
+  std::vector<uint8_t> out;
+  emit_marker(out, 0xd8);  // SOI
+  // APP0 JFIF
+  emit_marker(out, 0xe0);
+  emit_u16(out, 16);
+  const char jfif[] = "JFIF";
+  out.insert(out.end(), jfif, jfif + 5);
+  out.push_back(1); out.push_back(1);   // version
+  out.push_back(0);                     // units
+  emit_u16(out, 1); emit_u16(out, 1);   // density
+  out.push_back(0); out.push_back(0);   // thumbnail
+
+  // DQT
+  for (int t = 0; t < (channels == 3 ? 2 : 1); ++t) {
+    emit_marker(out, 0xdb);
+    emit_u16(out, 67);
+    out.push_back(t);
+    const int* q = t == 0 ? qluma : qchroma;
+    for (int i = 0; i < 64; ++i) out.push_back(q[kZigzag[i]]);
+  }
+  // SOF0
+  emit_marker(out, 0xc0);
+  emit_u16(out, 8 + 3 * channels);
+  out.push_back(8);
+  emit_u16(out, h);
+  emit_u16(out, w);
+  out.push_back(channels);
+  for (int ci = 0; ci < channels; ++ci) {
+    out.push_back(ci + 1);
+    out.push_back(0x11);                // 1x1 sampling (4:4:4)
+    out.push_back(ci == 0 ? 0 : 1);
+  }
+  // DHT (4 tables for color, 2 for gray)
+  struct Spec { uint8_t cls, id; const uint8_t* bits; const uint8_t* vals;
+                int nvals; };
+  std::vector<Spec> specs = {{0, 0, kDcLumaBits, kDcLumaVals, 12},
+                             {1, 0, kAcLumaBits, kAcLumaVals, 162}};
+  if (channels == 3) {
+    specs.push_back({0, 1, kDcChromaBits, kDcChromaVals, 12});
+    specs.push_back({1, 1, kAcChromaBits, kAcChromaVals, 162});
+  }
+  for (auto& s : specs) {
+    emit_marker(out, 0xc4);
+    emit_u16(out, 19 + s.nvals);
+    out.push_back((s.cls << 4) | s.id);
+    for (int i = 1; i <= 16; ++i) out.push_back(s.bits[i]);
+    for (int i = 0; i < s.nvals; ++i) out.push_back(s.vals[i]);
+  }
+  // SOS
+  emit_marker(out, 0xda);
+  emit_u16(out, 6 + 2 * channels);
+  out.push_back(channels);
+  for (int ci = 0; ci < channels; ++ci) {
+    out.push_back(ci + 1);
+    out.push_back(ci == 0 ? 0x00 : 0x11);
+  }
+  out.push_back(0); out.push_back(63); out.push_back(0);
+
+  HuffEncTable dcl, acl, dcc, acc_t;
+  build_enc_table(kDcLumaBits, kDcLumaVals, dcl);
+  build_enc_table(kAcLumaBits, kAcLumaVals, acl);
+  build_enc_table(kDcChromaBits, kDcChromaVals, dcc);
+  build_enc_table(kAcChromaBits, kAcChromaVals, acc_t);
+
+  BitWriter bw(out);
+  int pred[3] = {0, 0, 0};
+  int mcus_y = (h + 7) / 8, mcus_x = (w + 7) / 8;
+  float block[64], coef[64];
+  for (int by = 0; by < mcus_y; ++by) {
+    for (int bx = 0; bx < mcus_x; ++bx) {
+      for (int ci = 0; ci < channels; ++ci) {
+        const int* q = ci == 0 ? qluma : qchroma;
+        HuffEncTable& dct = ci == 0 ? dcl : dcc;
+        HuffEncTable& act = ci == 0 ? acl : acc_t;
+        // Gather the 8x8 block in this component's color space.
+        for (int y = 0; y < 8; ++y) {
+          int sy = by * 8 + y;
+          if (sy >= h) sy = h - 1;
+          for (int x = 0; x < 8; ++x) {
+            int sx = bx * 8 + x;
+            if (sx >= w) sx = w - 1;
+            const uint8_t* px = rgb + (sy * (size_t)w + sx) * channels;
+            float v;
+            if (channels == 1) {
+              v = px[0];
+            } else {
+              float r = px[0], g = px[1], b = px[2];
+              if (ci == 0)
+                v = 0.299f * r + 0.587f * g + 0.114f * b;
+              else if (ci == 1)
+                v = -0.168736f * r - 0.331264f * g + 0.5f * b + 128.0f;
+              else
+                v = 0.5f * r - 0.418688f * g - 0.081312f * b + 128.0f;
+            }
+            block[y * 8 + x] = v - 128.0f;
+          }
+        }
+        fdct8x8(block, coef);
+        int zz[64];
+        for (int i = 0; i < 64; ++i) {
+          int nat = kZigzag[i];
+          zz[i] = (int)std::lround(coef[nat] / q[nat]);
+        }
+        // DC
+        int diff = zz[0] - pred[ci];
+        pred[ci] = zz[0];
+        int mag = diff < 0 ? -diff : diff;
+        int nbits = bit_length(mag);
+        bw.put(dct.code[nbits], dct.size[nbits]);
+        if (nbits)
+          bw.put(diff < 0 ? diff + (1 << nbits) - 1 : diff, nbits);
+        // AC
+        int run = 0;
+        for (int i = 1; i < 64; ++i) {
+          if (zz[i] == 0) {
+            ++run;
+            continue;
+          }
+          while (run > 15) {
+            bw.put(act.code[0xf0], act.size[0xf0]);  // ZRL
+            run -= 16;
+          }
+          int amag = zz[i] < 0 ? -zz[i] : zz[i];
+          int abits = bit_length(amag);
+          int sym = (run << 4) | abits;
+          bw.put(act.code[sym], act.size[sym]);
+          bw.put(zz[i] < 0 ? zz[i] + (1 << abits) - 1 : zz[i], abits);
+          run = 0;
+        }
+        if (run > 0) bw.put(act.code[0x00], act.size[0x00]);  // EOB
+      }
+    }
+  }
+  bw.flush();
+  emit_marker(out, 0xd9);  // EOI
+  return out;
+}
+
+// ---------------------------------------------------------------------------
+// Decoder
+// ---------------------------------------------------------------------------
+
+struct HuffDecTable {
+  // Canonical decode: per length, first code and value index.
+  int32_t mincode[17], maxcode[17], valptr[17];
+  uint8_t vals[256];
+  bool present = false;
+};
+
+static void build_dec_table(const uint8_t* bits, const uint8_t* vals,
+                            int nvals, HuffDecTable& t) {
+  std::memcpy(t.vals, vals, nvals);
+  int code = 0, k = 0;
+  for (int len = 1; len <= 16; ++len) {
+    t.valptr[len] = k;
+    t.mincode[len] = code;
+    code += bits[len];
+    k += bits[len];
+    t.maxcode[len] = bits[len] ? code - 1 : -1;
+    code <<= 1;
+  }
+  t.present = true;
+}
+
+struct BitReader {
+  const uint8_t* data;
+  size_t size, pos;
+  uint32_t acc = 0;
+  int nbits = 0;
+  BitReader(const uint8_t* d, size_t s, size_t p)
+      : data(d), size(s), pos(p) {}
+  int next_bit() {
+    if (nbits == 0) {
+      if (pos >= size) return -1;
+      uint8_t b = data[pos++];
+      if (b == 0xff) {
+        if (pos < size && data[pos] == 0x00) {
+          ++pos;  // stuffed byte
+        } else {
+          // Marker hit (e.g. RST/EOI): signal end of entropy data.
+          --pos;
+          return -1;
+        }
+      }
+      acc = b;
+      nbits = 8;
+    }
+    --nbits;
+    return (acc >> nbits) & 1;
+  }
+  void reset_to_byte() { nbits = 0; }
+};
+
+static int huff_decode(BitReader& br, const HuffDecTable& t) {
+  int code = 0;
+  for (int len = 1; len <= 16; ++len) {
+    int bit = br.next_bit();
+    if (bit < 0) return -1;
+    code = (code << 1) | bit;
+    if (t.maxcode[len] >= 0 && code <= t.maxcode[len] &&
+        code >= t.mincode[len])
+      return t.vals[t.valptr[len] + (code - t.mincode[len])];
+  }
+  throw std::runtime_error("JPEG decode: bad huffman code");
+}
+
+static int receive_extend(BitReader& br, int nbits) {
+  if (nbits == 0) return 0;
+  int v = 0;
+  for (int i = 0; i < nbits; ++i) {
+    int bit = br.next_bit();
+    if (bit < 0) throw std::runtime_error("JPEG decode: truncated");
+    v = (v << 1) | bit;
+  }
+  if (v < (1 << (nbits - 1))) v += ((-1) << nbits) + 1;
+  return v;
+}
+
+struct Component {
+  int id = 0, hs = 1, vs = 1, tq = 0, td = 0, ta = 0;
+  int dc_pred = 0;
+  std::vector<float> plane;  // component-resolution samples
+  int pw = 0, ph = 0;
+};
+
+std::vector<uint8_t> decode(const uint8_t* data, size_t size, int& out_h,
+                            int& out_w, int& out_c) {
+  size_t pos = 0;
+  auto rd_u16 = [&](size_t p) -> int {
+    return (data[p] << 8) | data[p + 1];
+  };
+  if (size < 4 || data[0] != 0xff || data[1] != 0xd8)
+    throw std::runtime_error("JPEG decode: missing SOI");
+  pos = 2;
+  uint16_t qt[4][64] = {};
+  HuffDecTable dc_tabs[4], ac_tabs[4];
+  Component comps[4];
+  int ncomp = 0, height = 0, width = 0, restart_interval = 0;
+  int hmax = 1, vmax = 1;
+
+  while (pos + 4 <= size) {
+    if (data[pos] != 0xff) throw std::runtime_error("JPEG: bad marker");
+    uint8_t marker = data[pos + 1];
+    pos += 2;
+    if (marker == 0xd9) break;  // EOI
+    if (marker == 0x01 || (marker >= 0xd0 && marker <= 0xd7)) continue;
+    int seglen = rd_u16(pos);
+    size_t seg_end = pos + seglen;
+    if (marker == 0xdb) {  // DQT
+      size_t p = pos + 2;
+      while (p < seg_end) {
+        int pq = data[p] >> 4, tq_id = data[p] & 15;
+        ++p;
+        for (int i = 0; i < 64; ++i) {
+          int v = pq ? rd_u16(p + 2 * i) : data[p + i];
+          qt[tq_id][kZigzag[i]] = v;
+        }
+        p += pq ? 128 : 64;
+      }
+    } else if (marker == 0xc4) {  // DHT
+      size_t p = pos + 2;
+      while (p < seg_end) {
+        int cls = data[p] >> 4, id = data[p] & 15;
+        ++p;
+        uint8_t bits[17] = {0};
+        int nvals = 0;
+        for (int i = 1; i <= 16; ++i) {
+          bits[i] = data[p + i - 1];
+          nvals += bits[i];
+        }
+        p += 16;
+        if (cls == 0)
+          build_dec_table(bits, data + p, nvals, dc_tabs[id]);
+        else
+          build_dec_table(bits, data + p, nvals, ac_tabs[id]);
+        p += nvals;
+      }
+    } else if (marker == 0xc0 || marker == 0xc1) {  // SOF0/1 baseline
+      height = rd_u16(pos + 3);
+      width = rd_u16(pos + 5);
+      ncomp = data[pos + 7];
+      if (ncomp > 4) throw std::runtime_error("JPEG: too many components");
+      for (int i = 0; i < ncomp; ++i) {
+        size_t p = pos + 8 + 3 * i;
+        comps[i].id = data[p];
+        comps[i].hs = data[p + 1] >> 4;
+        comps[i].vs = data[p + 1] & 15;
+        comps[i].tq = data[p + 2];
+        hmax = std::max(hmax, comps[i].hs);
+        vmax = std::max(vmax, comps[i].vs);
+      }
+    } else if (marker == 0xc2) {
+      throw std::runtime_error("JPEG: progressive not supported");
+    } else if (marker == 0xdd) {  // DRI
+      restart_interval = rd_u16(pos + 2);
+    } else if (marker == 0xda) {  // SOS
+      int ns = data[pos + 2];
+      for (int i = 0; i < ns; ++i) {
+        int cid = data[pos + 3 + 2 * i];
+        int tables = data[pos + 4 + 2 * i];
+        for (int c = 0; c < ncomp; ++c)
+          if (comps[c].id == cid) {
+            comps[c].td = tables >> 4;
+            comps[c].ta = tables & 15;
+          }
+      }
+      pos = seg_end;
+      // --- entropy-coded scan ---
+      int mcux = (width + 8 * hmax - 1) / (8 * hmax);
+      int mcuy = (height + 8 * vmax - 1) / (8 * vmax);
+      for (int c = 0; c < ncomp; ++c) {
+        comps[c].pw = mcux * 8 * comps[c].hs;
+        comps[c].ph = mcuy * 8 * comps[c].vs;
+        comps[c].plane.assign((size_t)comps[c].pw * comps[c].ph, 0.0f);
+        comps[c].dc_pred = 0;
+      }
+      BitReader br(data, size, pos);
+      float coef[64], pix[64];
+      int mcu_count = 0;
+      for (int my = 0; my < mcuy; ++my) {
+        for (int mx = 0; mx < mcux; ++mx) {
+          if (restart_interval && mcu_count &&
+              mcu_count % restart_interval == 0) {
+            br.reset_to_byte();
+            // Expect RSTn marker.
+            while (br.pos + 1 < size && data[br.pos] == 0xff &&
+                   data[br.pos + 1] >= 0xd0 && data[br.pos + 1] <= 0xd7) {
+              br.pos += 2;
+              for (int c = 0; c < ncomp; ++c) comps[c].dc_pred = 0;
+            }
+          }
+          ++mcu_count;
+          for (int c = 0; c < ncomp; ++c) {
+            Component& comp = comps[c];
+            const uint16_t* q = qt[comp.tq];
+            for (int v = 0; v < comp.vs; ++v) {
+              for (int hh = 0; hh < comp.hs; ++hh) {
+                std::memset(coef, 0, sizeof(coef));
+                int sym = huff_decode(br, dc_tabs[comp.td]);
+                if (sym < 0) throw std::runtime_error("JPEG: truncated");
+                int diff = receive_extend(br, sym);
+                comp.dc_pred += diff;
+                coef[0] = (float)comp.dc_pred * q[0];
+                for (int k = 1; k < 64;) {
+                  int rs = huff_decode(br, ac_tabs[comp.ta]);
+                  if (rs < 0) throw std::runtime_error("JPEG: truncated");
+                  int run = rs >> 4, sbits = rs & 15;
+                  if (sbits == 0) {
+                    if (run != 15) break;  // EOB
+                    k += 16;
+                    continue;
+                  }
+                  k += run;
+                  if (k > 63)
+                    throw std::runtime_error("JPEG: AC overflow");
+                  int nat = kZigzag[k];
+                  coef[nat] = (float)receive_extend(br, sbits) * q[nat];
+                  ++k;
+                }
+                idct8x8(coef, pix);
+                int ox = (mx * comp.hs + hh) * 8;
+                int oy = (my * comp.vs + v) * 8;
+                for (int y = 0; y < 8; ++y)
+                  for (int x = 0; x < 8; ++x)
+                    comp.plane[(size_t)(oy + y) * comp.pw + ox + x] =
+                        pix[y * 8 + x] + 128.0f;
+              }
+            }
+          }
+        }
+      }
+      // --- color convert / upsample ---
+      out_h = height;
+      out_w = width;
+      out_c = ncomp == 1 ? 1 : 3;
+      std::vector<uint8_t> img((size_t)height * width * out_c);
+      auto clamp8 = [](float v) -> uint8_t {
+        return v < 0 ? 0 : (v > 255 ? 255 : (uint8_t)std::lround(v));
+      };
+      for (int y = 0; y < height; ++y) {
+        for (int x = 0; x < width; ++x) {
+          if (ncomp == 1) {
+            img[(size_t)y * width + x] =
+                clamp8(comps[0].plane[(size_t)y * comps[0].pw + x]);
+            continue;
+          }
+          auto sample = [&](const Component& cp) -> float {
+            int sx = x * cp.hs / hmax, sy = y * cp.vs / vmax;
+            return cp.plane[(size_t)sy * cp.pw + sx];
+          };
+          float Y = sample(comps[0]);
+          float cb = sample(comps[1]) - 128.0f;
+          float cr = sample(comps[2]) - 128.0f;
+          uint8_t* px = &img[((size_t)y * width + x) * 3];
+          px[0] = clamp8(Y + 1.402f * cr);
+          px[1] = clamp8(Y - 0.344136f * cb - 0.714136f * cr);
+          px[2] = clamp8(Y + 1.772f * cb);
+        }
+      }
+      return img;
+    } else {
+      // Skip APPn/COM/unknown.
+    }
+    pos = seg_end;
+  }
+  throw std::runtime_error("JPEG decode: no scan found");
+}
+
+}  // namespace t2r_jpeg

@@ -15,6 +15,13 @@ std::vector<at::Tensor> fused_bn_relu_backward(
     at::Tensor dy, at::Tensor x, at::Tensor gamma, at::Tensor beta,
     at::Tensor stats, bool fused_relu);
 
+std::vector<at::Tensor> maxpool_nhwc_forward(at::Tensor x, int64_t kh,
+                                             int64_t kw, bool ceil_mode);
+
+at::Tensor maxpool_nhwc_backward(at::Tensor dy, at::Tensor argmax,
+                                 int64_t N, int64_t C, int64_t H,
+                                 int64_t W, int64_t kh, int64_t kw);
+
 at::Tensor fused_preprocess(at::Tensor raw, int64_t oy, int64_t ox,
                             int64_t th, int64_t tw,
                             c10::optional<at::Tensor> delta_b,
@@ -28,6 +35,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "BN inference apply (+ReLU) (NHWC bf16)");
   m.def("fused_bn_relu_backward", &fused_bn_relu_backward,
         "Fused BN(+ReLU) backward (NHWC bf16)");
+  m.def("maxpool_nhwc_forward", &maxpool_nhwc_forward,
+        "Non-overlapping NHWC bf16 max-pool forward (+argmax)");
+  m.def("maxpool_nhwc_backward", &maxpool_nhwc_backward,
+        "Non-overlapping NHWC bf16 max-pool backward (gather)");
   m.def("fused_preprocess", &fused_preprocess,
         "Fused crop+convert+photometric distortion (uint8 NHWC -> f32/bf16)");
 }

@@ -27,6 +27,7 @@ from torch import nn
 
 from tensor2robot_amd import gin
 from tensor2robot_amd.ops import fused_bn
+from tensor2robot_amd.ops import maxpool as fused_maxpool
 
 
 def _conv_bn_relu(in_ch: int, out_ch: int, kernel: int, stride: int = 1,
@@ -51,10 +52,10 @@ class Grasping44(nn.Module):
     ch = channels
     self.conv1 = nn.Conv2d(3, ch, 6, stride=2, padding=2, bias=False)
     self.bn1 = fused_bn.FusedBatchNormReLU(ch, eps=0.001, momentum=0.003)
-    self.pool1 = nn.MaxPool2d(3, stride=3, ceil_mode=True)
+    self.pool1 = fused_maxpool.FusedMaxPool2d(3, ceil_mode=True)
     self.block1 = nn.Sequential(*[
         _conv_bn_relu(ch, ch, 5, padding=2) for _ in range(num_convs[0])])
-    self.pool2 = nn.MaxPool2d(3, stride=3, ceil_mode=True)
+    self.pool2 = fused_maxpool.FusedMaxPool2d(3, ceil_mode=True)
     self.fc_action1 = nn.Linear(action_dim, 256)
     self.bn_action = fused_bn.FusedBatchNormReLU(256, eps=0.001,
                                                  momentum=0.003)
@@ -63,7 +64,7 @@ class Grasping44(nn.Module):
                                                   momentum=0.003)
     self.block2 = nn.Sequential(*[
         _conv_bn_relu(ch, ch, 3, padding=1) for _ in range(num_convs[1])])
-    self.pool3 = nn.MaxPool2d(2, stride=2, ceil_mode=True)
+    self.pool3 = fused_maxpool.FusedMaxPool2d(2, ceil_mode=True)
     self.block3 = nn.Sequential(*[
         _conv_bn_relu(ch, ch, 3, padding=0) for _ in range(num_convs[2])])
     # 472 -> 236 -> 79 -> 27 -> 14 -> 12 -> 10 -> 8 spatial.

@@ -134,3 +134,40 @@ def test_grasping44_uses_hip_bn():
   with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
     out = net(x, a)
   assert out.shape == (2,)
+
+
+@requires_gpu
+def test_fused_maxpool_matches_torch():
+  from tensor2robot_amd.ops import maxpool as fmp
+  torch.manual_seed(0)
+  for n, c, h, w, k in [(2, 64, 27, 27, 2), (2, 64, 79, 79, 3),
+                        (1, 64, 236, 236, 3), (2, 32, 14, 14, 2)]:
+    x = torch.randn(n, c, h, w, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    pool = fmp.FusedMaxPool2d(k, ceil_mode=True)
+    y = pool(x)
+    y_ref = torch.nn.functional.max_pool2d(
+        x.detach().clone(), k, stride=k, ceil_mode=True)
+    assert y.shape == y_ref.shape
+    assert torch.equal(y.float(), y_ref.float()), (n, c, h, w, k)
+    # Backward gather vs torch's scatter.
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    x_ref = x.detach().clone().requires_grad_(True)
+    torch.nn.functional.max_pool2d(
+        x_ref, k, stride=k, ceil_mode=True).backward(dy)
+    assert torch.equal(x.grad.float(), x_ref.grad.float()), (n, c, h, w, k)
+
+
+@requires_gpu
+def test_fused_maxpool_floor_mode_zero_grads_outside():
+  from tensor2robot_amd.ops import maxpool as fmp
+  x = torch.randn(1, 8, 7, 7, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+  pool = fmp.FusedMaxPool2d(2, ceil_mode=False)
+  y = pool(x)
+  assert y.shape[-2:] == (3, 3)
+  y.sum().backward()
+  # Last row/col (outside any window) get zero grads.
+  assert torch.all(x.grad[:, :, 6, :] == 0)
+  assert torch.all(x.grad[:, :, :, 6] == 0)
