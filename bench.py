@@ -36,6 +36,8 @@ def parse_args():
   p.add_argument("--no-ema", action="store_true")
   p.add_argument("--no-preprocess", action="store_true",
                  help="feed pre-cropped 472x472 f32 (ablation only)")
+  p.add_argument("--no-hipgraph", action="store_true",
+                 help="disable hipGraph capture of the train step")
   return p.parse_args()
 
 
@@ -109,6 +111,17 @@ def main():
   autocast = torch.autocast(device_type="cuda", dtype=torch.bfloat16,
                             enabled=use_cuda)
 
+  def preprocess_to(images, action, static_image, static_action):
+    """Dynamic (host-RNG) preprocessing -> static graph inputs."""
+    with autocast:
+      features = build_features(model, images, action, run_modes.TRAIN,
+                                preprocess=not args.no_preprocess)
+      image = features["state/image"].permute(0, 3, 1, 2).contiguous(
+          memory_format=torch.channels_last)
+      actions = model.pack_action_vector(features)
+    static_image.copy_(image)
+    static_action.copy_(actions)
+
   def step(i, global_step):
     images, action, labels_t = pool[i % len(pool)]
     optimizer.zero_grad(set_to_none=True)
@@ -129,6 +142,57 @@ def main():
     if ema is not None:
       ema.update()
     return loss
+
+  # -- hipGraph capture (single graph launch per step; preprocess stays
+  # outside the graph because its distortion params come from host RNG) --
+  graphed = None
+  static = {}
+  if use_cuda and not args.no_hipgraph:
+    try:
+      from tensor2robot_amd.parallel import graph_step
+      for i in range(3):  # settle MIOpen algo find before capture
+        step(i, i)
+      torch.cuda.synchronize()
+      images0, action0, labels0 = pool[0]
+      dtype = torch.bfloat16
+      static["image"] = torch.zeros(
+          (args.batch_size, 3, t2r_models.CROP_HEIGHT,
+           t2r_models.CROP_WIDTH), dtype=dtype, device=device
+      ).contiguous(memory_format=torch.channels_last)
+      static["action"] = torch.zeros((args.batch_size,
+                                      t2r_models.ACTION_DIM),
+                                     dtype=dtype, device=device)
+      static["labels"] = torch.zeros((args.batch_size, 1), device=device)
+      preprocess_to(images0, action0, static["image"], static["action"])
+      static["labels"].copy_(labels0.reshape(-1, 1))
+
+      def graph_body():
+        optimizer.zero_grad(set_to_none=False)
+        with autocast:
+          logit = network(static["image"], static["action"])
+        loss = torch.nn.functional.binary_cross_entropy_with_logits(
+            logit.float(), static["labels"])
+        if dp_engine is not None:
+          dp_engine.backward(loss)
+        else:
+          loss.backward()
+        optimizer.step(0)
+        if ema is not None:
+          ema.update()
+        return loss
+
+      graphed = graph_step.GraphedTrainStep(graph_body)
+
+      def step(i, global_step):  # noqa: F811 (graph-replay fast path)
+        images, action, labels_t = pool[i % len(pool)]
+        preprocess_to(images, action, static["image"], static["action"])
+        static["labels"].copy_(labels_t.reshape(-1, 1))
+        return graphed.replay()
+    except Exception as e:  # pragma: no cover - depends on runtime
+      if rank == 0:
+        print(f"# hipGraph capture unavailable, running eager: {e!r}",
+              flush=True)
+      graphed = None
 
   def barrier_sync():
     if distributed:

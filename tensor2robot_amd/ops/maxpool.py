@@ -53,6 +53,9 @@ class FusedMaxPool2d(nn.Module):
     self.ceil_mode = ceil_mode
 
   def _use_hip(self, x: torch.Tensor) -> bool:
+    import os
+    if os.environ.get("T2R_DISABLE_FUSED_MAXPOOL"):
+      return False
     return (x.is_cuda and x.dtype == torch.bfloat16 and x.dim() == 4 and
             x.shape[1] % 8 == 0)
 
