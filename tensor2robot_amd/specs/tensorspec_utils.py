@@ -1020,3 +1020,61 @@ def write_t2r_assets_to_file(t2r_assets: T2RAssets, path: str):
 def load_t2r_assets_from_file(path: str) -> T2RAssets:
   with open(path) as f:
     return T2RAssets.from_pbtxt(f.read())
+
+
+# ---------------------------------------------------------------------------
+# Legacy pickle assets IO + converter (reference :1703-1732 and
+# utils/convert_pkl_assets_to_proto_assets.py:35-57)
+# ---------------------------------------------------------------------------
+
+T2R_ASSETS_FILENAME = "t2r_assets.pbtxt"
+INPUT_SPEC_PKL_FILENAME = "input_specifications.pkl"
+
+
+def write_input_spec_to_pkl_file(path: str, feature_spec, label_spec):
+  """Legacy pickle of (feature, label) specs (reference :1703-1714)."""
+  import pickle
+  payload = {
+      "feature_spec": [
+          (k, tuple(v.shape), str(v.dtype), v.name, v.is_optional,
+           v.is_sequence, v.data_format, v.dataset_key)
+          for k, v in flatten_spec_structure(feature_spec).items()],
+      "label_spec": [
+          (k, tuple(v.shape), str(v.dtype), v.name, v.is_optional,
+           v.is_sequence, v.data_format, v.dataset_key)
+          for k, v in flatten_spec_structure(label_spec).items()],
+  }
+  tmp = path + ".tmp"
+  with open(tmp, "wb") as f:
+    pickle.dump(payload, f)
+  import os
+  os.replace(tmp, path)
+
+
+def load_input_spec_from_pkl_file(path: str):
+  """Loads the legacy pickle into (feature, label) spec structs."""
+  import pickle
+  with open(path, "rb") as f:
+    payload = pickle.load(f)
+
+  def unpack(entries):
+    out = TensorSpecStruct()
+    for (k, shape, dtype, name, is_optional, is_sequence, data_format,
+         dataset_key) in entries:
+      out[k] = ExtendedTensorSpec(
+          shape, canonical_dtype(dtype.replace("torch.", "")), name=name,
+          is_optional=is_optional, is_sequence=is_sequence,
+          data_format=data_format, dataset_key=dataset_key)
+    return out
+
+  return unpack(payload["feature_spec"]), unpack(payload["label_spec"])
+
+
+def convert_pkl_assets_to_proto_assets(pkl_path: str, assets_path: str,
+                                       global_step: int = 0):
+  """Pickle -> pbtxt assets converter (reference converter binary)."""
+  feature_spec, label_spec = load_input_spec_from_pkl_file(pkl_path)
+  assets = T2RAssets(feature_spec=feature_spec, label_spec=label_spec,
+                     global_step=global_step)
+  write_t2r_assets_to_file(assets, assets_path)
+  return assets

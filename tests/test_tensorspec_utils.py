@@ -329,3 +329,24 @@ def test_t2r_assets_roundtrip(tmp_path):
   assert loaded.feature_spec["state/img"].shape == (8, 8, 3)
   assert loaded.feature_spec["state/img"].data_format == "JPEG"
   assert loaded.label_spec["target"].dtype == torch.float32
+
+
+def test_legacy_pkl_assets_roundtrip(tmp_path):
+  import torch
+  spec = tsu.TensorSpecStruct()
+  spec["state/image"] = tsu.ExtendedTensorSpec((64, 64, 3), torch.uint8,
+                                               name="img",
+                                               data_format="jpeg")
+  labels = tsu.TensorSpecStruct()
+  labels["pose"] = tsu.ExtendedTensorSpec((2,), torch.float32, name="pose")
+  pkl = str(tmp_path / "input_specifications.pkl")
+  tsu.write_input_spec_to_pkl_file(pkl, spec, labels)
+  f, l = tsu.load_input_spec_from_pkl_file(pkl)
+  assert tuple(f["state/image"].shape) == (64, 64, 3)
+  assert f["state/image"].data_format == "jpeg"
+  assert l["pose"].dtype == torch.float32
+  out = str(tmp_path / "t2r_assets.pbtxt")
+  assets = tsu.convert_pkl_assets_to_proto_assets(pkl, out, global_step=7)
+  assert assets.global_step == 7
+  loaded = tsu.load_t2r_assets_from_file(out)
+  assert "state/image" in loaded.feature_spec
