@@ -171,7 +171,7 @@ def main():
         with autocast:
           logit = network(static["image"], static["action"])
         loss = torch.nn.functional.binary_cross_entropy_with_logits(
-            logit.float(), static["labels"])
+            logit.float(), static["labels"].reshape(logit.shape))
         if dp_engine is not None:
           dp_engine.backward(loss)
         else:
