@@ -233,3 +233,20 @@ def test_exploration_policies():
       predictor=predictor, state_key="state/image", seed=0)
   a2 = sched.SelectAction(state)
   assert a2.shape == a0.shape
+
+
+def test_exported_predictor_serialized_examples(tmp_path):
+  """tf_example receiver parity: feed serialized Examples to a servable."""
+  import numpy as np
+  from tensor2robot_amd.data import example as example_mod
+  from tensor2robot_amd.export_generators import default_export_generator
+  from tensor2robot_amd.utils import modes as run_modes
+
+  model, export_dir = _train_and_export(tmp_path)
+  predictor = esp.ExportedSavedModelPredictor(export_dir, timeout=5)
+  assert predictor.is_loaded
+  records = [example_mod.encode_example(
+      {"measured_position": np.array([0.1, -0.2, 0.3], np.float32)})
+      for _ in range(4)]
+  out = predictor.predict_serialized(records)
+  assert out["prediction"].shape[0] == 4
