@@ -208,10 +208,19 @@ class GoldenValuesHook(TrainHook):
     self._values: List[Dict[str, np.ndarray]] = []
 
   def after_step(self, context, outputs):
-    if GOLDEN_COLLECTION:
-      self._values.append(
-          {k: v.numpy().copy() for k, v in GOLDEN_COLLECTION.items()})
-      GOLDEN_COLLECTION.clear()
+    step_values = {k: v.numpy().copy()
+                   for k, v in GOLDEN_COLLECTION.items()}
+    GOLDEN_COLLECTION.clear()
+    # The loss is always a golden value: even models that register no
+    # tensors get data->checkpoint determinism coverage.
+    loss = getattr(outputs, "loss", None)
+    if loss is None and isinstance(outputs, dict):
+      loss = outputs.get("loss")
+    if loss is not None:
+      step_values.setdefault(
+          "loss", np.asarray(float(loss), dtype=np.float64))
+    if step_values:
+      self._values.append(step_values)
 
   def end(self, context):
     os.makedirs(self._log_dir, exist_ok=True)
