@@ -74,7 +74,7 @@ class T2RModelFixture:
     t2r_model = getattr(module, model_name)(**module_kwargs)
     model_dir = self._tmpdir(model_dir)
     gen = input_generators.DefaultRecordInputGenerator(
-        file_patterns=file_patterns, batch_size=batch_size)
+        file_patterns=file_patterns, batch_size=batch_size, seed=123)
     result = train_eval.train_eval_model(
         t2r_model=t2r_model, input_generator_train=gen,
         input_generator_eval=None, max_train_steps=max_train_steps,
