@@ -190,8 +190,9 @@ def main():
         return graphed.replay()
     except Exception as e:  # pragma: no cover - depends on runtime
       if rank == 0:
+        import sys as _sys
         print(f"# hipGraph capture unavailable, running eager: {e!r}",
-              flush=True)
+              file=_sys.stderr, flush=True)
       graphed = None
 
   def barrier_sync():
