@@ -147,7 +147,12 @@ def main():
   # outside the graph because its distortion params come from host RNG) --
   graphed = None
   static = {}
-  if use_cuda and not args.no_hipgraph:
+  # hipGraph is single-GPU only this round: RCCL collectives inside a
+  # captured graph are unvalidated on this pool (no multi-GPU box to
+  # test on), and a hung capture on the 8-GPU scaling run is worse than
+  # the launch-overhead win.  The DP path keeps eager overlapped
+  # all-reduce.
+  if use_cuda and not distributed and not args.no_hipgraph:
     try:
       from tensor2robot_amd.parallel import graph_step
       for i in range(3):  # settle MIOpen algo find before capture

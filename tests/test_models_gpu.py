@@ -159,3 +159,52 @@ def test_maml_inner_loop_gpu():
   assert float(inner_losses[-1]) < float(inner_losses[0])
   outputs[1]["prediction"].pow(2).mean().backward()
   assert base.network.weight.grad is not None
+
+
+@requires_gpu
+def test_graphed_step_trains_grasping44():
+  """hipGraph replay must actually update weights (loss decreases)."""
+  from tensor2robot_amd.parallel import graph_step
+  from tensor2robot_amd.models import optimizers as optimizers_mod
+  from tensor2robot_amd.research.qtopt import t2r_models, networks
+  torch.manual_seed(0)
+  device = torch.device("cuda:0")
+  network = networks.Grasping44(action_dim=t2r_models.ACTION_DIM).to(
+      device).to(memory_format=torch.channels_last).train()
+  opt_factory = optimizers_mod.create_momentum_optimizer(
+      learning_rate=1e-3, momentum=0.9)
+  optimizer = opt_factory(network.parameters())
+  bs = 8
+  image = torch.rand(bs, 3, 472, 472, device=device).to(
+      torch.bfloat16).contiguous(memory_format=torch.channels_last)
+  action = torch.rand(bs, t2r_models.ACTION_DIM, device=device).to(
+      torch.bfloat16)
+  # Learnable rule: success iff mean pixel > 0.5 region — use fixed labels.
+  labels = (torch.rand(bs, device=device) > 0.5).float()
+  autocast = torch.autocast("cuda", dtype=torch.bfloat16)
+
+  def body():
+    optimizer.zero_grad(set_to_none=False)
+    with autocast:
+      logit = network(image, action)
+    loss = torch.nn.functional.binary_cross_entropy_with_logits(
+        logit.float(), labels.reshape(logit.shape))
+    loss.backward()
+    optimizer.step(0)
+    return loss
+
+  for _ in range(3):
+    body()
+  torch.cuda.synchronize()
+  graphed = graph_step.GraphedTrainStep(body)
+  first = float(graphed.replay())
+  params_before = [p.detach().clone() for p in network.parameters()]
+  for _ in range(30):
+    graphed.replay()
+  last = float(graphed.replay())
+  torch.cuda.synchronize()
+  assert last < first, (first, last)  # memorizes fixed labels
+  changed = sum(
+      0 if torch.equal(p.detach(), q) else 1
+      for p, q in zip(network.parameters(), params_before))
+  assert changed > 0
