@@ -15,6 +15,8 @@ std::vector<at::Tensor> fused_bn_relu_backward(
     at::Tensor dy, at::Tensor x, at::Tensor gamma, at::Tensor beta,
     at::Tensor stats, bool fused_relu);
 
+at::Tensor mfma_probe(at::Tensor A, at::Tensor B);
+
 std::vector<at::Tensor> maxpool_nhwc_forward(at::Tensor x, int64_t kh,
                                              int64_t kw, bool ceil_mode);
 
@@ -35,6 +37,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "BN inference apply (+ReLU) (NHWC bf16)");
   m.def("fused_bn_relu_backward", &fused_bn_relu_backward,
         "Fused BN(+ReLU) backward (NHWC bf16)");
+  m.def("mfma_probe", &mfma_probe, "32x32x16 bf16 MFMA layout probe");
   m.def("maxpool_nhwc_forward", &maxpool_nhwc_forward,
         "Non-overlapping NHWC bf16 max-pool forward (+argmax)");
   m.def("maxpool_nhwc_backward", &maxpool_nhwc_backward,
