@@ -22,6 +22,7 @@ hip_sources = [
     os.path.join(HIP_DIR, "preprocess.hip"),
     os.path.join(HIP_DIR, "maxpool.hip"),
     os.path.join(HIP_DIR, "mfma_probe.hip"),
+    os.path.join(HIP_DIR, "conv_s1.hip"),
 ]
 
 import pybind11

@@ -1,0 +1,107 @@
+"""MFMAConv2d: hand-written MFMA conv for stride-1 NHWC bf16.
+
+Wraps ops/hip/conv_s1.hip. Forward and backward-data both run the MFMA
+kernel (backward-data is the same convolution with spatially-flipped,
+channel-transposed weights); the weight gradient uses
+aten.convolution_backward (MIOpen wrw) — wrw is a [64 x 64] x 200k-K
+reduction where MIOpen's split-K igemm is already reasonable.
+
+Weights are prepacked per call to the kernel's LDS-friendly layout
+[rs][c/16][k][24] (16 used + 8 pad for conflict-free B-fragment reads);
+the pack is a few tensor ops on a 200 KB tensor and is captured inside
+hipGraphs along with the conv.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+from torch import nn
+import torch.nn.functional as F
+
+from tensor2robot_amd import ops as ops_mod
+
+WPAD = 24
+
+
+def pack_weights(w: torch.Tensor) -> torch.Tensor:
+  """[K, C, R, S] -> flat [R*S, C/16, K, 24] bf16 (kernel layout)."""
+  k, c, r, s = w.shape
+  wp = w.permute(2, 3, 1, 0).reshape(r * s, c // 16, 16, k)
+  wp = wp.permute(0, 1, 3, 2).contiguous()          # [rs][c16][k][16]
+  out = torch.zeros(r * s, c // 16, k, WPAD, dtype=torch.bfloat16,
+                    device=w.device)
+  out[..., :16] = wp.to(torch.bfloat16)
+  return out
+
+
+def pack_weights_bwd(w: torch.Tensor) -> torch.Tensor:
+  """Pack for backward-data: flip spatially, swap in/out channels."""
+  wb = w.flip(2, 3).permute(1, 0, 2, 3).contiguous()
+  return pack_weights(wb)
+
+
+def _supported(x: torch.Tensor, weight: torch.Tensor, stride,
+               padding) -> bool:
+  import os
+  if os.environ.get("T2R_DISABLE_MFMA_CONV"):
+    return False
+  if not (x.is_cuda and x.dtype == torch.bfloat16):
+    return False
+  k, c, r, s = weight.shape
+  return (stride == (1, 1) and r <= 5 and s <= 5 and
+          c % 16 == 0 and c <= 64 and k % 32 == 0 and k <= 64 and
+          padding[0] == padding[1])
+
+
+class _MFMAConvFunction(torch.autograd.Function):
+
+  @staticmethod
+  def forward(ctx, x, weight, pad):
+    ext = ops_mod.require_hip()
+    k, c, r, s = weight.shape
+    if not x.is_contiguous(memory_format=torch.channels_last):
+      x = x.contiguous(memory_format=torch.channels_last)
+    wpk = pack_weights(weight)
+    y = ext.conv_s1_nhwc(x, wpk, k, r, s, pad)
+    ctx.save_for_backward(x, weight)
+    ctx.pad = pad
+    return y
+
+  @staticmethod
+  def backward(ctx, dy):
+    ext = ops_mod.require_hip()
+    x, weight = ctx.saved_tensors
+    k, c, r, s = weight.shape
+    dy = dy.contiguous(memory_format=torch.channels_last)
+    dx = dw = None
+    if ctx.needs_input_grad[0]:
+      # SAME-pad duality: the dy->dx conv pad is (R-1-pad).
+      bpad = r - 1 - ctx.pad
+      wpk_b = pack_weights_bwd(weight)
+      dx = ext.conv_s1_nhwc(dy.to(torch.bfloat16), wpk_b, c, r, s, bpad)
+    if ctx.needs_input_grad[1]:
+      dw = torch.ops.aten.convolution_backward(
+          dy.float(), x.float(), weight.float(), None, (1, 1),
+          (ctx.pad, ctx.pad), (1, 1), False, (0, 0), 1,
+          (False, True, False))[1].to(weight.dtype)
+    return dx, dw, None
+
+
+class MFMAConv2d(nn.Conv2d):
+  """Conv2d that runs the MFMA kernel on supported GPU bf16 shapes."""
+
+  def __init__(self, in_channels, out_channels, kernel_size, stride=1,
+               padding=0, bias=False):
+    super().__init__(in_channels, out_channels, kernel_size,
+                     stride=stride, padding=padding, bias=bias)
+
+  def forward(self, x):
+    if self.bias is None and _supported(x, self.weight, self.stride,
+                                        self.padding):
+      w = self.weight
+      if w.dtype != torch.bfloat16:
+        w = w.to(torch.bfloat16)
+      return _MFMAConvFunction.apply(x, w, self.padding[0])
+    return super().forward(x)

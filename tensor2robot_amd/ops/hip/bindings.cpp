@@ -17,6 +17,9 @@ std::vector<at::Tensor> fused_bn_relu_backward(
 
 at::Tensor mfma_probe(at::Tensor A, at::Tensor B);
 
+at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
+                        int64_t R, int64_t S, int64_t pad);
+
 std::vector<at::Tensor> maxpool_nhwc_forward(at::Tensor x, int64_t kh,
                                              int64_t kw, bool ceil_mode);
 
@@ -38,6 +41,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_bn_relu_backward", &fused_bn_relu_backward,
         "Fused BN(+ReLU) backward (NHWC bf16)");
   m.def("mfma_probe", &mfma_probe, "32x32x16 bf16 MFMA layout probe");
+  m.def("conv_s1_nhwc", &conv_s1_nhwc,
+        "MFMA stride-1 NHWC bf16 conv (prepacked weights)");
   m.def("maxpool_nhwc_forward", &maxpool_nhwc_forward,
         "Non-overlapping NHWC bf16 max-pool forward (+argmax)");
   m.def("maxpool_nhwc_backward", &maxpool_nhwc_backward,

@@ -26,3 +26,67 @@ def test_mfma_32x32x16_bf16_layout():
   D2 = _t2r_hip.mfma_probe(I, B2)
   ref2 = I.float() @ B2.float()
   assert torch.allclose(D2, ref2, atol=2e-2, rtol=2e-2)
+
+
+@requires_gpu
+@pytest.mark.parametrize("shape", [
+    # (N, C, H, W, K, R, pad)
+    (2, 64, 79, 79, 64, 5, 2),    # Grasping44 block1
+    (2, 64, 27, 27, 64, 3, 1),    # block2 SAME
+    (2, 64, 14, 14, 64, 3, 0),    # block3 VALID
+    (1, 64, 33, 17, 64, 5, 2),    # ragged edges
+    (1, 32, 16, 16, 32, 3, 1),    # smaller C/K
+])
+def test_mfma_conv_forward_matches_torch(shape):
+  import torch.nn.functional as F
+  from tensor2robot_amd.ops import conv as mconv
+  n, c, h, w, k, r, pad = shape
+  torch.manual_seed(0)
+  x = torch.randn(n, c, h, w, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last)
+  weight = (torch.randn(k, c, r, r, device="cuda") * 0.1).to(
+      torch.bfloat16)
+  y = mconv._MFMAConvFunction.apply(x, weight, pad)
+  ref = F.conv2d(x.float(), weight.float(), padding=pad)
+  err = (y.float() - ref).abs().max().item()
+  scale = ref.abs().max().item()
+  assert err < 0.02 * max(scale, 1.0), (shape, err, scale)
+
+
+@requires_gpu
+def test_mfma_conv_backward_matches_torch():
+  import torch.nn.functional as F
+  from tensor2robot_amd.ops import conv as mconv
+  torch.manual_seed(0)
+  n, c, h, w, k, r, pad = 2, 64, 27, 27, 64, 3, 1
+  x = torch.randn(n, c, h, w, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+  weight = (torch.randn(k, c, r, r, device="cuda") * 0.1).to(
+      torch.bfloat16).requires_grad_(True)
+  y = mconv._MFMAConvFunction.apply(x, weight, pad)
+  dy = torch.randn_like(y)
+  y.backward(dy)
+
+  x2 = x.detach().float().requires_grad_(True)
+  w2 = weight.detach().float().requires_grad_(True)
+  F.conv2d(x2, w2, padding=pad).backward(dy.float())
+  dx_err = (x.grad.float() - x2.grad).abs().max().item()
+  dx_scale = x2.grad.abs().max().item()
+  assert dx_err < 0.03 * max(dx_scale, 1.0), (dx_err, dx_scale)
+  dw_err = (weight.grad.float() - w2.grad).abs().max().item()
+  assert dw_err < 0.03 * max(w2.grad.abs().max().item(), 1.0)
+
+
+@requires_gpu
+def test_mfma_conv_module_dispatch():
+  from tensor2robot_amd.ops import conv as mconv
+  m = mconv.MFMAConv2d(64, 64, 5, padding=2).cuda().to(
+      memory_format=torch.channels_last)
+  x = torch.randn(2, 64, 33, 33, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last)
+  y = m(x)
+  assert y.shape == (2, 64, 33, 33)
+  # Unsupported shape falls back to torch conv.
+  m2 = mconv.MFMAConv2d(3, 64, 6, stride=2, padding=2).cuda()
+  x2 = torch.randn(2, 3, 64, 64, device="cuda")
+  assert m2(x2).shape[1] == 64
