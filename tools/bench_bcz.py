@@ -1,0 +1,102 @@
+"""BC-Z training throughput on one MI355X (BASELINE config #3 evidence).
+
+Full step: on-GPU preprocess (crop 472^2 of 512x640 -> resize 100^2 ->
+distort) + FiLM-ResNet forward + component losses + backward + Adam.
+
+  python tools/bench_bcz.py [--steps 50] [--warmup 15] [--batch-size 32]
+"""
+
+import argparse
+import functools
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from tensor2robot_amd.models import optimizers
+from tensor2robot_amd.research.bcz import model as bcz_model
+from tensor2robot_amd.specs import tensorspec_utils as tsu
+from tensor2robot_amd.utils import modes as run_modes
+
+
+def main():
+  p = argparse.ArgumentParser()
+  p.add_argument("--steps", type=int, default=50)
+  p.add_argument("--warmup", type=int, default=15)
+  p.add_argument("--batch-size", type=int, default=32)
+  p.add_argument("--resnet-size", type=int, default=18)
+  args = p.parse_args()
+  assert torch.cuda.is_available()
+  torch.backends.cudnn.benchmark = True
+  device = torch.device("cuda:0")
+
+  model = bcz_model.BCZModel(
+      image_size=(100, 100), input_size=(512, 640), num_waypoints=10,
+      resnet_size=args.resnet_size,
+      preprocessor_cls=functools.partial(
+          bcz_model.BCZPreprocessor, image_size=(100, 100),
+          input_size=(512, 640), crop_size=(472, 472), mock_subtask=True),
+      create_optimizer_fn=lambda: optimizers.create_adam_optimizer(1e-4),
+      device_type="gpu", compute_dtype="bfloat16")
+  model.to_device(device)
+  model.network.to(memory_format=torch.channels_last)
+  optimizer = model.create_optimizer()
+
+  bs = args.batch_size
+  g = torch.Generator().manual_seed(0)
+  features = tsu.TensorSpecStruct()
+  features["image"] = torch.randint(0, 256, (bs, 512, 640, 3),
+                                    dtype=torch.uint8,
+                                    generator=g).to(device)
+  features["subtask_id"] = torch.zeros(bs, 1, dtype=torch.int64,
+                                       device=device)
+  for name, size, _, _ in model._action_components:
+    features["present/" + name] = torch.rand(bs, size,
+                                             generator=g).to(device)
+  labels = tsu.TensorSpecStruct()
+  labels["future/xyz_residual"] = torch.randn(bs, 10, 3,
+                                              generator=g).to(device)
+  labels["future/quaternion"] = torch.randn(bs, 10, 4,
+                                            generator=g).to(device)
+  labels["future/target_close"] = (torch.rand(bs, 10, 1, generator=g)
+                                   > 0.5).float().to(device)
+  autocast = torch.autocast("cuda", dtype=torch.bfloat16)
+
+  def step(i):
+    optimizer.zero_grad(set_to_none=True)
+    with autocast:
+      f = tsu.TensorSpecStruct()
+      for k, v in features.items():
+        f[k] = v
+      l = tsu.TensorSpecStruct()
+      for k, v in labels.items():
+        l[k] = v
+      f, l = model.preprocessor.preprocess(f, l, run_modes.TRAIN)
+      ops = model.model_fn(f, l, run_modes.TRAIN)
+    ops.loss.backward()
+    optimizer.step(i)
+    return ops.loss
+
+  for i in range(args.warmup):
+    step(i)
+  torch.cuda.synchronize()
+  t0 = time.perf_counter()
+  for i in range(args.steps):
+    step(args.warmup + i)
+  torch.cuda.synchronize()
+  elapsed = time.perf_counter() - t0
+  print(json.dumps({
+      "metric": "images/sec BC-Z FiLM-ResNet%d train, 100x100 "
+                "(512x640 raw), bs=%d" % (args.resnet_size, bs),
+      "value": round(bs * args.steps / elapsed, 2),
+      "ms_per_step": round(elapsed / args.steps * 1000, 3),
+      "dtype": "bf16", "data": "synthetic", "n_gpus": 1,
+  }))
+
+
+if __name__ == "__main__":
+  main()
