@@ -120,10 +120,34 @@ conv_s1_nhwc_kernel(const cbf16_t* __restrict__ x,
   const int prow = (wave * 32 + mrow) / TILE_W;
   const int pcol = (wave * 32 + mrow) % TILE_W;
 
+  // T14-style register prefetch: issue chunk rs+1's global loads BEFORE
+  // the rs MFMAs (latency hides under compute), write them to the spare
+  // LDS buffer after the consumers' barrier.
   const int RS = R * S;
+  uint4 wreg[3];                    // 3 x 16 B per thread covers 12 KiB
+  const int wpieces = glds_per_chunk;  // <= 768
+  auto fetch_w = [&](int rs) {
+    const cbf16_t* src = wpk + (long)rs * wchunk_bf16;
+#pragma unroll
+    for (int j = 0; j < 3; ++j) {
+      const int i = tid + j * 256;
+      if (i < wpieces)
+        wreg[j] = *reinterpret_cast<const uint4*>(&src[i * 8]);
+    }
+  };
+  auto write_w = [&](int buf) {
+    short* dst = wbuf(buf);
+#pragma unroll
+    for (int j = 0; j < 3; ++j) {
+      const int i = tid + j * 256;
+      if (i < wpieces)
+        *reinterpret_cast<uint4*>(&dst[i * 8]) = wreg[j];
+    }
+  };
   for (int rs = 0; rs < RS; ++rs) {
     const int r = rs / S, s = rs % S;
     const int buf = rs & 1;
+    if (rs + 1 < RS) fetch_w(rs + 1);  // loads in flight over the MFMAs
     for (int c16 = 0; c16 < c16n; ++c16) {
       cbf16x8 a_frag = *reinterpret_cast<const cbf16x8*>(
           &xtile[((prow + r) * HALO_W + (pcol + s)) * XPITCH
@@ -140,7 +164,7 @@ conv_s1_nhwc_kernel(const cbf16_t* __restrict__ x,
     }
     if (rs + 1 < RS) {
       __syncthreads();              // wbuf[buf^1] consumers done
-      stage_w(rs + 1, buf ^ 1);
+      write_w(buf ^ 1);
       __syncthreads();              // wbuf[buf^1] ready
     }
   }
