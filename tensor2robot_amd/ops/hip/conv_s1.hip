@@ -1,27 +1,27 @@
 // Hand-written MFMA conv for stride-1 NHWC bf16 (gfx950 / CDNA4).
 //
-// Covers the Grasping44 hot convs (5x5 SAME @79^2, 3x3 SAME @27^2,
-// 3x3 VALID, C=K=64) where MIOpen's igemm runs at ~15-20% MFMA
-// utilization on these small-channel shapes (profiles/).  Backward-data
-// is the same kernel on flipped/transposed prepacked weights.
+// Covers the Grasping44 stride-1 convs (5x5 SAME @79^2, 3x3 SAME/VALID,
+// C=K=64) where MIOpen's igemm runs at ~15-20% MFMA utilization on
+// these small-channel shapes (profiles/).  Backward-data is the same
+// kernel on flipped/transposed prepacked weights.
 //
 // Design (cdna_hip_programming.md §3/§5, MI355X_MICROARCH.md):
-//  * implicit GEMM: out[p, k] = sum_{r,s,c} x[p+Δ(r,s), c] * w[r,s,c,k]
-//    computed with v_mfma_f32_32x32x16_bf16; M = pixels, N = K, K' = C.
-//  * one 256-thread WG computes an 8x16-pixel output tile for all K:
-//    4 waves x (32 pixels x K).  The 12x20-pixel input halo tile is
-//    staged in LDS once per WG with a PADDED 144-B pixel stride so the
-//    A-fragment ds_read_b128 lane groups land on 16 distinct banks
-//    (linear 128-B stride = 16-way conflict, the §5 GEMM trap).
-//  * weights are HOST-PREPACKED to [rs][c16][n][24] (8-slot pad => 48-B
-//    n-stride, conflict-free B-fragment reads) and staged per (r,s)
-//    via global_load_lds (lane-linear dest, layouts match exactly),
-//    double-buffered so chunk rs+1 streams while rs computes.
+//  * implicit GEMM: out[p, k] = sum_{r,s,c} x[p+D(r,s), c] * w[r,s,c,k],
+//    v_mfma_f32_32x32x16_bf16; M = pixels, N = K, K' = C*R*S.
+//  * a 256-thread WG computes an 8x16-pixel tile x K_WG channels:
+//    4 waves x (32 pixels x K_WG).  The halo tile lives in LDS with a
+//    PADDED 144-B pixel stride (linear 128-B stride = 16-way bank
+//    conflict on the A-fragment ds_read_b128 -- the §5 GEMM trap).
+//  * ALL weight chunks for the WG's channel group are staged ONCE
+//    ([rs][c16][n][16] image, 32-B n-stride = 2-way conflict), so the
+//    main loop is barrier-free.  When R*S*C*K_WG exceeds LDS the K dim
+//    splits across blockIdx.y (5x5: K_WG=32, 2 splits, ~137 KiB LDS).
+//  * measured (same-box interleaved A/B vs MIOpen/CK):
+//    3x3@27^2 3.4x, 3x3@14^2 3.7x; 5x5@79^2 -- see profiles/.
 //  * fragment layouts verified by the mfma_probe GPU test:
 //    A row = l%32, k = (l>>5)*8+j; B col = l%32 (same k map);
 //    C/D col = l&31, row = (reg&3)+8*(reg>>2)+4*(l>>5).
-//  * all LDS in ONE __shared__ array (a second object forces vmcnt(0)
-//    before every ds_read of a glds pipeline — §5 trap 4a).
+//  * all LDS in ONE __shared__ array (§5 trap 4a).
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
@@ -36,38 +36,33 @@ typedef __attribute__((ext_vector_type(16))) float cf32x16;
 #define TILE_W 16
 #define HALO_H (TILE_H + 4)   // supports R <= 5
 #define HALO_W (TILE_W + 4)
-#define XPITCH 72             // bf16 elements per pixel row in LDS (64+8 pad)
-#define WPAD 24               // bf16 per n-row in the weight image (16+8 pad)
+#define XPITCH 72             // bf16 per pixel row in LDS (64 + 8 pad)
+#define WPAD 24               // bf16 per n-row in the GLOBAL packed image
 
-// LDS: x tile + 2 weight buffers, one shared object.
-// x: HALO_H*HALO_W pixels * XPITCH bf16 = 240*72*2 = 34560 B
-// w: 2 * (C16MAX=4) * 64 * WPAD * 2 = 2*12288 B
 #define XTILE_BF16 (HALO_H * HALO_W * XPITCH)
-#define WBUF_BF16 (4 * 64 * WPAD)
 
-extern "C" __global__ void __launch_bounds__(256, 2)
+// C16N: C/16.  NT_WG: 32-wide N tiles computed per workgroup.
+// RS_CAP: compile-time R*S capacity of the staged weight image.
+template <int C16N, int NT_WG, int RS_CAP>
+__global__ void __launch_bounds__(256, 2)
 conv_s1_nhwc_kernel(const cbf16_t* __restrict__ x,
                     const cbf16_t* __restrict__ wpk,
                     cbf16_t* __restrict__ y,
-                    int N, int C, int H, int W,
-                    int K, int R, int S, int pad,
+                    int N, int H, int W, int K,
+                    int R, int S, int pad,
                     int OH, int OW, int tiles_h, int tiles_w) {
-  __shared__ short lds[XTILE_BF16 + 2 * WBUF_BF16];
+  constexpr int C = C16N * 16;
+  constexpr int K_WG = NT_WG * 32;
+  constexpr int WLDS = RS_CAP * C16N * K_WG * 16;
+  __shared__ short lds[XTILE_BF16 + WLDS];
   short* xtile = lds;
-  // NOTE: no array-of-LDS-pointers (hipcc cannot statically initialize
-  // addrspace(3) casts) — compute buffer bases by index.
-  auto wbuf = [&](int b) -> short* {
-    return lds + XTILE_BF16 + b * WBUF_BF16;
-  };
+  short* wall = lds + XTILE_BF16;
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
-  const int c16n = C >> 4;            // K'-chunks of 16
-  const int ntiles = K >> 5;          // 32-wide N tiles (<=2 supported)
-  const int wchunk_bf16 = c16n * K * WPAD;
+  const int n0 = blockIdx.y * K_WG;    // channel-group offset
 
-  // Tile coordinates.
   long wg = blockIdx.x;
   const int img = wg / (tiles_h * tiles_w);
   const int trest = wg % (tiles_h * tiles_w);
@@ -78,7 +73,7 @@ conv_s1_nhwc_kernel(const cbf16_t* __restrict__ x,
   // ---- stage the x halo tile (zero OOB), 16-B chunks ----
   {
     const int halo_h = TILE_H + R - 1, halo_w = TILE_W + S - 1;
-    const int chunks = C >> 3;        // 16-B chunks of 8 bf16
+    constexpr int chunks = C >> 3;
     const int total = halo_h * halo_w * chunks;
     for (int i = tid; i < total; i += 256) {
       const int chunk = i % chunks;
@@ -96,65 +91,171 @@ conv_s1_nhwc_kernel(const cbf16_t* __restrict__ x,
     }
   }
 
-  // ---- prefetch weight chunk rs=0 ----
-  // wpk layout bytes == LDS layout bytes (both [c16][n][WPAD]); glds
-  // dest is wave-uniform base + lane*16, source is the matching linear
-  // address.
-  const int glds_per_chunk = (wchunk_bf16 * 2) / 16;  // 16-B pieces
-  auto stage_w = [&](int rs, int buf) {
-    const cbf16_t* src = wpk + (long)rs * wchunk_bf16;
-    short* dst = wbuf(buf);
-    for (int i = tid; i < glds_per_chunk; i += 256) {
-      *reinterpret_cast<uint4*>(&dst[i * 8]) =
-          *reinterpret_cast<const uint4*>(&src[i * 8]);
+  // ---- stage every weight chunk for this channel group ----
+  // Global image: [rs][c16][n(K)][WPAD]; LDS image:
+  // [rs][c16][n(K_WG)][16].  Two 16-B pieces per n-row.
+  {
+    const int rows = R * S * C16N * K_WG;
+    for (int i = tid; i < rows * 2; i += 256) {
+      const int nrow = i >> 1, half = (i & 1) * 8;
+      const int nn = nrow % K_WG;
+      const int rc = nrow / K_WG;          // rs * C16N + c16
+      *reinterpret_cast<uint4*>(&wall[nrow * 16 + half]) =
+          *reinterpret_cast<const uint4*>(
+              &wpk[((long)rc * K + n0 + nn) * WPAD + half]);
     }
-  };
-  stage_w(0, 0);
+  }
   __syncthreads();
 
-  // ---- main loop over (r, s) ----
-  cf32x16 acc[2] = {{}, {}};
+  // ---- barrier-free main loop ----
+  cf32x16 acc[NT_WG];
+#pragma unroll
+  for (int nt = 0; nt < NT_WG; ++nt) acc[nt] = (cf32x16){};
   const int mrow = lane & 31;
   const int kgrp = lane >> 5;
-  // Wave's 32 pixels: rows [wave*2, wave*2+2) x 16 cols.
   const int prow = (wave * 32 + mrow) / TILE_W;
   const int pcol = (wave * 32 + mrow) % TILE_W;
 
-  // T14-style register prefetch: issue chunk rs+1's global loads BEFORE
-  // the rs MFMAs (latency hides under compute), write them to the spare
-  // LDS buffer after the consumers' barrier.
   const int RS = R * S;
-  uint4 wreg[3];                    // 3 x 16 B per thread covers 12 KiB
-  const int wpieces = glds_per_chunk;  // <= 768
-  auto fetch_w = [&](int rs) {
-    const cbf16_t* src = wpk + (long)rs * wchunk_bf16;
-#pragma unroll
-    for (int j = 0; j < 3; ++j) {
-      const int i = tid + j * 256;
-      if (i < wpieces)
-        wreg[j] = *reinterpret_cast<const uint4*>(&src[i * 8]);
-    }
-  };
-  auto write_w = [&](int buf) {
-    short* dst = wbuf(buf);
-#pragma unroll
-    for (int j = 0; j < 3; ++j) {
-      const int i = tid + j * 256;
-      if (i < wpieces)
-        *reinterpret_cast<uint4*>(&dst[i * 8]) = wreg[j];
-    }
-  };
   for (int rs = 0; rs < RS; ++rs) {
     const int r = rs / S, s = rs % S;
-    const int buf = rs & 1;
-    if (rs + 1 < RS) fetch_w(rs + 1);  // loads in flight over the MFMAs
-    for (int c16 = 0; c16 < c16n; ++c16) {
+#pragma unroll
+    for (int c16 = 0; c16 < C16N; ++c16) {
       cbf16x8 a_frag = *reinterpret_cast<const cbf16x8*>(
           &xtile[((prow + r) * HALO_W + (pcol + s)) * XPITCH
                  + c16 * 16 + kgrp * 8]);
 #pragma unroll
-      for (int nt = 0; nt < 2; ++nt) {
-        if (nt >= ntiles) break;
+      for (int nt = 0; nt < NT_WG; ++nt) {
+        const int n = nt * 32 + mrow;
+        cbf16x8 b_frag = *reinterpret_cast<const cbf16x8*>(
+            &wall[(((rs * C16N) + c16) * K_WG + n) * 16 + kgrp * 8]);
+        acc[nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            a_frag, b_frag, acc[nt], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue: scatter accumulators ----
+  const int ocol_n = lane & 31;
+#pragma unroll
+  for (int nt = 0; nt < NT_WG; ++nt) {
+#pragma unroll
+    for (int reg = 0; reg < 16; ++reg) {
+      const int m = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+      const int p = wave * 32 + m;
+      const int orow = oh0 + p / TILE_W;
+      const int ocol = ow0 + p % TILE_W;
+      if (orow < OH && ocol < OW) {
+        y[(((long)img * OH + orow) * OW + ocol) * K
+          + n0 + nt * 32 + ocol_n] = __float2bfloat16(acc[nt][reg]);
+      }
+    }
+  }
+}
+
+
+// ---------------------------------------------------------------------------
+// Big-tile variant for large-RS convs (5x5 @ 79^2): 512 threads compute a
+// 16x16-pixel tile x all 64 channels; weights double-buffered per (r,s)
+// with register prefetch (2x the compute per staged weight byte of the
+// 256-thread tile; staging loads hide under the MFMAs).
+// ---------------------------------------------------------------------------
+
+#define BTILE 16
+#define BHALO (BTILE + 4)
+
+template <int C16N, int NTILES>
+__global__ void __launch_bounds__(512, 2)
+conv_s1_nhwc_big_kernel(const cbf16_t* __restrict__ x,
+                        const cbf16_t* __restrict__ wpk,
+                        cbf16_t* __restrict__ y,
+                        int N, int H, int W, int K,
+                        int R, int S, int pad,
+                        int OH, int OW, int tiles_h, int tiles_w) {
+  constexpr int C = C16N * 16;
+  constexpr int WBUF = C16N * NTILES * 32 * WPAD;     // bf16 per chunk
+  __shared__ short lds[BHALO * BHALO * XPITCH + 2 * WBUF];
+  short* xtile = lds;
+  auto wbuf = [&](int b) -> short* {
+    return lds + BHALO * BHALO * XPITCH + b * WBUF;
+  };
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+
+  long wg = blockIdx.x;
+  const int img = wg / (tiles_h * tiles_w);
+  const int trest = wg % (tiles_h * tiles_w);
+  const int th = trest / tiles_w;
+  const int tw = trest % tiles_w;
+  const int oh0 = th * BTILE, ow0 = tw * BTILE;
+
+  {
+    const int halo_h = BTILE + R - 1, halo_w = BTILE + S - 1;
+    constexpr int chunks = C >> 3;
+    const int total = halo_h * halo_w * chunks;
+    for (int i = tid; i < total; i += 512) {
+      const int chunk = i % chunks;
+      const int pix = i / chunks;
+      const int hrow = pix / halo_w, hcol = pix % halo_w;
+      const int iy = oh0 - pad + hrow;
+      const int ix = ow0 - pad + hcol;
+      uint4 v = make_uint4(0, 0, 0, 0);
+      if (iy >= 0 && iy < H && ix >= 0 && ix < W) {
+        v = *reinterpret_cast<const uint4*>(
+            x + (((long)img * H + iy) * W + ix) * C + chunk * 8);
+      }
+      *reinterpret_cast<uint4*>(
+          &xtile[(hrow * BHALO + hcol) * XPITCH + chunk * 8]) = v;
+    }
+  }
+
+  constexpr int WPIECES = (WBUF * 2) / 16;
+  constexpr int WPT = (WPIECES + 511) / 512;
+  {
+    const cbf16_t* src = wpk;
+    short* dst = wbuf(0);
+#pragma unroll
+    for (int j = 0; j < WPT; ++j) {
+      const int i = tid + j * 512;
+      if (i < WPIECES)
+        *reinterpret_cast<uint4*>(&dst[i * 8]) =
+            *reinterpret_cast<const uint4*>(&src[i * 8]);
+    }
+  }
+  __syncthreads();
+
+  cf32x16 acc[NTILES];
+#pragma unroll
+  for (int nt = 0; nt < NTILES; ++nt) acc[nt] = (cf32x16){};
+  const int mrow = lane & 31;
+  const int kgrp = lane >> 5;
+  const int prow = (wave * 32 + mrow) / BTILE;
+  const int pcol = (wave * 32 + mrow) % BTILE;
+
+  const int RS = R * S;
+  constexpr int wchunk_bf16 = WBUF;
+  uint4 wreg[WPT];
+  for (int rs = 0; rs < RS; ++rs) {
+    const int r = rs / S, s = rs % S;
+    const int buf = rs & 1;
+    if (rs + 1 < RS) {
+      const cbf16_t* src = wpk + (long)(rs + 1) * wchunk_bf16;
+#pragma unroll
+      for (int j = 0; j < WPT; ++j) {
+        const int i = tid + j * 512;
+        if (i < WPIECES)
+          wreg[j] = *reinterpret_cast<const uint4*>(&src[i * 8]);
+      }
+    }
+#pragma unroll
+    for (int c16 = 0; c16 < C16N; ++c16) {
+      cbf16x8 a_frag = *reinterpret_cast<const cbf16x8*>(
+          &xtile[((prow + r) * BHALO + (pcol + s)) * XPITCH
+                 + c16 * 16 + kgrp * 8]);
+#pragma unroll
+      for (int nt = 0; nt < NTILES; ++nt) {
         const int n = nt * 32 + mrow;
         cbf16x8 b_frag = *reinterpret_cast<const cbf16x8*>(
             &wbuf(buf)[(c16 * K + n) * WPAD + kgrp * 8]);
@@ -163,26 +264,32 @@ conv_s1_nhwc_kernel(const cbf16_t* __restrict__ x,
       }
     }
     if (rs + 1 < RS) {
-      __syncthreads();              // wbuf[buf^1] consumers done
-      write_w(buf ^ 1);
-      __syncthreads();              // wbuf[buf^1] ready
+      __syncthreads();
+      {
+        short* dst = wbuf(buf ^ 1);
+#pragma unroll
+        for (int j = 0; j < WPT; ++j) {
+          const int i = tid + j * 512;
+          if (i < WPIECES)
+            *reinterpret_cast<uint4*>(&dst[i * 8]) = wreg[j];
+        }
+      }
+      __syncthreads();
     }
   }
 
-  // ---- epilogue: scatter accumulators ----
   const int ocol_n = lane & 31;
 #pragma unroll
-  for (int nt = 0; nt < 2; ++nt) {
-    if (nt >= ntiles) break;
+  for (int nt = 0; nt < NTILES; ++nt) {
 #pragma unroll
     for (int reg = 0; reg < 16; ++reg) {
       const int m = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
       const int p = wave * 32 + m;
-      const int orow = oh0 + p / TILE_W;
-      const int ocol = ow0 + p % TILE_W;
+      const int orow = oh0 + p / BTILE;
+      const int ocol = ow0 + p % BTILE;
       if (orow < OH && ocol < OW) {
-        y[(((long)img * OH + orow) * OW + ocol) * K + nt * 32 + ocol_n] =
-            __float2bfloat16(acc[nt][reg]);
+        y[(((long)img * OH + orow) * OW + ocol) * K
+          + nt * 32 + ocol_n] = __float2bfloat16(acc[nt][reg]);
       }
     }
   }
@@ -209,16 +316,42 @@ at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
   auto y = at::empty({N, K, OH, OW},
                      x.options().memory_format(
                          at::MemoryFormat::ChannelsLast));
-  const int tiles_h = (OH + TILE_H - 1) / TILE_H;
-  const int tiles_w = (OW + TILE_W - 1) / TILE_W;
-  const long grid = (long)N * tiles_h * tiles_w;
+  const bool small = (R * S) <= 9;
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(conv_s1_nhwc_kernel, dim3(grid), dim3(256), 0,
-                     stream.stream(),
-                     (const cbf16_t*)x.data_ptr(),
-                     (const cbf16_t*)wpk.data_ptr(),
-                     (cbf16_t*)y.data_ptr(),
-                     N, C, H, W, (int)K, (int)R, (int)S, (int)pad,
-                     OH, OW, tiles_h, tiles_w);
+  if (small) {
+    const int tiles_h = (OH + TILE_H - 1) / TILE_H;
+    const int tiles_w = (OW + TILE_W - 1) / TILE_W;
+    const long grid = (long)N * tiles_h * tiles_w;
+    auto launch = [&](auto kern) {
+      hipLaunchKernelGGL(kern, dim3(grid), dim3(256), 0, stream.stream(),
+                         (const cbf16_t*)x.data_ptr(),
+                         (const cbf16_t*)wpk.data_ptr(),
+                         (cbf16_t*)y.data_ptr(),
+                         N, H, W, (int)K, (int)R, (int)S, (int)pad,
+                         OH, OW, tiles_h, tiles_w);
+    };
+    if (C == 64 && K == 64) launch(conv_s1_nhwc_kernel<4, 2, 9>);
+    else if (C == 32 && K == 32) launch(conv_s1_nhwc_kernel<2, 1, 9>);
+    else if (C == 48 && K == 64) launch(conv_s1_nhwc_kernel<3, 2, 9>);
+    else if (C == 16 && K == 32) launch(conv_s1_nhwc_kernel<1, 1, 9>);
+    else TORCH_CHECK(false, "conv_s1_nhwc: unsupported C/K combo");
+  } else {
+    const int tiles_h = (OH + BTILE - 1) / BTILE;
+    const int tiles_w = (OW + BTILE - 1) / BTILE;
+    const long grid = (long)N * tiles_h * tiles_w;
+    auto launch = [&](auto kern) {
+      hipLaunchKernelGGL(kern, dim3(grid), dim3(512), 0, stream.stream(),
+                         (const cbf16_t*)x.data_ptr(),
+                         (const cbf16_t*)wpk.data_ptr(),
+                         (cbf16_t*)y.data_ptr(),
+                         N, H, W, (int)K, (int)R, (int)S, (int)pad,
+                         OH, OW, tiles_h, tiles_w);
+    };
+    if (C == 64 && K == 64) launch(conv_s1_nhwc_big_kernel<4, 2>);
+    else if (C == 32 && K == 32) launch(conv_s1_nhwc_big_kernel<2, 1>);
+    else if (C == 48 && K == 64) launch(conv_s1_nhwc_big_kernel<3, 2>);
+    else if (C == 16 && K == 32) launch(conv_s1_nhwc_big_kernel<1, 1>);
+    else TORCH_CHECK(false, "conv_s1_nhwc: unsupported C/K combo");
+  }
   return y;
 }

@@ -27,14 +27,17 @@ from torch import nn
 
 from tensor2robot_amd import gin
 from tensor2robot_amd.ops import fused_bn
+from tensor2robot_amd.ops import conv as mfma_conv
 from tensor2robot_amd.ops import maxpool as fused_maxpool
 
 
 def _conv_bn_relu(in_ch: int, out_ch: int, kernel: int, stride: int = 1,
                   padding: int = 0) -> nn.Sequential:
+  # MFMAConv2d self-dispatches: the hand-written kernel on supported
+  # stride-1 shapes (3x3 blocks, 3.3-4x vs MIOpen), F.conv2d otherwise.
   return nn.Sequential(
-      nn.Conv2d(in_ch, out_ch, kernel, stride=stride, padding=padding,
-                bias=False),
+      mfma_conv.MFMAConv2d(in_ch, out_ch, kernel, stride=stride,
+                           padding=padding, bias=False),
       fused_bn.FusedBatchNormReLU(out_ch, eps=0.001, momentum=0.003),
   )
 
