@@ -53,3 +53,22 @@ def test_golden_values_roundtrip(tmp_path):
   fixture.train_and_check_golden_predictions(
       pose_env_models, "PoseEnvRegressionModel", FIXTURE, golden_file,
       generate_golden_data=False)
+
+
+def test_train_eval_test_utils(tmp_path):
+  from tensor2robot_amd.utils import train_eval_test_utils as tet
+  fixture = t2r_test_fixture.T2RModelFixture()
+  fixture.random_train(mocks, "MockT2RModel", model_dir=str(tmp_path))
+  tet.assert_output_files(str(tmp_path))
+  with pytest.raises(AssertionError):
+    tet.assert_output_files(str(tmp_path), ("no_such_file_*",))
+
+
+def test_train_eval_gin_helper(tmp_path):
+  import os
+  from tensor2robot_amd.utils import train_eval_test_utils as tet
+  repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+  cfg = os.path.join(repo, "tensor2robot_amd", "research", "pose_env",
+                     "configs", "run_train_reg_model.gin")
+  result = tet.test_train_eval_gin(str(tmp_path), cfg)
+  assert result["global_step"] == 1
