@@ -107,4 +107,9 @@ class MFMAConv2d(nn.Conv2d):
       if w.dtype != torch.bfloat16:
         w = w.to(torch.bfloat16)
       return _MFMAConvFunction.apply(x, w, self.padding[0])
+    if x.is_cuda and x.dtype != self.weight.dtype and \
+        not torch.is_autocast_enabled():
+      # bf16 activations outside autocast: run the fallback in bf16 too.
+      return F.conv2d(x, self.weight.to(x.dtype), None, self.stride,
+                      self.padding, self.dilation, self.groups)
     return super().forward(x)
