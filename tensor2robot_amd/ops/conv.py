@@ -66,7 +66,7 @@ class _MFMAConvFunction(torch.autograd.Function):
     k, c, r, s = weight.shape
     if not x.is_contiguous(memory_format=torch.channels_last):
       x = x.contiguous(memory_format=torch.channels_last)
-    wpk = pack_weights(weight)
+    wpk = ext.pack_conv_w(weight, False)
     y = ext.conv_s1_nhwc(x, wpk, k, r, s, pad)
     ctx.save_for_backward(x, weight)
     ctx.pad = pad
@@ -82,7 +82,7 @@ class _MFMAConvFunction(torch.autograd.Function):
     if ctx.needs_input_grad[0]:
       # SAME-pad duality: the dy->dx conv pad is (R-1-pad).
       bpad = r - 1 - ctx.pad
-      wpk_b = pack_weights_bwd(weight)
+      wpk_b = ext.pack_conv_w(weight, True)
       dx = ext.conv_s1_nhwc(dy.to(torch.bfloat16), wpk_b, c, r, s, bpad)
     if ctx.needs_input_grad[1]:
       # bf16 wrw via MIOpen (same dtype the autocast path uses; an f32

@@ -20,6 +20,8 @@ at::Tensor mfma_probe(at::Tensor A, at::Tensor B);
 at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
                         int64_t R, int64_t S, int64_t pad);
 
+at::Tensor pack_conv_w(at::Tensor w, bool transpose);
+
 std::vector<at::Tensor> maxpool_nhwc_forward(at::Tensor x, int64_t kh,
                                              int64_t kw, bool ceil_mode);
 
@@ -43,6 +45,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "32x32x16 bf16 MFMA layout probe");
   m.def("conv_s1_nhwc", &conv_s1_nhwc,
         "MFMA stride-1 NHWC bf16 conv (prepacked weights)");
+  m.def("pack_conv_w", &pack_conv_w,
+        "single-kernel conv weight pack (transpose=bwd-data layout)");
   m.def("maxpool_nhwc_forward", &maxpool_nhwc_forward,
         "Non-overlapping NHWC bf16 max-pool forward (+argmax)");
   m.def("maxpool_nhwc_backward", &maxpool_nhwc_backward,

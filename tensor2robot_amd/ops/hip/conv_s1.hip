@@ -355,3 +355,72 @@ at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
   }
   return y;
 }
+
+// ---------------------------------------------------------------------------
+// Single-kernel weight pack: [K,C,R,S] (torch, any float dtype widened to
+// bf16 host-side = the .to(bf16) cast folds in here) -> the conv kernel's
+// [rs][c16][n][WPAD] image.  transpose=true additionally swaps the C/K
+// roles and flips r,s — the backward-data pack — so the python wrapper
+// launches ONE kernel instead of ~7 tensor ops per conv per step.
+// ---------------------------------------------------------------------------
+
+template <bool TRANSPOSE>
+__global__ void __launch_bounds__(256)
+pack_conv_w_kernel(const cbf16_t* __restrict__ w,   // [K][C][R][S]
+                   cbf16_t* __restrict__ out,       // [rs][c16][n][WPAD]
+                   int K, int C, int R, int S) {
+  // Output dims: C-role = TRANSPOSE ? K : C; K-role = TRANSPOSE ? C : K.
+  const int crole = TRANSPOSE ? K : C;
+  const int krole = TRANSPOSE ? C : K;
+  const long total = (long)R * S * (crole / 16) * krole * WPAD;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const int cc = i % WPAD;
+    long rest = i / WPAD;
+    const int n = rest % krole;
+    rest /= krole;
+    const int c16 = rest % (crole / 16);
+    const int rs = rest / (crole / 16);
+    cbf16_t v = __float2bfloat16(0.0f);
+    if (cc < 16) {
+      const int c = c16 * 16 + cc;
+      int r = rs / S, s = rs % S;
+      int kk, ci;
+      if (TRANSPOSE) {
+        kk = c;          // original K index comes from the C-role dim
+        ci = n;          // original C index comes from the K-role dim
+        r = R - 1 - r;
+        s = S - 1 - s;
+      } else {
+        kk = n;
+        ci = c;
+      }
+      v = w[(((long)kk * C + ci) * R + r) * S + s];
+    }
+    out[i] = v;
+  }
+}
+
+at::Tensor pack_conv_w(at::Tensor w, bool transpose) {
+  TORCH_CHECK(w.is_cuda() && w.dim() == 4, "pack_conv_w: 4D CUDA");
+  w = w.contiguous();
+  if (w.scalar_type() != at::kBFloat16) w = w.to(at::kBFloat16);
+  const int K = w.size(0), C = w.size(1), R = w.size(2), S = w.size(3);
+  const int crole = transpose ? K : C;
+  const int krole = transpose ? C : K;
+  TORCH_CHECK(crole % 16 == 0, "pack_conv_w: C-role % 16");
+  auto out = at::empty({(long)R * S, crole / 16, krole, WPAD},
+                       w.options());
+  const long total = (long)R * S * (crole / 16) * krole * WPAD;
+  const int grid = (int)std::min<long>((total + 255) / 256, 4096);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (transpose)
+    hipLaunchKernelGGL(pack_conv_w_kernel<true>, dim3(grid), dim3(256),
+                       0, stream.stream(), (const cbf16_t*)w.data_ptr(),
+                       (cbf16_t*)out.data_ptr(), K, C, R, S);
+  else
+    hipLaunchKernelGGL(pack_conv_w_kernel<false>, dim3(grid), dim3(256),
+                       0, stream.stream(), (const cbf16_t*)w.data_ptr(),
+                       (cbf16_t*)out.data_ptr(), K, C, R, S);
+  return out;
+}

@@ -80,13 +80,32 @@ def test_mfma_conv_backward_matches_torch():
 @requires_gpu
 def test_mfma_conv_module_dispatch():
   from tensor2robot_amd.ops import conv as mconv
-  m = mconv.MFMAConv2d(64, 64, 5, padding=2).cuda().to(
+  m = mconv.MFMAConv2d(64, 64, 3, padding=1).cuda().to(
       memory_format=torch.channels_last)
   x = torch.randn(2, 64, 33, 33, device="cuda").to(torch.bfloat16) \
       .contiguous(memory_format=torch.channels_last)
   y = m(x)
   assert y.shape == (2, 64, 33, 33)
+  # 5x5 falls back to torch conv (bf16 weights cast inside).
+  m5 = mconv.MFMAConv2d(64, 64, 5, padding=2).cuda().to(
+      memory_format=torch.channels_last)
+  assert m5(x).shape == (2, 64, 33, 33)
   # Unsupported shape falls back to torch conv.
   m2 = mconv.MFMAConv2d(3, 64, 6, stride=2, padding=2).cuda()
   x2 = torch.randn(2, 3, 64, 64, device="cuda")
   assert m2(x2).shape[1] == 64
+
+
+@requires_gpu
+def test_pack_kernel_matches_python():
+  from tensor2robot_amd.ops import conv as mconv
+  from tensor2robot_amd.ops import _t2r_hip
+  torch.manual_seed(0)
+  w = torch.randn(64, 64, 3, 3, device="cuda").to(torch.bfloat16)
+  assert torch.equal(_t2r_hip.pack_conv_w(w, False),
+                     mconv.pack_weights(w))
+  assert torch.equal(_t2r_hip.pack_conv_w(w, True),
+                     mconv.pack_weights_bwd(w))
+  w5 = torch.randn(64, 64, 5, 5, device="cuda").to(torch.bfloat16)
+  assert torch.equal(_t2r_hip.pack_conv_w(w5, True),
+                     mconv.pack_weights_bwd(w5))
