@@ -17,15 +17,22 @@ from typing import Dict, Optional
 
 class SummaryWriter:
 
-  def __init__(self, log_dir: str, filename: str = "events.jsonl"):
+  def __init__(self, log_dir: str, filename: str = "events.jsonl",
+               tensorboard: bool = False):
     os.makedirs(log_dir, exist_ok=True)
     self._path = os.path.join(log_dir, filename)
     self._file = open(self._path, "a")
+    self._tb = None
+    if tensorboard or os.environ.get("T2R_TENSORBOARD_EVENTS"):
+      from tensor2robot_amd.utils import tb_events
+      self._tb = tb_events.TBEventWriter(log_dir)
 
   def add_scalar(self, tag: str, value: float, step: int):
     rec = {"step": int(step), "tag": tag, "value": float(value),
            "wall_time": time.time()}
     self._file.write(json.dumps(rec) + "\n")
+    if self._tb is not None:
+      self._tb.add_scalar(tag, value, step)
 
   def add_scalars(self, scalars: Dict[str, float], step: int):
     for tag, value in scalars.items():
@@ -39,6 +46,9 @@ class SummaryWriter:
       self._file.flush()
       self._file.close()
       self._file = None
+    if self._tb is not None:
+      self._tb.close()
+      self._tb = None
 
   def __enter__(self):
     return self
