@@ -111,7 +111,7 @@ def bench_full_step():
   autocast = torch.autocast("cuda", dtype=torch.bfloat16)
 
   def eager_step():
-    optimizer.zero_grad(set_to_none=False)
+    optimizer.zero_grad(set_to_none=True)
     with autocast:
       logit = network(image, action)
     loss = F.binary_cross_entropy_with_logits(
