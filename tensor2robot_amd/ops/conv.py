@@ -50,10 +50,10 @@ def _supported(x: torch.Tensor, weight: torch.Tensor, stride,
   if not (x.is_cuda and x.dtype == torch.bfloat16):
     return False
   k, c, r, s = weight.shape
-  # Measured dispatch rule (profiles/): the MFMA kernel wins 3.3-4x on
-  # r*s <= 9 (all-staged weights, barrier-free loop); MIOpen's igemm
-  # still wins on 5x5 @ large maps, so those fall back.
-  return (stride == (1, 1) and r * s <= 9 and
+  # Measured dispatch rule (profiles/): 3.3-4x vs MIOpen on r*s <= 9
+  # (all-staged weights, barrier-free loop); 1.07x on the 5x5 via the
+  # glds-ring variant (async LDS-DMA weight stream).
+  return (stride == (1, 1) and r <= 5 and s <= 5 and
           c % 16 == 0 and c <= 64 and k % 32 == 0 and k <= 64 and
           padding[0] == padding[1])
 

@@ -295,6 +295,12 @@ conv_s1_nhwc_big_kernel(const cbf16_t* __restrict__ x,
   }
 }
 
+template <int C16N, int NTILES>
+__global__ void conv_s1_nhwc_ring_kernel(
+    const cbf16_t* __restrict__ x, const cbf16_t* __restrict__ wpk,
+    cbf16_t* __restrict__ y, int N, int H, int W, int K, int R, int S,
+    int pad, int OH, int OW, int tiles_h, int tiles_w);
+
 // ---------------------------------------------------------------------------
 // Host wrapper
 // ---------------------------------------------------------------------------
@@ -347,10 +353,19 @@ at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
                          N, H, W, (int)K, (int)R, (int)S, (int)pad,
                          OH, OW, tiles_h, tiles_w);
     };
-    if (C == 64 && K == 64) launch(conv_s1_nhwc_big_kernel<4, 2>);
-    else if (C == 32 && K == 32) launch(conv_s1_nhwc_big_kernel<2, 1>);
-    else if (C == 48 && K == 64) launch(conv_s1_nhwc_big_kernel<3, 2>);
-    else if (C == 16 && K == 32) launch(conv_s1_nhwc_big_kernel<1, 1>);
+    static const bool use_ring = std::getenv("T2R_CONV_NO_RING") == nullptr;
+    if (C == 64 && K == 64)
+      use_ring ? launch(conv_s1_nhwc_ring_kernel<4, 2>)
+               : launch(conv_s1_nhwc_big_kernel<4, 2>);
+    else if (C == 32 && K == 32)
+      use_ring ? launch(conv_s1_nhwc_ring_kernel<2, 1>)
+               : launch(conv_s1_nhwc_big_kernel<2, 1>);
+    else if (C == 48 && K == 64)
+      use_ring ? launch(conv_s1_nhwc_ring_kernel<3, 2>)
+               : launch(conv_s1_nhwc_big_kernel<3, 2>);
+    else if (C == 16 && K == 32)
+      use_ring ? launch(conv_s1_nhwc_ring_kernel<1, 1>)
+               : launch(conv_s1_nhwc_big_kernel<1, 1>);
     else TORCH_CHECK(false, "conv_s1_nhwc: unsupported C/K combo");
   }
   return y;
@@ -423,4 +438,145 @@ at::Tensor pack_conv_w(at::Tensor w, bool transpose) {
                        0, stream.stream(), (const cbf16_t*)w.data_ptr(),
                        (cbf16_t*)out.data_ptr(), K, C, R, S);
   return out;
+}
+
+// ---------------------------------------------------------------------------
+// glds-ring variant for the 5x5: weight chunks stream via LDS-DMA
+// (global_load_lds) through a 3-slot ring with counted vmcnt + raw
+// barriers (cdna_hip_programming.md §5 "glds, 2-3 LDS buffers" rows) —
+// no register round-trip, no write pass, loads 2 chunks ahead.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ void conv_waitcnt_vm(int ahead) {
+  // s_waitcnt imm: vmcnt[3:0], expcnt[6:4]=7, lgkmcnt[11:8]=15.
+  switch (ahead) {
+    case 0: __builtin_amdgcn_s_waitcnt(0 | (7 << 4) | (15 << 8)); break;
+    case 1: __builtin_amdgcn_s_waitcnt(2 | (7 << 4) | (15 << 8)); break;
+    default: __builtin_amdgcn_s_waitcnt(4 | (7 << 4) | (15 << 8)); break;
+  }
+}
+
+template <int C16N, int NTILES>
+__global__ void __launch_bounds__(512, 2)
+conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
+                         const cbf16_t* __restrict__ wpk,
+                         cbf16_t* __restrict__ y,
+                         int N, int H, int W, int K,
+                         int R, int S, int pad,
+                         int OH, int OW, int tiles_h, int tiles_w) {
+  constexpr int C = C16N * 16;
+  constexpr int WBUF = C16N * NTILES * 32 * WPAD;      // bf16 per chunk
+  constexpr int PIECES = (WBUF * 2) / 1024;            // 1-KiB DMA pieces
+  __shared__ short lds[BHALO * BHALO * XPITCH + 3 * WBUF];
+  short* xtile = lds;
+  auto wbuf = [&](int slot) -> short* {
+    return lds + BHALO * BHALO * XPITCH + slot * WBUF;
+  };
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+
+  long wg = blockIdx.x;
+  const int img = wg / (tiles_h * tiles_w);
+  const int trest = wg % (tiles_h * tiles_w);
+  const int th = trest / tiles_w;
+  const int tw = trest % tiles_w;
+  const int oh0 = th * BTILE, ow0 = tw * BTILE;
+
+  // Wave w issues DMA pieces {w, w+8, ...} of each chunk; with
+  // PIECES=12 waves 0-3 own 2 pieces, 4-7 own 1: NOT uniform.  Use a
+  // 6-wave x 2-piece assignment so the counted vmcnt is uniform among
+  // issuers (waves 6,7 issue nothing and skip the waits).
+  const bool issuer = wave < 6;
+  auto issue_chunk = [&](int rs, int slot) {
+    if (!issuer) return;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int piece = wave * 2 + j;
+      if (piece < PIECES) {
+        const cbf16_t* src = wpk + (long)rs * WBUF + piece * 512
+                             + lane * 8;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) uint32_t*)src,
+            (__attribute__((address_space(3))) uint32_t*)
+                (wbuf(slot) + piece * 512),
+            16, 0, 0);
+      }
+    }
+  };
+
+  {
+    const int halo_h = BTILE + R - 1, halo_w = BTILE + S - 1;
+    constexpr int chunks = C >> 3;
+    const int total = halo_h * halo_w * chunks;
+    for (int i = tid; i < total; i += 512) {
+      const int chunk = i % chunks;
+      const int pix = i / chunks;
+      const int hrow = pix / halo_w, hcol = pix % halo_w;
+      const int iy = oh0 - pad + hrow;
+      const int ix = ow0 - pad + hcol;
+      uint4 v = make_uint4(0, 0, 0, 0);
+      if (iy >= 0 && iy < H && ix >= 0 && ix < W) {
+        v = *reinterpret_cast<const uint4*>(
+            x + (((long)img * H + iy) * W + ix) * C + chunk * 8);
+      }
+      *reinterpret_cast<uint4*>(
+          &xtile[(hrow * BHALO + hcol) * XPITCH + chunk * 8]) = v;
+    }
+  }
+  issue_chunk(0, 0);
+  if (R * S > 1) issue_chunk(1, 1);
+  // One full drain in the prologue (also covers the x-tile loads).
+  __syncthreads();
+
+  cf32x16 acc[NTILES];
+#pragma unroll
+  for (int nt = 0; nt < NTILES; ++nt) acc[nt] = (cf32x16){};
+  const int mrow = lane & 31;
+  const int kgrp = lane >> 5;
+  const int prow = (wave * 32 + mrow) / BTILE;
+  const int pcol = (wave * 32 + mrow) % BTILE;
+
+  const int RS = R * S;
+  for (int rs = 0; rs < RS; ++rs) {
+    const int r = rs / S, s = rs % S;
+    const int slot = rs % 3;
+    if (rs + 2 < RS) issue_chunk(rs + 2, (rs + 2) % 3);
+    if (issuer) conv_waitcnt_vm(min(RS - 1 - rs, 2));
+    __builtin_amdgcn_s_barrier();     // chunk rs landed for everyone
+#pragma unroll
+    for (int c16 = 0; c16 < C16N; ++c16) {
+      cbf16x8 a_frag = *reinterpret_cast<const cbf16x8*>(
+          &xtile[((prow + r) * BHALO + (pcol + s)) * XPITCH
+                 + c16 * 16 + kgrp * 8]);
+#pragma unroll
+      for (int nt = 0; nt < NTILES; ++nt) {
+        const int n = nt * 32 + mrow;
+        cbf16x8 b_frag = *reinterpret_cast<const cbf16x8*>(
+            &wbuf(slot)[(c16 * K + n) * WPAD + kgrp * 8]);
+        acc[nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            a_frag, b_frag, acc[nt], 0, 0, 0);
+      }
+    }
+    // Slot rs%3 is refilled at iteration rs+1 (chunk rs+3): everyone
+    // must be done reading before that DMA can land.
+    __builtin_amdgcn_s_barrier();
+  }
+
+  const int ocol_n = lane & 31;
+#pragma unroll
+  for (int nt = 0; nt < NTILES; ++nt) {
+#pragma unroll
+    for (int reg = 0; reg < 16; ++reg) {
+      const int m = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+      const int p = wave * 32 + m;
+      const int orow = oh0 + p / BTILE;
+      const int ocol = ow0 + p % BTILE;
+      if (orow < OH && ocol < OW) {
+        y[(((long)img * OH + orow) * OW + ocol) * K
+          + nt * 32 + ocol_n] = __float2bfloat16(acc[nt][reg]);
+      }
+    }
+  }
 }
