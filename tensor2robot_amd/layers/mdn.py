@@ -12,7 +12,7 @@ loss(labels) = mean NLL :164-167).
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 from torch import nn

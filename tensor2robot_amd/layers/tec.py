@@ -10,7 +10,7 @@ both_directions / reverse_direction / cross_entropy / triplet
 
 from __future__ import annotations
 
-from typing import Optional, Sequence, Tuple
+from typing import Optional, Sequence
 
 import torch
 from torch import nn

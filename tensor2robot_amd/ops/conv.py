@@ -14,8 +14,6 @@ hipGraphs along with the conv.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 from torch import nn
 import torch.nn.functional as F

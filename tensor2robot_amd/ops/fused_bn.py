@@ -8,8 +8,6 @@ On CPU it runs the plain torch reference (also used by numerics tests).
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 from torch import nn
 

@@ -20,7 +20,7 @@ cheap in practice).
 from __future__ import annotations
 
 import logging
-from typing import Callable, Dict, Optional
+from typing import Callable, Optional
 
 import torch
 

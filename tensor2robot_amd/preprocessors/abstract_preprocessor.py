@@ -13,7 +13,7 @@ kernels in tensor2robot_amd/ops).
 from __future__ import annotations
 
 import abc
-from typing import Optional, Tuple
+from typing import Tuple
 
 from tensor2robot_amd import gin
 from tensor2robot_amd.specs import tensorspec_utils as tsu

@@ -18,7 +18,7 @@ preprocessing (crop/resize/distort) runs on-GPU after H2D of raw uint8.
 from __future__ import annotations
 
 import enum
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, List, Optional, Sequence
 
 import torch
 from torch import nn

@@ -13,7 +13,6 @@ import logging
 import os
 from typing import Callable, Optional
 
-import numpy as np
 
 from tensor2robot_amd import gin
 from tensor2robot_amd.utils import summaries as summaries_mod

@@ -12,7 +12,6 @@ from __future__ import annotations
 
 import torch
 
-from tensor2robot_amd import gin
 
 
 def tile_to_match_context(net: torch.Tensor,

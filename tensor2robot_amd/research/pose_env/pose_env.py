@@ -9,7 +9,7 @@ target pose per-task for meta-learning (reference :84-89).
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import numpy as np
 

@@ -7,7 +7,6 @@ PoseEnvRegressionModel :231 (image -> pose), PoseEnvContinuousMCModel :92
 
 from __future__ import annotations
 
-from typing import Dict, Optional
 
 import numpy as np
 import torch

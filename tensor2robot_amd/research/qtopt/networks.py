@@ -20,7 +20,7 @@ on the hot path.
 
 from __future__ import annotations
 
-from typing import Dict, Optional, Tuple
+from typing import Tuple
 
 import torch
 from torch import nn

@@ -12,7 +12,7 @@ checkpoint semantics (models/optimizers.ExponentialMovingAverage).
 
 from __future__ import annotations
 
-from typing import Dict, Optional, Tuple
+from typing import Dict, Optional
 
 from tensor2robot_amd import gin
 from tensor2robot_amd.models import optimizers as optimizers_mod

@@ -17,7 +17,7 @@ from __future__ import annotations
 
 import fnmatch
 import random
-from typing import Iterable, List, Optional, Sequence, Tuple
+from typing import Iterable, Optional, Sequence
 
 import torch
 

@@ -12,7 +12,7 @@ height_to_bottom), DefaultGrasping44ImagePreprocessor :242 (512x640 uint8
 
 from __future__ import annotations
 
-from typing import Dict, Optional
+from typing import Optional
 
 import torch
 
@@ -23,7 +23,6 @@ from tensor2robot_amd.preprocessors import abstract_preprocessor
 from tensor2robot_amd.preprocessors import distortion
 from tensor2robot_amd.research.qtopt import networks
 from tensor2robot_amd.specs import tensorspec_utils as tsu
-from tensor2robot_amd.utils import modes as run_modes
 
 TSPEC = tsu.ExtendedTensorSpec
 

@@ -12,7 +12,7 @@ task embedding, optional end-token head :299-311).
 from __future__ import annotations
 
 import collections
-from typing import Dict, Optional
+from typing import Dict
 
 import numpy as np
 import torch

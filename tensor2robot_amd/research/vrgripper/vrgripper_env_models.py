@@ -11,7 +11,7 @@ conditioning, learned conv1d loss for the MAML inner loop).
 
 from __future__ import annotations
 
-from typing import Dict, Optional, Sequence, Tuple
+from typing import Dict, List, Optional, Sequence
 
 import numpy as np
 import torch

@@ -12,8 +12,7 @@ validate_and_pack :1244, feature-dict generation :1558-1629, varlen pad/clip
 from __future__ import annotations
 
 import collections
-import copy as _copy
-from typing import Any, Dict, Iterable, Mapping, Optional, Sequence, Tuple, Union
+from typing import Any, Dict, Mapping, Optional, Sequence, Tuple
 
 import numpy as np
 import torch

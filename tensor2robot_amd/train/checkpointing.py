@@ -14,7 +14,6 @@ from __future__ import annotations
 import os
 import re
 import shutil
-import tempfile
 import time
 from typing import Dict, List, Optional
 
