@@ -14,6 +14,8 @@ hipGraphs along with the conv.
 
 from __future__ import annotations
 
+import os
+
 import torch
 from torch import nn
 import torch.nn.functional as F
@@ -42,7 +44,6 @@ def pack_weights_bwd(w: torch.Tensor) -> torch.Tensor:
 
 def _supported(x: torch.Tensor, weight: torch.Tensor, stride,
                padding) -> bool:
-  import os
   if os.environ.get("T2R_DISABLE_MFMA_CONV"):
     return False
   if not (x.is_cuda and x.dtype == torch.bfloat16):
@@ -83,12 +84,19 @@ class _MFMAConvFunction(torch.autograd.Function):
       wpk_b = ext.pack_conv_w(weight, True)
       dx = ext.conv_s1_nhwc(dy.to(torch.bfloat16), wpk_b, c, r, s, bpad)
     if ctx.needs_input_grad[1]:
-      # bf16 wrw via MIOpen (same dtype the autocast path uses; an f32
-      # upcast here cost 30% whole-step throughput).
-      dw = torch.ops.aten.convolution_backward(
-          dy.to(torch.bfloat16), x, weight, None, (1, 1),
-          (ctx.pad, ctx.pad), (1, 1), False, (0, 0), 1,
-          (False, True, False))[1].to(weight.dtype)
+      if c % 32 == 0 and not os.environ.get("T2R_DISABLE_MFMA_WRW"):
+        # MFMA wrw: fp32 LDS-accumulated [rs][c][k] -> [k][c][r][s].
+        dw_f32 = ext.conv_s1_wrw(x, dy.to(torch.bfloat16), r, s,
+                                 ctx.pad)
+        dw = dw_f32.reshape(r, s, c, k).permute(3, 2, 0, 1) \
+            .contiguous().to(weight.dtype)
+      else:
+        # bf16 wrw via MIOpen (an f32 upcast here cost 30% whole-step
+        # throughput).
+        dw = torch.ops.aten.convolution_backward(
+            dy.to(torch.bfloat16), x, weight, None, (1, 1),
+            (ctx.pad, ctx.pad), (1, 1), False, (0, 0), 1,
+            (False, True, False))[1].to(weight.dtype)
     return dx, dw, None
 
 

@@ -22,6 +22,9 @@ at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
 
 at::Tensor pack_conv_w(at::Tensor w, bool transpose);
 
+at::Tensor conv_s1_wrw(at::Tensor x, at::Tensor dy, int64_t R, int64_t S,
+                       int64_t pad);
+
 std::vector<at::Tensor> maxpool_nhwc_forward(at::Tensor x, int64_t kh,
                                              int64_t kw, bool ceil_mode);
 
@@ -47,6 +50,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA stride-1 NHWC bf16 conv (prepacked weights)");
   m.def("pack_conv_w", &pack_conv_w,
         "single-kernel conv weight pack (transpose=bwd-data layout)");
+  m.def("conv_s1_wrw", &conv_s1_wrw,
+        "MFMA stride-1 conv weight gradient (LDS-accumulated rs groups)");
   m.def("maxpool_nhwc_forward", &maxpool_nhwc_forward,
         "Non-overlapping NHWC bf16 max-pool forward (+argmax)");
   m.def("maxpool_nhwc_backward", &maxpool_nhwc_backward,
