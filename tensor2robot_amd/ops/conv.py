@@ -84,7 +84,10 @@ class _MFMAConvFunction(torch.autograd.Function):
       wpk_b = ext.pack_conv_w(weight, True)
       dx = ext.conv_s1_nhwc(dy.to(torch.bfloat16), wpk_b, c, r, s, bpad)
     if ctx.needs_input_grad[1]:
-      if c % 32 == 0 and not os.environ.get("T2R_DISABLE_MFMA_WRW"):
+      # OPT-IN only: the MFMA wrw measured 0.25-0.44x MIOpen (the
+      # transposed scalar LDS staging dominates; profiles/); kept as
+      # groundwork for a tr_b16-based round-2 variant.
+      if c % 32 == 0 and os.environ.get("T2R_ENABLE_MFMA_WRW"):
         # MFMA wrw: fp32 LDS-accumulated [rs][c][k] -> [k][c][r][s].
         dw_f32 = ext.conv_s1_wrw(x, dy.to(torch.bfloat16), r, s,
                                  ctx.pad)
