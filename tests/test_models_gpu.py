@@ -184,7 +184,9 @@ def test_graphed_step_trains_grasping44():
   autocast = torch.autocast("cuda", dtype=torch.bfloat16)
 
   def body():
-    optimizer.zero_grad(set_to_none=False)
+    # set_to_none=True inside capture: backward writes stable
+    # graph-pool grad buffers (the bench.py graph-body contract).
+    optimizer.zero_grad(set_to_none=True)
     with autocast:
       logit = network(image, action)
     loss = torch.nn.functional.binary_cross_entropy_with_logits(
