@@ -103,6 +103,46 @@ class _MFMAConvFunction(torch.autograd.Function):
     return dx, dw, None
 
 
+class _StemConvFunction(torch.autograd.Function):
+  """6x6/2 C=3 stem forward on the MFMA kernel; dw via MIOpen.
+
+  The stem is the network's first layer, so no input gradient is
+  produced (dx would be the raw-image gradient).
+  """
+
+  @staticmethod
+  def forward(ctx, x, weight):
+    ext = ops_mod.require_hip()
+    if not x.is_contiguous(memory_format=torch.channels_last):
+      x = x.contiguous(memory_format=torch.channels_last)
+    wpk = ext.pack_stem_w(weight)
+    y = ext.conv_stem_nhwc(x, wpk)
+    ctx.save_for_backward(x, weight)
+    return y
+
+  @staticmethod
+  def backward(ctx, dy):
+    x, weight = ctx.saved_tensors
+    dy = dy.contiguous(memory_format=torch.channels_last)
+    dw = None
+    if ctx.needs_input_grad[1]:
+      dw = torch.ops.aten.convolution_backward(
+          dy.to(torch.bfloat16), x, weight, None, (2, 2), (2, 2),
+          (1, 1), False, (0, 0), 1,
+          (False, True, False))[1].to(weight.dtype)
+    return None, dw
+
+
+def _stem_supported(x, weight, stride, padding) -> bool:
+  if os.environ.get("T2R_DISABLE_MFMA_CONV"):
+    return False
+  if not (x.is_cuda and x.dtype == torch.bfloat16):
+    return False
+  k, c, r, s = weight.shape
+  return (stride == (2, 2) and padding == (2, 2) and (k, c, r, s) ==
+          (64, 3, 6, 6) and not x.requires_grad)
+
+
 class MFMAConv2d(nn.Conv2d):
   """Conv2d that runs the MFMA kernel on supported GPU bf16 shapes."""
 
@@ -118,6 +158,12 @@ class MFMAConv2d(nn.Conv2d):
       if w.dtype != torch.bfloat16:
         w = w.to(torch.bfloat16)
       return _MFMAConvFunction.apply(x, w, self.padding[0])
+    if self.bias is None and _stem_supported(x, self.weight, self.stride,
+                                             self.padding):
+      w = self.weight
+      if w.dtype != torch.bfloat16:
+        w = w.to(torch.bfloat16)
+      return _StemConvFunction.apply(x, w)
     if x.is_cuda and x.dtype != self.weight.dtype and \
         not torch.is_autocast_enabled():
       # bf16 activations outside autocast: run the fallback in bf16 too.

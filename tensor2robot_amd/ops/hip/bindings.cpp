@@ -25,6 +25,10 @@ at::Tensor pack_conv_w(at::Tensor w, bool transpose);
 at::Tensor conv_s1_wrw(at::Tensor x, at::Tensor dy, int64_t R, int64_t S,
                        int64_t pad);
 
+at::Tensor conv_stem_nhwc(at::Tensor x, at::Tensor wpk);
+
+at::Tensor pack_stem_w(at::Tensor w);
+
 std::vector<at::Tensor> maxpool_nhwc_forward(at::Tensor x, int64_t kh,
                                              int64_t kw, bool ceil_mode);
 
@@ -52,6 +56,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "single-kernel conv weight pack (transpose=bwd-data layout)");
   m.def("conv_s1_wrw", &conv_s1_wrw,
         "MFMA stride-1 conv weight gradient (LDS-accumulated rs groups)");
+  m.def("conv_stem_nhwc", &conv_stem_nhwc,
+        "MFMA 6x6/2 C=3 stem conv (implicit im2col)");
+  m.def("pack_stem_w", &pack_stem_w, "stem weight pack [12][64][24]");
   m.def("maxpool_nhwc_forward", &maxpool_nhwc_forward,
         "Non-overlapping NHWC bf16 max-pool forward (+argmax)");
   m.def("maxpool_nhwc_backward", &maxpool_nhwc_backward,

@@ -53,7 +53,8 @@ class Grasping44(nn.Module):
     self.action_dim = action_dim
     self.num_classes = num_classes
     ch = channels
-    self.conv1 = nn.Conv2d(3, ch, 6, stride=2, padding=2, bias=False)
+    self.conv1 = mfma_conv.MFMAConv2d(3, ch, 6, stride=2, padding=2,
+                                      bias=False)
     self.bn1 = fused_bn.FusedBatchNormReLU(ch, eps=0.001, momentum=0.003)
     self.pool1 = fused_maxpool.FusedMaxPool2d(3, ceil_mode=True)
     self.block1 = nn.Sequential(*[

@@ -138,3 +138,37 @@ def test_mfma_wrw_matches_torch(shape):
   err = (dw_t - ref).abs().max().item()
   scale = ref.abs().max().item()
   assert err < 0.01 * max(scale, 1.0), (shape, err, scale)
+
+
+@requires_gpu
+def test_stem_conv_matches_torch():
+  import torch.nn.functional as F
+  from tensor2robot_amd.ops import _t2r_hip
+  torch.manual_seed(0)
+  for h, w in [(472, 472), (96, 130)]:
+    x = torch.randn(2, 3, h, w, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    weight = (torch.randn(64, 3, 6, 6, device="cuda") * 0.1).to(
+        torch.bfloat16)
+    wpk = _t2r_hip.pack_stem_w(weight)
+    y = _t2r_hip.conv_stem_nhwc(x, wpk)
+    ref = F.conv2d(x.float(), weight.float(), stride=2, padding=2)
+    assert y.shape == ref.shape, (y.shape, ref.shape)
+    err = (y.float() - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err < 0.02 * max(scale, 1.0), (h, w, err, scale)
+
+
+@requires_gpu
+def test_stem_conv_module_and_grads():
+  from tensor2robot_amd.ops import conv as mconv
+  torch.manual_seed(0)
+  m = mconv.MFMAConv2d(3, 64, 6, stride=2, padding=2).cuda().to(
+      torch.bfloat16).to(memory_format=torch.channels_last)
+  x = torch.randn(2, 3, 96, 96, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last)
+  y = m(x)
+  assert y.shape == (2, 64, 48, 48)
+  y.float().pow(2).mean().backward()
+  assert m.weight.grad is not None
+  assert torch.isfinite(m.weight.grad).all()
