@@ -103,11 +103,40 @@ class _MFMAConvFunction(torch.autograd.Function):
     return dx, dw, None
 
 
-class _StemConvFunction(torch.autograd.Function):
-  """6x6/2 C=3 stem forward on the MFMA kernel; dw via MIOpen.
+def _space_to_depth_nhwc(x: torch.Tensor, pad_to: int = 16
+                         ) -> torch.Tensor:
+  """[N,3,H,W] cl -> [N,pad_to,H/2,W/2] cl, c-order (dr, ds, c)."""
+  n, c, h, w = x.shape
+  nhwc = x.permute(0, 2, 3, 1)          # view of channels_last
+  s2d = nhwc.reshape(n, h // 2, 2, w // 2, 2, c) \
+      .permute(0, 1, 3, 2, 4, 5).reshape(n, h // 2, w // 2, 4 * c)
+  out = torch.zeros(n, h // 2, w // 2, pad_to, dtype=x.dtype,
+                    device=x.device)
+  out[..., :4 * c] = s2d
+  return out.permute(0, 3, 1, 2).contiguous(
+      memory_format=torch.channels_last)
 
-  The stem is the network's first layer, so no input gradient is
-  produced (dx would be the raw-image gradient).
+
+def _stem_weight_s2d(weight: torch.Tensor, pad_to: int = 16
+                     ) -> torch.Tensor:
+  """[64,3,6,6] -> [64,pad_to,3,3]: w2[k, (dr,ds,c), r2, s2] =
+  w[k, c, 2*r2+dr, 2*s2+ds]."""
+  k, c, r, s = weight.shape
+  w6 = weight.reshape(k, c, 3, 2, 3, 2)         # [k,c,r2,dr,s2,ds]
+  w2 = w6.permute(0, 3, 5, 1, 2, 4).reshape(k, 4 * c, 3, 3)
+  out = torch.zeros(k, pad_to, 3, 3, dtype=weight.dtype,
+                    device=weight.device)
+  out[:, :4 * c] = w2
+  return out
+
+
+class _StemConvFunction(torch.autograd.Function):
+  """6x6/2 C=3 stem via space-to-depth + the fast 3x3/1 MFMA path.
+
+  The 2x2-block reshape turns the strided 6x6 into an exactly
+  equivalent 3x3 stride-1 conv with C=12 (zero-padded to 16) — the
+  STAGE_ALL conv_s1 kernel's sweet spot.  The stem is the network's
+  first layer, so no input gradient is produced.
   """
 
   @staticmethod
@@ -115,8 +144,10 @@ class _StemConvFunction(torch.autograd.Function):
     ext = ops_mod.require_hip()
     if not x.is_contiguous(memory_format=torch.channels_last):
       x = x.contiguous(memory_format=torch.channels_last)
-    wpk = ext.pack_stem_w(weight)
-    y = ext.conv_stem_nhwc(x, wpk)
+    x_s2d = _space_to_depth_nhwc(x)
+    w_s2d = _stem_weight_s2d(weight)
+    wpk = ext.pack_conv_w(w_s2d, False)
+    y = ext.conv_s1_nhwc(x_s2d, wpk, weight.shape[0], 3, 3, 1)
     ctx.save_for_backward(x, weight)
     return y
 

@@ -339,6 +339,7 @@ at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
     if (C == 64 && K == 64) launch(conv_s1_nhwc_kernel<4, 2, 9>);
     else if (C == 32 && K == 32) launch(conv_s1_nhwc_kernel<2, 1, 9>);
     else if (C == 48 && K == 64) launch(conv_s1_nhwc_kernel<3, 2, 9>);
+    else if (C == 16 && K == 64) launch(conv_s1_nhwc_kernel<1, 2, 9>);
     else if (C == 16 && K == 32) launch(conv_s1_nhwc_kernel<1, 1, 9>);
     else TORCH_CHECK(false, "conv_s1_nhwc: unsupported C/K combo");
   } else {
