@@ -165,10 +165,9 @@ class _StemConvFunction(torch.autograd.Function):
 
 
 def _stem_supported(x, weight, stride, padding) -> bool:
-  # OPT-IN: measured 0.71x MIOpen (C=3 makes the A-image build
-  # staging-bound; profiles/). Groundwork for a space-to-depth round-2
-  # variant.
-  if not os.environ.get("T2R_ENABLE_MFMA_STEM"):
+  # Space-to-depth route measured 1.13x MIOpen end-to-end (profiles/).
+  if os.environ.get("T2R_DISABLE_MFMA_CONV") or \
+      os.environ.get("T2R_DISABLE_MFMA_STEM"):
     return False
   if not (x.is_cuda and x.dtype == torch.bfloat16):
     return False
