@@ -82,6 +82,11 @@ def main():
   distributed = world_size > 1
 
   use_cuda = torch.cuda.is_available()
+  if use_cuda:
+    # Pin MIOpen to the packaged searched-best kernels (its runtime
+    # find is a per-process lottery on this pool - profiles/).
+    from tensor2robot_amd.utils import miopen_db
+    miopen_db.use_packaged_db()
   torch.backends.cudnn.benchmark = True  # let MIOpen autotune conv algos
   if distributed:
     backend = "nccl" if use_cuda else "gloo"
