@@ -51,8 +51,10 @@ def _supported(x: torch.Tensor, weight: torch.Tensor, stride,
   k, c, r, s = weight.shape
   # Measured dispatch rule (profiles/): 3.3-4x vs MIOpen on r*s <= 9
   # (all-staged weights, barrier-free loop); 1.07x on the 5x5 via the
-  # glds-ring variant (async LDS-DMA weight stream).
-  return (stride == (1, 1) and r <= 5 and s <= 5 and
+  # glds-ring variant.  T2R_MFMA_MAX_RS caps which kernels dispatch
+  # here (e.g. 9 = 3x3 only) for A/B against a tuned MIOpen.
+  max_rs = int(os.environ.get("T2R_MFMA_MAX_RS", "25"))
+  return (stride == (1, 1) and r <= 5 and s <= 5 and r * s <= max_rs and
           c % 16 == 0 and c <= 64 and k % 32 == 0 and k <= 64 and
           padding[0] == padding[1])
 
