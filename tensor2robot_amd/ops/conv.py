@@ -146,7 +146,7 @@ class _StemConvFunction(torch.autograd.Function):
     ext = ops_mod.require_hip()
     if not x.is_contiguous(memory_format=torch.channels_last):
       x = x.contiguous(memory_format=torch.channels_last)
-    x_s2d = _space_to_depth_nhwc(x)
+    x_s2d = ext.s2d_stem(x)
     w_s2d = _stem_weight_s2d(weight)
     wpk = ext.pack_conv_w(w_s2d, False)
     y = ext.conv_s1_nhwc(x_s2d, wpk, weight.shape[0], 3, 3, 1)

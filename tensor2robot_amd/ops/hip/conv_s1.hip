@@ -581,3 +581,57 @@ conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// Fused space-to-depth for the 6x6/2 stem: [N][H][W][3] bf16 (NHWC) ->
+// [N][H/2][W/2][16] with c' = dr*6 + ds*3 + c and slots 12..15 zero.
+// One global read + one write replaces the torch zeros/permute/copy
+// chain (4 kernels, ~3x the bytes).  The 6 taps of each (dr) row are
+// contiguous in NHWC (x[n, 2oy+dr, 2ox..2ox+1, 0..2]) and every
+// offset is u32-aligned (element offsets are multiples of 6), so the
+// read is 2x3 dwords and the write is 2 dwordx4 per output pixel.
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(4))) unsigned int cuint4v;
+
+__global__ void __launch_bounds__(256)
+s2d_stem_kernel(const unsigned int* __restrict__ x,
+                cuint4v* __restrict__ y,
+                long total, int H, int W, int OH, int OW) {
+  const int row_u32 = (W * 3) >> 1;   // u32 per input row (W even)
+  for (long p = blockIdx.x * 256L + threadIdx.x; p < total;
+       p += (long)gridDim.x * 256) {
+    const int ox = (int)(p % OW);
+    long rest = p / OW;
+    const int oy = (int)(rest % OH);
+    const long n = rest / OH;
+    const long base = (((n * H + 2L * oy) * W) * 3 >> 1) + 3L * ox;
+    const unsigned int a0 = x[base], a1 = x[base + 1], a2 = x[base + 2];
+    const long b = base + row_u32;
+    const unsigned int b0 = x[b], b1 = x[b + 1], b2 = x[b + 2];
+    y[p * 2] = cuint4v{a0, a1, a2, b0};
+    y[p * 2 + 1] = cuint4v{b1, b2, 0u, 0u};
+  }
+}
+
+at::Tensor s2d_stem(at::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16,
+              "s2d_stem: bf16 CUDA input required");
+  TORCH_CHECK(x.dim() == 4 && x.size(1) == 3 &&
+              x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "s2d_stem: [N,3,H,W] channels_last required");
+  const int N = x.size(0), H = x.size(2), W = x.size(3);
+  TORCH_CHECK(H % 2 == 0 && W % 2 == 0, "s2d_stem: even H, W required");
+  const int OH = H / 2, OW = W / 2;
+  auto y = at::empty({N, 16, OH, OW},
+                     x.options().memory_format(
+                         at::MemoryFormat::ChannelsLast));
+  const long total = (long)N * OH * OW;
+  const long blocks = std::min((total + 255) / 256, 8192L);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(s2d_stem_kernel, dim3(blocks), dim3(256), 0,
+                     stream.stream(),
+                     (const unsigned int*)x.data_ptr(),
+                     (cuint4v*)y.data_ptr(), total, H, W, OH, OW);
+  return y;
+}

@@ -172,3 +172,17 @@ def test_stem_conv_module_and_grads():
   y.float().pow(2).mean().backward()
   assert m.weight.grad is not None
   assert torch.isfinite(m.weight.grad).all()
+
+
+@pytest.mark.gpu
+def test_s2d_stem_kernel_matches_reference():
+  from tensor2robot_amd import ops as ops_mod
+  from tensor2robot_amd.ops import conv as conv_mod
+  ext = ops_mod.require_hip()
+  x = torch.randn(3, 3, 472, 472, device="cuda", dtype=torch.bfloat16)
+  x = x.contiguous(memory_format=torch.channels_last)
+  got = ext.s2d_stem(x)
+  want = conv_mod._space_to_depth_nhwc(x)
+  assert got.shape == want.shape == (3, 16, 236, 236)
+  assert got.is_contiguous(memory_format=torch.channels_last)
+  assert torch.equal(got, want)
