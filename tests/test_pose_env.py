@@ -73,3 +73,20 @@ def test_mc_critic_megabatch_q():
   features["action/pose"] = torch.rand(2, 7, 2)
   out = model.q_func(features, "predict")
   assert out["q_predicted"].shape == (2, 7)
+
+
+def test_regression_model_trains_from_weighted_mix(tmp_path):
+  """Dataset mixing end-to-end: weighted sampling over two tfrecord
+  sources through the full train loop (reference bc mixing workflow,
+  input_generators.py:229-301)."""
+  model = pose_env_models.PoseEnvRegressionModel(
+      create_optimizer_fn=lambda: optimizers.create_adam_optimizer(1e-3))
+  train_gen = input_generators.WeightedRecordInputGenerator(
+      file_patterns=[FIXTURE, FIXTURE], weights=[0.75, 0.25],
+      batch_size=8, seed=2)
+  result = train_eval.train_eval_model(
+      t2r_model=model, input_generator_train=train_gen,
+      input_generator_eval=None, max_train_steps=5,
+      model_dir=str(tmp_path))
+  assert result["global_step"] == 5
+  assert np.isfinite(result["loss"])
