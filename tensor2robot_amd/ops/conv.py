@@ -86,10 +86,17 @@ class _MFMAConvFunction(torch.autograd.Function):
       wpk_b = ext.pack_conv_w(weight, True)
       dx = ext.conv_s1_nhwc(dy.to(torch.bfloat16), wpk_b, c, r, s, bpad)
     if ctx.needs_input_grad[1]:
-      # OPT-IN only: the MFMA wrw measured 0.25-0.44x MIOpen (the
-      # transposed scalar LDS staging dominates; profiles/); kept as
-      # groundwork for a tr_b16-based round-2 variant.
-      if c % 32 == 0 and os.environ.get("T2R_ENABLE_MFMA_WRW"):
+      # OPT-IN: T2R_ENABLE_MFMA_WRW=2 selects the v2 register-
+      # accumulator kernel (C=K=64, 3x3/5x5); =1 the v1 LDS-accumulator
+      # (0.25-0.44x MIOpen, kept as groundwork); unset -> MIOpen.
+      wrw_mode = os.environ.get("T2R_ENABLE_MFMA_WRW", "")
+      if wrw_mode == "2" and c == 64 and k == 64 and r == s and \
+          r in (3, 5):
+        dw_f32 = ext.conv_s1_wrw2(x, dy.to(torch.bfloat16), r, s,
+                                  ctx.pad)
+        dw = dw_f32.reshape(r, s, c, k).permute(3, 2, 0, 1) \
+            .contiguous().to(weight.dtype)
+      elif c % 32 == 0 and wrw_mode:
         # MFMA wrw: fp32 LDS-accumulated [rs][c][k] -> [k][c][r][s].
         dw_f32 = ext.conv_s1_wrw(x, dy.to(torch.bfloat16), r, s,
                                  ctx.pad)

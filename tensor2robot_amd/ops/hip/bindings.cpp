@@ -25,6 +25,8 @@ at::Tensor s2d_stem(at::Tensor x);
 
 at::Tensor conv_s1_wrw(at::Tensor x, at::Tensor dy, int64_t R, int64_t S,
                        int64_t pad);
+at::Tensor conv_s1_wrw2(at::Tensor x, at::Tensor dy, int64_t R, int64_t S,
+                        int64_t pad);
 
 at::Tensor conv_stem_nhwc(at::Tensor x, at::Tensor wpk);
 
@@ -57,6 +59,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "single-kernel conv weight pack (transpose=bwd-data layout)");
   m.def("s2d_stem", &s2d_stem,
         "fused space-to-depth [N,3,H,W]->[N,16,H/2,W/2] for the stem");
+  m.def("conv_s1_wrw2", &conv_s1_wrw2,
+        "MFMA wrw v2: register-accumulator dW, single staging pass");
   m.def("conv_s1_wrw", &conv_s1_wrw,
         "MFMA stride-1 conv weight gradient (LDS-accumulated rs groups)");
   m.def("conv_stem_nhwc", &conv_stem_nhwc,

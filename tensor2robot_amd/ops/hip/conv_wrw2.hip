@@ -1,0 +1,189 @@
+// MFMA weight-gradient (wrw) kernel, v2: register accumulation.
+//
+// Same GEMM view as v1 (conv_wrw.hip): dW[rs][c][k] = sum_p
+// x[p + D(rs), c] * dy[p, k] with the pixel axis as contraction; both
+// operands staged TRANSPOSED ([c][p] / [k][p]) in LDS per spatial
+// window.  v1 measured 0.25-0.44x MIOpen because each (window, rs)
+// iteration did an LDS read-modify-write of the dW slice (48 LDS ops
+// per rs per wave) and the RS_GROUP grid dimension re-staged every
+// window up to 5x.  v2 removes both:
+//   - 8 waves per WG = (c-tile x k-tile quadrant) x (rs half); each
+//     wave keeps ITS rs-half's accumulators entirely in VGPRs
+//     (ceil(25/2)=13 x 16 f32 = 208 VGPRs for the 5x5 -> 2 waves/SIMD)
+//   - every window is staged exactly once; all rs offsets consume the
+//     same LDS image (the halo covers every shift)
+//   - one atomicAdd flush per WG at the end, straight from VGPRs.
+//
+// LDS: xT 64 x 248 x 2B = 31.7 KiB + dyT 64 x 136 x 2B = 17.4 KiB =
+// 49.1 KiB -> one 512-thread WG per CU (VGPR-bound anyway).
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+typedef __hip_bfloat16 w2bf16_t;
+typedef __attribute__((ext_vector_type(8))) short w2bf16x8;
+typedef __attribute__((ext_vector_type(16))) float w2f32x16;
+
+#define W2TILE_H 8
+#define W2TILE_W 16
+#define W2WIN_P (W2TILE_H * W2TILE_W)      // 128 window pixels
+#define W2HALO_H (W2TILE_H + 4)
+#define W2HALO_W (W2TILE_W + 4)
+#define W2HALO_P (W2HALO_H * W2HALO_W)     // 240
+#define W2XT_PITCH 248                     // halo pixels + pad
+#define W2DYT_PITCH 136                    // window pixels + pad
+
+template <int R, int S>
+__global__ void __launch_bounds__(512, 1)
+conv_s1_wrw2_kernel(const w2bf16_t* __restrict__ x,
+                    const w2bf16_t* __restrict__ dy,
+                    float* __restrict__ dw,   // [R*S][C][K] f32, zeroed
+                    int N, int H, int W, int pad,
+                    int OH, int OW, int tiles_h, int tiles_w,
+                    int window_groups) {
+  constexpr int C = 64, K = 64;
+  constexpr int RS = R * S;
+  constexpr int RS_PER = (RS + 1) / 2;     // rs offsets per wave half
+  __shared__ short lds_xt[C * W2XT_PITCH];
+  __shared__ short lds_dyt[K * W2DYT_PITCH];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int mtile = wave & 1;              // c-tile
+  const int ntile = (wave >> 1) & 1;       // k-tile
+  const int rshalf = wave >> 2;            // 0 or 1
+  const int rs_base = rshalf * RS_PER;
+
+  w2f32x16 acc[RS_PER];
+#pragma unroll
+  for (int g = 0; g < RS_PER; ++g) acc[g] = w2f32x16{};
+
+  const int total_windows = (int)((long)N * tiles_h * tiles_w);
+  const int mrow = lane & 31;
+  const int kgrp = lane >> 5;
+
+  for (int win = blockIdx.x; win < total_windows; win += window_groups) {
+    const int img = win / (tiles_h * tiles_w);
+    const int trest = win % (tiles_h * tiles_w);
+    const int oh0 = (trest / tiles_w) * W2TILE_H;
+    const int ow0 = (trest % tiles_w) * W2TILE_W;
+
+    // ---- stage x halo transposed: lds_xt[c][halo_p] ----
+    {
+      constexpr int halo_h = W2TILE_H + R - 1, halo_w = W2TILE_W + S - 1;
+      constexpr int chunks = C >> 3;
+      for (int i = tid; i < W2HALO_P * chunks; i += 512) {
+        const int chunk = i % chunks;
+        const int p = i / chunks;
+        const int hrow = p / W2HALO_W, hcol = p % W2HALO_W;
+        const int iy = oh0 - pad + hrow;
+        const int ix = ow0 - pad + hcol;
+        w2bf16x8 v = {};
+        if (hrow < halo_h && hcol < halo_w &&
+            iy >= 0 && iy < H && ix >= 0 && ix < W) {
+          v = *reinterpret_cast<const w2bf16x8*>(
+              x + (((long)img * H + iy) * W + ix) * C + chunk * 8);
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          lds_xt[(chunk * 8 + j) * W2XT_PITCH + p] = v[j];
+      }
+    }
+    // ---- stage dy window transposed: lds_dyt[k][win_p] ----
+    {
+      constexpr int kchunks = K >> 3;
+      for (int i = tid; i < W2WIN_P * kchunks; i += 512) {
+        const int chunk = i % kchunks;
+        const int p = i / kchunks;
+        const int orow = oh0 + p / W2TILE_W;
+        const int ocol = ow0 + p % W2TILE_W;
+        w2bf16x8 v = {};
+        if (orow < OH && ocol < OW) {
+          v = *reinterpret_cast<const w2bf16x8*>(
+              dy + (((long)img * OH + orow) * OW + ocol) * K + chunk * 8);
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          lds_dyt[(chunk * 8 + j) * W2DYT_PITCH + p] = v[j];
+      }
+    }
+    __syncthreads();
+
+    // ---- all rs offsets of this wave's half, window staged once ----
+#pragma unroll
+    for (int g = 0; g < RS_PER; ++g) {
+      const int rs = rs_base + g;
+      if (rs < RS) {
+        const int r = rs / S, s = rs % S;
+        const int c = mtile * 32 + mrow;
+        const int k = ntile * 32 + mrow;
+#pragma unroll
+        for (int kstep = 0; kstep < W2WIN_P / 16; ++kstep) {
+          // 16 contraction pixels = window row kstep; the shifted
+          // halo indices are contiguous within the halo row.
+          const int halo_base = (kstep + r) * W2HALO_W + s;
+          w2bf16x8 a_frag = *reinterpret_cast<const w2bf16x8*>(
+              &lds_xt[c * W2XT_PITCH + halo_base + kgrp * 8]);
+          w2bf16x8 b_frag = *reinterpret_cast<const w2bf16x8*>(
+              &lds_dyt[k * W2DYT_PITCH + kstep * 16 + kgrp * 8]);
+          acc[g] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a_frag, b_frag, acc[g], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();   // all waves done before restaging
+  }
+
+  // ---- flush straight from VGPRs; ~RS*16 atomics per lane, once ----
+#pragma unroll
+  for (int g = 0; g < RS_PER; ++g) {
+    const int rs = rs_base + g;
+    if (rs < RS) {
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int c = mtile * 32
+            + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+        const int k = ntile * 32 + (lane & 31);
+        atomicAdd(&dw[((long)rs * C + c) * K + k], acc[g][reg]);
+      }
+    }
+  }
+}
+
+at::Tensor conv_s1_wrw2(at::Tensor x, at::Tensor dy, int64_t R,
+                        int64_t S, int64_t pad) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16,
+              "conv_s1_wrw2: bf16 x required");
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == at::kBFloat16,
+              "conv_s1_wrw2: bf16 dy required");
+  x = x.contiguous(at::MemoryFormat::ChannelsLast);
+  dy = dy.contiguous(at::MemoryFormat::ChannelsLast);
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = dy.size(1), OH = dy.size(2), OW = dy.size(3);
+  TORCH_CHECK(C == 64 && K == 64, "conv_s1_wrw2: C == K == 64 only");
+  TORCH_CHECK(OH == H + 2 * pad - R + 1 && OW == W + 2 * pad - S + 1,
+              "conv_s1_wrw2: dy shape mismatch");
+  auto dw = at::zeros({(long)R * S, C, K},
+                      x.options().dtype(at::kFloat));
+  const int tiles_h = (OH + W2TILE_H - 1) / W2TILE_H;
+  const int tiles_w = (OW + W2TILE_W - 1) / W2TILE_W;
+  const int total_windows = (int)((long)N * tiles_h * tiles_w);
+  const int window_groups = std::min(total_windows, 512);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(window_groups), dim3(512), 0,
+                       stream.stream(),
+                       (const w2bf16_t*)x.data_ptr(),
+                       (const w2bf16_t*)dy.data_ptr(),
+                       (float*)dw.data_ptr(),
+                       N, H, W, (int)pad, OH, OW, tiles_h, tiles_w,
+                       window_groups);
+  };
+  if (R == 3 && S == 3) launch(conv_s1_wrw2_kernel<3, 3>);
+  else if (R == 5 && S == 5) launch(conv_s1_wrw2_kernel<5, 5>);
+  else TORCH_CHECK(false, "conv_s1_wrw2: R/S must be 3x3 or 5x5");
+  return dw;
+}

@@ -186,3 +186,31 @@ def test_s2d_stem_kernel_matches_reference():
   assert got.shape == want.shape == (3, 16, 236, 236)
   assert got.is_contiguous(memory_format=torch.channels_last)
   assert torch.equal(got, want)
+
+
+@requires_gpu
+@pytest.mark.parametrize("shape", [
+    (2, 64, 79, 79, 64, 5, 2),
+    (2, 64, 27, 27, 64, 3, 1),
+    (1, 64, 33, 17, 64, 3, 0),
+    (3, 64, 78, 78, 64, 5, 2),
+])
+def test_mfma_wrw2_matches_torch(shape):
+  import torch.nn.functional as F
+  from tensor2robot_amd.ops import _t2r_hip
+  n, c, h, w, k, r, pad = shape
+  torch.manual_seed(0)
+  x = torch.randn(n, c, h, w, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last)
+  oh, ow = h + 2 * pad - r + 1, w + 2 * pad - r + 1
+  dy = torch.randn(n, k, oh, ow, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last)
+  dw = _t2r_hip.conv_s1_wrw2(x, dy, r, r, pad)
+  dw_t = dw.reshape(r, r, c, k).permute(3, 2, 0, 1).contiguous()
+  x32 = x.float().requires_grad_(False)
+  w32 = torch.zeros(k, c, r, r, device="cuda", requires_grad=True)
+  F.conv2d(x32, w32, padding=pad).backward(dy.float())
+  ref = w32.grad
+  err = (dw_t - ref).abs().max().item()
+  scale = ref.abs().max().item()
+  assert err < 0.01 * max(scale, 1.0), (shape, err, scale)
