@@ -112,32 +112,65 @@ conv_s1_wrw2_kernel(const w2bf16_t* __restrict__ x,
     }
     __syncthreads();
 
-    // ---- all rs offsets of this wave's half, window staged once ----
-#pragma unroll
-    for (int g = 0; g < RS_PER; ++g) {
-      const int rs = rs_base + g;
-      if (rs < RS) {
-        const int r = rs / S, s = rs % S;
-        const int c = mtile * 32 + mrow;
-        const int k = ntile * 32 + mrow;
+    // ---- all rs offsets of this wave's half, window staged once.
+    // kstep outer so the dy fragment (independent of rs) is read once
+    // and reused across the whole rs half: 8 b-reads + RS_PER*8
+    // a-reads per wave per window instead of 2*RS_PER*8. ----
+    {
+      const int c = mtile * 32 + mrow;
+      const int k = ntile * 32 + mrow;
+      if constexpr (RS_PER <= 5) {
+        // kstep outer: the dy fragment (independent of rs) is read
+        // once per kstep and reused across the rs half (8 b-reads +
+        // RS_PER*8 a-reads instead of 2*RS_PER*8).  Register-safe at
+        // RS_PER<=5 (149 VGPRs, no spills).
 #pragma unroll
         for (int kstep = 0; kstep < W2WIN_P / 16; ++kstep) {
-          // 16 contraction pixels = window row kstep; the shifted
-          // halo indices are contiguous within the halo row.
-          const int halo_base = (kstep + r) * W2HALO_W + s;
-          w2bf16x8 a_frag = *reinterpret_cast<const w2bf16x8*>(
-              &lds_xt[c * W2XT_PITCH + halo_base + kgrp * 8]);
           w2bf16x8 b_frag = *reinterpret_cast<const w2bf16x8*>(
               &lds_dyt[k * W2DYT_PITCH + kstep * 16 + kgrp * 8]);
-          acc[g] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              a_frag, b_frag, acc[g], 0, 0, 0);
+#pragma unroll
+          for (int g = 0; g < RS_PER; ++g) {
+            const int rs = rs_base + g;
+            if (rs < RS) {
+              const int r = rs / S, s = rs % S;
+              const int halo_base = (kstep + r) * W2HALO_W + s;
+              w2bf16x8 a_frag = *reinterpret_cast<const w2bf16x8*>(
+                  &lds_xt[c * W2XT_PITCH + halo_base + kgrp * 8]);
+              acc[g] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                  a_frag, b_frag, acc[g], 0, 0, 0);
+            }
+          }
+        }
+      } else {
+        // g outer: at RS_PER=13 the hoisted-b order spills 82 VGPRs
+        // (13 concurrent a-load chains); one-g-at-a-time keeps the
+        // 208 accumulators + one chain live (1 spill).
+#pragma unroll
+        for (int g = 0; g < RS_PER; ++g) {
+          const int rs = rs_base + g;
+          if (rs < RS) {
+            const int r = rs / S, s = rs % S;
+#pragma unroll
+            for (int kstep = 0; kstep < W2WIN_P / 16; ++kstep) {
+              const int halo_base = (kstep + r) * W2HALO_W + s;
+              w2bf16x8 a_frag = *reinterpret_cast<const w2bf16x8*>(
+                  &lds_xt[c * W2XT_PITCH + halo_base + kgrp * 8]);
+              w2bf16x8 b_frag = *reinterpret_cast<const w2bf16x8*>(
+                  &lds_dyt[k * W2DYT_PITCH + kstep * 16 + kgrp * 8]);
+              acc[g] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                  a_frag, b_frag, acc[g], 0, 0, 0);
+            }
+          }
         }
       }
     }
     __syncthreads();   // all waves done before restaging
   }
 
-  // ---- flush straight from VGPRs; ~RS*16 atomics per lane, once ----
+  // ---- flush: non-atomic per-WG partial (atomics measured ~120 us
+  // chip-wide: every WG RMWs the same 3200 cache lines; partials +
+  // a reduce kernel are plain streaming traffic instead) ----
+  float* part = dw + (long)blockIdx.x * RS * C * K;
 #pragma unroll
   for (int g = 0; g < RS_PER; ++g) {
     const int rs = rs_base + g;
@@ -147,9 +180,20 @@ conv_s1_wrw2_kernel(const w2bf16_t* __restrict__ x,
         const int c = mtile * 32
             + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
         const int k = ntile * 32 + (lane & 31);
-        atomicAdd(&dw[((long)rs * C + c) * K + k], acc[g][reg]);
+        part[((long)rs * C + c) * K + k] = acc[g][reg];
       }
     }
+  }
+}
+
+__global__ void __launch_bounds__(256)
+wrw2_reduce_kernel(const float* __restrict__ part,
+                   float* __restrict__ dw, long cells, int nparts) {
+  for (long i = blockIdx.x * 256L + threadIdx.x; i < cells;
+       i += (long)gridDim.x * 256) {
+    float s = 0.0f;
+    for (int p = 0; p < nparts; ++p) s += part[p * cells + i];
+    dw[i] = s;
   }
 }
 
@@ -166,24 +210,33 @@ at::Tensor conv_s1_wrw2(at::Tensor x, at::Tensor dy, int64_t R,
   TORCH_CHECK(C == 64 && K == 64, "conv_s1_wrw2: C == K == 64 only");
   TORCH_CHECK(OH == H + 2 * pad - R + 1 && OW == W + 2 * pad - S + 1,
               "conv_s1_wrw2: dy shape mismatch");
-  auto dw = at::zeros({(long)R * S, C, K},
-                      x.options().dtype(at::kFloat));
+  const long cells = (long)R * S * C * K;
   const int tiles_h = (OH + W2TILE_H - 1) / W2TILE_H;
   const int tiles_w = (OW + W2TILE_W - 1) / W2TILE_W;
   const int total_windows = (int)((long)N * tiles_h * tiles_w);
-  const int window_groups = std::min(total_windows, 512);
+  // One WG per CU (the kernel is VGPR-bound at 1 WG/CU anyway); each
+  // WG owns a disjoint partial slice, summed by the reduce kernel.
+  const int window_groups = std::min(total_windows, 256);
+  auto part = at::empty({window_groups, (long)R * S, C, K},
+                        x.options().dtype(at::kFloat));
+  auto dw = at::empty({(long)R * S, C, K},
+                      x.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
   auto launch = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(window_groups), dim3(512), 0,
                        stream.stream(),
                        (const w2bf16_t*)x.data_ptr(),
                        (const w2bf16_t*)dy.data_ptr(),
-                       (float*)dw.data_ptr(),
+                       (float*)part.data_ptr(),
                        N, H, W, (int)pad, OH, OW, tiles_h, tiles_w,
                        window_groups);
   };
   if (R == 3 && S == 3) launch(conv_s1_wrw2_kernel<3, 3>);
   else if (R == 5 && S == 5) launch(conv_s1_wrw2_kernel<5, 5>);
   else TORCH_CHECK(false, "conv_s1_wrw2: R/S must be 3x3 or 5x5");
+  const int rblocks = (int)std::min((cells + 255) / 256, 1024L);
+  hipLaunchKernelGGL(wrw2_reduce_kernel, dim3(rblocks), dim3(256), 0,
+                     stream.stream(), (const float*)part.data_ptr(),
+                     (float*)dw.data_ptr(), cells, window_groups);
   return dw;
 }
