@@ -67,23 +67,32 @@ class _MFMAConvFunction(torch.autograd.Function):
     k, c, r, s = weight.shape
     if not x.is_contiguous(memory_format=torch.channels_last):
       x = x.contiguous(memory_format=torch.channels_last)
-    wpk = ext.pack_conv_w(weight, False)
+    if x.requires_grad:
+      # One dispatch for both packs; the dgrad pack rides to backward.
+      wpk, wpk_b = ext.pack_conv_w_pair(weight)
+    else:
+      wpk, wpk_b = ext.pack_conv_w(weight, False), None
     y = ext.conv_s1_nhwc(x, wpk, k, r, s, pad)
-    ctx.save_for_backward(x, weight)
+    if wpk_b is None:
+      ctx.save_for_backward(x, weight)
+    else:
+      ctx.save_for_backward(x, weight, wpk_b)
     ctx.pad = pad
     return y
 
   @staticmethod
   def backward(ctx, dy):
     ext = ops_mod.require_hip()
-    x, weight = ctx.saved_tensors
+    x, weight = ctx.saved_tensors[:2]
+    wpk_b = ctx.saved_tensors[2] if len(ctx.saved_tensors) > 2 else None
     k, c, r, s = weight.shape
     dy = dy.contiguous(memory_format=torch.channels_last)
     dx = dw = None
     if ctx.needs_input_grad[0]:
       # SAME-pad duality: the dy->dx conv pad is (R-1-pad).
       bpad = r - 1 - ctx.pad
-      wpk_b = ext.pack_conv_w(weight, True)
+      if wpk_b is None:
+        wpk_b = ext.pack_conv_w(weight, True)
       dx = ext.conv_s1_nhwc(dy.to(torch.bfloat16), wpk_b, c, r, s, bpad)
     if ctx.needs_input_grad[1]:
       # OPT-IN: T2R_ENABLE_MFMA_WRW=2 selects the v2 register-

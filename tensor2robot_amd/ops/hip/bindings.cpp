@@ -21,6 +21,7 @@ at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
                         int64_t R, int64_t S, int64_t pad);
 
 at::Tensor pack_conv_w(at::Tensor w, bool transpose);
+std::vector<at::Tensor> pack_conv_w_pair(at::Tensor w);
 at::Tensor s2d_stem(at::Tensor x);
 
 at::Tensor conv_s1_wrw(at::Tensor x, at::Tensor dy, int64_t R, int64_t S,
@@ -57,6 +58,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA stride-1 NHWC bf16 conv (prepacked weights)");
   m.def("pack_conv_w", &pack_conv_w,
         "single-kernel conv weight pack (transpose=bwd-data layout)");
+  m.def("pack_conv_w_pair", &pack_conv_w_pair,
+        "fwd + dgrad weight packs in one dispatch");
   m.def("s2d_stem", &s2d_stem,
         "fused space-to-depth [N,3,H,W]->[N,16,H/2,W/2] for the stem");
   m.def("conv_s1_wrw2", &conv_s1_wrw2,

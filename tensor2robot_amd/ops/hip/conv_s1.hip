@@ -441,6 +441,61 @@ at::Tensor pack_conv_w(at::Tensor w, bool transpose) {
   return out;
 }
 
+__global__ void __launch_bounds__(256)
+pack_conv_w_pair_kernel(const cbf16_t* __restrict__ w,  // [K][C][R][S]
+                        cbf16_t* __restrict__ outf,
+                        cbf16_t* __restrict__ outb,
+                        int K, int C, int R, int S,
+                        long totalf, long totalb) {
+  for (long ii = blockIdx.x * (long)blockDim.x + threadIdx.x;
+       ii < totalf + totalb; ii += (long)gridDim.x * blockDim.x) {
+    const bool tr = ii >= totalf;
+    const long i = tr ? ii - totalf : ii;
+    const int crole = tr ? K : C;
+    const int krole = tr ? C : K;
+    const int cc = i % WPAD;
+    long rest = i / WPAD;
+    const int n = rest % krole;
+    rest /= krole;
+    const int c16 = rest % (crole / 16);
+    const int rs = rest / (crole / 16);
+    cbf16_t v = __float2bfloat16(0.0f);
+    if (cc < 16) {
+      const int c = c16 * 16 + cc;
+      int r = rs / S, s = rs % S;
+      int kk, ci;
+      if (tr) { kk = c; ci = n; r = R - 1 - r; s = S - 1 - s; }
+      else    { kk = n; ci = c; }
+      v = w[(((long)kk * C + ci) * R + r) * S + s];
+    }
+    (tr ? outb : outf)[i] = v;
+  }
+}
+
+std::vector<at::Tensor> pack_conv_w_pair(at::Tensor w) {
+  // One dispatch emitting BOTH the forward pack and the dgrad pack
+  // (transposed c/k roles, rotated rs) — halves the per-step pack
+  // launches and takes the pack off the backward critical path.
+  TORCH_CHECK(w.is_cuda() && w.dim() == 4, "pack_conv_w_pair: 4D CUDA");
+  w = w.contiguous();
+  if (w.scalar_type() != at::kBFloat16) w = w.to(at::kBFloat16);
+  const int K = w.size(0), C = w.size(1), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(C % 16 == 0 && K % 16 == 0, "pack_conv_w_pair: C,K % 16");
+  auto outf = at::empty({(long)R * S, C / 16, K, WPAD}, w.options());
+  auto outb = at::empty({(long)R * S, K / 16, C, WPAD}, w.options());
+  const long totalf = (long)R * S * (C / 16) * K * WPAD;
+  const long totalb = (long)R * S * (K / 16) * C * WPAD;
+  const int grid =
+      (int)std::min<long>((totalf + totalb + 255) / 256, 4096);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(pack_conv_w_pair_kernel, dim3(grid), dim3(256), 0,
+                     stream.stream(), (const cbf16_t*)w.data_ptr(),
+                     (cbf16_t*)outf.data_ptr(),
+                     (cbf16_t*)outb.data_ptr(), K, C, R, S,
+                     totalf, totalb);
+  return {outf, outb};
+}
+
 // ---------------------------------------------------------------------------
 // glds-ring variant for the 5x5: weight chunks stream via LDS-DMA
 // (global_load_lds) through a 3-slot ring with counted vmcnt + raw

@@ -26,16 +26,14 @@ typedef __hip_bfloat16 w2bf16_t;
 typedef __attribute__((ext_vector_type(8))) short w2bf16x8;
 typedef __attribute__((ext_vector_type(16))) float w2f32x16;
 
-#define W2TILE_H 8
 #define W2TILE_W 16
-#define W2WIN_P (W2TILE_H * W2TILE_W)      // 128 window pixels
-#define W2HALO_H (W2TILE_H + 4)
 #define W2HALO_W (W2TILE_W + 4)
-#define W2HALO_P (W2HALO_H * W2HALO_W)     // 240
-#define W2XT_PITCH 248                     // halo pixels + pad
-#define W2DYT_PITCH 136                    // window pixels + pad
 
-template <int R, int S>
+// TH = window tile height: 8 (128-pixel window, 49 KiB LDS) for the
+// 3x3; 16 (256-pixel window, 86 KiB LDS) for the 5x5 — the bigger
+// window doubles MFMAs per staged byte, which the 25-offset reuse
+// needs to go compute-bound.
+template <int R, int S, int TH>
 __global__ void __launch_bounds__(512, 1)
 conv_s1_wrw2_kernel(const w2bf16_t* __restrict__ x,
                     const w2bf16_t* __restrict__ dy,
@@ -46,6 +44,12 @@ conv_s1_wrw2_kernel(const w2bf16_t* __restrict__ x,
   constexpr int C = 64, K = 64;
   constexpr int RS = R * S;
   constexpr int RS_PER = (RS + 1) / 2;     // rs offsets per wave half
+  constexpr int W2TILE_H = TH;
+  constexpr int W2WIN_P = W2TILE_H * W2TILE_W;
+  constexpr int W2HALO_H = W2TILE_H + 4;
+  constexpr int W2HALO_P = W2HALO_H * W2HALO_W;
+  constexpr int W2XT_PITCH = W2HALO_P + 8;     // multiple of 8 (b128)
+  constexpr int W2DYT_PITCH = W2WIN_P + 8;
   __shared__ short lds_xt[C * W2XT_PITCH];
   __shared__ short lds_dyt[K * W2DYT_PITCH];
 
@@ -211,7 +215,8 @@ at::Tensor conv_s1_wrw2(at::Tensor x, at::Tensor dy, int64_t R,
   TORCH_CHECK(OH == H + 2 * pad - R + 1 && OW == W + 2 * pad - S + 1,
               "conv_s1_wrw2: dy shape mismatch");
   const long cells = (long)R * S * C * K;
-  const int tiles_h = (OH + W2TILE_H - 1) / W2TILE_H;
+  const int tile_h = (R == 5) ? 16 : 8;
+  const int tiles_h = (OH + tile_h - 1) / tile_h;
   const int tiles_w = (OW + W2TILE_W - 1) / W2TILE_W;
   const int total_windows = (int)((long)N * tiles_h * tiles_w);
   // One WG per CU (the kernel is VGPR-bound at 1 WG/CU anyway); each
@@ -231,8 +236,8 @@ at::Tensor conv_s1_wrw2(at::Tensor x, at::Tensor dy, int64_t R,
                        N, H, W, (int)pad, OH, OW, tiles_h, tiles_w,
                        window_groups);
   };
-  if (R == 3 && S == 3) launch(conv_s1_wrw2_kernel<3, 3>);
-  else if (R == 5 && S == 5) launch(conv_s1_wrw2_kernel<5, 5>);
+  if (R == 3 && S == 3) launch(conv_s1_wrw2_kernel<3, 3, 8>);
+  else if (R == 5 && S == 5) launch(conv_s1_wrw2_kernel<5, 5, 16>);
   else TORCH_CHECK(false, "conv_s1_wrw2: R/S must be 3x3 or 5x5");
   const int rblocks = (int)std::min((cells + 255) / 256, 1024L);
   hipLaunchKernelGGL(wrw2_reduce_kernel, dim3(rblocks), dim3(256), 0,
