@@ -7,9 +7,11 @@ aten.convolution_backward (MIOpen wrw) — wrw is a [64 x 64] x 200k-K
 reduction where MIOpen's split-K igemm is already reasonable.
 
 Weights are prepacked per call to the kernel's LDS-friendly layout
-[rs][c/16][k][24] (16 used + 8 pad for conflict-free B-fragment reads);
-the pack is a few tensor ops on a 200 KB tensor and is captured inside
-hipGraphs along with the conv.
+[rs][c/16][k][24] (16 used + 8 pad for conflict-free B-fragment reads)
+by one HIP kernel (pack_conv_w); when the input needs a gradient, a
+single paired dispatch (pack_conv_w_pair) emits the forward AND the
+flipped/transposed dgrad pack together, and the dgrad pack is saved to
+backward.  Everything is captured inside hipGraphs along with the conv.
 """
 
 from __future__ import annotations
