@@ -1,15 +1,14 @@
 // Fused training BatchNorm + ReLU for NHWC bf16 tensors (gfx950 / CDNA4).
 //
 // Replaces the MIOpen BN kernel chain (MeanVariance, FinalMeanVariance,
-// Norm) + separate ReLU clamp + their backward counterparts with 2 forward
-// and 2 backward kernels, fewer HBM passes and no contended atomics:
+// Norm) + separate ReLU clamp + their backward counterparts with 3 forward
+// and 3 backward kernels, fewer HBM passes and no f32 atomics:
 //
-//   fwd: stats (per-WG partial slabs; the LAST block — ticket counter —
-//        reduces the slabs and emits mean/invstd/scale/shift + running
-//        stats in the same launch) -> apply (normalize + ReLU, one pass)
+//   fwd: stats (per-WG partial slabs) -> finalize (reduce + scale/shift +
+//        running stats) -> apply (normalize + ReLU, one pass)
 //   bwd: reduce (partial dbeta/dgamma slabs; ReLU mask RECOMPUTED from x,
-//        so only x and dy are read - y is never saved; last block emits
-//        dbeta/dgamma + k1/k2/k3) -> dx = k1*g + k2*x + k3 (one pass)
+//        so only x and dy are read - y is never saved) -> coeffs ->
+//        dx = k1*g + k2*x + k3 (one pass)
 //
 // Design notes (cdna_hip_programming.md):
 //  * wave64; 256-thread workgroups; each thread owns 8 consecutive
@@ -44,17 +43,6 @@ union Vec8 {
 
 // Workgroups for the streaming kernels: enough to fill 256 CUs several
 // times over, small enough that the partial-slab reduction stays cheap.
-// One self-resetting ticket per direction (fwd=0, bwd=1); allocated
-// lazily OUTSIDE graph capture (warmup always precedes capture).
-static int* bn_ticket(int which) {
-  static int* buf = nullptr;
-  if (buf == nullptr) {
-    (void)hipMalloc(&buf, 2 * sizeof(int));
-    (void)hipMemset(buf, 0, 2 * sizeof(int));
-  }
-  return buf + which;
-}
-
 static int pick_grid(long M, int rows, int cap = 1024) {
   long wgs = (M + rows - 1) / rows;
   if (wgs > cap) wgs = cap;
@@ -197,6 +185,50 @@ bn_stats_kernel(const bf16_t* __restrict__ x, float* __restrict__ partial,
 // Finalize: reduce slabs; mean/invstd + scale/shift + running stats.
 // ---------------------------------------------------------------------------
 
+// One block per channel; 256 threads stride the WG slabs, LDS tree-reduce.
+extern "C" __global__ void __launch_bounds__(256)
+bn_finalize_kernel(
+    const float* __restrict__ partial, int n_wgs,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ mean_out, float* __restrict__ invstd_out,
+    float* __restrict__ scale_out, float* __restrict__ shift_out,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    long M, int C, float eps, float momentum) {
+  const int c = blockIdx.x;
+  float s = 0.f, q = 0.f;
+  for (int w = threadIdx.x; w < n_wgs; w += 256) {
+    s += partial[(long)w * 2 * C + c];
+    q += partial[(long)w * 2 * C + C + c];
+  }
+  __shared__ float ls[256], lq[256];
+  ls[threadIdx.x] = s;
+  lq[threadIdx.x] = q;
+  __syncthreads();
+  for (int st = 128; st > 0; st >>= 1) {
+    if (threadIdx.x < st) {
+      ls[threadIdx.x] += ls[threadIdx.x + st];
+      lq[threadIdx.x] += lq[threadIdx.x + st];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    s = ls[0];
+    q = lq[0];
+    float mean = s / (float)M;
+    float var = fmaxf(q / (float)M - mean * mean, 0.f);
+    float invstd = rsqrtf(var + eps);
+    float sc = gamma[c] * invstd;
+    mean_out[c] = mean;
+    invstd_out[c] = invstd;
+    scale_out[c] = sc;
+    shift_out[c] = beta[c] - mean * sc;
+    if (running_mean != nullptr) {
+      float unbiased = (M > 1) ? var * (float)M / (float)(M - 1) : var;
+      running_mean[c] += momentum * (mean - running_mean[c]);
+      running_var[c] += momentum * (unbiased - running_var[c]);
+    }
+  }
+}
 
 // ---------------------------------------------------------------------------
 // Apply: y = relu?(x * scale + shift), one coalesced pass.
@@ -236,19 +268,6 @@ bn_apply_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
 // relu_mask recomputed as (xhat*gamma + beta > 0) - no y stream.
 // ---------------------------------------------------------------------------
 
-__device__ __forceinline__ void bn_coeffs_one(
-    int c, float db, float dg, const float* gamma, const float* mean,
-    const float* invstd, float* dbeta_out, float* dgamma_out, float* k1,
-    float* k2, float* k3, long M) {
-  dbeta_out[c] = db;
-  dgamma_out[c] = dg;
-  const float gs = gamma[c] * invstd[c];
-  const float t = dg / (float)M * invstd[c];
-  k1[c] = gs;
-  k2[c] = -gs * t;
-  k3[c] = gs * (mean[c] * t - db / (float)M);
-}
-
 extern "C" __global__ void __launch_bounds__(256)
 bn_bwd_reduce_kernel(const bf16_t* __restrict__ x,
                      const bf16_t* __restrict__ dy,
@@ -257,11 +276,7 @@ bn_bwd_reduce_kernel(const bf16_t* __restrict__ x,
                      const float* __restrict__ gamma,
                      const float* __restrict__ beta,
                      float* __restrict__ partial, long M, int C,
-                     int fused_relu,
-                     float* __restrict__ dbeta_out,
-                     float* __restrict__ dgamma_out,
-                     float* __restrict__ k1, float* __restrict__ k2,
-                     float* __restrict__ k3, int* __restrict__ ticket) {
+                     int fused_relu) {
   const int tpr = C >> 3;
   const int rows = 256 / tpr;
   const int rg = threadIdx.x / tpr;
@@ -309,53 +324,6 @@ bn_bwd_reduce_kernel(const bf16_t* __restrict__ x,
     out[c] = a;
     out[C + c] = b;
   }
-
-  // ---- last-block coeffs (replaces bn_bwd_coeffs_kernel) ----
-  __shared__ int lb;
-  if (threadIdx.x == 0) {
-    __threadfence();
-    const int prev = atomicAdd(ticket, 1);
-    lb = (prev == (int)gridDim.x - 1);
-    if (lb) *ticket = 0;
-  }
-  __syncthreads();
-  if (!lb) return;
-  __threadfence();
-  const int n_wgs = (int)gridDim.x;
-  float* red = lds;
-  const int tpc = (C <= 256) ? (256 / C) : 1;
-  if (C <= 256) {
-    const int c = threadIdx.x / tpc;
-    const int sub = threadIdx.x % tpc;
-    float da = 0.f, ga2 = 0.f;
-    if (c < C) {
-      for (int w = sub; w < n_wgs; w += tpc) {
-        da += partial[(long)w * 2 * C + c];
-        ga2 += partial[(long)w * 2 * C + C + c];
-      }
-    }
-    red[threadIdx.x] = da;
-    red[256 + threadIdx.x] = ga2;
-    __syncthreads();
-    if (c < C && sub == 0) {
-      for (int j = 1; j < tpc; ++j) {
-        da += red[threadIdx.x + j];
-        ga2 += red[256 + threadIdx.x + j];
-      }
-      bn_coeffs_one(c, da, ga2, gamma, mean, invstd, dbeta_out,
-                    dgamma_out, k1, k2, k3, M);
-    }
-  } else {
-    for (int cc = threadIdx.x; cc < C; cc += 256) {
-      float da = 0.f, ga2 = 0.f;
-      for (int w = 0; w < n_wgs; ++w) {
-        da += partial[(long)w * 2 * C + cc];
-        ga2 += partial[(long)w * 2 * C + C + cc];
-      }
-      bn_coeffs_one(cc, da, ga2, gamma, mean, invstd, dbeta_out,
-                    dgamma_out, k1, k2, k3, M);
-    }
-  }
 }
 
 // ---------------------------------------------------------------------------
@@ -364,6 +332,42 @@ bn_bwd_reduce_kernel(const bf16_t* __restrict__ x,
 //      = k1*g + k2*x + k3
 // ---------------------------------------------------------------------------
 
+extern "C" __global__ void __launch_bounds__(256)
+bn_bwd_coeffs_kernel(
+    const float* __restrict__ partial, int n_wgs,
+    const float* __restrict__ gamma, const float* __restrict__ mean,
+    const float* __restrict__ invstd, float* __restrict__ dbeta_out,
+    float* __restrict__ dgamma_out, float* __restrict__ k1,
+    float* __restrict__ k2, float* __restrict__ k3, long M, int C) {
+  const int c = blockIdx.x;
+  float db = 0.f, dg = 0.f;
+  for (int w = threadIdx.x; w < n_wgs; w += 256) {
+    db += partial[(long)w * 2 * C + c];
+    dg += partial[(long)w * 2 * C + C + c];
+  }
+  __shared__ float ls[256], lq[256];
+  ls[threadIdx.x] = db;
+  lq[threadIdx.x] = dg;
+  __syncthreads();
+  for (int st = 128; st > 0; st >>= 1) {
+    if (threadIdx.x < st) {
+      ls[threadIdx.x] += ls[threadIdx.x + st];
+      lq[threadIdx.x] += lq[threadIdx.x + st];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    db = ls[0];
+    dg = lq[0];
+    dbeta_out[c] = db;
+    dgamma_out[c] = dg;
+    float gs = gamma[c] * invstd[c];
+    float t = dg / (float)M * invstd[c];
+    k1[c] = gs;
+    k2[c] = -gs * t;
+    k3[c] = gs * (mean[c] * t - db / (float)M);
+  }
+}
 
 extern "C" __global__ void __launch_bounds__(256)
 bn_bwd_dx_kernel(const bf16_t* __restrict__ x,
@@ -438,19 +442,21 @@ std::vector<at::Tensor> fused_bn_relu_forward(
   const size_t lds_bytes = 2l * rows * C * sizeof(float);
   auto stream = at::cuda::getCurrentCUDAStream();
 
+  hipLaunchKernelGGL(bn_stats_kernel, dim3(grid), dim3(256), lds_bytes,
+                     stream.stream(),
+                     reinterpret_cast<const bf16_t*>(x.data_ptr()),
+                     partial.data_ptr<float>(), M, C);
   float* rm = running_mean.has_value()
                   ? running_mean->data_ptr<float>() : nullptr;
   float* rv = running_var.has_value()
                   ? running_var->data_ptr<float>() : nullptr;
   float* stats_ptr = stats.data_ptr<float>();
-  hipLaunchKernelGGL(bn_stats_kernel, dim3(grid), dim3(256), lds_bytes,
-                     stream.stream(),
-                     reinterpret_cast<const bf16_t*>(x.data_ptr()),
-                     partial.data_ptr<float>(), M, C,
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(256),
+                     0, stream.stream(), partial.data_ptr<float>(), grid,
                      gamma.data_ptr<float>(), beta.data_ptr<float>(),
                      stats_ptr, stats_ptr + C, stats_ptr + 2 * C,
-                     stats_ptr + 3 * C, rm, rv, (float)eps,
-                     (float)momentum, bn_ticket(0));
+                     stats_ptr + 3 * C, rm, rv, M, C, (float)eps,
+                     (float)momentum);
   hipLaunchKernelGGL(bn_apply_kernel, dim3(grid), dim3(256), 0,
                      stream.stream(),
                      reinterpret_cast<const bf16_t*>(x.data_ptr()),
@@ -505,9 +511,12 @@ std::vector<at::Tensor> fused_bn_relu_backward(
                      reinterpret_cast<const bf16_t*>(dy.data_ptr()),
                      stats_ptr, stats_ptr + C, gamma.data_ptr<float>(),
                      beta.data_ptr<float>(), partial.data_ptr<float>(), M,
-                     C, fused_relu ? 1 : 0, grads_ptr, grads_ptr + C,
-                     coeffs_ptr, coeffs_ptr + C, coeffs_ptr + 2 * C,
-                     bn_ticket(1));
+                     C, fused_relu ? 1 : 0);
+  hipLaunchKernelGGL(bn_bwd_coeffs_kernel, dim3(C), dim3(256),
+                     0, stream.stream(), partial.data_ptr<float>(), grid,
+                     gamma.data_ptr<float>(), stats_ptr, stats_ptr + C,
+                     grads_ptr, grads_ptr + C, coeffs_ptr, coeffs_ptr + C,
+                     coeffs_ptr + 2 * C, M, C);
   hipLaunchKernelGGL(bn_bwd_dx_kernel, dim3(grid), dim3(256), 0,
                      stream.stream(),
                      reinterpret_cast<const bf16_t*>(x.data_ptr()),
