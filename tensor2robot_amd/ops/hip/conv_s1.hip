@@ -503,12 +503,19 @@ std::vector<at::Tensor> pack_conv_w_pair(at::Tensor w) {
 // no register round-trip, no write pass, loads 2 chunks ahead.
 // ---------------------------------------------------------------------------
 
+#define RING_DEPTH 6
+
 __device__ __forceinline__ void conv_waitcnt_vm(int ahead) {
   // s_waitcnt imm: vmcnt[3:0], expcnt[6:4]=7, lgkmcnt[11:8]=15.
+  // Each outstanding chunk is 2 glds pieces per issuer wave.
   switch (ahead) {
     case 0: __builtin_amdgcn_s_waitcnt(0 | (7 << 4) | (15 << 8)); break;
     case 1: __builtin_amdgcn_s_waitcnt(2 | (7 << 4) | (15 << 8)); break;
-    default: __builtin_amdgcn_s_waitcnt(4 | (7 << 4) | (15 << 8)); break;
+    case 2: __builtin_amdgcn_s_waitcnt(4 | (7 << 4) | (15 << 8)); break;
+    case 3: __builtin_amdgcn_s_waitcnt(6 | (7 << 4) | (15 << 8)); break;
+    case 4: __builtin_amdgcn_s_waitcnt(8 | (7 << 4) | (15 << 8)); break;
+    default:
+      __builtin_amdgcn_s_waitcnt(10 | (7 << 4) | (15 << 8)); break;
   }
 }
 
@@ -523,7 +530,11 @@ conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
   constexpr int C = C16N * 16;
   constexpr int WBUF = C16N * NTILES * 32 * WPAD;      // bf16 per chunk
   constexpr int PIECES = (WBUF * 2) / 1024;            // 1-KiB DMA pieces
-  __shared__ short lds[BHALO * BHALO * XPITCH + 3 * WBUF];
+  // 6-deep ring (131 KiB LDS with the x halo): the MFMA phase per rs
+  // is only ~250 cycles at the 1-WG/CU occupancy this LDS footprint
+  // forces, so a 2-chunk lead cannot hide L2/HBM latency — 5 chunks
+  // in flight can.
+  __shared__ short lds[BHALO * BHALO * XPITCH + RING_DEPTH * WBUF];
   short* xtile = lds;
   auto wbuf = [&](int slot) -> short* {
     return lds + BHALO * BHALO * XPITCH + slot * WBUF;
@@ -581,8 +592,9 @@ conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
           &xtile[(hrow * BHALO + hcol) * XPITCH + chunk * 8]) = v;
     }
   }
-  issue_chunk(0, 0);
-  if (R * S > 1) issue_chunk(1, 1);
+  for (int cpre = 0; cpre < RING_DEPTH - 1 && cpre < R * S; ++cpre) {
+    issue_chunk(cpre, cpre % RING_DEPTH);
+  }
   // One full drain in the prologue (also covers the x-tile loads).
   __syncthreads();
 
@@ -597,9 +609,10 @@ conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
   const int RS = R * S;
   for (int rs = 0; rs < RS; ++rs) {
     const int r = rs / S, s = rs % S;
-    const int slot = rs % 3;
-    if (rs + 2 < RS) issue_chunk(rs + 2, (rs + 2) % 3);
-    if (issuer) conv_waitcnt_vm(min(RS - 1 - rs, 2));
+    const int slot = rs % RING_DEPTH;
+    if (rs + RING_DEPTH - 1 < RS)
+      issue_chunk(rs + RING_DEPTH - 1, (rs + RING_DEPTH - 1) % RING_DEPTH);
+    if (issuer) conv_waitcnt_vm(min(RS - 1 - rs, RING_DEPTH - 1));
     __builtin_amdgcn_s_barrier();     // chunk rs landed for everyone
 #pragma unroll
     for (int c16 = 0; c16 < C16N; ++c16) {
@@ -615,8 +628,9 @@ conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
             a_frag, b_frag, acc[nt], 0, 0, 0);
       }
     }
-    // Slot rs%3 is refilled at iteration rs+1 (chunk rs+3): everyone
-    // must be done reading before that DMA can land.
+    // Slot rs%RING_DEPTH is refilled by the DMA issued at iteration
+    // rs+1 (chunk rs+RING_DEPTH): everyone must be done reading
+    // before that DMA can be issued.
     __builtin_amdgcn_s_barrier();
   }
 
