@@ -54,44 +54,9 @@ static int pick_grid(long M, int rows, int cap = 1024) {
 // Forward stats: per-WG partial sum / sumsq slabs [n_wgs, 2, C].
 // ---------------------------------------------------------------------------
 
-__device__ __forceinline__ void bn_finalize_one(
-    int c, float s, float q, const float* gamma, const float* beta,
-    float* mean_out, float* invstd_out, float* scale_out,
-    float* shift_out, float* running_mean, float* running_var, long M,
-    float eps, float momentum) {
-  const float mean = s / (float)M;
-  const float var = fmaxf(q / (float)M - mean * mean, 0.f);
-  const float invstd = rsqrtf(var + eps);
-  const float sc = gamma[c] * invstd;
-  mean_out[c] = mean;
-  invstd_out[c] = invstd;
-  scale_out[c] = sc;
-  shift_out[c] = beta[c] - mean * sc;
-  if (running_mean != nullptr) {
-    const float unbiased = (M > 1) ? var * (float)M / (float)(M - 1) : var;
-    running_mean[c] += momentum * (mean - running_mean[c]);
-    running_var[c] += momentum * (unbiased - running_var[c]);
-  }
-}
-
-// The ticket counter lets the LAST workgroup of the streaming reduce
-// fold the finalize step into the same launch (one fewer kernel per BN
-// per step).  Self-resetting: the last block zeroes it for the next
-// call, so one statically-allocated int per direction suffices
-// (single-stream assumption: all BN layers of a net run on one
-// stream, as the train step and hipGraph capture do).
 extern "C" __global__ void __launch_bounds__(256)
 bn_stats_kernel(const bf16_t* __restrict__ x, float* __restrict__ partial,
-                long M, int C,
-                const float* __restrict__ gamma,
-                const float* __restrict__ beta,
-                float* __restrict__ mean_out,
-                float* __restrict__ invstd_out,
-                float* __restrict__ scale_out,
-                float* __restrict__ shift_out,
-                float* __restrict__ running_mean,
-                float* __restrict__ running_var,
-                float eps, float momentum, int* __restrict__ ticket) {
+                long M, int C) {
   const int tpr = C >> 3;
   const int rows = 256 / tpr;
   const int rg = threadIdx.x / tpr;
@@ -129,55 +94,6 @@ bn_stats_kernel(const bf16_t* __restrict__ x, float* __restrict__ partial,
     }
     out[c] = a;
     out[C + c] = b;
-  }
-
-  // ---- last-block finalize (replaces bn_finalize_kernel) ----
-  __shared__ int lb;
-  if (threadIdx.x == 0) {
-    __threadfence();                       // slab visible before ticket
-    const int prev = atomicAdd(ticket, 1);
-    lb = (prev == (int)gridDim.x - 1);
-    if (lb) *ticket = 0;                   // reset for the next call
-  }
-  __syncthreads();
-  if (!lb) return;
-  __threadfence();                         // see every block's slab
-  const int n_wgs = (int)gridDim.x;
-  float* red = lds;                        // lds is >= 4096 floats
-  const int tpc = (C <= 256) ? (256 / C) : 1;
-  const int c = (C <= 256) ? (int)threadIdx.x / tpc : -1;
-  if (C <= 256) {
-    const int sub = threadIdx.x % tpc;
-    float sa = 0.f, qa = 0.f;
-    if (c < C) {
-      for (int w = sub; w < n_wgs; w += tpc) {
-        sa += partial[(long)w * 2 * C + c];
-        qa += partial[(long)w * 2 * C + C + c];
-      }
-    }
-    red[threadIdx.x] = sa;
-    red[256 + threadIdx.x] = qa;
-    __syncthreads();
-    if (c < C && (threadIdx.x % tpc) == 0) {
-      for (int j = 1; j < tpc; ++j) {
-        sa += red[threadIdx.x + j];
-        qa += red[256 + threadIdx.x + j];
-      }
-      bn_finalize_one(c, sa, qa, gamma, beta, mean_out, invstd_out,
-                      scale_out, shift_out, running_mean, running_var,
-                      M, eps, momentum);
-    }
-  } else {
-    for (int cc = threadIdx.x; cc < C; cc += 256) {
-      float sa = 0.f, qa = 0.f;
-      for (int w = 0; w < n_wgs; ++w) {
-        sa += partial[(long)w * 2 * C + cc];
-        qa += partial[(long)w * 2 * C + C + cc];
-      }
-      bn_finalize_one(cc, sa, qa, gamma, beta, mean_out, invstd_out,
-                      scale_out, shift_out, running_mean, running_var,
-                      M, eps, momentum);
-    }
   }
 }
 
