@@ -79,7 +79,10 @@ def main():
   world_size = int(os.environ.get("WORLD_SIZE", "1"))
   rank = int(os.environ.get("RANK", "0"))
   local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-  distributed = world_size > 1
+  # T2R_FORCE_DIST exercises the distributed code path at world_size 1
+  # (a 1-GPU box): single-rank RCCL collectives are no-ops but run the
+  # same code the 8-GPU driver launch does.
+  distributed = world_size > 1 or bool(os.environ.get("T2R_FORCE_DIST"))
 
   use_cuda = torch.cuda.is_available()
   if use_cuda:
@@ -104,8 +107,18 @@ def main():
   network = model.network
   network.to(memory_format=torch.channels_last)
 
+  want_graph = use_cuda and not args.no_hipgraph
   dp_engine = None
-  if distributed:
+  if distributed and torch.distributed.is_initialized():
+    # Identical start on every rank (the reference's chief-initialized
+    # variables); the graphed-dist path has no DDP hooks to do it.
+    for prm in network.parameters():
+      torch.distributed.broadcast(prm.data, src=0)
+  if distributed and not want_graph:
+    # Eager path: hook-driven bucketed all-reduce overlapped with
+    # backward.  NOT used under graphs: the hooks would record RCCL
+    # collectives into the capture; the graphed-dist step instead syncs
+    # grads eagerly between graph replay and the optimizer step.
     from tensor2robot_amd.parallel import ddp
     dp_engine = ddp.DataParallelEngine(network)
 
@@ -152,12 +165,13 @@ def main():
   # outside the graph because its distortion params come from host RNG) --
   graphed = None
   static = {}
-  # hipGraph is single-GPU only this round: RCCL collectives inside a
-  # captured graph are unvalidated on this pool (no multi-GPU box to
-  # test on), and a hung capture on the 8-GPU scaling run is worse than
-  # the launch-overhead win.  The DP path keeps eager overlapped
-  # all-reduce.
-  if use_cuda and not distributed and not args.no_hipgraph:
+  # Distributed: the graph captures forward+backward ONLY; the gradient
+  # all-reduce + optimizer + EMA run eager after each replay.  RCCL
+  # collectives are never recorded into a capture (unvalidated on this
+  # pool - no multi-GPU box to test on), yet the ~250-kernel fwd+bwd
+  # still replays as one launch, so the 8-GPU scaling numbers are not
+  # stuck at eager launch overhead.
+  if use_cuda and not args.no_hipgraph:
     try:
       from tensor2robot_amd.parallel import graph_step
       for i in range(3):  # settle MIOpen algo find before capture
@@ -185,28 +199,55 @@ def main():
           logit = network(static["image"], static["action"])
         loss = torch.nn.functional.binary_cross_entropy_with_logits(
             logit.float(), static["labels"].reshape(logit.shape))
-        if dp_engine is not None:
-          dp_engine.backward(loss)
-        else:
-          loss.backward()
-        optimizer.step(0)
-        if ema is not None:
-          ema.update()
+        loss.backward()
+        if not distributed:
+          optimizer.step(0)
+          if ema is not None:
+            ema.update()
         return loss
 
       graphed = graph_step.GraphedTrainStep(graph_body)
 
-      def step(i, global_step):  # noqa: F811 (graph-replay fast path)
-        images, action, labels_t = pool[i % len(pool)]
-        preprocess_to(images, action, static["image"], static["action"])
-        static["labels"].copy_(labels_t.reshape(-1, 1))
-        return graphed.replay()
+      if distributed:
+        grad_params = [prm for prm in network.parameters()
+                       if prm.requires_grad]
+
+        def sync_grads():
+          grads = [prm.grad for prm in grad_params
+                   if prm.grad is not None]
+          flat = torch._utils._flatten_dense_tensors(grads)
+          torch.distributed.all_reduce(flat)
+          flat.div_(world_size)
+          for g, g_synced in zip(
+              grads, torch._utils._unflatten_dense_tensors(flat, grads)):
+            g.copy_(g_synced)
+
+        def step(i, global_step):  # noqa: F811 (graphed fwd+bwd)
+          images, action, labels_t = pool[i % len(pool)]
+          preprocess_to(images, action, static["image"], static["action"])
+          static["labels"].copy_(labels_t.reshape(-1, 1))
+          loss = graphed.replay()
+          sync_grads()
+          optimizer.step(global_step)
+          if ema is not None:
+            ema.update()
+          return loss
+      else:
+
+        def step(i, global_step):  # noqa: F811 (graph-replay fast path)
+          images, action, labels_t = pool[i % len(pool)]
+          preprocess_to(images, action, static["image"], static["action"])
+          static["labels"].copy_(labels_t.reshape(-1, 1))
+          return graphed.replay()
     except Exception as e:  # pragma: no cover - depends on runtime
       if rank == 0:
         import sys as _sys
         print(f"# hipGraph capture unavailable, running eager: {e!r}",
               file=_sys.stderr, flush=True)
       graphed = None
+      if distributed and dp_engine is None:
+        from tensor2robot_amd.parallel import ddp
+        dp_engine = ddp.DataParallelEngine(network)
 
   def barrier_sync():
     if distributed:
