@@ -295,7 +295,7 @@ conv_s1_nhwc_big_kernel(const cbf16_t* __restrict__ x,
   }
 }
 
-template <int C16N, int NTILES>
+template <int C16N, int NTILES, int RING_DEPTH>
 __global__ void conv_s1_nhwc_ring_kernel(
     const cbf16_t* __restrict__ x, const cbf16_t* __restrict__ wpk,
     cbf16_t* __restrict__ y, int N, int H, int W, int K, int R, int S,
@@ -355,17 +355,24 @@ at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
                          OH, OW, tiles_h, tiles_w);
     };
     static const bool use_ring = std::getenv("T2R_CONV_NO_RING") == nullptr;
+    // T2R_RING_DEPTH: 3 (shallow, 94.5 KiB LDS) or 6 (deep, 131 KiB;
+    // default) weight-chunk ring for same-box A/Bs.
+    static const bool deep_ring = []() {
+      const char* v = std::getenv("T2R_RING_DEPTH");
+      return v == nullptr || v[0] != '3';
+    }();
     if (C == 64 && K == 64)
-      use_ring ? launch(conv_s1_nhwc_ring_kernel<4, 2>)
+      use_ring ? (deep_ring ? launch(conv_s1_nhwc_ring_kernel<4, 2, 6>)
+                            : launch(conv_s1_nhwc_ring_kernel<4, 2, 3>))
                : launch(conv_s1_nhwc_big_kernel<4, 2>);
     else if (C == 32 && K == 32)
-      use_ring ? launch(conv_s1_nhwc_ring_kernel<2, 1>)
+      use_ring ? launch(conv_s1_nhwc_ring_kernel<2, 1, 6>)
                : launch(conv_s1_nhwc_big_kernel<2, 1>);
     else if (C == 48 && K == 64)
-      use_ring ? launch(conv_s1_nhwc_ring_kernel<3, 2>)
+      use_ring ? launch(conv_s1_nhwc_ring_kernel<3, 2, 6>)
                : launch(conv_s1_nhwc_big_kernel<3, 2>);
     else if (C == 16 && K == 32)
-      use_ring ? launch(conv_s1_nhwc_ring_kernel<1, 1>)
+      use_ring ? launch(conv_s1_nhwc_ring_kernel<1, 1, 6>)
                : launch(conv_s1_nhwc_big_kernel<1, 1>);
     else TORCH_CHECK(false, "conv_s1_nhwc: unsupported C/K combo");
   }
@@ -503,8 +510,6 @@ std::vector<at::Tensor> pack_conv_w_pair(at::Tensor w) {
 // no register round-trip, no write pass, loads 2 chunks ahead.
 // ---------------------------------------------------------------------------
 
-#define RING_DEPTH 6
-
 __device__ __forceinline__ void conv_waitcnt_vm(int ahead) {
   // s_waitcnt imm: vmcnt[3:0], expcnt[6:4]=7, lgkmcnt[11:8]=15.
   // Each outstanding chunk is 2 glds pieces per issuer wave.
@@ -519,7 +524,7 @@ __device__ __forceinline__ void conv_waitcnt_vm(int ahead) {
   }
 }
 
-template <int C16N, int NTILES>
+template <int C16N, int NTILES, int RING_DEPTH>
 __global__ void __launch_bounds__(512, 2)
 conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
                          const cbf16_t* __restrict__ wpk,
