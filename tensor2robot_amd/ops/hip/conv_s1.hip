@@ -355,11 +355,12 @@ at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
                          OH, OW, tiles_h, tiles_w);
     };
     static const bool use_ring = std::getenv("T2R_CONV_NO_RING") == nullptr;
-    // T2R_RING_DEPTH: 3 (shallow, 94.5 KiB LDS) or 6 (deep, 131 KiB;
-    // default) weight-chunk ring for same-box A/Bs.
+    // T2R_RING_DEPTH: 3 (shallow, 94.5 KiB LDS; default — measured
+    // 477 vs 446 TF against the 6-deep ring: the extra LDS slots buy
+    // no latency cover at 1 WG/CU and cost L2 locality) or 6.
     static const bool deep_ring = []() {
       const char* v = std::getenv("T2R_RING_DEPTH");
-      return v == nullptr || v[0] != '3';
+      return v != nullptr && v[0] == '6';
     }();
     if (C == 64 && K == 64)
       use_ring ? (deep_ring ? launch(conv_s1_nhwc_ring_kernel<4, 2, 6>)
