@@ -190,6 +190,16 @@ def main():
       preprocess_to(images0, action0, static["image"], static["action"])
       static["labels"].copy_(labels0.reshape(-1, 1))
 
+      grad_params = [prm for prm in network.parameters()
+                     if prm.requires_grad]
+      flat = None
+      if distributed:
+        # One flat f32 comm buffer; graph1 gathers grads into it, the
+        # all-reduce runs EAGER between the two graph replays (the one
+        # RCCL call per step), graph2 scatters + optimizer + EMA.
+        total = sum(prm.numel() for prm in grad_params)
+        flat = torch.zeros(total, dtype=torch.float32, device=device)
+
       def graph_body():
         # set_to_none inside capture: backward then WRITES fresh
         # graph-pool buffers (stable across replays) instead of
@@ -200,7 +210,13 @@ def main():
         loss = torch.nn.functional.binary_cross_entropy_with_logits(
             logit.float(), static["labels"].reshape(logit.shape))
         loss.backward()
-        if not distributed:
+        if distributed:
+          off = 0
+          for prm in grad_params:
+            n = prm.numel()
+            flat[off:off + n].copy_(prm.grad.view(-1))
+            off += n
+        else:
           optimizer.step(0)
           if ema is not None:
             ema.update()
@@ -209,28 +225,29 @@ def main():
       graphed = graph_step.GraphedTrainStep(graph_body)
 
       if distributed:
-        grad_params = [prm for prm in network.parameters()
-                       if prm.requires_grad]
+        # p.grad tensors are stable graph-pool buffers after capture;
+        # graph2 references them directly.
+        def opt_body():
+          flat.div_(float(world_size))
+          off = 0
+          for prm in grad_params:
+            n = prm.numel()
+            prm.grad.view(-1).copy_(flat[off:off + n])
+            off += n
+          optimizer.step(0)
+          if ema is not None:
+            ema.update()
+          return None
 
-        def sync_grads():
-          grads = [prm.grad for prm in grad_params
-                   if prm.grad is not None]
-          flat = torch._utils._flatten_dense_tensors(grads)
-          torch.distributed.all_reduce(flat)
-          flat.div_(world_size)
-          for g, g_synced in zip(
-              grads, torch._utils._unflatten_dense_tensors(flat, grads)):
-            g.copy_(g_synced)
+        opt_graphed = graph_step.GraphedTrainStep(opt_body)
 
         def step(i, global_step):  # noqa: F811 (graphed fwd+bwd)
           images, action, labels_t = pool[i % len(pool)]
           preprocess_to(images, action, static["image"], static["action"])
           static["labels"].copy_(labels_t.reshape(-1, 1))
           loss = graphed.replay()
-          sync_grads()
-          optimizer.step(global_step)
-          if ema is not None:
-            ema.update()
+          torch.distributed.all_reduce(flat)
+          opt_graphed.replay()
           return loss
       else:
 

@@ -192,10 +192,10 @@ conv_s1_nhwc_big_kernel(const cbf16_t* __restrict__ x,
   const int oh0 = th * BTILE, ow0 = tw * BTILE;
 
   {
-    const int halo_h = BTILE + R - 1, halo_w = BTILE + S - 1;
+    const int halo_h = TILE_HH + R - 1, halo_w = BTILE + S - 1;
     constexpr int chunks = C >> 3;
     const int total = halo_h * halo_w * chunks;
-    for (int i = tid; i < total; i += 512) {
+    for (int i = tid; i < total; i += NTHREADS) {
       const int chunk = i % chunks;
       const int pix = i / chunks;
       const int hrow = pix / halo_w, hcol = pix % halo_w;
@@ -362,7 +362,22 @@ at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
       const char* v = std::getenv("T2R_RING_DEPTH");
       return v != nullptr && v[0] == '6';
     }();
-    if (C == 64 && K == 64)
+    // T2R_RING_WAVES=4: 256-thread WGs on 8x16 tiles (2 WGs/CU).
+    static const bool small_waves = []() {
+      const char* v = std::getenv("T2R_RING_WAVES");
+      return v != nullptr && v[0] == '4';
+    }();
+    if (C == 64 && K == 64 && use_ring && small_waves) {
+      const int th8 = (OH + 7) / 8;
+      const long grid8 = (long)N * th8 * tiles_w;
+      hipLaunchKernelGGL((conv_s1_nhwc_ring_kernel<4, 2, 3, 4>),
+                         dim3(grid8), dim3(256), 0, stream.stream(),
+                         (const cbf16_t*)x.data_ptr(),
+                         (const cbf16_t*)wpk.data_ptr(),
+                         (cbf16_t*)y.data_ptr(),
+                         N, H, W, (int)K, (int)R, (int)S, (int)pad,
+                         OH, OW, th8, tiles_w);
+    } else if (C == 64 && K == 64)
       use_ring ? (deep_ring ? launch(conv_s1_nhwc_ring_kernel<4, 2, 6>)
                             : launch(conv_s1_nhwc_ring_kernel<4, 2, 3>))
                : launch(conv_s1_nhwc_big_kernel<4, 2>);
@@ -511,22 +526,29 @@ std::vector<at::Tensor> pack_conv_w_pair(at::Tensor w) {
 // no register round-trip, no write pass, loads 2 chunks ahead.
 // ---------------------------------------------------------------------------
 
-__device__ __forceinline__ void conv_waitcnt_vm(int ahead) {
+__device__ __forceinline__ void conv_waitcnt_vm(int count) {
   // s_waitcnt imm: vmcnt[3:0], expcnt[6:4]=7, lgkmcnt[11:8]=15.
-  // Each outstanding chunk is 2 glds pieces per issuer wave.
-  switch (ahead) {
+  // `count` = allowed outstanding glds pieces for this wave.
+  switch (count) {
     case 0: __builtin_amdgcn_s_waitcnt(0 | (7 << 4) | (15 << 8)); break;
-    case 1: __builtin_amdgcn_s_waitcnt(2 | (7 << 4) | (15 << 8)); break;
-    case 2: __builtin_amdgcn_s_waitcnt(4 | (7 << 4) | (15 << 8)); break;
-    case 3: __builtin_amdgcn_s_waitcnt(6 | (7 << 4) | (15 << 8)); break;
-    case 4: __builtin_amdgcn_s_waitcnt(8 | (7 << 4) | (15 << 8)); break;
+    case 1: __builtin_amdgcn_s_waitcnt(1 | (7 << 4) | (15 << 8)); break;
+    case 2: __builtin_amdgcn_s_waitcnt(2 | (7 << 4) | (15 << 8)); break;
+    case 3: __builtin_amdgcn_s_waitcnt(3 | (7 << 4) | (15 << 8)); break;
+    case 4: __builtin_amdgcn_s_waitcnt(4 | (7 << 4) | (15 << 8)); break;
+    case 5: __builtin_amdgcn_s_waitcnt(5 | (7 << 4) | (15 << 8)); break;
+    case 6: __builtin_amdgcn_s_waitcnt(6 | (7 << 4) | (15 << 8)); break;
+    case 8: __builtin_amdgcn_s_waitcnt(8 | (7 << 4) | (15 << 8)); break;
     default:
       __builtin_amdgcn_s_waitcnt(10 | (7 << 4) | (15 << 8)); break;
   }
 }
 
-template <int C16N, int NTILES, int RING_DEPTH>
-__global__ void __launch_bounds__(512, 2)
+// WAVES=8: one 512-thread WG owns a 16x16 tile (1 WG/CU).  WAVES=4:
+// a 256-thread WG owns an 8x16 tile -> 2 independent WGs/CU with
+// uncoupled barriers, at the cost of streaming each weight chunk
+// twice per 256 output pixels.
+template <int C16N, int NTILES, int RING_DEPTH, int WAVES = 8>
+__global__ void __launch_bounds__(WAVES * 64, 2)
 conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
                          const cbf16_t* __restrict__ wpk,
                          cbf16_t* __restrict__ y,
@@ -536,14 +558,13 @@ conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
   constexpr int C = C16N * 16;
   constexpr int WBUF = C16N * NTILES * 32 * WPAD;      // bf16 per chunk
   constexpr int PIECES = (WBUF * 2) / 1024;            // 1-KiB DMA pieces
-  // 6-deep ring (131 KiB LDS with the x halo): the MFMA phase per rs
-  // is only ~250 cycles at the 1-WG/CU occupancy this LDS footprint
-  // forces, so a 2-chunk lead cannot hide L2/HBM latency — 5 chunks
-  // in flight can.
-  __shared__ short lds[BHALO * BHALO * XPITCH + RING_DEPTH * WBUF];
+  constexpr int NTHREADS = WAVES * 64;
+  constexpr int TILE_HH = WAVES * 2;                   // tile = TH x 16
+  constexpr int HALO_HH = TILE_HH + 4;
+  __shared__ short lds[HALO_HH * BHALO * XPITCH + RING_DEPTH * WBUF];
   short* xtile = lds;
   auto wbuf = [&](int slot) -> short* {
-    return lds + BHALO * BHALO * XPITCH + slot * WBUF;
+    return lds + HALO_HH * BHALO * XPITCH + slot * WBUF;
   };
 
   const int tid = threadIdx.x;
@@ -555,18 +576,19 @@ conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
   const int trest = wg % (tiles_h * tiles_w);
   const int th = trest / tiles_w;
   const int tw = trest % tiles_w;
-  const int oh0 = th * BTILE, ow0 = tw * BTILE;
+  const int oh0 = th * TILE_HH, ow0 = tw * BTILE;
 
-  // Wave w issues DMA pieces {w, w+8, ...} of each chunk; with
-  // PIECES=12 waves 0-3 own 2 pieces, 4-7 own 1: NOT uniform.  Use a
-  // 6-wave x 2-piece assignment so the counted vmcnt is uniform among
-  // issuers (waves 6,7 issue nothing and skip the waits).
-  const bool issuer = wave < 6;
+  // Uniform piece assignment so the counted vmcnt is the same for all
+  // issuer waves: WAVES=8 -> 6 issuer waves x 2 pieces (PIECES=12);
+  // WAVES=4 -> 4 issuer waves x PIECES/4.
+  constexpr int ISSUERS = (WAVES == 8) ? 6 : WAVES;
+  constexpr int PPW = (PIECES + ISSUERS - 1) / ISSUERS;
+  const bool issuer = wave < ISSUERS;
   auto issue_chunk = [&](int rs, int slot) {
     if (!issuer) return;
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      const int piece = wave * 2 + j;
+    for (int j = 0; j < PPW; ++j) {
+      const int piece = wave * PPW + j;
       if (piece < PIECES) {
         const cbf16_t* src = wpk + (long)rs * WBUF + piece * 512
                              + lane * 8;
@@ -580,10 +602,10 @@ conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
   };
 
   {
-    const int halo_h = BTILE + R - 1, halo_w = BTILE + S - 1;
+    const int halo_h = TILE_HH + R - 1, halo_w = BTILE + S - 1;
     constexpr int chunks = C >> 3;
     const int total = halo_h * halo_w * chunks;
-    for (int i = tid; i < total; i += 512) {
+    for (int i = tid; i < total; i += NTHREADS) {
       const int chunk = i % chunks;
       const int pix = i / chunks;
       const int hrow = pix / halo_w, hcol = pix % halo_w;
@@ -618,7 +640,8 @@ conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
     const int slot = rs % RING_DEPTH;
     if (rs + RING_DEPTH - 1 < RS)
       issue_chunk(rs + RING_DEPTH - 1, (rs + RING_DEPTH - 1) % RING_DEPTH);
-    if (issuer) conv_waitcnt_vm(min(RS - 1 - rs, RING_DEPTH - 1));
+    if (issuer)
+      conv_waitcnt_vm(min(RS - 1 - rs, RING_DEPTH - 1) * PPW);
     __builtin_amdgcn_s_barrier();     // chunk rs landed for everyone
 #pragma unroll
     for (int c16 = 0; c16 < C16N; ++c16) {
