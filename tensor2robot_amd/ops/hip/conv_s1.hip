@@ -192,10 +192,10 @@ conv_s1_nhwc_big_kernel(const cbf16_t* __restrict__ x,
   const int oh0 = th * BTILE, ow0 = tw * BTILE;
 
   {
-    const int halo_h = TILE_HH + R - 1, halo_w = BTILE + S - 1;
+    const int halo_h = BTILE + R - 1, halo_w = BTILE + S - 1;
     constexpr int chunks = C >> 3;
     const int total = halo_h * halo_w * chunks;
-    for (int i = tid; i < total; i += NTHREADS) {
+    for (int i = tid; i < total; i += 512) {
       const int chunk = i % chunks;
       const int pix = i / chunks;
       const int hrow = pix / halo_w, hcol = pix % halo_w;
@@ -295,7 +295,7 @@ conv_s1_nhwc_big_kernel(const cbf16_t* __restrict__ x,
   }
 }
 
-template <int C16N, int NTILES, int RING_DEPTH>
+template <int C16N, int NTILES, int RING_DEPTH, int WAVES = 8>
 __global__ void conv_s1_nhwc_ring_kernel(
     const cbf16_t* __restrict__ x, const cbf16_t* __restrict__ wpk,
     cbf16_t* __restrict__ y, int N, int H, int W, int K, int R, int S,
@@ -547,7 +547,7 @@ __device__ __forceinline__ void conv_waitcnt_vm(int count) {
 // a 256-thread WG owns an 8x16 tile -> 2 independent WGs/CU with
 // uncoupled barriers, at the cost of streaming each weight chunk
 // twice per 256 output pixels.
-template <int C16N, int NTILES, int RING_DEPTH, int WAVES = 8>
+template <int C16N, int NTILES, int RING_DEPTH, int WAVES>
 __global__ void __launch_bounds__(WAVES * 64, 2)
 conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
                          const cbf16_t* __restrict__ wpk,
