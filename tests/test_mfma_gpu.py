@@ -214,3 +214,13 @@ def test_mfma_wrw2_matches_torch(shape):
   err = (dw_t - ref).abs().max().item()
   scale = ref.abs().max().item()
   assert err < 0.01 * max(scale, 1.0), (shape, err, scale)
+
+
+@requires_gpu
+def test_pack_conv_w_pair_matches_single():
+  from tensor2robot_amd.ops import _t2r_hip
+  for shape in [(64, 64, 3, 3), (64, 64, 5, 5), (32, 64, 3, 3)]:
+    w = torch.randn(*shape, device="cuda").to(torch.bfloat16)
+    outf, outb = _t2r_hip.pack_conv_w_pair(w)
+    assert torch.equal(outf, _t2r_hip.pack_conv_w(w, False))
+    assert torch.equal(outb, _t2r_hip.pack_conv_w(w, True))
