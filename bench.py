@@ -214,7 +214,10 @@ def main():
           off = 0
           for prm in grad_params:
             n = prm.numel()
-            flat[off:off + n].copy_(prm.grad.view(-1))
+            # conv grads are channels_last: view(-1) is illegal, but a
+            # shaped view of the contiguous flat slice can be copy_
+            # target/source with layout conversion in one kernel.
+            flat[off:off + n].view_as(prm.grad).copy_(prm.grad)
             off += n
         else:
           optimizer.step(0)
@@ -232,7 +235,7 @@ def main():
           off = 0
           for prm in grad_params:
             n = prm.numel()
-            prm.grad.view(-1).copy_(flat[off:off + n])
+            prm.grad.copy_(flat[off:off + n].view_as(prm.grad))
             off += n
           optimizer.step(0)
           if ema is not None:
