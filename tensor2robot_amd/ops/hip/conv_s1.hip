@@ -362,10 +362,13 @@ at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
       const char* v = std::getenv("T2R_RING_DEPTH");
       return v != nullptr && v[0] == '6';
     }();
-    // T2R_RING_WAVES=4: 256-thread WGs on 8x16 tiles (2 WGs/CU).
+    // 4-wave default (measured 490-497 vs 478 TF on the 5x5 @78^2):
+    // 256-thread WGs on 8x16 tiles -> 2 independent WGs/CU whose
+    // barriers don't couple, worth more than the 2x weight streaming.
+    // T2R_RING_WAVES=8 selects the single 512-thread-WG variant.
     static const bool small_waves = []() {
       const char* v = std::getenv("T2R_RING_WAVES");
-      return v != nullptr && v[0] == '4';
+      return v == nullptr || v[0] != '8';
     }();
     if (C == 64 && K == 64 && use_ring && small_waves) {
       const int th8 = (OH + 7) / 8;
