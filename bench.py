@@ -194,49 +194,57 @@ def main():
                      if prm.requires_grad]
       flat = None
       if distributed:
-        # One flat f32 comm buffer; graph1 gathers grads into it, the
+        # One flat f32 comm buffer.  Every p.grad is pre-assigned as a
+        # strided VIEW into it (channels_last strides for 4D params),
+        # so backward ACCUMULATES straight into the comm buffer: no
+        # gather/scatter kernels at all.  graph1 = zero+fwd+bwd, the
         # all-reduce runs EAGER between the two graph replays (the one
-        # RCCL call per step), graph2 scatters + optimizer + EMA.
+        # RCCL call per step), graph2 = div + optimizer + EMA.
         total = sum(prm.numel() for prm in grad_params)
         flat = torch.zeros(total, dtype=torch.float32, device=device)
 
+        def assign_grad_views():
+          off = 0
+          for prm in grad_params:
+            n = prm.numel()
+            sl = flat[off:off + n]
+            if prm.dim() == 4 and prm.is_contiguous(
+                memory_format=torch.channels_last):
+              no, c, h, w = prm.shape
+              g = sl.view(no, h, w, c).permute(0, 3, 1, 2)
+            else:
+              g = sl.view(prm.shape)
+            prm.grad = g
+            off += n
+
       def graph_body():
-        # set_to_none inside capture: backward then WRITES fresh
-        # graph-pool buffers (stable across replays) instead of
-        # zero-fill + accumulate-add per param (~165 kernels/step).
-        optimizer.zero_grad(set_to_none=True)
+        if distributed:
+          # grads are views of `flat`: one fill clears them all, and
+          # backward accumulates into the comm buffer directly.
+          flat.zero_()
+        else:
+          # set_to_none inside capture: backward then WRITES fresh
+          # graph-pool buffers (stable across replays) instead of
+          # zero-fill + accumulate-add per param (~165 kernels/step).
+          optimizer.zero_grad(set_to_none=True)
         with autocast:
           logit = network(static["image"], static["action"])
         loss = torch.nn.functional.binary_cross_entropy_with_logits(
             logit.float(), static["labels"].reshape(logit.shape))
         loss.backward()
-        if distributed:
-          off = 0
-          for prm in grad_params:
-            n = prm.numel()
-            # conv grads are channels_last: view(-1) is illegal, but a
-            # shaped view of the contiguous flat slice can be copy_
-            # target/source with layout conversion in one kernel.
-            flat[off:off + n].view_as(prm.grad).copy_(prm.grad)
-            off += n
-        else:
+        if not distributed:
           optimizer.step(0)
           if ema is not None:
             ema.update()
         return loss
 
+      if distributed:
+        assign_grad_views()
       graphed = graph_step.GraphedTrainStep(graph_body)
 
       if distributed:
-        # p.grad tensors are stable graph-pool buffers after capture;
-        # graph2 references them directly.
         def opt_body():
           flat.div_(float(world_size))
-          off = 0
-          for prm in grad_params:
-            n = prm.numel()
-            prm.grad.copy_(flat[off:off + n].view_as(prm.grad))
-            off += n
           optimizer.step(0)
           if ema is not None:
             ema.update()
