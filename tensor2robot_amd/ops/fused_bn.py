@@ -73,6 +73,11 @@ class FusedBatchNormReLU(nn.Module):
     self.register_buffer("running_var", torch.ones(num_features))
     self.register_buffer("num_batches_tracked",
                          torch.tensor(0, dtype=torch.long))
+    # Python-side mirror: the per-forward GPU increment was a captured
+    # kernel per BN layer per step (~2% of the train step); the count
+    # is only bookkeeping (momentum is explicit), so it lives on the
+    # host and is synced into the buffer when a state_dict is taken.
+    self._batches_tracked_py = 0
 
   def _use_hip(self, x: torch.Tensor) -> bool:
     return (x.is_cuda and x.dtype == torch.bfloat16 and
@@ -86,7 +91,7 @@ class FusedBatchNormReLU(nn.Module):
             flat, self.weight, self.bias,
             self.running_mean, self.running_var, self.eps, self.momentum,
             self.fuse_relu)
-        self.num_batches_tracked += 1
+        self._batches_tracked_py += 1
       else:
         invstd = torch.rsqrt(self.running_var + self.eps)
         scale = (self.weight * invstd).float()
@@ -107,6 +112,13 @@ class FusedBatchNormReLU(nn.Module):
     if self.fuse_relu:
       y = torch.relu(y)
     return y
+
+  def _save_to_state_dict(self, destination, prefix, keep_vars):
+    if self._batches_tracked_py:
+      self.num_batches_tracked.fill_(
+          int(self.num_batches_tracked) + self._batches_tracked_py)
+      self._batches_tracked_py = 0
+    super()._save_to_state_dict(destination, prefix, keep_vars)
 
   def extra_repr(self):
     return (f"{self.num_features}, eps={self.eps}, "
