@@ -171,6 +171,7 @@ def main():
   # pool - no multi-GPU box to test on), yet the ~250-kernel fwd+bwd
   # still replays as one launch, so the 8-GPU scaling numbers are not
   # stuck at eager launch overhead.
+  eager_step = step
   if use_cuda and not args.no_hipgraph:
     try:
       from tensor2robot_amd.parallel import graph_step
@@ -273,9 +274,26 @@ def main():
         print(f"# hipGraph capture unavailable, running eager: {e!r}",
               file=_sys.stderr, flush=True)
       graphed = None
-      if distributed and dp_engine is None:
-        from tensor2robot_amd.parallel import ddp
-        dp_engine = ddp.DataParallelEngine(network)
+
+  if distributed and torch.distributed.is_initialized():
+    # The capture-or-eager decision must be COLLECTIVE: one rank
+    # replaying graphs (one all-reduce of `flat` per step) while
+    # another runs the hook-bucketed eager engine would mismatch
+    # collectives and hang the job.  Demote everyone if anyone failed.
+    ok = torch.tensor(
+        [0.0 if graphed is None and use_cuda and not args.no_hipgraph
+         else 1.0],
+        device=device if use_cuda else "cpu")
+    torch.distributed.all_reduce(ok, op=torch.distributed.ReduceOp.MIN)
+    if float(ok.item()) < 1.0 and graphed is not None:
+      graphed = None
+      step = eager_step
+  if distributed and graphed is None and dp_engine is None:
+    from tensor2robot_amd.parallel import ddp
+    dp_engine = ddp.DataParallelEngine(network)
+    # Eager fallback must not leave grads aliased into the comm buffer.
+    for prm in network.parameters():
+      prm.grad = None
 
   def barrier_sync():
     if distributed:
