@@ -21,6 +21,7 @@ hip_sources = [
     os.path.join(HIP_DIR, "fused_bn_relu.hip"),
     os.path.join(HIP_DIR, "preprocess.hip"),
     os.path.join(HIP_DIR, "maxpool.hip"),
+    os.path.join(HIP_DIR, "spatial_softmax.hip"),
     os.path.join(HIP_DIR, "mfma_probe.hip"),
     os.path.join(HIP_DIR, "conv_s1.hip"),
     os.path.join(HIP_DIR, "conv_wrw.hip"),

@@ -5,11 +5,13 @@ Reference `layers/spatial_softmax.py:29-89` BuildSpatialSoftmax: reshape
 against precomputed x/y position grids in [-1, 1], output [N, 2C] feature
 points plus the softmax attention map.
 
-Torch-native NCHW.  The softmax + two weighted reductions over H*W is a
-single fused pass on GPU bandwidth terms; at the feature-map sizes the
-robot nets use it is a minor cost next to the conv tower, so the torch
-composition (one softmax + one matmul against the [HW, 2] grid) is kept —
-it already fuses the x/y expectations into one GEMM.
+On GPU bf16 channels_last inputs this dispatches to the fused CDNA4
+kernel (ops/hip/spatial_softmax.hip): one online-softmax pass per
+(image, channel) accumulating max/sum/x/y-expectation in registers plus
+one map-write pass, and a single fused backward — replacing the torch
+reshape + softmax + [HW,2] matmul chain (5-6 kernels with f32 casts).
+The torch composition below remains the CPU/reference path and the
+Gumbel-sampling path.
 """
 
 from __future__ import annotations
@@ -20,6 +22,35 @@ import torch
 from torch import nn
 
 from tensor2robot_amd import gin
+from tensor2robot_amd import ops as ops_mod
+
+
+class _FusedSpatialSoftmax(torch.autograd.Function):
+  """HIP fused soft arg-max; see ops/hip/spatial_softmax.hip."""
+
+  @staticmethod
+  def forward(ctx, x, temperature):
+    ext = ops_mod.require_hip()
+    if not x.is_contiguous(memory_format=torch.channels_last):
+      x = x.contiguous(memory_format=torch.channels_last)
+    points, smap = ext.spatial_softmax_fwd(x, float(temperature))
+    ctx.save_for_backward(smap, points)
+    ctx.temperature = float(temperature)
+    return points, smap
+
+  @staticmethod
+  def backward(ctx, dpoints, dmap):
+    ext = ops_mod.require_hip()
+    smap, points = ctx.saved_tensors
+    dp = None
+    if dpoints is not None:
+      dp = dpoints.to(torch.bfloat16).contiguous()
+    dm = None
+    if dmap is not None:
+      dm = dmap.to(torch.bfloat16).contiguous(
+          memory_format=torch.channels_last)
+    dx = ext.spatial_softmax_bwd(smap, points, dp, dm, ctx.temperature)
+    return dx, None
 
 
 def _position_grid(height: int, width: int, device, dtype) -> torch.Tensor:
@@ -50,8 +81,18 @@ class SpatialSoftmax(nn.Module):
       self._grid_cache = (key, _position_grid(h, w, device, dtype))
     return self._grid_cache[1]
 
+  def _fused_supported(self, x: torch.Tensor) -> bool:
+    import os
+    if os.environ.get("T2R_DISABLE_FUSED_SPATIAL_SOFTMAX"):
+      return False
+    return (x.is_cuda and x.dtype == torch.bfloat16 and x.dim() == 4 and
+            x.shape[1] <= 256 and
+            not (self.use_gumbel and self.training))
+
   def forward(self, features: torch.Tensor
               ) -> Tuple[torch.Tensor, torch.Tensor]:
+    if self._fused_supported(features):
+      return _FusedSpatialSoftmax.apply(features, self.temperature)
     n, c, h, w = features.shape
     logits = features.reshape(n * c, h * w).float() / self.temperature
     if self.use_gumbel and self.training:

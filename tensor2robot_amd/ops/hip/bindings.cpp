@@ -22,6 +22,11 @@ at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
 
 at::Tensor pack_conv_w(at::Tensor w, bool transpose);
 std::vector<at::Tensor> pack_conv_w_pair(at::Tensor w);
+std::vector<at::Tensor> spatial_softmax_fwd(at::Tensor x, double temp);
+at::Tensor spatial_softmax_bwd(at::Tensor map, at::Tensor points,
+                               c10::optional<at::Tensor> dpoints,
+                               c10::optional<at::Tensor> dmap,
+                               double temp);
 at::Tensor s2d_stem(at::Tensor x);
 
 at::Tensor conv_s1_wrw(at::Tensor x, at::Tensor dy, int64_t R, int64_t S,
@@ -69,6 +74,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_stem_nhwc", &conv_stem_nhwc,
         "MFMA 6x6/2 C=3 stem conv (implicit im2col)");
   m.def("pack_stem_w", &pack_stem_w, "stem weight pack [12][64][24]");
+  m.def("spatial_softmax_fwd", &spatial_softmax_fwd,
+        "fused online-softmax soft arg-max: points [N,2C] + map");
+  m.def("spatial_softmax_bwd", &spatial_softmax_bwd,
+        "fused spatial softmax backward");
   m.def("maxpool_nhwc_forward", &maxpool_nhwc_forward,
         "Non-overlapping NHWC bf16 max-pool forward (+argmax)");
   m.def("maxpool_nhwc_backward", &maxpool_nhwc_backward,

@@ -224,3 +224,45 @@ def test_pack_conv_w_pair_matches_single():
     outf, outb = _t2r_hip.pack_conv_w_pair(w)
     assert torch.equal(outf, _t2r_hip.pack_conv_w(w, False))
     assert torch.equal(outb, _t2r_hip.pack_conv_w(w, True))
+
+
+@requires_gpu
+def test_fused_spatial_softmax_matches_torch():
+  import os
+  from tensor2robot_amd.layers import spatial_softmax as ss
+  torch.manual_seed(3)
+  x = torch.randn(4, 64, 13, 17, device="cuda") * 3.0
+  xb = x.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+  mod = ss.SpatialSoftmax(temperature=0.7)
+
+  # fp32 torch reference (fused path disabled).
+  os.environ["T2R_DISABLE_FUSED_SPATIAL_SOFTMAX"] = "1"
+  try:
+    xr = x.clone().requires_grad_(True)
+    pts_ref, map_ref = mod(xr)
+    (pts_ref.sum() + (map_ref * 0.01).sum()).backward()
+  finally:
+    del os.environ["T2R_DISABLE_FUSED_SPATIAL_SOFTMAX"]
+
+  xf = xb.clone().requires_grad_(True)
+  pts, smap = mod(xf)
+  assert pts.shape == (4, 128) and smap.shape == (4, 64, 13, 17)
+  assert (pts.float() - pts_ref).abs().max().item() < 2e-2
+  assert (smap.float() - map_ref).abs().max().item() < 2e-2
+  (pts.sum() + (smap.float() * 0.01).sum()).backward()
+  scale = xr.grad.abs().max().item()
+  assert (xf.grad.float() - xr.grad).abs().max().item() < 0.05 * scale
+
+  # points-only gradient path (dmap is None).
+  xg = xb.clone().requires_grad_(True)
+  pts2, _ = mod(xg)
+  pts2.sum().backward()
+  xr2 = x.clone().requires_grad_(True)
+  os.environ["T2R_DISABLE_FUSED_SPATIAL_SOFTMAX"] = "1"
+  try:
+    p_ref2, _ = mod(xr2)
+    p_ref2.sum().backward()
+  finally:
+    del os.environ["T2R_DISABLE_FUSED_SPATIAL_SOFTMAX"]
+  scale2 = xr2.grad.abs().max().item()
+  assert (xg.grad.float() - xr2.grad).abs().max().item() < 0.05 * scale2
