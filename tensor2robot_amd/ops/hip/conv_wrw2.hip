@@ -201,6 +201,205 @@ wrw2_reduce_kernel(const float* __restrict__ part,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// v3: occupancy-first split.  The v2 monolith (8 waves, 208 accum
+// VGPRs, 2 waves/SIMD) is latency-bound; v3 follows the 4-wave ring
+// lesson: 256-thread WGs where each wave is one (c,k) quadrant and
+// blockIdx.y picks an rs GROUP of <=5 offsets -> 80 accumulator VGPRs
+// and 3 WGs/CU (LDS-bound), so 3 waves/SIMD hide the MFMA + LDS
+// latency.  Price: each window is staged once PER rs group (up to 5x
+// global re-read, ~47 us chip-wide for the 5x5 @78^2 layer - cheap
+// next to the latency win).  Flush: non-atomic per-WG partial slices.
+// ---------------------------------------------------------------------------
+
+#define W3_GROUP 5
+
+template <int R, int S>
+__global__ void __launch_bounds__(256, 2)
+conv_s1_wrw3_kernel(const w2bf16_t* __restrict__ x,
+                    const w2bf16_t* __restrict__ dy,
+                    float* __restrict__ dw_part,  // [gx][gy][G][C][K]
+                    int N, int H, int W, int pad,
+                    int OH, int OW, int tiles_h, int tiles_w,
+                    int window_groups) {
+  constexpr int C = 64, K = 64;
+  constexpr int RS = R * S;
+  constexpr int TH = 8;
+  constexpr int WIN_P = TH * W2TILE_W;         // 128
+  constexpr int HALO_H = TH + 4;
+  constexpr int HALO_P = HALO_H * W2HALO_W;    // 240
+  constexpr int XT_P = HALO_P + 8;
+  constexpr int DYT_P = WIN_P + 8;
+  __shared__ short lds_xt[C * XT_P];
+  __shared__ short lds_dyt[K * DYT_P];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int mtile = wave & 1;
+  const int ntile = wave >> 1;
+  const int rs_base = blockIdx.y * W3_GROUP;
+  const int rs_count = min(RS - rs_base, W3_GROUP);
+
+  w2f32x16 acc[W3_GROUP];
+#pragma unroll
+  for (int g = 0; g < W3_GROUP; ++g) acc[g] = w2f32x16{};
+
+  const int total_windows = (int)((long)N * tiles_h * tiles_w);
+  const int mrow = lane & 31;
+  const int kgrp = lane >> 5;
+
+  for (int win = blockIdx.x; win < total_windows; win += window_groups) {
+    const int img = win / (tiles_h * tiles_w);
+    const int trest = win % (tiles_h * tiles_w);
+    const int oh0 = (trest / tiles_w) * TH;
+    const int ow0 = (trest % tiles_w) * W2TILE_W;
+
+    {
+      constexpr int halo_h = TH + R - 1, halo_w = W2TILE_W + S - 1;
+      constexpr int chunks = C >> 3;
+      for (int i = tid; i < HALO_P * chunks; i += 256) {
+        const int chunk = i % chunks;
+        const int p = i / chunks;
+        const int hrow = p / W2HALO_W, hcol = p % W2HALO_W;
+        const int iy = oh0 - pad + hrow;
+        const int ix = ow0 - pad + hcol;
+        w2bf16x8 v = {};
+        if (hrow < halo_h && hcol < halo_w &&
+            iy >= 0 && iy < H && ix >= 0 && ix < W) {
+          v = *reinterpret_cast<const w2bf16x8*>(
+              x + (((long)img * H + iy) * W + ix) * C + chunk * 8);
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          lds_xt[(chunk * 8 + j) * XT_P + p] = v[j];
+      }
+    }
+    {
+      constexpr int kchunks = K >> 3;
+      for (int i = tid; i < WIN_P * kchunks; i += 256) {
+        const int chunk = i % kchunks;
+        const int p = i / kchunks;
+        const int orow = oh0 + p / W2TILE_W;
+        const int ocol = ow0 + p % W2TILE_W;
+        w2bf16x8 v = {};
+        if (orow < OH && ocol < OW) {
+          v = *reinterpret_cast<const w2bf16x8*>(
+              dy + (((long)img * OH + orow) * OW + ocol) * K + chunk * 8);
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          lds_dyt[(chunk * 8 + j) * DYT_P + p] = v[j];
+      }
+    }
+    __syncthreads();
+
+    {
+      const int c = mtile * 32 + mrow;
+      const int k = ntile * 32 + mrow;
+#pragma unroll
+      for (int kstep = 0; kstep < WIN_P / 16; ++kstep) {
+        w2bf16x8 b_frag = *reinterpret_cast<const w2bf16x8*>(
+            &lds_dyt[k * DYT_P + kstep * 16 + kgrp * 8]);
+#pragma unroll
+        for (int g = 0; g < W3_GROUP; ++g) {
+          const int rs = rs_base + g;
+          if (rs < RS) {
+            const int r = rs / S, s = rs % S;
+            const int halo_base = (kstep + r) * W2HALO_W + s;
+            w2bf16x8 a_frag = *reinterpret_cast<const w2bf16x8*>(
+                &lds_xt[c * XT_P + halo_base + kgrp * 8]);
+            acc[g] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                a_frag, b_frag, acc[g], 0, 0, 0);
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  float* part = dw_part
+      + ((long)blockIdx.x * gridDim.y + blockIdx.y) * W3_GROUP * C * K;
+#pragma unroll
+  for (int g = 0; g < W3_GROUP; ++g) {
+    if (rs_base + g < RS) {
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int c = mtile * 32
+            + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+        const int k = ntile * 32 + (lane & 31);
+        part[((long)g * C + c) * K + k] = acc[g][reg];
+      }
+    }
+  }
+}
+
+// Reduce: dw[rs][c][k] = sum over gx of part[gx][rs/G][rs%G][c][k].
+__global__ void __launch_bounds__(256)
+wrw3_reduce_kernel(const float* __restrict__ part,
+                   float* __restrict__ dw, int RS, long ck,
+                   int ngx, int ngy) {
+  const long cells = (long)RS * ck;
+  for (long i = blockIdx.x * 256L + threadIdx.x; i < cells;
+       i += (long)gridDim.x * 256) {
+    const int rs = (int)(i / ck);
+    const long rest = i % ck;
+    const int gy = rs / W3_GROUP, g = rs % W3_GROUP;
+    float s = 0.0f;
+    for (int gx = 0; gx < ngx; ++gx) {
+      s += part[(((long)gx * ngy + gy) * W3_GROUP + g) * ck + rest];
+    }
+    dw[i] = s;
+  }
+}
+
+at::Tensor conv_s1_wrw3(at::Tensor x, at::Tensor dy, int64_t R,
+                        int64_t S, int64_t pad) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16,
+              "conv_s1_wrw3: bf16 x required");
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == at::kBFloat16,
+              "conv_s1_wrw3: bf16 dy required");
+  x = x.contiguous(at::MemoryFormat::ChannelsLast);
+  dy = dy.contiguous(at::MemoryFormat::ChannelsLast);
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = dy.size(1), OH = dy.size(2), OW = dy.size(3);
+  TORCH_CHECK(C == 64 && K == 64, "conv_s1_wrw3: C == K == 64 only");
+  TORCH_CHECK(OH == H + 2 * pad - R + 1 && OW == W + 2 * pad - S + 1,
+              "conv_s1_wrw3: dy shape mismatch");
+  const int tiles_h = (OH + 7) / 8;
+  const int tiles_w = (OW + W2TILE_W - 1) / W2TILE_W;
+  const int total_windows = (int)((long)N * tiles_h * tiles_w);
+  const int ngy = (int)((R * S + W3_GROUP - 1) / W3_GROUP);
+  // 3 WGs/CU resident; split windows so the whole grid fills the chip
+  // ~3x over without inflating the reduce.
+  const int window_groups = std::min(total_windows, 768 / ngy * 1);
+  auto part = at::empty({(long)window_groups * ngy, W3_GROUP,
+                         (long)C, K}, x.options().dtype(at::kFloat));
+  auto dw = at::empty({(long)R * S, C, K},
+                      x.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(window_groups, ngy), dim3(256),
+                       0, stream.stream(),
+                       (const w2bf16_t*)x.data_ptr(),
+                       (const w2bf16_t*)dy.data_ptr(),
+                       (float*)part.data_ptr(),
+                       N, H, W, (int)pad, OH, OW, tiles_h, tiles_w,
+                       window_groups);
+  };
+  if (R == 3 && S == 3) launch(conv_s1_wrw3_kernel<3, 3>);
+  else if (R == 5 && S == 5) launch(conv_s1_wrw3_kernel<5, 5>);
+  else TORCH_CHECK(false, "conv_s1_wrw3: R/S must be 3x3 or 5x5");
+  const long cells = (long)R * S * C * K;
+  const int rblocks = (int)std::min((cells + 255) / 256, 1024L);
+  hipLaunchKernelGGL(wrw3_reduce_kernel, dim3(rblocks), dim3(256), 0,
+                     stream.stream(), (const float*)part.data_ptr(),
+                     (float*)dw.data_ptr(), (int)(R * S), (long)C * K,
+                     window_groups, ngy);
+  return dw;
+}
+
 at::Tensor conv_s1_wrw2(at::Tensor x, at::Tensor dy, int64_t R,
                         int64_t S, int64_t pad) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16,

@@ -266,3 +266,30 @@ def test_fused_spatial_softmax_matches_torch():
     del os.environ["T2R_DISABLE_FUSED_SPATIAL_SOFTMAX"]
   scale2 = xr2.grad.abs().max().item()
   assert (xg.grad.float() - xr2.grad).abs().max().item() < 0.05 * scale2
+
+
+@requires_gpu
+@pytest.mark.parametrize("shape", [
+    (2, 64, 79, 79, 64, 5, 2),
+    (2, 64, 27, 27, 64, 3, 1),
+    (3, 64, 78, 78, 64, 5, 2),
+])
+def test_mfma_wrw3_matches_torch(shape):
+  import torch.nn.functional as F
+  from tensor2robot_amd.ops import _t2r_hip
+  n, c, h, w, k, r, pad = shape
+  torch.manual_seed(0)
+  x = torch.randn(n, c, h, w, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last)
+  oh, ow = h + 2 * pad - r + 1, w + 2 * pad - r + 1
+  dy = torch.randn(n, k, oh, ow, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last)
+  dw = _t2r_hip.conv_s1_wrw3(x, dy, r, r, pad)
+  dw_t = dw.reshape(r, r, c, k).permute(3, 2, 0, 1).contiguous()
+  x32 = x.float()
+  w32 = torch.zeros(k, c, r, r, device="cuda", requires_grad=True)
+  F.conv2d(x32, w32, padding=pad).backward(dy.float())
+  ref = w32.grad
+  err = (dw_t - ref).abs().max().item()
+  scale = ref.abs().max().item()
+  assert err < 0.01 * max(scale, 1.0), (shape, err, scale)
