@@ -256,41 +256,70 @@ conv_s1_wrw3_kernel(const w2bf16_t* __restrict__ x,
     const int oh0 = (trest / tiles_w) * TH;
     const int ow0 = (trest % tiles_w) * W2TILE_W;
 
+    // Two-phase staging: ALL guarded global loads are issued first
+    // (independent, overlapped latencies), then the transposed LDS
+    // write pass drains them with one wait — the single-loop
+    // load->8-scattered-writes form serialized a full HBM/L2 latency
+    // per iteration under this kernel's register pressure.
     {
       constexpr int halo_h = TH + R - 1, halo_w = W2TILE_W + S - 1;
       constexpr int chunks = C >> 3;
-      for (int i = tid; i < HALO_P * chunks; i += 256) {
+      constexpr int XITER = (HALO_P * chunks + 255) / 256;
+      w2bf16x8 vx[XITER];
+#pragma unroll
+      for (int t = 0; t < XITER; ++t) {
+        const int i = tid + t * 256;
         const int chunk = i % chunks;
         const int p = i / chunks;
         const int hrow = p / W2HALO_W, hcol = p % W2HALO_W;
         const int iy = oh0 - pad + hrow;
         const int ix = ow0 - pad + hcol;
-        w2bf16x8 v = {};
-        if (hrow < halo_h && hcol < halo_w &&
+        vx[t] = w2bf16x8{};
+        if (i < HALO_P * chunks && hrow < halo_h && hcol < halo_w &&
             iy >= 0 && iy < H && ix >= 0 && ix < W) {
-          v = *reinterpret_cast<const w2bf16x8*>(
+          vx[t] = *reinterpret_cast<const w2bf16x8*>(
               x + (((long)img * H + iy) * W + ix) * C + chunk * 8);
         }
+      }
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          lds_xt[(chunk * 8 + j) * XT_P + p] = v[j];
+      for (int t = 0; t < XITER; ++t) {
+        const int i = tid + t * 256;
+        if (i < HALO_P * chunks) {
+          const int chunk = i % chunks;
+          const int p = i / chunks;
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            lds_xt[(chunk * 8 + j) * XT_P + p] = vx[t][j];
+        }
       }
     }
     {
       constexpr int kchunks = K >> 3;
-      for (int i = tid; i < WIN_P * kchunks; i += 256) {
+      constexpr int YITER = (WIN_P * kchunks + 255) / 256;
+      w2bf16x8 vy[YITER];
+#pragma unroll
+      for (int t = 0; t < YITER; ++t) {
+        const int i = tid + t * 256;
         const int chunk = i % kchunks;
         const int p = i / kchunks;
         const int orow = oh0 + p / W2TILE_W;
         const int ocol = ow0 + p % W2TILE_W;
-        w2bf16x8 v = {};
-        if (orow < OH && ocol < OW) {
-          v = *reinterpret_cast<const w2bf16x8*>(
+        vy[t] = w2bf16x8{};
+        if (i < WIN_P * kchunks && orow < OH && ocol < OW) {
+          vy[t] = *reinterpret_cast<const w2bf16x8*>(
               dy + (((long)img * OH + orow) * OW + ocol) * K + chunk * 8);
         }
+      }
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          lds_dyt[(chunk * 8 + j) * DYT_P + p] = v[j];
+      for (int t = 0; t < YITER; ++t) {
+        const int i = tid + t * 256;
+        if (i < WIN_P * kchunks) {
+          const int chunk = i % kchunks;
+          const int p = i / kchunks;
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            lds_dyt[(chunk * 8 + j) * DYT_P + p] = vy[t][j];
+        }
       }
     }
     __syncthreads();
