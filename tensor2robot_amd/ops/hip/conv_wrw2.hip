@@ -337,8 +337,14 @@ conv_s1_wrw3_kernel(const w2bf16_t* __restrict__ x,
           if (rs < RS) {
             const int r = rs / S, s = rs % S;
             const int halo_base = (kstep + r) * W2HALO_W + s;
-            w2bf16x8 a_frag = *reinterpret_cast<const w2bf16x8*>(
-                &lds_xt[c * XT_P + halo_base + kgrp * 8]);
+            // Element-wise u16 reads: the (r,s)-shifted start is NOT
+            // 16B aligned for most offsets, and a misaligned
+            // ds_read_b128 replays at 64 cycles/instr (guide G17) —
+            // the dominant stall of wrw v1-v3.
+            const short* arow = &lds_xt[c * XT_P + halo_base + kgrp * 8];
+            w2bf16x8 a_frag;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) a_frag[j] = arow[j];
             acc[g] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 a_frag, b_frag, acc[g], 0, 0, 0);
           }
