@@ -17,6 +17,8 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
+from tensor2robot_amd.utils import miopen_db
+
 from tensor2robot_amd.models import optimizers
 from tensor2robot_amd.research.bcz import model as bcz_model
 from tensor2robot_amd.specs import tensorspec_utils as tsu
@@ -31,6 +33,7 @@ def main():
   p.add_argument("--resnet-size", type=int, default=18)
   args = p.parse_args()
   assert torch.cuda.is_available()
+  miopen_db.use_packaged_db()
   torch.backends.cudnn.benchmark = True
   device = torch.device("cuda:0")
 

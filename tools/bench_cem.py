@@ -21,6 +21,8 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import numpy as np
 import torch
 
+from tensor2robot_amd.utils import miopen_db
+
 from tensor2robot_amd.predictors import checkpoint_predictor
 from tensor2robot_amd.research.qtopt import t2r_models
 
@@ -33,6 +35,7 @@ def main():
   p.add_argument("--warmup", type=int, default=10)
   args = p.parse_args()
   assert torch.cuda.is_available()
+  miopen_db.use_packaged_db()
   torch.backends.cudnn.benchmark = True
 
   model = t2r_models.GraspingModel(
