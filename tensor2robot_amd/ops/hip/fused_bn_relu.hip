@@ -43,7 +43,16 @@ union Vec8 {
 
 // Workgroups for the streaming kernels: enough to fill 256 CUs several
 // times over, small enough that the partial-slab reduction stays cheap.
-static int pick_grid(long M, int rows, int cap = 1024) {
+// T2R_BN_GRID_CAP overrides for A/Bs (big 78^2 layers measured ~1/3 of
+// HBM roof at cap 1024 = half the chip's wave slots).
+static int pick_grid(long M, int rows, int cap = 0) {
+  if (cap == 0) {
+    static const int env_cap = []() {
+      const char* v = std::getenv("T2R_BN_GRID_CAP");
+      return v ? atoi(v) : 2048;
+    }();
+    cap = env_cap;
+  }
   long wgs = (M + rows - 1) / rows;
   if (wgs > cap) wgs = cap;
   if (wgs < 1) wgs = 1;
