@@ -43,13 +43,14 @@ union Vec8 {
 
 // Workgroups for the streaming kernels: enough to fill 256 CUs several
 // times over, small enough that the partial-slab reduction stays cheap.
-// T2R_BN_GRID_CAP overrides for A/Bs (big 78^2 layers measured ~1/3 of
-// HBM roof at cap 1024 = half the chip's wave slots).
+// T2R_BN_GRID_CAP overrides for A/Bs: caps 1024/2048/4096 measured
+// IDENTICAL (1281 GB/s whole-loop) — the ~3 TB/s per-kernel stream
+// rate is not a wave-count problem; PMC follow-up is round-2 work.
 static int pick_grid(long M, int rows, int cap = 0) {
   if (cap == 0) {
     static const int env_cap = []() {
       const char* v = std::getenv("T2R_BN_GRID_CAP");
-      return v ? atoi(v) : 2048;
+      return v ? atoi(v) : 1024;
     }();
     cap = env_cap;
   }
