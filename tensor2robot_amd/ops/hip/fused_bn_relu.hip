@@ -78,15 +78,29 @@ bn_stats_kernel(const bf16_t* __restrict__ x, float* __restrict__ partial,
   float sum[8], sq[8];
 #pragma unroll
   for (int i = 0; i < 8; ++i) { sum[i] = 0.f; sq[i] = 0.f; }
-  for (long r = (long)blockIdx.x * rows + rg; r < M;
-       r += (long)gridDim.x * rows) {
-    Vec8 vec;
-    vec.raw = *reinterpret_cast<const uint4*>(x + r * C + cbase);
+  // 4x unrolled grid-stride: 4 independent loads in flight per thread
+  // (measured ~2-3 TB/s before: the stream was memory-level-
+  // parallelism-bound, not wave-bound).
+  const long stride = (long)gridDim.x * rows;
+  for (long r = (long)blockIdx.x * rows + rg; r < M; r += stride * 4) {
+    Vec8 vec[4];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      float f = bf2f(vec.v[i]);
-      sum[i] += f;
-      sq[i] += f * f;
+    for (int u = 0; u < 4; ++u) {
+      const long rr = r + u * stride;
+      if (rr < M)
+        vec[u].raw = *reinterpret_cast<const uint4*>(x + rr * C + cbase);
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const long rr = r + u * stride;
+      if (rr < M) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          float f = bf2f(vec[u].v[i]);
+          sum[i] += f;
+          sq[i] += f * f;
+        }
+      }
     }
   }
 #pragma unroll
@@ -175,17 +189,29 @@ bn_apply_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
     sc[i] = scale[cbase + i];
     sh[i] = shift[cbase + i];
   }
-  for (long r = (long)blockIdx.x * rows + rg; r < M;
-       r += (long)gridDim.x * rows) {
-    Vec8 in, out;
-    in.raw = *reinterpret_cast<const uint4*>(x + r * C + cbase);
+  const long stride = (long)gridDim.x * rows;
+  for (long r = (long)blockIdx.x * rows + rg; r < M; r += stride * 4) {
+    Vec8 in[4];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      float f = bf2f(in.v[i]) * sc[i] + sh[i];
-      if (fuse_relu) f = fmaxf(f, 0.f);
-      out.v[i] = f2bf(f);
+    for (int u = 0; u < 4; ++u) {
+      const long rr = r + u * stride;
+      if (rr < M)
+        in[u].raw = *reinterpret_cast<const uint4*>(x + rr * C + cbase);
     }
-    *reinterpret_cast<uint4*>(y + r * C + cbase) = out.raw;
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const long rr = r + u * stride;
+      if (rr < M) {
+        Vec8 out;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          float f = bf2f(in[u].v[i]) * sc[i] + sh[i];
+          if (fuse_relu) f = fmaxf(f, 0.f);
+          out.v[i] = f2bf(f);
+        }
+        *reinterpret_cast<uint4*>(y + rr * C + cbase) = out.raw;
+      }
+    }
   }
 }
 
@@ -220,18 +246,30 @@ bn_bwd_reduce_kernel(const bf16_t* __restrict__ x,
     db[i] = 0.f;
     dg[i] = 0.f;
   }
-  for (long r = (long)blockIdx.x * rows + rg; r < M;
-       r += (long)gridDim.x * rows) {
-    Vec8 vx, vdy;
-    vx.raw = *reinterpret_cast<const uint4*>(x + r * C + cbase);
-    vdy.raw = *reinterpret_cast<const uint4*>(dy + r * C + cbase);
+  const long stride = (long)gridDim.x * rows;
+  for (long r = (long)blockIdx.x * rows + rg; r < M; r += stride * 4) {
+    Vec8 vx[4], vdy[4];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      float xhat = (bf2f(vx.v[i]) - mu[i]) * is[i];
-      float g = bf2f(vdy.v[i]);
-      if (fused_relu && (xhat * ga[i] + be[i]) <= 0.f) g = 0.f;
-      db[i] += g;
-      dg[i] += g * xhat;
+    for (int u = 0; u < 4; ++u) {
+      const long rr = r + u * stride;
+      if (rr < M) {
+        vx[u].raw = *reinterpret_cast<const uint4*>(x + rr * C + cbase);
+        vdy[u].raw = *reinterpret_cast<const uint4*>(dy + rr * C + cbase);
+      }
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const long rr = r + u * stride;
+      if (rr < M) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          float xhat = (bf2f(vx[u].v[i]) - mu[i]) * is[i];
+          float g = bf2f(vdy[u].v[i]);
+          if (fused_relu && (xhat * ga[i] + be[i]) <= 0.f) g = 0.f;
+          db[i] += g;
+          dg[i] += g * xhat;
+        }
+      }
     }
   }
 #pragma unroll
@@ -320,22 +358,35 @@ bn_bwd_dx_kernel(const bf16_t* __restrict__ x,
     ga[i] = gamma[cbase + i];
     be[i] = beta[cbase + i];
   }
-  for (long r = (long)blockIdx.x * rows + rg; r < M;
-       r += (long)gridDim.x * rows) {
-    Vec8 vx, vdy, out;
-    vx.raw = *reinterpret_cast<const uint4*>(x + r * C + cbase);
-    vdy.raw = *reinterpret_cast<const uint4*>(dy + r * C + cbase);
+  const long stride = (long)gridDim.x * rows;
+  for (long r = (long)blockIdx.x * rows + rg; r < M; r += stride * 4) {
+    Vec8 vx[4], vdy[4];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      float xv = bf2f(vx.v[i]);
-      float g = bf2f(vdy.v[i]);
-      if (fused_relu) {
-        float xhat = (xv - mu[i]) * is[i];
-        if (xhat * ga[i] + be[i] <= 0.f) g = 0.f;
+    for (int u = 0; u < 4; ++u) {
+      const long rr = r + u * stride;
+      if (rr < M) {
+        vx[u].raw = *reinterpret_cast<const uint4*>(x + rr * C + cbase);
+        vdy[u].raw = *reinterpret_cast<const uint4*>(dy + rr * C + cbase);
       }
-      out.v[i] = f2bf(a[i] * g + b[i] * xv + c3[i]);
     }
-    *reinterpret_cast<uint4*>(dx + r * C + cbase) = out.raw;
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const long rr = r + u * stride;
+      if (rr < M) {
+        Vec8 out;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          float xv = bf2f(vx[u].v[i]);
+          float g = bf2f(vdy[u].v[i]);
+          if (fused_relu) {
+            float xhat = (xv - mu[i]) * is[i];
+            if (xhat * ga[i] + be[i] <= 0.f) g = 0.f;
+          }
+          out.v[i] = f2bf(a[i] * g + b[i] * xv + c3[i]);
+        }
+        *reinterpret_cast<uint4*>(dx + rr * C + cbase) = out.raw;
+      }
+    }
   }
 }
 
