@@ -119,3 +119,54 @@ def test_predict_from_model(tmp_path):
       model_dir=str(tmp_path)))
   assert len(preds) == 4
   assert "prediction" in preds[0]
+
+
+def test_backup_checkpoint_for_eval(tmp_path):
+  """Reference train_eval.py:616-684: copy-aside before eval, retry on
+  in-flight tmp files, GC old backups."""
+  model_dir = str(tmp_path)
+  # No checkpoint yet -> gives up after retries.
+  assert checkpointing.create_backup_checkpoint_for_eval(
+      model_dir, max_retries=1, retry_sleep=0.01) is None
+  for step in (3, 7):
+    torch.save({"model_state": {}, "global_step": step},
+               os.path.join(model_dir, f"model.ckpt-{step}.pt"))
+  dst = checkpointing.create_backup_checkpoint_for_eval(
+      model_dir, max_retries=2, retry_sleep=0.01)
+  assert dst is not None and dst.endswith("model.ckpt-7.pt")
+  assert os.path.exists(dst)
+  # A newer checkpoint replaces the backup; the old backup is GC'd.
+  torch.save({"model_state": {}, "global_step": 9},
+             os.path.join(model_dir, "model.ckpt-9.pt"))
+  dst2 = checkpointing.create_backup_checkpoint_for_eval(
+      model_dir, max_retries=2, retry_sleep=0.01)
+  assert dst2.endswith("model.ckpt-9.pt")
+  backup_dir = os.path.dirname(dst2)
+  assert sorted(os.listdir(backup_dir)) == ["model.ckpt-9.pt"]
+  # An in-flight .tmp file blocks the copy until retries run out.
+  open(os.path.join(model_dir, "model.ckpt-11.pt.tmp"), "w").close()
+  torch.save({"model_state": {}, "global_step": 11},
+             os.path.join(model_dir, "model.ckpt-11.pt"))
+  assert checkpointing.create_backup_checkpoint_for_eval(
+      model_dir, max_retries=1, retry_sleep=0.01) is None
+
+
+def test_multi_eval_input_generator(tmp_path, monkeypatch):
+  """Reference default_input_generator.py:128-141: the eval dataset is
+  selected by the eval job name from the cluster env."""
+  from tensor2robot_amd.data import input_generators
+  monkeypatch.setenv("T2R_MULTI_EVAL_NAME", "holdout")
+  gen = input_generators.MultiEvalRecordInputGenerator(
+      eval_map={"train": "/data/a*.tfrecord",
+                "holdout": "/data/b*.tfrecord"},
+      batch_size=2)
+  assert gen._file_patterns == "/data/b*.tfrecord"
+  monkeypatch.delenv("T2R_MULTI_EVAL_NAME")
+  # TF_CONFIG-style naming.
+  monkeypatch.setenv(
+      "TF_CONFIG", '{"task": {"type": "eval_train", "index": 0}}')
+  gen2 = input_generators.MultiEvalRecordInputGenerator(
+      eval_map={"train": "/data/a*.tfrecord",
+                "holdout": "/data/b*.tfrecord"},
+      batch_size=2)
+  assert gen2._file_patterns == "/data/a*.tfrecord"
