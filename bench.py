@@ -294,6 +294,11 @@ def main():
     # Eager fallback must not leave grads aliased into the comm buffer.
     for prm in network.parameters():
       prm.grad = None
+  if distributed and torch.distributed.is_initialized():
+    # The 3 eager settle steps before capture ran unsynced; re-align
+    # ranks (in-place writes - the captured graphs read these tensors).
+    for prm in network.parameters():
+      torch.distributed.broadcast(prm.data, src=0)
 
   def barrier_sync():
     if distributed:
