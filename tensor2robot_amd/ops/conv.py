@@ -1,5 +1,10 @@
 """MFMAConv2d: hand-written MFMA conv for stride-1 NHWC bf16.
 
+Replaces the reference's TF Conv2D call sites (SURVEY 2.10 item 1:
+research/qtopt/networks.py:444-580 5x5/3x3 SAME convs,
+film_resnet_model.py:100-105, vision_layers.py:30-160) with CDNA4
+MFMA kernels.
+
 Wraps ops/hip/conv_s1.hip. Forward and backward-data both run the MFMA
 kernel (backward-data is the same convolution with spatially-flipped,
 channel-transposed weights); the weight gradient uses

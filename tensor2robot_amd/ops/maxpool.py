@@ -1,7 +1,8 @@
 """FusedMaxPool2d: non-overlapping NHWC bf16 max-pool on HIP.
 
 Drop-in for nn.MaxPool2d(k, stride=k, ceil_mode=...) in the grasping
-nets (ops/hip/maxpool.hip): forward stores a window-local uint8 argmax;
+nets (reference slim max_pool2d call sites, qtopt/networks.py:460,534;
+SURVEY 2.10 item 4) via ops/hip/maxpool.hip: forward stores a window-local uint8 argmax;
 backward is a conflict-free gather — torch's atomic max_pool_backward
 was 7.4% of the QT-Opt steady-state step (profiles/).
 

@@ -2,7 +2,9 @@
 
 Single-kernel-pair replacement for the torch op chain in
 `preprocessors/distortion.py` when the raw uint8 batch is already on the
-GPU.  Distortion parameter ranges match
+GPU (reference preprocessors/distortion.py:56-133 crop+distort and
+image_transformations.py:176-265).  Distortion
+parameter ranges match
 `image_transformations.ApplyPhotometricImageDistortions` defaults
 (brightness ±0.125, saturation [0.5,1.5], contrast [0.5,1.5]).
 """

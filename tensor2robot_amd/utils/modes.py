@@ -1,4 +1,5 @@
-"""Run-mode constants (the Estimator ModeKeys analog)."""
+"""Run-mode constants (the reference's tf.estimator.ModeKeys
+analog; see abstract_model.py model_fn mode handling)."""
 
 TRAIN = "train"
 EVAL = "eval"
