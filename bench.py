@@ -244,6 +244,18 @@ def main():
       graphed = graph_step.GraphedTrainStep(graph_body)
 
       if distributed:
+        # The step depends on autograd ACCUMULATING into the grad
+        # views (the documented gradient_as_bucket_view mechanism); if
+        # any p.grad got rebound to a fresh tensor during capture, the
+        # all-reduce would sync a dead buffer — assert aliasing, and
+        # let the failure demote every rank to the eager engine.
+        base = flat.data_ptr()
+        end = base + flat.numel() * flat.element_size()
+        for prm in grad_params:
+          if not (base <= prm.grad.data_ptr() < end):
+            raise RuntimeError(
+                "grad view rebound during capture; eager fallback")
+
         def opt_body():
           flat.div_(float(world_size))
           optimizer.step(0)
