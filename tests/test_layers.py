@@ -301,3 +301,33 @@ def test_embed_modules():
   assert full(torch.randn(3, 9)).shape == (3, 12)
   img = tec.EmbedConditionImages(fc_layers=(32, 16))
   assert img(torch.rand(2, 3, 64, 64)).shape == (2, 16)
+
+
+def test_spatial_softmax_temperature_sharpens():
+  feat = torch.zeros(1, 1, 5, 5)
+  feat[0, 0, 4, 4] = 2.0   # mild peak at (+1, +1)
+  soft_pts, _ = spatial_softmax.SpatialSoftmax(temperature=10.0)(feat)
+  sharp_pts, _ = spatial_softmax.SpatialSoftmax(temperature=0.1)(feat)
+  # Lower temperature moves the expectation toward the arg-max.
+  assert sharp_pts[0, 0] > soft_pts[0, 0]
+  assert sharp_pts[0, 1] > soft_pts[0, 1]
+  torch.testing.assert_close(sharp_pts[0],
+                             torch.tensor([1.0, 1.0]), atol=1e-2, rtol=0)
+
+
+def test_spatial_softmax_gumbel_training_path():
+  # Reference :69-73 RelaxedOneHotCategorical sampling (train only).
+  torch.manual_seed(0)
+  mod = spatial_softmax.SpatialSoftmax(use_gumbel=True,
+                                       gumbel_temperature=1.0)
+  mod.train()
+  feat = torch.randn(2, 3, 6, 6)
+  points, smap = mod(feat)
+  assert points.shape == (2, 6)
+  torch.testing.assert_close(smap.sum(dim=(2, 3)),
+                             torch.ones(2, 3), atol=1e-4, rtol=0)
+  # Eval mode ignores gumbel -> deterministic.
+  mod.eval()
+  p1, _ = mod(feat)
+  p2, _ = mod(feat)
+  torch.testing.assert_close(p1, p2)
