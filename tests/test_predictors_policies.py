@@ -266,3 +266,28 @@ def test_grasping_cem_policy_cpu():
           np.uint8)
   action = policy.SelectAction(state)
   assert action.shape == (t2r_models.ACTION_DIM,)
+
+
+def test_warmup_requests_roundtrip(tmp_path):
+  """Reference abstract_export_generator.py:109-142: zero-filled warmup
+  request records a server replays before traffic."""
+  from tensor2robot_amd.data import tfrecord
+  from tensor2robot_amd.export_generators import default_export_generator
+  model = mocks.MockT2RModel(device_type="cpu")
+  gen = default_export_generator.DefaultExportGenerator()
+  gen.set_specification_from_model(model)
+  path = gen.create_warmup_requests_numpy([1, 4], str(tmp_path))
+  assert os.path.basename(path) == "warmup_requests.tfrecord"
+  records = list(tfrecord.read_records(path))
+  assert len(records) == 2
+  spec = gen.serving_input_spec()
+  for i, bs in enumerate((1, 4)):
+    feats = example_codec.decode_example(records[i])
+    # Every required serving input appears, zero-filled, batch bs.
+    for key, sp in spec.items():
+      name = sp.name or key
+      assert name in feats, (name, sorted(feats))
+      arr = feats[name]
+      import numpy as _np
+      assert _np.asarray(arr).size % bs == 0
+      assert float(_np.abs(_np.asarray(arr, dtype=_np.float64)).sum()) == 0.0
