@@ -149,3 +149,39 @@ def test_bcz_eval_metric_key_for_exporter():
   model = _small_model()
   assert model.is_xyz_space
   assert not model.is_joint_space
+
+
+def test_preprocessor_mixup_blends_images_and_future_labels():
+  """Reference model.py:166-173: one Beta(alpha, alpha) coefficient per
+  batch, image blended against the batch-reversed pairing, only
+  future/* labels blended."""
+  torch.manual_seed(0)
+  model = _small_model()
+  prep = type(model.preprocessor)(
+      model_feature_specification_fn=model.get_feature_specification,
+      model_label_specification_fn=model.get_label_specification,
+      image_size=(64, 64), input_size=(80, 100), crop_size=(72, 90),
+      mock_subtask=True, mixup_alpha=2.0, binarize_gripper=False,
+      rescale_gripper=False)
+  features = tsu.TensorSpecStruct()
+  features["image"] = torch.randint(0, 256, (2, 80, 100, 3),
+                                    dtype=torch.uint8)
+  labels = tsu.TensorSpecStruct()
+  labels["future/xyz"] = torch.tensor([[0.0], [1.0]])
+  labels["gripper_now"] = torch.tensor([[0.0], [1.0]])
+  f, l = prep._preprocess_fn(features, labels, run_modes.TRAIN)
+  # future/* blended: the two rows become m and (1-m); they sum to 1.
+  blended = l["future/xyz"]
+  torch.testing.assert_close(blended[0] + blended[1],
+                             torch.tensor([1.0]))
+  # Non-future labels untouched.
+  torch.testing.assert_close(l["gripper_now"],
+                             torch.tensor([[0.0], [1.0]]))
+  # Eval mode: no mixup.
+  f2, l2 = prep._preprocess_fn(
+      tsu.TensorSpecStruct([("image", features["original_image"])]),
+      tsu.TensorSpecStruct([("future/xyz",
+                             torch.tensor([[0.0], [1.0]]))]),
+      run_modes.EVAL)
+  torch.testing.assert_close(l2["future/xyz"],
+                             torch.tensor([[0.0], [1.0]]))
