@@ -59,3 +59,32 @@ def test_sequence_example_roundtrip(context, sequences):
   assert set(seqs) == set(sequences)
   for k, steps in sequences.items():
     assert len(seqs[k]) == len(steps)
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.lists(st.binary(min_size=0, max_size=300), min_size=0,
+                max_size=10))
+def test_tfrecord_roundtrip_property(tmp_path_factory, payloads):
+  from tensor2robot_amd.data import tfrecord
+  import os
+  d = tmp_path_factory.mktemp("tfr")
+  path = os.path.join(str(d), "r.tfrecord")
+  with tfrecord.TFRecordWriter(path) as w:
+    for p in payloads:
+      w.write(p)
+  got = list(tfrecord.read_records(path, verify_crc=True))
+  assert got == payloads
+
+
+def test_tfrecord_rejects_corrupt_crc(tmp_path):
+  from tensor2robot_amd.data import tfrecord
+  import os
+  path = os.path.join(str(tmp_path), "r.tfrecord")
+  with tfrecord.TFRecordWriter(path) as w:
+    w.write(b"hello world")
+  raw = bytearray(open(path, "rb").read())
+  raw[-2] ^= 0xFF            # flip a bit inside the data CRC
+  open(path, "wb").write(bytes(raw))
+  import pytest as _pytest
+  with _pytest.raises(IOError):
+    list(tfrecord.read_records(path, verify_crc=True))
