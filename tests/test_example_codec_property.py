@@ -88,3 +88,40 @@ def test_tfrecord_rejects_corrupt_crc(tmp_path):
   import pytest as _pytest
   with _pytest.raises(IOError):
     list(tfrecord.read_records(path, verify_crc=True))
+
+
+# ---- spec-structure flatten/pack property ----
+
+_path_keys = st.lists(
+    st.text(alphabet="abcdefgh", min_size=1, max_size=4),
+    min_size=1, max_size=3).map(lambda parts: "/".join(parts))
+
+
+@settings(max_examples=100, deadline=None)
+@given(st.dictionaries(_path_keys, st.integers(0, 100),
+                       min_size=1, max_size=8))
+def test_spec_struct_flatten_pack_roundtrip(flat):
+  """flatten(pack(flat)) == flat for any prefix-free path set
+  (reference tensorspec_utils flatten/pack contract)."""
+  from tensor2robot_amd.specs import tensorspec_utils as tsu
+  import torch
+  # Drop keys where one path is a strict prefix of another (invalid
+  # hierarchies: 'a' cannot be both a leaf and a subtree).
+  keys = sorted(flat)
+  pruned = {}
+  for k in keys:
+    if any(k != other and k.startswith(other + "/") for other in keys):
+      continue
+    pruned[k] = flat[k]
+  keys = sorted(pruned)
+  pruned = {k: v for k, v in pruned.items()
+            if not any(k != o and o.startswith(k + "/") for o in keys)}
+  if not pruned:
+    return
+  s = tsu.TensorSpecStruct()
+  for k, v in pruned.items():
+    s[k] = torch.tensor([v])
+  flat_again = tsu.flatten_spec_structure(s)
+  assert set(flat_again.keys()) == set(pruned.keys())
+  for k, v in pruned.items():
+    assert int(flat_again[k][0]) == v
