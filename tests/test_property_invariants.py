@@ -1,9 +1,8 @@
-"""Property-based roundtrip tests of the native tf.Example wire codec.
+"""Property-based invariants (hypothesis): wire codec, TFRecord
+framing, spec-structure flatten/pack, crop/distortion bounds.
 
-The hand-written varint/length-delimited encoder (data/example.py,
-reference tf.train.Example wire format) must decode every value it can
-encode — hypothesis sweeps dtypes, shapes, extreme values and unicode
-names far past the hand-picked cases in test_data_pipeline.
+These sweep dtypes, shapes, extreme values and unicode names far past
+the hand-picked cases in the per-module suites.
 """
 
 import numpy as np
