@@ -85,8 +85,11 @@ class SpatialSoftmax(nn.Module):
     import os
     if os.environ.get("T2R_DISABLE_FUSED_SPATIAL_SOFTMAX"):
       return False
+    # H*W cap: the kernel loops pixels serially per (image, channel)
+    # thread — right for the robot nets' small maps, wrong for huge
+    # ones (those fall back to the softmax+GEMM composition).
     return (x.is_cuda and x.dtype == torch.bfloat16 and x.dim() == 4 and
-            x.shape[1] <= 256 and
+            x.shape[1] <= 256 and x.shape[2] * x.shape[3] <= 4096 and
             not (self.use_gumbel and self.training))
 
   def forward(self, features: torch.Tensor
