@@ -296,3 +296,19 @@ def test_mfma_wrw3_matches_torch(shape):
   err = (dw_t - ref).abs().max().item()
   scale = ref.abs().max().item()
   assert err < 0.01 * max(scale, 1.0), (shape, err, scale)
+
+
+@requires_gpu
+def test_spatial_softmax_large_map_fallback():
+  """Maps past the 4096-pixel fused cap use the torch composition on
+  GPU with matching numerics."""
+  from tensor2robot_amd.layers import spatial_softmax as ss
+  torch.manual_seed(1)
+  mod = ss.SpatialSoftmax()
+  x = torch.randn(2, 8, 80, 80, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last)
+  assert not mod._fused_supported(x)
+  points, smap = mod(x)
+  assert points.shape == (2, 16) and smap.shape == (2, 8, 80, 80)
+  ref_pts, _ = mod(x.float())
+  assert (points.float() - ref_pts).abs().max().item() < 2e-2
