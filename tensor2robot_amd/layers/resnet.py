@@ -78,6 +78,7 @@ class _BuildingBlockV1(nn.Module):
   def __init__(self, in_ch: int, filters: int, stride: int,
                use_projection: bool):
     super().__init__()
+    self.film_width = 2 * filters  # apply site: post-bn2, `filters` ch
     self.conv1 = _conv_fixed_padding(in_ch, filters, 3, stride)
     self.bn1 = _bn(filters, relu=True)
     self.conv2 = _conv_fixed_padding(filters, filters, 3, 1)
@@ -105,6 +106,7 @@ class _BottleneckV1(nn.Module):
                use_projection: bool):
     super().__init__()
     out_ch = filters * self.expansion
+    self.film_width = 2 * out_ch  # apply site: post-bn3, expanded ch
     self.conv1 = _conv_fixed_padding(in_ch, filters, 1, 1)
     self.bn1 = _bn(filters, relu=True)
     self.conv2 = _conv_fixed_padding(filters, filters, 3, stride)
@@ -134,6 +136,7 @@ class _BuildingBlockV2(nn.Module):
   def __init__(self, in_ch: int, filters: int, stride: int,
                use_projection: bool):
     super().__init__()
+    self.film_width = 2 * filters  # apply site: post-bn2 pre-ReLU
     self.bn1 = _bn(in_ch, relu=True)
     self.conv1 = _conv_fixed_padding(in_ch, filters, 3, stride)
     self.bn2 = _bn(filters, relu=True)
@@ -146,8 +149,12 @@ class _BuildingBlockV2(nn.Module):
     pre = self.bn1(x)
     shortcut = self.shortcut(pre) if self.shortcut is not None else x
     y = self.conv1(pre)
-    y = self.bn2(y)
-    y = apply_film(y, gamma_beta)
+    if gamma_beta is not None:
+      # Reference order: batch_norm -> _apply_film -> relu -> conv
+      # (film_resnet_model.py:210-213) — split the BN+ReLU fusion.
+      y = F.relu(apply_film(self.bn2(y, relu=False), gamma_beta))
+    else:
+      y = self.bn2(y)
     y = self.conv2(y)
     return y + shortcut
 
@@ -161,6 +168,7 @@ class _BottleneckV2(nn.Module):
                use_projection: bool):
     super().__init__()
     out_ch = filters * self.expansion
+    self.film_width = 2 * filters  # apply site: post-bn3, pre-expansion
     self.bn1 = _bn(in_ch, relu=True)
     self.conv1 = _conv_fixed_padding(in_ch, filters, 1, 1)
     self.bn2 = _bn(filters, relu=True)
@@ -177,8 +185,12 @@ class _BottleneckV2(nn.Module):
     y = self.conv1(pre)
     y = self.bn2(y)
     y = self.conv2(y)
-    y = self.bn3(y)
-    y = apply_film(y, gamma_beta)
+    if gamma_beta is not None:
+      # Reference order: batch_norm -> _apply_film -> relu -> conv3
+      # (film_resnet_model.py:333-336) — split the BN+ReLU fusion.
+      y = F.relu(apply_film(self.bn3(y, relu=False), gamma_beta))
+    else:
+      y = self.bn3(y)
     y = self.conv3(y)
     return y + shortcut
 
@@ -263,11 +275,16 @@ class ResNet(nn.Module):
 
   @property
   def film_channels(self) -> List[int]:
-    """Per-block FiLM width 2C, ordered block_layer1..4 (flat)."""
+    """Per-block FiLM width 2C at the apply site, block_layer1..4 flat.
+
+    The apply-site channel count depends on the block type (pre-expansion
+    `filters` for v2 blocks, expanded channels for v1 bottleneck) — the
+    reference generator sizes FiLM as 2*filters per block
+    (resnet.py:129-139 with filter_sizes = num_filters*2^i)."""
     dims = []
     for layer in self.block_layers:
-      for _ in layer.blocks:
-        dims.append(2 * layer.out_channels)
+      for block in layer.blocks:
+        dims.append(block.film_width)
     return dims
 
   def forward(self, images: torch.Tensor,
@@ -320,7 +337,8 @@ class LinearFiLMGenerator(nn.Module):
       raise ValueError("enabled_block_layers must have one entry per "
                        "block layer")
     self._blocks_per_layer = [len(l.blocks) for l in resnet.block_layers]
-    self._widths = [2 * l.out_channels for l in resnet.block_layers]
+    # Apply-site FiLM width per block (uniform within a layer).
+    self._widths = [l.blocks[0].film_width for l in resnet.block_layers]
     fcs = []
     for enabled, n_blocks, width in zip(self._enabled,
                                         self._blocks_per_layer,

@@ -146,15 +146,19 @@ class MAMLInnerLoopGradientDescent:
       inner_losses.append(loss)
 
     # Monitor adaptation: final forward on the last condition step.
+    # The loss call stays INSIDE the swap so a train_fn that reads
+    # network parameters (e.g. regularizers) sees the adapted fast
+    # weights, matching the preceding inner steps (reference :290-306).
     final_features, final_labels = inputs_list[-2]
     with swap_parameters(network, fast):
       final_outputs = inference_network_fn(features=final_features,
                                            labels=final_labels, mode=mode,
                                            params=params)
+      final_loss = self._extract_train_loss(model_train_fn(
+          features=final_features, labels=final_labels,
+          inference_outputs=final_outputs, mode=mode, params=params))
     inner_outputs.append(final_outputs)
-    inner_losses.append(self._extract_train_loss(model_train_fn(
-        features=final_features, labels=final_labels,
-        inference_outputs=final_outputs, mode=mode, params=params)))
+    inner_losses.append(final_loss)
 
     with swap_parameters(network, fast):
       params_cond = dict(params)

@@ -61,7 +61,13 @@ def _supported(x: torch.Tensor, weight: torch.Tensor, stride,
   # glds-ring variant.  T2R_MFMA_MAX_RS caps which kernels dispatch
   # here (e.g. 9 = 3x3 only) for A/B against a tuned MIOpen.
   max_rs = int(os.environ.get("T2R_MFMA_MAX_RS", "25"))
-  return (stride == (1, 1) and r <= 5 and s <= 5 and r * s <= max_rs and
+  # r == s: the dgrad path derives its pad as (r-1-pad) for BOTH dims
+  # and the wrw kernels assume square windows.  c % 32 when the input
+  # needs a gradient: dgrad swaps channel roles (output channels = c)
+  # and conv_s1_nhwc requires its output-channel count % 32.
+  if x.requires_grad and c % 32 != 0:
+    return False
+  return (stride == (1, 1) and r == s and r * s <= max_rs and
           c % 16 == 0 and c <= 64 and k % 32 == 0 and k <= 64 and
           padding[0] == padding[1])
 
@@ -102,18 +108,28 @@ class _MFMAConvFunction(torch.autograd.Function):
         wpk_b = ext.pack_conv_w(weight, True)
       dx = ext.conv_s1_nhwc(dy.to(torch.bfloat16), wpk_b, c, r, s, bpad)
     if ctx.needs_input_grad[1]:
-      # OPT-IN: T2R_ENABLE_MFMA_WRW=2 selects the v2 register-
-      # accumulator kernel (C=K=64, 3x3/5x5); =1 the v1 LDS-accumulator
-      # (0.25-0.44x MIOpen, kept as groundwork); unset -> MIOpen.
+      # OPT-IN: T2R_ENABLE_MFMA_WRW selects a hand wrw kernel
+      # (=1 v1 LDS-accumulator, =2 v2 register-accumulator, =3 v3
+      # rs-split, =4 v4 tr_b16 pixel-major); unset -> MIOpen.
       wrw_mode = os.environ.get("T2R_ENABLE_MFMA_WRW", "")
-      if wrw_mode == "2" and c == 64 and k == 64 and r == s and \
-          r in (3, 5):
+      sq35 = c == 64 and k == 64 and r == s and r in (3, 5)
+      if wrw_mode == "4" and sq35:
+        dw_f32 = ext.conv_s1_wrw4(x, dy.to(torch.bfloat16), r, s,
+                                  ctx.pad)
+        dw = dw_f32.reshape(r, s, c, k).permute(3, 2, 0, 1) \
+            .contiguous().to(weight.dtype)
+      elif wrw_mode == "3" and sq35:
+        dw_f32 = ext.conv_s1_wrw3(x, dy.to(torch.bfloat16), r, s,
+                                  ctx.pad)
+        dw = dw_f32.reshape(r, s, c, k).permute(3, 2, 0, 1) \
+            .contiguous().to(weight.dtype)
+      elif wrw_mode == "2" and sq35:
         dw_f32 = ext.conv_s1_wrw2(x, dy.to(torch.bfloat16), r, s,
                                   ctx.pad)
         dw = dw_f32.reshape(r, s, c, k).permute(3, 2, 0, 1) \
             .contiguous().to(weight.dtype)
-      elif c % 32 == 0 and wrw_mode:
-        # MFMA wrw: fp32 LDS-accumulated [rs][c][k] -> [k][c][r][s].
+      elif c % 32 == 0 and wrw_mode == "1":
+        # MFMA wrw v1: fp32 LDS-accumulated [rs][c][k] -> [k][c][r][s].
         dw_f32 = ext.conv_s1_wrw(x, dy.to(torch.bfloat16), r, s,
                                  ctx.pad)
         dw = dw_f32.reshape(r, s, c, k).permute(3, 2, 0, 1) \

@@ -83,14 +83,20 @@ class FusedBatchNormReLU(nn.Module):
     return (x.is_cuda and x.dtype == torch.bfloat16 and
             self.num_features % 8 == 0 and self.num_features <= 2048)
 
-  def forward(self, x: torch.Tensor) -> torch.Tensor:
+  def forward(self, x: torch.Tensor,
+              relu: "bool | None" = None) -> torch.Tensor:
+    """Run BN(+ReLU).  `relu` overrides the constructor's fuse_relu for
+    this call — FiLM sites need BN -> film -> ReLU (reference
+    film_resnet_model.py:210-212,333-335), so the block splits the
+    fusion only when a gamma_beta is actually present."""
+    fuse_relu = self.fuse_relu if relu is None else relu
     if self._use_hip(x):
       flat, shape_info = _flat_nhwc(x)
       if self.training:
         y = _FusedBNReLUFunction.apply(
             flat, self.weight, self.bias,
             self.running_mean, self.running_var, self.eps, self.momentum,
-            self.fuse_relu)
+            fuse_relu)
         self._batches_tracked_py += 1
       else:
         invstd = torch.rsqrt(self.running_var + self.eps)
@@ -98,18 +104,13 @@ class FusedBatchNormReLU(nn.Module):
         shift = (self.bias - self.running_mean * self.weight * invstd
                  ).float()
         y = ops_mod.require_hip().bn_inference_apply(
-            flat, scale.contiguous(), shift.contiguous(), self.fuse_relu)
+            flat, scale.contiguous(), shift.contiguous(), fuse_relu)
       return _unflat(y, shape_info)
     # Torch reference path (CPU / non-bf16): identical math.
-    if x.dim() == 4:
-      y = torch.nn.functional.batch_norm(
-          x, self.running_mean, self.running_var, self.weight, self.bias,
-          self.training, self.momentum, self.eps)
-    else:
-      y = torch.nn.functional.batch_norm(
-          x, self.running_mean, self.running_var, self.weight, self.bias,
-          self.training, self.momentum, self.eps)
-    if self.fuse_relu:
+    y = torch.nn.functional.batch_norm(
+        x, self.running_mean, self.running_var, self.weight, self.bias,
+        self.training, self.momentum, self.eps)
+    if fuse_relu:
       y = torch.relu(y)
     return y
 

@@ -67,6 +67,34 @@ def test_resnet_film_disabled_layers():
   assert out.shape == (3, 512)
 
 
+def test_resnet50_v2_film_runs():
+  """FiLM on bottleneck v2: widths must match the pre-expansion apply
+  site (2*filters, reference film_resnet_model.py:333-336) — the
+  round-1 2*out_channels sizing broadcast-crashed here."""
+  net = resnet.ResNet(resnet_size=50, num_classes=0, version=2)
+  gen = resnet.LinearFiLMGenerator(embedding_dim=6, resnet=net)
+  gbs = gen(torch.randn(2, 6))
+  # block_layer1 filters=64 -> width 128 (not 2*256).
+  assert gbs[0].shape == (2, 128)
+  out, _ = net(torch.randn(2, 3, 32, 32), film_gamma_betas=gbs)
+  assert out.shape == (2, 2048)
+
+
+def test_film_v2_order_bn_film_relu():
+  """V2 block applies FiLM between BN and ReLU (reference
+  film_resnet_model.py:210-212): with a strongly negative beta the
+  post-ReLU activations must be able to reach exact zero, which the
+  wrong order relu-then-film cannot produce."""
+  torch.manual_seed(0)
+  block = resnet._BuildingBlockV2(8, 8, 1, use_projection=False).eval()
+  x = torch.randn(2, 8, 6, 6)
+  # gamma = 0, beta = -1000: film output = xhat - 1000, so post-ReLU is
+  # all zeros and conv2's output is exactly zero -> block out == x.
+  gb = torch.cat([torch.zeros(2, 8), torch.full((2, 8), -1000.0)], dim=1)
+  out = block(x, gb)
+  torch.testing.assert_close(out, x)
+
+
 def test_resnet_warm_start_skips_head(tmp_path):
   net = resnet.ResNet(resnet_size=18, num_classes=5)
   path = str(tmp_path / "ckpt.pt")
