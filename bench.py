@@ -5,8 +5,15 @@ QT-Opt critic train, 472x472 input, bs=32/GPU, bf16, on 1..8 MI355X.
 
 Each timed step is the full training step: on-GPU preprocessing of the raw
 synthetic 512x640 uint8 batch (convert + crop + photometric distortion),
-Grasping44 forward, sigmoid log-loss, backward, bucketed RCCL gradient
+Grasping44 forward, sigmoid log-loss, backward, flat-buffer RCCL gradient
 all-reduce (N>1), momentum optimizer step and EMA update — nothing skipped.
+
+This file is a THIN caller: the performance-engineered step (hipGraph
+capture, flat-grad-view distributed comm, channels_last, MIOpen DB
+pinning) lives in the framework — tensor2robot_amd/train/train_eval.py
+Trainer + tensor2robot_amd/parallel/fast_step.py — and is what every
+`train_eval_model()` user gets.  The bench constructs the model, a
+synthetic on-device batch pool, and times Trainer-driven steps.
 
 Launch (driver contract):
   python bench.py --gpus 1 --steps K --warmup W
@@ -15,15 +22,17 @@ Launch (driver contract):
 """
 
 import argparse
+import functools
+import itertools
 import json
 import os
 import time
 
 import torch
 
-from tensor2robot_amd.models import optimizers as optimizers_mod
 from tensor2robot_amd.research.qtopt import t2r_models
 from tensor2robot_amd.specs import tensorspec_utils as tsu
+from tensor2robot_amd.train import train_eval
 from tensor2robot_amd.utils import modes as run_modes
 
 
@@ -42,7 +51,7 @@ def parse_args():
 
 
 def make_synthetic_pool(batch_size, device, n_batches=4, seed=0):
-  """Fixed pool of raw uint8 batches, resident on the GPU."""
+  """Fixed pool of raw uint8 (features, labels) batches on the device."""
   g = torch.Generator(device="cpu").manual_seed(seed)
   pool = []
   for _ in range(n_batches):
@@ -51,27 +60,18 @@ def make_synthetic_pool(batch_size, device, n_batches=4, seed=0):
                             t2r_models.RAW_WIDTH, 3),
                            generator=g, dtype=torch.uint8)
     action = torch.rand((batch_size, t2r_models.ACTION_DIM), generator=g)
-    labels = (torch.rand((batch_size, 1), generator=g) > 0.5).float()
-    pool.append((images.to(device), action.to(device), labels.to(device)))
+    success = (torch.rand((batch_size, 1), generator=g) > 0.5).float()
+    features = tsu.TensorSpecStruct()
+    features["state/image"] = images.to(device)
+    offset = 0
+    for name, size in t2r_models.ACTION_COMPONENTS:
+      features["action/" + name] = action[:, offset: offset + size].to(
+          device)
+      offset += size
+    labels = tsu.TensorSpecStruct()
+    labels["grasp_success"] = success.to(device)
+    pool.append((features, labels))
   return pool
-
-
-def build_features(model, images, action, mode, preprocess=True):
-  features = tsu.TensorSpecStruct()
-  if preprocess:
-    features["state/image"] = images
-    offset = 0
-    for name, size in t2r_models.ACTION_COMPONENTS:
-      features["action/" + name] = action[:, offset: offset + size]
-      offset += size
-    features, _ = model.preprocessor.preprocess(features, None, mode)
-  else:
-    features["state/image"] = images
-    offset = 0
-    for name, size in t2r_models.ACTION_COMPONENTS:
-      features["action/" + name] = action[:, offset: offset + size]
-      offset += size
-  return features
 
 
 def main():
@@ -85,238 +85,47 @@ def main():
   distributed = world_size > 1 or bool(os.environ.get("T2R_FORCE_DIST"))
 
   use_cuda = torch.cuda.is_available()
-  if use_cuda:
-    # Pin MIOpen to the packaged searched-best kernels (its runtime
-    # find is a per-process lottery on this pool - profiles/).
-    from tensor2robot_amd.utils import miopen_db
-    miopen_db.use_packaged_db()
-  torch.backends.cudnn.benchmark = True  # let MIOpen autotune conv algos
   if distributed:
     backend = "nccl" if use_cuda else "gloo"
     torch.distributed.init_process_group(backend=backend)
-  device = torch.device(f"cuda:{local_rank}") if use_cuda else \
-      torch.device("cpu")
   if use_cuda:
-    torch.cuda.set_device(device)
+    torch.cuda.set_device(torch.device(f"cuda:{local_rank}"))
 
   model = t2r_models.GraspingModel(
       device_type="gpu" if use_cuda else "cpu",
       compute_dtype="bfloat16" if use_cuda else "float32",
       use_avg_model_params=not args.no_ema)
-  model.to_device(device)
-  network = model.network
-  network.to(memory_format=torch.channels_last)
-
-  want_graph = use_cuda and not args.no_hipgraph
-  dp_engine = None
-  if distributed and torch.distributed.is_initialized():
-    # Identical start on every rank (the reference's chief-initialized
-    # variables); the graphed-dist path has no DDP hooks to do it.
-    for prm in network.parameters():
-      torch.distributed.broadcast(prm.data, src=0)
-  if distributed and not want_graph:
-    # Eager path: hook-driven bucketed all-reduce overlapped with
-    # backward.  NOT used under graphs: the hooks would record RCCL
-    # collectives into the capture; the graphed-dist step instead syncs
-    # grads eagerly between graph replay and the optimizer step.
-    from tensor2robot_amd.parallel import ddp
-    dp_engine = ddp.DataParallelEngine(network)
-
-  optimizer = model.create_optimizer()
-  ema = model.create_ema()
+  trainer = train_eval.Trainer(model, model_dir="",
+                               use_hip_graph=use_cuda and
+                               not args.no_hipgraph)
+  device = trainer.device
 
   pool = make_synthetic_pool(args.batch_size, device, seed=1234 + rank)
-  autocast = torch.autocast(device_type="cuda", dtype=torch.bfloat16,
-                            enabled=use_cuda)
+  preprocess_fn = None
+  if not args.no_preprocess:
+    preprocess_fn = functools.partial(model.preprocessor.preprocess,
+                                      mode=run_modes.TRAIN)
+  else:
+    # Ablation: pre-preprocess the pool once; the timed step consumes
+    # ready 472x472 f32 crops.
+    with torch.no_grad():
+      pool = [model.preprocessor.preprocess(f, l, run_modes.EVAL)
+              for f, l in pool]
 
-  def preprocess_to(images, action, static_image, static_action):
-    """Dynamic (host-RNG) preprocessing -> static graph inputs."""
-    with autocast:
-      features = build_features(model, images, action, run_modes.TRAIN,
-                                preprocess=not args.no_preprocess)
-      image = features["state/image"].permute(0, 3, 1, 2).contiguous(
-          memory_format=torch.channels_last)
-      actions = model.pack_action_vector(features)
-    static_image.copy_(image)
-    static_action.copy_(actions)
+  pool_iter = itertools.cycle(pool)
 
-  def step(i, global_step):
-    images, action, labels_t = pool[i % len(pool)]
-    optimizer.zero_grad(set_to_none=True)
-    with autocast:
-      features = build_features(model, images, action, run_modes.TRAIN,
-                                preprocess=not args.no_preprocess)
-      image = features["state/image"].permute(0, 3, 1, 2).contiguous(
-          memory_format=torch.channels_last)
-      actions = model.pack_action_vector(features)
-      logit = network(image, actions)
-    loss = torch.nn.functional.binary_cross_entropy_with_logits(
-        logit.float(), labels_t.reshape(logit.shape))
-    if dp_engine is not None:
-      dp_engine.backward(loss)
-    else:
-      loss.backward()
-    optimizer.step(global_step)
-    if ema is not None:
-      ema.update()
-    return loss
-
-  # -- hipGraph capture (single graph launch per step; preprocess stays
-  # outside the graph because its distortion params come from host RNG) --
-  graphed = None
-  static = {}
-  # Distributed: the graph captures forward+backward ONLY; the gradient
-  # all-reduce + optimizer + EMA run eager after each replay.  RCCL
-  # collectives are never recorded into a capture (unvalidated on this
-  # pool - no multi-GPU box to test on), yet the ~250-kernel fwd+bwd
-  # still replays as one launch, so the 8-GPU scaling numbers are not
-  # stuck at eager launch overhead.
-  eager_step = step
-  if use_cuda and not args.no_hipgraph:
-    try:
-      from tensor2robot_amd.parallel import graph_step
-      for i in range(3):  # settle MIOpen algo find before capture
-        step(i, i)
-      torch.cuda.synchronize()
-      images0, action0, labels0 = pool[0]
-      dtype = torch.bfloat16
-      static["image"] = torch.zeros(
-          (args.batch_size, 3, t2r_models.CROP_HEIGHT,
-           t2r_models.CROP_WIDTH), dtype=dtype, device=device
-      ).contiguous(memory_format=torch.channels_last)
-      static["action"] = torch.zeros((args.batch_size,
-                                      t2r_models.ACTION_DIM),
-                                     dtype=dtype, device=device)
-      static["labels"] = torch.zeros((args.batch_size, 1), device=device)
-      preprocess_to(images0, action0, static["image"], static["action"])
-      static["labels"].copy_(labels0.reshape(-1, 1))
-
-      grad_params = [prm for prm in network.parameters()
-                     if prm.requires_grad]
-      flat = None
-      if distributed:
-        # One flat f32 comm buffer.  Every p.grad is pre-assigned as a
-        # strided VIEW into it (channels_last strides for 4D params),
-        # so backward ACCUMULATES straight into the comm buffer: no
-        # gather/scatter kernels at all.  graph1 = zero+fwd+bwd, the
-        # all-reduce runs EAGER between the two graph replays (the one
-        # RCCL call per step), graph2 = div + optimizer + EMA.
-        total = sum(prm.numel() for prm in grad_params)
-        flat = torch.zeros(total, dtype=torch.float32, device=device)
-
-        def assign_grad_views():
-          off = 0
-          for prm in grad_params:
-            n = prm.numel()
-            sl = flat[off:off + n]
-            if prm.dim() == 4 and prm.is_contiguous(
-                memory_format=torch.channels_last):
-              no, c, h, w = prm.shape
-              g = sl.view(no, h, w, c).permute(0, 3, 1, 2)
-            else:
-              g = sl.view(prm.shape)
-            prm.grad = g
-            off += n
-
-      def graph_body():
-        if distributed:
-          # grads are views of `flat`: one fill clears them all, and
-          # backward accumulates into the comm buffer directly.
-          flat.zero_()
-        else:
-          # set_to_none inside capture: backward then WRITES fresh
-          # graph-pool buffers (stable across replays) instead of
-          # zero-fill + accumulate-add per param (~165 kernels/step).
-          optimizer.zero_grad(set_to_none=True)
-        with autocast:
-          logit = network(static["image"], static["action"])
-        loss = torch.nn.functional.binary_cross_entropy_with_logits(
-            logit.float(), static["labels"].reshape(logit.shape))
-        loss.backward()
-        if not distributed:
-          optimizer.step(0)
-          if ema is not None:
-            ema.update()
-        return loss
-
-      if distributed:
-        assign_grad_views()
-      graphed = graph_step.GraphedTrainStep(graph_body)
-
-      if distributed:
-        # The step depends on autograd ACCUMULATING into the grad
-        # views (the documented gradient_as_bucket_view mechanism); if
-        # any p.grad got rebound to a fresh tensor during capture, the
-        # all-reduce would sync a dead buffer — assert aliasing, and
-        # let the failure demote every rank to the eager engine.
-        base = flat.data_ptr()
-        end = base + flat.numel() * flat.element_size()
-        for prm in grad_params:
-          if not (base <= prm.grad.data_ptr() < end):
-            raise RuntimeError(
-                "grad view rebound during capture; eager fallback")
-
-        def opt_body():
-          flat.div_(float(world_size))
-          optimizer.step(0)
-          if ema is not None:
-            ema.update()
-          return None
-
-        opt_graphed = graph_step.GraphedTrainStep(opt_body)
-
-        def step(i, global_step):  # noqa: F811 (graphed fwd+bwd)
-          images, action, labels_t = pool[i % len(pool)]
-          preprocess_to(images, action, static["image"], static["action"])
-          static["labels"].copy_(labels_t.reshape(-1, 1))
-          loss = graphed.replay()
-          torch.distributed.all_reduce(flat)
-          opt_graphed.replay()
-          return loss
-      else:
-
-        def step(i, global_step):  # noqa: F811 (graph-replay fast path)
-          images, action, labels_t = pool[i % len(pool)]
-          preprocess_to(images, action, static["image"], static["action"])
-          static["labels"].copy_(labels_t.reshape(-1, 1))
-          return graphed.replay()
-    except Exception as e:  # pragma: no cover - depends on runtime
-      if rank == 0:
-        import sys as _sys
-        print(f"# hipGraph capture unavailable, running eager: {e!r}",
-              file=_sys.stderr, flush=True)
-      graphed = None
-
-  if distributed and torch.distributed.is_initialized():
-    # The capture-or-eager decision must be COLLECTIVE: one rank
-    # replaying graphs (one all-reduce of `flat` per step) while
-    # another runs the hook-bucketed eager engine would mismatch
-    # collectives and hang the job.  Demote everyone if anyone failed.
-    ok = torch.tensor(
-        [0.0 if graphed is None and use_cuda and not args.no_hipgraph
-         else 1.0],
-        device=device if use_cuda else "cpu")
-    torch.distributed.all_reduce(ok, op=torch.distributed.ReduceOp.MIN)
-    if float(ok.item()) < 1.0 and graphed is not None:
-      graphed = None
-      step = eager_step
-  if distributed and graphed is None and dp_engine is None:
-    from tensor2robot_amd.parallel import ddp
-    dp_engine = ddp.DataParallelEngine(network)
-    # Eager fallback must not leave grads aliased into the comm buffer.
-    for prm in network.parameters():
-      prm.grad = None
-  if distributed and torch.distributed.is_initialized():
-    # The 3 eager settle steps before capture ran unsynced; re-align
-    # ranks (in-place writes - the captured graphs read these tensors).
-    for prm in network.parameters():
-      torch.distributed.broadcast(prm.data, src=0)
+  def input_fn():
+    return pool_iter  # device-resident batches; H2D already done
 
   def barrier_sync():
     if distributed:
       torch.distributed.barrier()
     if use_cuda:
       torch.cuda.synchronize()
+
+  def run_steps(n):
+    trainer.train(input_fn, trainer.global_step + n,
+                  preprocess_fn=preprocess_fn)
 
   warmup_iters = args.warmup
   if use_cuda:
@@ -326,20 +135,16 @@ def main():
     # Settling is untimed, so cover the ramp regardless of the caller's
     # warmup count, then probe until a short timing window stabilizes.
     warmup_iters = max(args.warmup, 30)
-  for i in range(warmup_iters):
-    step(i, i)
+  run_steps(warmup_iters)
   barrier_sync()
   if use_cuda:
-    # Settle probe: run 5-step windows until two consecutive windows
-    # agree within 3% (or a hard cap), so the timed region below
-    # measures steady state even on a cold box.  All untimed.
+    # Settle probe: 5-step windows until two consecutive windows agree
+    # within 3% (or a hard cap), so the timed region below measures
+    # steady state even on a cold box.  All untimed.
     prev_win = None
-    settle_step = warmup_iters
     for _ in range(24):  # cap: 120 extra steps (~0.5 s at steady state)
       w0 = time.perf_counter()
-      for _ in range(5):
-        step(settle_step, settle_step)
-        settle_step += 1
+      run_steps(5)
       barrier_sync()
       win = time.perf_counter() - w0
       stable = (prev_win is not None
@@ -347,7 +152,9 @@ def main():
       if distributed:
         # The continue/break decision must be COLLECTIVE (every window
         # ends in a barrier): stop only when every rank is stable.
-        flag = torch.tensor([1.0 if stable else 0.0], device=device)
+        flag = torch.tensor(
+            [1.0 if stable else 0.0],
+            device=device if use_cuda else "cpu")
         torch.distributed.all_reduce(flag,
                                      op=torch.distributed.ReduceOp.MIN)
         stable = float(flag.item()) >= 1.0
@@ -356,8 +163,7 @@ def main():
       prev_win = win
   barrier_sync()
   t0 = time.perf_counter()
-  for i in range(args.steps):
-    step(i, args.warmup + i)
+  run_steps(args.steps)
   barrier_sync()
   elapsed = time.perf_counter() - t0
 

@@ -75,6 +75,18 @@ class AbstractInputGenerator:
   def label_spec(self):
     return self._label_spec
 
+  def defer_preprocessing(self):
+    """Detach the bound preprocess_fn for on-device execution.
+
+    After this, the pipeline yields RAW parsed batches and the caller
+    (Trainer) applies the returned fn after H2D transfer — the deferred
+    device-preprocess path (preprocess rides HIP, dynamic RNG stays
+    outside the hipGraph).  Returns None if nothing is bound.
+    """
+    fn = self._preprocess_fn
+    self._preprocess_fn = None
+    return fn
+
   def set_preprocess_fn(self, preprocess_fn):
     """Mode must already be bound (reference :100-129 enforcement)."""
     if isinstance(preprocess_fn, functools.partial):

@@ -88,7 +88,8 @@ class Trainer:
                hooks: Optional[List[hooks_mod.TrainHook]] = None,
                max_to_keep: int = 5,
                keep_checkpoint_every_n_hours: Optional[float] = None,
-               log_every_n_steps: int = 100):
+               log_every_n_steps: int = 100,
+               use_hip_graph: Optional[bool] = None):
     self.model = model
     self.model_dir = model_dir
     self.hooks = hooks or []
@@ -96,14 +97,28 @@ class Trainer:
     self.optimizer = None
     self.ema = None
     self._dp_engine = None
+    self._fast_engine = None
+    self._captured_ops = None
     self._log_every = log_every_n_steps
     self.rank, self.world_size = _dist_info()
     self.is_chief = self.rank == 0
     self.device = resolve_device(model)
+    if self.device.type == "cuda":
+      # Bench-grade device setup is the DEFAULT training path, not a
+      # benchmark-only trick: pin MIOpen to the packaged searched-best
+      # kernels (runtime find is a per-process lottery on this pool)
+      # and let it autotune the rest.
+      from tensor2robot_amd.utils import miopen_db
+      miopen_db.use_packaged_db()
+      torch.backends.cudnn.benchmark = True
     model.to_device(self.device)
     self._autocast_enabled = (
         self.device.type == "cuda" and
         model.compute_dtype in (torch.bfloat16, torch.float16))
+    if use_hip_graph is None:
+      use_hip_graph = self.device.type == "cuda" and \
+          not os.environ.get("T2R_DISABLE_HIPGRAPH")
+    self._use_hip_graph = use_hip_graph
     self.checkpointer = checkpointing.Checkpointer(
         model_dir, max_to_keep=max_to_keep,
         keep_checkpoint_every_n_hours=keep_checkpoint_every_n_hours) \
@@ -121,9 +136,11 @@ class Trainer:
     if self.optimizer is not None:
       return
     _ = self.model.network  # lazy construction + device placement
-    if self.world_size > 1:
-      from tensor2robot_amd.parallel import ddp
-      self._dp_engine = ddp.DataParallelEngine(self.model.network)
+    if self.device.type == "cuda":
+      self.model.network.to(memory_format=torch.channels_last)
+    # Distributed gradient sync is owned by the FastStepEngine (flat-
+    # grad-view graphed step on GPU, bucketed eager engine otherwise) —
+    # see parallel/fast_step.py.  No separate hook engine here.
     self.optimizer = self.model.create_optimizer()
     self.ema = self.model.create_ema()
     if self.model_dir:
@@ -147,9 +164,38 @@ class Trainer:
                           dtype=self.model.compute_dtype,
                           enabled=self._autocast_enabled)
 
+  def _ensure_fast_engine(self, features, labels):
+    """Build the graphed/flat-grad step engine on the first batch."""
+    if self._fast_engine is not None:
+      return
+    from tensor2robot_amd.parallel import fast_step
+
+    def loss_fn(f, l):
+      ops = self.model.model_fn(f, l, TRAIN)
+      # The ops object's tensors ARE the graph-static outputs: they
+      # refresh in place on every replay, so hooks and summaries keep
+      # seeing live values.
+      self._captured_ops = ops
+      return ops.loss
+
+    lr_schedule = getattr(self.optimizer, "lr_schedule", None)
+    self._fast_engine = fast_step.FastStepEngine(
+        self.network, self.optimizer, self.ema, device=self.device,
+        use_graph=self._use_hip_graph,
+        autocast_dtype=self.model.compute_dtype
+        if self._autocast_enabled else torch.float32,
+        lr_schedule=lr_schedule)
+    self._fast_engine.build(loss_fn, features, labels,
+                            global_step=self.global_step)
+
   # -- train ---------------------------------------------------------------
   def train(self, input_fn, max_steps: int,
-            save_checkpoint_steps: Optional[int] = None) -> Dict[str, float]:
+            save_checkpoint_steps: Optional[int] = None,
+            preprocess_fn=None) -> Dict[str, float]:
+    """Step loop.  `preprocess_fn(features, labels)` runs ON DEVICE
+    right after H2D transfer when provided (deferred preprocessing:
+    uint8 crosses the bus, crop/distort rides HIP, and the dynamic
+    host-RNG work stays outside the hipGraph)."""
     self._ensure_built()
     self.network.train()
     for hook in self.hooks:
@@ -157,29 +203,24 @@ class Trainer:
         hook.begin(self.context)
     iterator = iter(input_fn())
     last_log = time.time()
-    last_loss = float("nan")
+    last_loss_t = None
     while self.global_step < max_steps:
       features, labels = next(iterator)
       features = pipeline.move_struct_to_device(features, self.device)
       labels = pipeline.move_struct_to_device(labels, self.device)
+      if preprocess_fn is not None:
+        with self._autocast():
+          features, labels = preprocess_fn(features, labels)
+      self._ensure_fast_engine(features, labels)
       for hook in self.hooks:
         if self.is_chief or hook.every_rank:
           hook.before_step(self.context)
-      self.optimizer.zero_grad(set_to_none=True)
-      with self._autocast():
-        ops = self.model.model_fn(features, labels, TRAIN)
-      loss = ops.loss
-      if self._dp_engine is not None:
-        self._dp_engine.backward(loss)
-      else:
-        loss.backward()
-      self.optimizer.step(self.global_step)
-      if self.ema is not None:
-        self.ema.update()
+      loss = self._fast_engine.step(features, labels, self.global_step)
+      ops = self._captured_ops
       self.global_step += 1
-      last_loss = float(loss.detach().float().cpu())
+      last_loss_t = loss  # host sync deferred: .item() forces a GPU wait
       if self.is_chief and self.summary_writer is not None and \
-          ops.scalar_summaries:
+          ops is not None and ops.scalar_summaries:
         self.summary_writer.add_scalars(ops.scalar_summaries,
                                         self.global_step)
       for hook in self.hooks:
@@ -191,8 +232,12 @@ class Trainer:
       if self.is_chief and self.global_step % self._log_every == 0:
         now = time.time()
         _log.info("step=%d loss=%.5f steps/s=%.2f", self.global_step,
-                  last_loss, self._log_every / max(now - last_log, 1e-9))
+                  float(last_loss_t.detach().float().cpu()),
+                  self._log_every / max(now - last_log, 1e-9))
         last_log = now
+    last_loss = float(last_loss_t.detach().float().cpu()) \
+        if isinstance(last_loss_t, torch.Tensor) else float("nan")
+    del last_loss_t
     if self.checkpointer is not None:
       self.save_checkpoint()
     for hook in self.hooks:
@@ -204,7 +249,12 @@ class Trainer:
 
   # -- eval ----------------------------------------------------------------
   def evaluate(self, input_fn, eval_steps: int,
-               eval_name: str = "") -> Dict[str, float]:
+               eval_name: str = "",
+               distributed_eval: bool = False) -> Dict[str, float]:
+    """Eval loop.  `distributed_eval=True` when EVERY rank runs this
+    with a sharded input_fn (eval-only mode): metrics are then reduced
+    cross-rank (sum totals / sum counts).  Leave False for chief-only
+    eval — the all_reduce would hang ranks that never enter."""
     self._ensure_built()
     self.network.eval()
     if self.ema is not None:
@@ -235,6 +285,16 @@ class Trainer:
       if self.ema is not None:
         self.ema.swap_out()
       self.network.train()
+    if distributed_eval and self.world_size > 1:
+      # Cross-rank metric reduction: each rank evaluated a shard, so
+      # the job-level metric is sum(totals)/sum(counts) over ranks
+      # (reference semantics: one evaluator sees all data).
+      keys = sorted(totals)
+      vec = torch.tensor([totals[k] for k in keys] + [float(count)],
+                         dtype=torch.float64)
+      torch.distributed.all_reduce(vec)
+      totals = {k: float(vec[i]) for i, k in enumerate(keys)}
+      count = int(vec[-1])
     metrics = {k: v / max(count, 1) for k, v in totals.items()}
     metrics["global_step"] = self.global_step
     if self.summary_writer is not None:
@@ -406,13 +466,14 @@ def train_eval_model(t2r_model=None,
 
   exporters = create_exporters_fn() if create_exporters_fn else []
 
-  def run_eval(eval_trainer) -> Dict[str, float]:
+  def run_eval(eval_trainer, distributed_eval=False) -> Dict[str, float]:
     if input_generator_eval is None:
       return {}
     input_generator_eval.set_specification_from_model(t2r_model, EVAL)
     eval_input_fn = input_generator_eval.create_dataset_input_fn(EVAL)
     result = eval_trainer.evaluate(eval_input_fn, eval_steps,
-                                   eval_name=multi_eval_name or "")
+                                   eval_name=multi_eval_name or "",
+                                   distributed_eval=distributed_eval)
     export_root = os.path.join(model_dir, "export")
     for exporter in exporters:
       exporter.export(eval_trainer, result, export_root)
@@ -422,8 +483,10 @@ def train_eval_model(t2r_model=None,
   if input_generator_train is None:
     if input_generator_eval is None:
       raise ValueError("Need at least one input generator")
+    # Eval-only: every rank participates, data is rank-sharded ->
+    # reduce metrics cross-rank.
     if not use_continuous_eval:
-      return run_eval(trainer)
+      return run_eval(trainer, distributed_eval=world_size > 1)
     last_seen = None
     result: Dict[str, float] = {}
     while True:
@@ -435,19 +498,26 @@ def train_eval_model(t2r_model=None,
       trainer._ensure_built()
       trainer.global_step = checkpointing.Checkpointer.restore(
           ckpt, trainer.network, trainer.optimizer, trainer.ema)
-      result = run_eval(trainer)
+      result = run_eval(trainer, distributed_eval=world_size > 1)
       if trainer.global_step >= max_train_steps:
         break
     return result
 
   # --- train (+eval) mode ---
   input_generator_train.set_specification_from_model(t2r_model, TRAIN)
+  device_preprocess_fn = None
+  if trainer.device.type == "cuda":
+    # Deferred preprocessing: the pipeline yields RAW parsed batches
+    # (uint8 crosses the bus at 1/4 the f32 bytes), and the Trainer
+    # runs the preprocessor on the GPU right after H2D transfer.
+    device_preprocess_fn = input_generator_train.defer_preprocessing()
   train_input_fn = input_generator_train.create_dataset_input_fn(
       TRAIN, pin_memory=trainer.device.type == "cuda")
 
   if input_generator_eval is None or not trainer.is_chief:
     result = trainer.train(train_input_fn, max_train_steps,
-                           save_checkpoint_steps)
+                           save_checkpoint_steps,
+                           preprocess_fn=device_preprocess_fn)
     return result
 
   # Interleaved train/eval on the chief (train_and_evaluate semantics).
@@ -455,7 +525,8 @@ def train_eval_model(t2r_model=None,
   result: Dict[str, float] = {}
   while trainer.global_step < max_train_steps:
     target = min(trainer.global_step + eval_interval, max_train_steps)
-    result = trainer.train(train_input_fn, target, save_checkpoint_steps)
+    result = trainer.train(train_input_fn, target, save_checkpoint_steps,
+                           preprocess_fn=device_preprocess_fn)
     eval_result = run_eval(trainer)
     result.update({f"eval_{k}": v for k, v in eval_result.items()})
     if eval_throttle_secs:
