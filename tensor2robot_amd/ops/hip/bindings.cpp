@@ -35,6 +35,8 @@ at::Tensor conv_s1_wrw2(at::Tensor x, at::Tensor dy, int64_t R, int64_t S,
                         int64_t pad);
 at::Tensor conv_s1_wrw3(at::Tensor x, at::Tensor dy, int64_t R, int64_t S,
                         int64_t pad);
+at::Tensor conv_s1_wrw4(at::Tensor x, at::Tensor dy, int64_t R, int64_t S,
+                        int64_t pad);
 
 at::Tensor conv_stem_nhwc(at::Tensor x, at::Tensor wpk);
 
@@ -69,6 +71,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fwd + dgrad weight packs in one dispatch");
   m.def("s2d_stem", &s2d_stem,
         "fused space-to-depth [N,3,H,W]->[N,16,H/2,W/2] for the stem");
+  m.def("conv_s1_wrw4", &conv_s1_wrw4,
+        "MFMA wrw v4: pixel-major x image + ds_read_b64_tr_b16 A-reads");
   m.def("conv_s1_wrw3", &conv_s1_wrw3,
         "MFMA wrw v3: occupancy-first rs-split (256-thr WGs, 3/CU)");
   m.def("conv_s1_wrw2", &conv_s1_wrw2,

@@ -312,3 +312,34 @@ def test_spatial_softmax_large_map_fallback():
   assert points.shape == (2, 16) and smap.shape == (2, 8, 80, 80)
   ref_pts, _ = mod(x.float())
   assert (points.float() - ref_pts).abs().max().item() < 2e-2
+
+
+@requires_gpu
+@pytest.mark.parametrize("shape", [
+    (2, 64, 37, 37, 64, 3, 1),
+    (2, 64, 78, 78, 64, 5, 2),
+    (1, 64, 33, 29, 64, 3, 1),     # non-square spatial + odd edges
+    (3, 64, 16, 16, 64, 5, 2),     # window == tile edge case
+])
+def test_mfma_wrw4_matches_torch(shape):
+  """v4 numerics: the G17 tr_b16 trap (base = 2/4/6 mod 8 shorts reads
+  the 8-aligned address's data with NO stall) makes this the first
+  gate on any layout change — run before timing anything."""
+  import torch.nn.functional as F
+  from tensor2robot_amd.ops import _t2r_hip
+  n, c, h, w, k, r, pad = shape
+  torch.manual_seed(0)
+  x = torch.randn(n, c, h, w, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last)
+  oh, ow = h + 2 * pad - r + 1, w + 2 * pad - r + 1
+  dy = torch.randn(n, k, oh, ow, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last)
+  dw = _t2r_hip.conv_s1_wrw4(x, dy, r, r, pad)
+  dw_t = dw.reshape(r, r, c, k).permute(3, 2, 0, 1).contiguous()
+  x32 = x.float()
+  w32 = torch.zeros(k, c, r, r, device="cuda", requires_grad=True)
+  F.conv2d(x32, w32, padding=pad).backward(dy.float())
+  ref = w32.grad
+  err = (dw_t - ref).abs().max().item()
+  scale = ref.abs().max().item()
+  assert err < 0.01 * max(scale, 1.0), (shape, err, scale)

@@ -46,14 +46,18 @@ def main():
     def v3():
       return _t2r_hip.conv_s1_wrw3(x, dy, r, r, pad)
 
-    for fn in (miopen, v2, v3):      # warmup + find
+    def v4():
+      return _t2r_hip.conv_s1_wrw4(x, dy, r, r, pad)
+
+    for fn in (miopen, v2, v3, v4):      # warmup + find
       for _ in range(5):
         fn()
     torch.cuda.synchronize()
-    times = {"miopen": 0.0, "v2": 0.0, "v3": 0.0}
+    times = {"miopen": 0.0, "v2": 0.0, "v3": 0.0, "v4": 0.0}
     iters = 50
     for _ in range(iters):           # interleaved
-      for name, fn in (("miopen", miopen), ("v2", v2), ("v3", v3)):
+      for name, fn in (("miopen", miopen), ("v2", v2), ("v3", v3),
+                       ("v4", v4)):
         torch.cuda.synchronize()
         t0 = time.perf_counter()
         fn()
@@ -64,7 +68,9 @@ def main():
       ms = times[name] / iters * 1000
       print(f"  {n}x{c}x{h}x{w} r={r}: {name:6s} {ms:7.3f} ms "
             f"{flops / (times[name] / iters) / 1e12:7.1f} TF")
-    print(f"  ratio v2={times['miopen'] / times['v2']:.2f}x v3={times['miopen'] / times['v3']:.2f}x vs miopen")
+    print(f"  ratio v2={times['miopen'] / times['v2']:.2f}x "
+          f"v3={times['miopen'] / times['v3']:.2f}x "
+          f"v4={times['miopen'] / times['v4']:.2f}x vs miopen")
 
 
 if __name__ == "__main__":
