@@ -180,8 +180,12 @@ class AbstractT2RModel(model_interface.ModelInterface):
     return dict(inference_outputs)
 
   def scalar_summary(self, name: str, value):
+    # Keep tensors AS tensors: a .cpu() here is a device sync, which is
+    # illegal inside hipGraph capture (the Trainer's graphed step runs
+    # model_fn under capture) and a per-step stall otherwise.  The
+    # Trainer materializes floats only at summary-write time.
     if isinstance(value, torch.Tensor):
-      value = float(value.detach().float().cpu())
+      value = value.detach()
     self._scalar_summaries[name] = value
 
   def pop_scalar_summaries(self) -> Dict[str, float]:

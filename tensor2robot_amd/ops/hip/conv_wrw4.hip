@@ -9,24 +9,33 @@
 // reads (8 instructions per fragment, conflicted).  v4 removes the
 // transpose entirely:
 //
-//  - x halo is staged PIXEL-major in packed [ph/4][c/16][4][16]
-//    subtiles (one 16-B vector write per 8-channel chunk, no scatter).
+//  - The rs grid dimension is aligned to WEIGHT ROWS: blockIdx.y = r,
+//    the in-wave accumulator group walks s = 0..S-1.  With r fixed,
+//    output-tile row kstep only ever contracts against x row
+//    (kstep + r), so the WG stages just the 8 x rows it needs
+//    (8 x 20 pixels = 20.5 KiB) instead of the full halo.
+//  - x is staged PIXEL-major in packed [ph/4][c/16][4][16] subtiles
+//    (one 16-B vector write per 8-channel chunk, no transpose
+//    scatter).  The row pitch is padded to 20 pixels for both kernel
+//    sizes, which makes the packed pixel index of a contraction slice
+//    pl = kstep*20 + s + lane_pixel: 20 = 0 (mod 4), so the LDS byte
+//    address decomposes as  addr_s[lane]  +  kstep*2560  — five
+//    precomputed per-lane address registers and a compile-time
+//    offset immediate; the MFMA inner loop has ZERO address VALU.
 //  - Each A fragment is TWO ds_read_b64_tr_b16: per 16-lane group the
 //    instruction reads a [4 pixel][16 channel] row-major block and
-//    delivers lane (l&15) its channel-column of 4 pixels; pixels +4..7
-//    come from the next subtile block, exactly +512 B, so the second
-//    read is the same address with offset:512.  Every lane address is
-//    8-B aligned by construction (the G17 tr_b16 trap: a base = 2/4/6
-//    mod 8 shorts silently returns the 8-aligned address's data).
-//  - The block layout makes the 16 lane addresses of a group cover 32
-//    distinct banks for EVERY (r,s) shift (pixel row = 32 B, block =
-//    128 B: bank = 32*(sc&1) + (ph&3)*8 + piece*2, all distinct), i.e.
-//    one of the guide's conflict-free tr subtilings.
+//    delivers lane (l&15) its channel-column of 4 pixels; pixels
+//    +4..7 live exactly +512 B away (next subtile block), so the
+//    second read is the same address register with offset +512.
+//    Every lane address is 8-B aligned by construction (the G17
+//    tr_b16 trap: a base = 2/4/6 mod 8 shorts silently returns the
+//    8-aligned address's data).  The 16 lane addresses of a group
+//    cover 32 distinct banks for EVERY s shift (pixel row = 32 B,
+//    block = 128 B), i.e. a conflict-free tr subtiling.
 //
 // dy stays in the v3 transposed image read with aligned b128 (its
-// fragments are always 16-B aligned).  Grid/flush structure follows v3
-// (256-thread WGs = 4 (c,k)-quadrant waves, blockIdx.y = rs group of
-// <=5 offsets, non-atomic per-WG partials + reduce kernel).
+// fragments are always 16-B aligned).  Flush follows v3 (non-atomic
+// per-WG partials + reduce kernel).
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
@@ -39,30 +48,31 @@ typedef __attribute__((ext_vector_type(4))) short w4short4;
 typedef __attribute__((ext_vector_type(16))) float w4f32x16;
 
 #define W4_TILE_W 16
-#define W4_GROUP 5
+#define W4_ROW_W 20            // padded x row pitch (pixels), % 4 == 0
 
+// 3x3 (48 acc VGPRs) holds 4 waves/SIMD spill-free; 5x5 (80 acc) needs
+// the 3-wave VGPR budget to stay spill-free — 10 inner-loop spills at
+// the 4-wave cap cost more than the lost wave.
 template <int R, int S>
-__global__ void __launch_bounds__(256, 3)
+__global__ void __launch_bounds__(256, (R * S > 9) ? 3 : 4)
 conv_s1_wrw4_kernel(const w4bf16_t* __restrict__ x,
                     const w4bf16_t* __restrict__ dy,
-                    float* __restrict__ dw_part,  // [gx][gy][G][C][K]
+                    float* __restrict__ dw_part,  // [gx][r][S][C][K]
                     int N, int H, int W, int pad,
                     int OH, int OW, int tiles_h, int tiles_w,
                     int window_groups) {
   constexpr int C = 64, K = 64;
-  constexpr int RS = R * S;
   constexpr int TH = 8;
   constexpr int WIN_P = TH * W4_TILE_W;          // 128
-  constexpr int HALO_H = TH + R - 1;             // 10 / 12
-  constexpr int HALO_W = W4_TILE_W + S - 1;      // 18 / 20
-  constexpr int HALO_P = HALO_H * HALO_W;        // 180 / 240 (%4 == 0)
-  static_assert(HALO_P % 4 == 0, "halo pixel count must be 4-aligned");
+  constexpr int HALO_W = W4_TILE_W + S - 1;      // cols actually used
+  constexpr int XROWS = TH;                      // one x row per kstep
+  constexpr int X_P = XROWS * W4_ROW_W;          // 160 pixels
   constexpr int DYT_P = WIN_P + 8;
   // x image: element (ph, c) lives at
   //   ((ph>>2)*4 + (c>>4))*64 + (ph&3)*16 + (c&15).
   // 16-B aligned base: tr_b16 lane addresses derive from it and a base
   // = 2/4/6 mod 8 shorts silently reads wrong data (guide G17).
-  __shared__ __attribute__((aligned(16))) short lds_x[HALO_P * C];
+  __shared__ __attribute__((aligned(16))) short lds_x[X_P * C];
   __shared__ __attribute__((aligned(16))) short lds_dyt[K * DYT_P];
 
   const int tid = threadIdx.x;
@@ -70,23 +80,34 @@ conv_s1_wrw4_kernel(const w4bf16_t* __restrict__ x,
   const int lane = tid & 63;
   const int mtile = wave & 1;              // c half
   const int ntile = wave >> 1;             // k half
-  const int rs_base = blockIdx.y * W4_GROUP;
+  const int r_fixed = blockIdx.y;          // weight row this WG owns
 
-  w4f32x16 acc[W4_GROUP];
+  w4f32x16 acc[S];
 #pragma unroll
-  for (int g = 0; g < W4_GROUP; ++g) acc[g] = w4f32x16{};
+  for (int g = 0; g < S; ++g) acc[g] = w4f32x16{};
 
   const int total_windows = (int)((long)N * tiles_h * tiles_w);
   const int mrow = lane & 31;
   const int kgrp = lane >> 5;
 
-  // tr_b16 lane address pieces (constant over the window loop): lane
-  // l of 16-lane group grp covers pixel ph_base + (grp>>1)*8 +
-  // ((l&15)>>2) at channel piece mtile*32 + (grp&1)*16 + ((l&15)&3)*4.
+  // tr_b16 per-lane addresses, one per s shift — constant across the
+  // window loop (the LDS image location is fixed).  Lane l of 16-lane
+  // group grp covers pixel pl = kstep*20 + s + (grp>>1)*8 + ((l&15)>>2)
+  // at channel piece mtile*32 + (grp&1)*16 + ((l&15)&3)*4; the kstep
+  // part is the offset immediate.
   const int grp = lane >> 4;
   const int m16 = lane & 15;
-  const int pix_lane = (grp >> 1) * 8 + (m16 >> 2);   // pixel offset
+  const int pix_lane = (grp >> 1) * 8 + (m16 >> 2);
   const int cpiece = mtile * 32 + (grp & 1) * 16 + (m16 & 3) * 4;
+  unsigned addr_s[S];
+#pragma unroll
+  for (int s = 0; s < S; ++s) {
+    const int pl = s + pix_lane;
+    const int el = ((pl >> 2) * 4 + (cpiece >> 4)) * 64 +
+                   (pl & 3) * 16 + (cpiece & 15);
+    addr_s[s] = (unsigned)(unsigned long long)
+        (__attribute__((address_space(3))) short*)&lds_x[el];
+  }
 
   for (int win = blockIdx.x; win < total_windows; win += window_groups) {
     const int img = win / (tiles_h * tiles_w);
@@ -94,17 +115,12 @@ conv_s1_wrw4_kernel(const w4bf16_t* __restrict__ x,
     const int oh0 = (trest / tiles_w) * TH;
     const int ow0 = (trest % tiles_w) * W4_TILE_W;
 
-    // ---- stage x halo pixel-major (two-phase: all guarded loads
-    // issued first, then the write pass drains them).  Lane->(ph,
+    // ---- stage the 8 x rows this r needs, pixel-major.  Lane->(ph,
     // chunk) mapping staggers (ph&3, chunk&1) across each 8-lane
     // write-service group so the b128 writes are bank-conflict-free.
     {
-      constexpr int ITEMS = HALO_P * 8;      // 8-channel chunks
-      constexpr int XITER = (ITEMS + 255) / 256;
-      // Blocks of 4 load-then-write iterations: full two-phase staging
-      // held 48+ loads in registers concurrently and spilled at the
-      // 3-waves/SIMD VGPR budget; 4-deep blocks keep the overlapped-
-      // latency benefit at 16 VGPRs of staging registers.
+      constexpr int ITEMS = X_P * 8;         // 8-channel chunks
+      constexpr int XITER = (ITEMS + 255) / 256;   // 5
       constexpr int XBLK = XITER > 4 ? 4 : XITER;
 #pragma unroll
       for (int t0 = 0; t0 < XITER; t0 += XBLK) {
@@ -117,9 +133,11 @@ conv_s1_wrw4_kernel(const w4bf16_t* __restrict__ x,
           const int ph = ((rest >> 2) << 2) | ((i >> 1) & 3);
           vx[u] = w4bf16x8{};
           if (i < ITEMS) {
-            const int iy = oh0 - pad + ph / HALO_W;
-            const int ix = ow0 - pad + ph % HALO_W;
-            if (iy >= 0 && iy < H && ix >= 0 && ix < W) {
+            const int hcol = ph % W4_ROW_W;
+            const int iy = oh0 - pad + r_fixed + ph / W4_ROW_W;
+            const int ix = ow0 - pad + hcol;
+            if (hcol < HALO_W && iy >= 0 && iy < H &&
+                ix >= 0 && ix < W) {
               vx[u] = *reinterpret_cast<const w4bf16x8*>(
                   x + (((long)img * H + iy) * W + ix) * C + chunk * 8);
             }
@@ -178,35 +196,32 @@ conv_s1_wrw4_kernel(const w4bf16_t* __restrict__ x,
 
     {
       const int k = ntile * 32 + mrow;
-#pragma unroll 2
+#pragma unroll
       for (int kstep = 0; kstep < WIN_P / 16; ++kstep) {
         w4bf16x8 b_frag = *reinterpret_cast<const w4bf16x8*>(
             &lds_dyt[k * DYT_P + kstep * 16 + kgrp * 8]);
 #pragma unroll
-        for (int g = 0; g < W4_GROUP; ++g) {
-          const int rs = rs_base + g;
-          if (rs < RS) {
-            const int r = rs / S, s = rs % S;
-            // Contraction pixels for (kstep, r, s) are the 16
-            // CONSECUTIVE packed halo pixels starting at
-            // (kstep+r)*HALO_W + s; this lane reads its 8 via two
-            // tr_b16 (pixels +0..3 and +4..7 = +512 B).
-            const int pl = (kstep + r) * HALO_W + s + pix_lane;
-            const int el = ((pl >> 2) * 4 + (cpiece >> 4)) * 64 +
-                           (pl & 3) * 16 + (cpiece & 15);
-            unsigned addr = (unsigned)(unsigned long long)
-                (__attribute__((address_space(3))) short*)&lds_x[el];
-            w4short4 lo, hi;
-            asm volatile(
-                "ds_read_b64_tr_b16 %0, %2\n\t"
-                "ds_read_b64_tr_b16 %1, %2 offset:512\n\t"
-                "s_waitcnt lgkmcnt(0)"
-                : "=&v"(lo), "=&v"(hi) : "v"(addr));
-            const w4bf16x8 a_frag = __builtin_shufflevector(
-                lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
-            acc[g] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                a_frag, b_frag, acc[g], 0, 0, 0);
+        for (int g = 0; g < S; ++g) {
+          // kstep lives in the compile-time offset immediate:
+          // +kstep*20 pixels = +kstep*5 blocks = +kstep*2560 bytes.
+          w4short4 lo, hi;
+          switch (kstep) {
+#define W4_CASE(KS) \
+            case KS: \
+              asm volatile( \
+                  "ds_read_b64_tr_b16 %0, %2 offset:" #KS "*2560\n\t" \
+                  "ds_read_b64_tr_b16 %1, %2 offset:" #KS "*2560+512\n\t" \
+                  "s_waitcnt lgkmcnt(0)" \
+                  : "=&v"(lo), "=&v"(hi) : "v"(addr_s[g])); \
+              break;
+            W4_CASE(0) W4_CASE(1) W4_CASE(2) W4_CASE(3)
+            W4_CASE(4) W4_CASE(5) W4_CASE(6) W4_CASE(7)
+#undef W4_CASE
           }
+          const w4bf16x8 a_frag = __builtin_shufflevector(
+              lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+          acc[g] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a_frag, b_frag, acc[g], 0, 0, 0);
         }
       }
     }
@@ -214,34 +229,33 @@ conv_s1_wrw4_kernel(const w4bf16_t* __restrict__ x,
   }
 
   float* part = dw_part
-      + ((long)blockIdx.x * gridDim.y + blockIdx.y) * W4_GROUP * C * K;
+      + ((long)blockIdx.x * gridDim.y + blockIdx.y) * S * C * K;
 #pragma unroll
-  for (int g = 0; g < W4_GROUP; ++g) {
-    if (rs_base + g < RS) {
+  for (int g = 0; g < S; ++g) {
 #pragma unroll
-      for (int reg = 0; reg < 16; ++reg) {
-        const int c = mtile * 32
-            + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
-        const int k = ntile * 32 + (lane & 31);
-        part[((long)g * C + c) * K + k] = acc[g][reg];
-      }
+    for (int reg = 0; reg < 16; ++reg) {
+      const int c = mtile * 32
+          + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+      const int k = ntile * 32 + (lane & 31);
+      part[((long)g * C + c) * K + k] = acc[g][reg];
     }
   }
 }
 
+// Reduce: dw[rs][c][k] = sum over gx of part[gx][rs/S][rs%S][c][k].
 __global__ void __launch_bounds__(256)
 wrw4_reduce_kernel(const float* __restrict__ part,
                    float* __restrict__ dw, int RS, long ck,
-                   int ngx, int ngy) {
+                   int ngx, int ngy, int group) {
   const long cells = (long)RS * ck;
   for (long i = blockIdx.x * 256L + threadIdx.x; i < cells;
        i += (long)gridDim.x * 256) {
     const int rs = (int)(i / ck);
     const long rest = i % ck;
-    const int gy = rs / W4_GROUP, g = rs % W4_GROUP;
+    const int gy = rs / group, g = rs % group;
     float s = 0.0f;
     for (int gx = 0; gx < ngx; ++gx) {
-      s += part[(((long)gx * ngy + gy) * W4_GROUP + g) * ck + rest];
+      s += part[(((long)gx * ngy + gy) * group + g) * ck + rest];
     }
     dw[i] = s;
   }
@@ -263,9 +277,12 @@ at::Tensor conv_s1_wrw4(at::Tensor x, at::Tensor dy, int64_t R,
   const int tiles_h = (OH + 7) / 8;
   const int tiles_w = (OW + W4_TILE_W - 1) / W4_TILE_W;
   const int total_windows = (int)((long)N * tiles_h * tiles_w);
-  const int ngy = (int)((R * S + W4_GROUP - 1) / W4_GROUP);
-  const int window_groups = std::min(total_windows, 768 / ngy);
-  auto part = at::empty({(long)window_groups * ngy, W4_GROUP,
+  const int ngy = (int)R;
+  // 4 WGs/CU resident -> 1024 concurrent WGs; split windows across
+  // gx so the grid fills the chip without inflating the reduce.
+  const int window_groups = std::min(total_windows,
+                                     std::max(1, 1024 / ngy));
+  auto part = at::empty({(long)window_groups * ngy, S,
                          (long)C, K}, x.options().dtype(at::kFloat));
   auto dw = at::empty({(long)R * S, C, K},
                       x.options().dtype(at::kFloat));
@@ -287,6 +304,6 @@ at::Tensor conv_s1_wrw4(at::Tensor x, at::Tensor dy, int64_t R,
   hipLaunchKernelGGL(wrw4_reduce_kernel, dim3(rblocks), dim3(256), 0,
                      stream.stream(), (const float*)part.data_ptr(),
                      (float*)dw.data_ptr(), (int)(R * S), (long)C * K,
-                     window_groups, ngy);
+                     window_groups, ngy, (int)S);
   return dw;
 }

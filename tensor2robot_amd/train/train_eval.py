@@ -220,9 +220,15 @@ class Trainer:
       self.global_step += 1
       last_loss_t = loss  # host sync deferred: .item() forces a GPU wait
       if self.is_chief and self.summary_writer is not None and \
-          ops is not None and ops.scalar_summaries:
-        self.summary_writer.add_scalars(ops.scalar_summaries,
-                                        self.global_step)
+          ops is not None and ops.scalar_summaries and \
+          self.global_step % self._log_every == 0:
+        # Materialize tensor summaries here (one sync per log interval,
+        # not per step; under graphs these are the live static tensors).
+        self.summary_writer.add_scalars(
+            {k: (float(v.detach().float().cpu())
+                 if isinstance(v, torch.Tensor) else v)
+             for k, v in ops.scalar_summaries.items()},
+            self.global_step)
       for hook in self.hooks:
         if self.is_chief or hook.every_rank:
           hook.after_step(self.context, ops)
