@@ -278,10 +278,13 @@ at::Tensor conv_s1_wrw4(at::Tensor x, at::Tensor dy, int64_t R,
   const int tiles_w = (OW + W4_TILE_W - 1) / W4_TILE_W;
   const int total_windows = (int)((long)N * tiles_h * tiles_w);
   const int ngy = (int)R;
-  // 4 WGs/CU resident -> 1024 concurrent WGs; split windows across
-  // gx so the grid fills the chip without inflating the reduce.
+  // Fill EXACTLY one residency wave: 3 WGs/CU for the 5x5 (VGPR
+  // budget), 4 for the 3x3 -> 768/1024 concurrent WGs.  A partial
+  // second wave of straggler WGs costs ~18% (measured 0.137 -> 0.162
+  // ms when the grid was 1020 WGs at 768 slots).
+  const int resident = (R * S > 9) ? 768 : 1024;
   const int window_groups = std::min(total_windows,
-                                     std::max(1, 1024 / ngy));
+                                     std::max(1, resident / ngy));
   auto part = at::empty({(long)window_groups * ngy, S,
                          (long)C, K}, x.options().dtype(at::kFloat));
   auto dw = at::empty({(long)R * S, C, K},

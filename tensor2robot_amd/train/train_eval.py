@@ -170,12 +170,22 @@ class Trainer:
       return
     from tensor2robot_amd.parallel import fast_step
 
+    def _detach(v):
+      if isinstance(v, torch.Tensor):
+        return v.detach()
+      if isinstance(v, dict):
+        return {k: _detach(x) for k, x in v.items()}
+      return v
+
     def loss_fn(f, l):
       ops = self.model.model_fn(f, l, TRAIN)
-      # The ops object's tensors ARE the graph-static outputs: they
-      # refresh in place on every replay, so hooks and summaries keep
-      # seeing live values.
-      self._captured_ops = ops
+      # Stash a DETACHED copy: the tensors are the graph-static outputs
+      # (refresh in place on every replay, so hooks/summaries see live
+      # values), but retaining the autograd graph here keeps default-
+      # stream AccumulateGrad nodes alive into the side-stream capture
+      # warmup — a stream mismatch that segfaults hipGraph capture_end.
+      self._captured_ops = ops.__class__(
+          *[_detach(getattr(ops, fld)) for fld in ops._fields])
       return ops.loss
 
     lr_schedule = getattr(self.optimizer, "lr_schedule", None)
@@ -218,7 +228,9 @@ class Trainer:
       loss = self._fast_engine.step(features, labels, self.global_step)
       ops = self._captured_ops
       self.global_step += 1
-      last_loss_t = loss  # host sync deferred: .item() forces a GPU wait
+      # Detached + host sync deferred (.item() forces a GPU wait, and a
+      # retained grad_fn would pin autograd nodes across a recapture).
+      last_loss_t = loss.detach()
       if self.is_chief and self.summary_writer is not None and \
           ops is not None and ops.scalar_summaries and \
           self.global_step % self._log_every == 0:
