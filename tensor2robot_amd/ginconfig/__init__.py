@@ -634,8 +634,23 @@ def parse_config_file(path):
   for base in _SEARCH_PATHS:
     candidate = os.path.join(base, path) if base else path
     if os.path.exists(candidate):
-      with open(candidate) as f:
-        parse_config(f.read())
+      # Includes resolve relative to the including file's directory
+      # (and the repo root two levels up covers the reference-style
+      # 'tensor2robot_amd/research/.../x.gin' include paths).
+      file_dir = os.path.dirname(os.path.abspath(candidate))
+      pushed = []
+      for extra in (file_dir,
+                    os.path.abspath(os.path.join(file_dir, "..", "..",
+                                                 "..", ".."))):
+        if extra not in _SEARCH_PATHS:
+          _SEARCH_PATHS.append(extra)
+          pushed.append(extra)
+      try:
+        with open(candidate) as f:
+          parse_config(f.read())
+      finally:
+        for extra in pushed:
+          _SEARCH_PATHS.remove(extra)
       return
   raise GinError(f"Config file not found: {path}")
 

@@ -62,7 +62,14 @@ class AbstractT2RModel(model_interface.ModelInterface):
                use_avg_model_params: bool = False,
                avg_model_params_decay: float = 0.9999,
                init_from_checkpoint_fn: Optional[Callable] = None,
+               use_sync_replicas_optimizer: bool = False,
                compute_dtype: str = "bfloat16"):
+    # use_sync_replicas_optimizer (reference abstract_model.py:201-203,
+    # 864-870): the reference's PS-architecture SyncReplicasOptimizer
+    # collapses into the always-synchronous RCCL all-reduce of the DP
+    # engine here — the flag is accepted for config parity and noted,
+    # but every multi-process run IS synchronous data-parallel.
+    self._use_sync_replicas_optimizer = use_sync_replicas_optimizer
     self._preprocessor_cls = preprocessor_cls
     self._create_optimizer_fn = create_optimizer_fn
     if device_type not in (DEVICE_TYPE_CPU, DEVICE_TYPE_GPU):

@@ -108,16 +108,21 @@ class _MFMAConvFunction(torch.autograd.Function):
         wpk_b = ext.pack_conv_w(weight, True)
       dx = ext.conv_s1_nhwc(dy.to(torch.bfloat16), wpk_b, c, r, s, bpad)
     if ctx.needs_input_grad[1]:
-      # OPT-IN: T2R_ENABLE_MFMA_WRW selects a hand wrw kernel
-      # (=1 v1 LDS-accumulator, =2 v2 register-accumulator, =3 v3
-      # rs-split, =4 v4 tr_b16 pixel-major); unset -> MIOpen.
+      # Default: the v4 tr_b16 kernel for the 5x5 (measured 1.05x
+      # MIOpen, profiles/); MIOpen for 3x3 (small shapes are launch/
+      # staging-bound, v4 0.5x there).  T2R_ENABLE_MFMA_WRW overrides:
+      # =1 v1 LDS-accumulator, =2 v2 register-accumulator, =3 v3
+      # rs-split, =4 v4 everywhere, =0/off -> MIOpen everywhere.
       wrw_mode = os.environ.get("T2R_ENABLE_MFMA_WRW", "")
       sq35 = c == 64 and k == 64 and r == s and r in (3, 5)
-      if wrw_mode == "4" and sq35:
-        dw_f32 = ext.conv_s1_wrw4(x, dy.to(torch.bfloat16), r, s,
-                                  ctx.pad)
-        dw = dw_f32.reshape(r, s, c, k).permute(3, 2, 0, 1) \
-            .contiguous().to(weight.dtype)
+      default_v4 = (wrw_mode == "" and c == 64 and k == 64 and
+                    r == 5 and s == 5)
+      if default_v4 or (wrw_mode == "4" and sq35):
+        # v4 emits [K,C,R,S] bf16 directly (permute+cast fused into
+        # its reduce kernel).
+        dw = ext.conv_s1_wrw4(x, dy.to(torch.bfloat16), r, s, ctx.pad)
+        if dw.dtype != weight.dtype:
+          dw = dw.to(weight.dtype)
       elif wrw_mode == "3" and sq35:
         dw_f32 = ext.conv_s1_wrw3(x, dy.to(torch.bfloat16), r, s,
                                   ctx.pad)

@@ -200,26 +200,60 @@ conv_s1_wrw4_kernel(const w4bf16_t* __restrict__ x,
       for (int kstep = 0; kstep < WIN_P / 16; ++kstep) {
         w4bf16x8 b_frag = *reinterpret_cast<const w4bf16x8*>(
             &lds_dyt[k * DYT_P + kstep * 16 + kgrp * 8]);
+        // All 2S tr reads of this kstep issue back-to-back in ONE asm
+        // block with a single waitcnt: LDS latency is paid once per
+        // kstep (reads pipeline through the LDS) instead of once per
+        // s.  kstep lives in the compile-time offset immediate:
+        // +kstep*20 pixels = +kstep*5 blocks = +kstep*2560 bytes.
+        w4short4 fr[2 * S];
+#define W4_CASE3(KS) \
+          case KS: \
+            asm volatile( \
+                "ds_read_b64_tr_b16 %0, %6 offset:" #KS "*2560\n\t" \
+                "ds_read_b64_tr_b16 %1, %6 offset:" #KS "*2560+512\n\t" \
+                "ds_read_b64_tr_b16 %2, %7 offset:" #KS "*2560\n\t" \
+                "ds_read_b64_tr_b16 %3, %7 offset:" #KS "*2560+512\n\t" \
+                "ds_read_b64_tr_b16 %4, %8 offset:" #KS "*2560\n\t" \
+                "ds_read_b64_tr_b16 %5, %8 offset:" #KS "*2560+512\n\t" \
+                "s_waitcnt lgkmcnt(0)" \
+                : "=&v"(fr[0]), "=&v"(fr[1]), "=&v"(fr[2]), \
+                  "=&v"(fr[3]), "=&v"(fr[4]), "=&v"(fr[5]) \
+                : "v"(addr_s[0]), "v"(addr_s[1]), "v"(addr_s[2])); \
+            break;
+#define W4_CASE5(KS) \
+          case KS: \
+            asm volatile( \
+                "ds_read_b64_tr_b16 %0, %10 offset:" #KS "*2560\n\t" \
+                "ds_read_b64_tr_b16 %1, %10 offset:" #KS "*2560+512\n\t" \
+                "ds_read_b64_tr_b16 %2, %11 offset:" #KS "*2560\n\t" \
+                "ds_read_b64_tr_b16 %3, %11 offset:" #KS "*2560+512\n\t" \
+                "ds_read_b64_tr_b16 %4, %12 offset:" #KS "*2560\n\t" \
+                "ds_read_b64_tr_b16 %5, %12 offset:" #KS "*2560+512\n\t" \
+                "ds_read_b64_tr_b16 %6, %13 offset:" #KS "*2560\n\t" \
+                "ds_read_b64_tr_b16 %7, %13 offset:" #KS "*2560+512\n\t" \
+                "ds_read_b64_tr_b16 %8, %14 offset:" #KS "*2560\n\t" \
+                "ds_read_b64_tr_b16 %9, %14 offset:" #KS "*2560+512\n\t" \
+                "s_waitcnt lgkmcnt(0)" \
+                : "=&v"(fr[0]), "=&v"(fr[1]), "=&v"(fr[2]), \
+                  "=&v"(fr[3]), "=&v"(fr[4]), "=&v"(fr[5]), \
+                  "=&v"(fr[6]), "=&v"(fr[7]), "=&v"(fr[8]), \
+                  "=&v"(fr[9]) \
+                : "v"(addr_s[0]), "v"(addr_s[1]), "v"(addr_s[2]), \
+                  "v"(addr_s[3]), "v"(addr_s[4])); \
+            break;
+#define W4_CASES(M) M(0) M(1) M(2) M(3) M(4) M(5) M(6) M(7)
+        if constexpr (S == 3) {
+          switch (kstep) { W4_CASES(W4_CASE3) }
+        } else {
+          switch (kstep) { W4_CASES(W4_CASE5) }
+        }
+#undef W4_CASES
+#undef W4_CASE3
+#undef W4_CASE5
 #pragma unroll
         for (int g = 0; g < S; ++g) {
-          // kstep lives in the compile-time offset immediate:
-          // +kstep*20 pixels = +kstep*5 blocks = +kstep*2560 bytes.
-          w4short4 lo, hi;
-          switch (kstep) {
-#define W4_CASE(KS) \
-            case KS: \
-              asm volatile( \
-                  "ds_read_b64_tr_b16 %0, %2 offset:" #KS "*2560\n\t" \
-                  "ds_read_b64_tr_b16 %1, %2 offset:" #KS "*2560+512\n\t" \
-                  "s_waitcnt lgkmcnt(0)" \
-                  : "=&v"(lo), "=&v"(hi) : "v"(addr_s[g])); \
-              break;
-            W4_CASE(0) W4_CASE(1) W4_CASE(2) W4_CASE(3)
-            W4_CASE(4) W4_CASE(5) W4_CASE(6) W4_CASE(7)
-#undef W4_CASE
-          }
           const w4bf16x8 a_frag = __builtin_shufflevector(
-              lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+              fr[2 * g], fr[2 * g + 1], 0, 1, 2, 3, 4, 5, 6, 7);
           acc[g] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               a_frag, b_frag, acc[g], 0, 0, 0);
         }
@@ -242,11 +276,16 @@ conv_s1_wrw4_kernel(const w4bf16_t* __restrict__ x,
   }
 }
 
-// Reduce: dw[rs][c][k] = sum over gx of part[gx][rs/S][rs%S][c][k].
+// Reduce: dw[k][c][r][s] (bf16, the autograd-facing [K,C,R,S] weight
+// gradient — layout permute and bf16 cast fused here instead of two
+// extra torch dispatches per conv) = sum over gx of
+// part[gx][rs/S][rs%S][c][k].  Loop order follows the partials (rs,
+// c, k fastest) so the ngx reads per cell stay coalesced; the single
+// output write scatters.
 __global__ void __launch_bounds__(256)
 wrw4_reduce_kernel(const float* __restrict__ part,
-                   float* __restrict__ dw, int RS, long ck,
-                   int ngx, int ngy, int group) {
+                   w4bf16_t* __restrict__ dw, int RS, long ck,
+                   int ngx, int ngy, int group, int C) {
   const long cells = (long)RS * ck;
   for (long i = blockIdx.x * 256L + threadIdx.x; i < cells;
        i += (long)gridDim.x * 256) {
@@ -257,7 +296,9 @@ wrw4_reduce_kernel(const float* __restrict__ part,
     for (int gx = 0; gx < ngx; ++gx) {
       s += part[(((long)gx * ngy + gy) * group + g) * ck + rest];
     }
-    dw[i] = s;
+    const int c = (int)(rest / (ck / C));
+    const int k = (int)(rest % (ck / C));
+    dw[((long)k * C + c) * RS + rs] = (w4bf16_t)s;
   }
 }
 
@@ -287,8 +328,8 @@ at::Tensor conv_s1_wrw4(at::Tensor x, at::Tensor dy, int64_t R,
                                      std::max(1, resident / ngy));
   auto part = at::empty({(long)window_groups * ngy, S,
                          (long)C, K}, x.options().dtype(at::kFloat));
-  auto dw = at::empty({(long)R * S, C, K},
-                      x.options().dtype(at::kFloat));
+  auto dw = at::empty({(long)K, C, R, S},
+                      x.options().dtype(at::kBFloat16));
   auto stream = at::cuda::getCurrentCUDAStream();
   auto launch = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(window_groups, ngy), dim3(256),
@@ -306,7 +347,7 @@ at::Tensor conv_s1_wrw4(at::Tensor x, at::Tensor dy, int64_t R,
   const int rblocks = (int)std::min((cells + 255) / 256, 1024L);
   hipLaunchKernelGGL(wrw4_reduce_kernel, dim3(rblocks), dim3(256), 0,
                      stream.stream(), (const float*)part.data_ptr(),
-                     (float*)dw.data_ptr(), (int)(R * S), (long)C * K,
-                     window_groups, ngy, (int)S);
+                     (w4bf16_t*)dw.data_ptr(), (int)(R * S),
+                     (long)C * K, window_groups, ngy, (int)S, (int)C);
   return dw;
 }

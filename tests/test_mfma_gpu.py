@@ -335,7 +335,7 @@ def test_mfma_wrw4_matches_torch(shape):
   dy = torch.randn(n, k, oh, ow, device="cuda").to(torch.bfloat16) \
       .contiguous(memory_format=torch.channels_last)
   dw = _t2r_hip.conv_s1_wrw4(x, dy, r, r, pad)
-  dw_t = dw.reshape(r, r, c, k).permute(3, 2, 0, 1).contiguous()
+  dw_t = dw.float()  # v4 emits [K,C,R,S] bf16 directly
   x32 = x.float()
   w32 = torch.zeros(k, c, r, r, device="cuda", requires_grad=True)
   F.conv2d(x32, w32, padding=pad).backward(dy.float())
