@@ -160,8 +160,11 @@ class AbstractExportGenerator:
                                    export_dir: str) -> str:
     """TFRecord of zero-filled request Examples for serving warmup.
 
-    Reference :109-142 (PredictionLog warmup records; here plain Examples
-    whose features are the zero-filled required inputs).
+    Reference :109-142 (PredictionLog warmup records).  Here a warmup
+    request of batch size B is B consecutive single-example records —
+    one Example per sample, exactly the wire format
+    `predict_serialized` consumes — so a server warms up by replaying
+    each group as one batched predict call.
     """
     from tensor2robot_amd.data import example as example_codec
     from tensor2robot_amd.data import tfrecord
@@ -171,9 +174,10 @@ class AbstractExportGenerator:
     with tfrecord.TFRecordWriter(path) as writer:
       for bs in batch_sizes:
         feed = tsu.make_constant_numpy(input_spec, 0.0, batch_size=bs)
-        features = {}
-        for key, arr in feed.items():
-          name = input_spec[key].name or key
-          features[name] = arr
-        writer.write(example_codec.encode_example(features))
+        for i in range(bs):
+          features = {}
+          for key, arr in feed.items():
+            name = input_spec[key].name or key
+            features[name] = arr[i]
+          writer.write(example_codec.encode_example(features))
     return path

@@ -279,15 +279,71 @@ def test_warmup_requests_roundtrip(tmp_path):
   path = gen.create_warmup_requests_numpy([1, 4], str(tmp_path))
   assert os.path.basename(path) == "warmup_requests.tfrecord"
   records = list(tfrecord.read_records(path))
-  assert len(records) == 2
+  # Batch-B request = B consecutive single-example records.
+  assert len(records) == 1 + 4
   spec = gen.serving_input_spec()
-  for i, bs in enumerate((1, 4)):
-    feats = example_codec.decode_example(records[i])
-    # Every required serving input appears, zero-filled, batch bs.
+  for rec in records:
+    feats = example_codec.decode_example(rec)
+    # Every required serving input appears, zero-filled, per example.
     for key, sp in spec.items():
       name = sp.name or key
       assert name in feats, (name, sorted(feats))
-      arr = feats[name]
       import numpy as _np
-      assert _np.asarray(arr).size % bs == 0
-      assert float(_np.abs(_np.asarray(arr, dtype=_np.float64)).sum()) == 0.0
+      arr = _np.asarray(feats[name], dtype=_np.float64)
+      assert float(_np.abs(arr).sum()) == 0.0
+
+
+def test_tf_example_export_embeds_warmup_and_roundtrips(tmp_path):
+  """a tf_example-mode export carries its parse
+  contract (t2r_assets spec) + warmup requests INSIDE the artifact, and
+  predict_serialized replays those records end to end."""
+  from tensor2robot_amd.data import tfrecord
+  from tensor2robot_amd.export_generators import default_export_generator
+  from tensor2robot_amd.specs import tensorspec_utils as tsu
+
+  model = mocks.MockT2RModel(device_type="cpu")
+  _ = model.network
+  gen = default_export_generator.DefaultExportGenerator()
+  gen.set_specification_from_model(model)
+  export_dir = gen.export(model, str(tmp_path / "export"),
+                          global_step=7, receiver_mode="tf_example",
+                          warmup_batch_sizes=[1, 3])
+  warmup_path = os.path.join(export_dir, tsu.EXTRA_ASSETS_DIRECTORY,
+                             "warmup_requests.tfrecord")
+  assert os.path.exists(warmup_path), "warmup requests not in artifact"
+
+  predictor = esp.ExportedSavedModelPredictor(
+      str(tmp_path / "export"), timeout=5)
+  assert predictor.is_loaded
+  records = list(tfrecord.read_records(warmup_path))
+  assert len(records) == 1 + 3
+  groups = [records[:1], records[1:]]
+  for bs, group in zip((1, 3), groups):
+    out = predictor.predict_serialized(group)
+    for key, arr in out.items():
+      assert arr.shape[0] == bs, (key, arr.shape)
+
+
+def test_tf_example_receiver_matches_numpy_receiver(tmp_path):
+  """The two receiver families produce the same feed for the same data
+  (reference default_export_generator.py:42-133 parity)."""
+  from tensor2robot_amd.data import example as example_mod
+  from tensor2robot_amd.export_generators import default_export_generator
+
+  model = mocks.MockT2RModel(device_type="cpu")
+  gen = default_export_generator.DefaultExportGenerator()
+  gen.set_specification_from_model(model)
+  numpy_receiver = gen.create_serving_input_receiver_numpy_fn()
+  example_receiver = gen.create_serving_input_receiver_tf_example_fn()
+
+  arr = np.array([[0.5, -1.0, 2.0]], np.float32)
+  spec = gen.serving_input_spec()
+  key = list(spec.keys())[0]
+  name = spec[key].name or key
+  feed_np = numpy_receiver({key: arr})
+  record = example_mod.encode_example({name: arr[0]})
+  feed_ex = example_receiver([record])
+  assert set(feed_np) == set(feed_ex)
+  for k in feed_np:
+    np.testing.assert_allclose(feed_np[k].numpy(),
+                               np.asarray(feed_ex[k]), rtol=1e-6)
