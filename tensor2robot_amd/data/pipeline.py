@@ -127,6 +127,76 @@ class WeightedRecordBatchIterator:
       yield {"": batch}
 
 
+class _PinSlot:
+  """One ring slot of reusable pinned host buffers.
+
+  `event` is recorded by move_struct_to_device on the consumer's
+  compute stream AFTER it enqueues the slot's H2D copies; the producer
+  waits on it before host-writing the slot again, so an in-flight DMA
+  never races a refill.
+  """
+
+  def __init__(self):
+    self.buffers: Dict[str, torch.Tensor] = {}
+    self.event = None
+
+  def mark_consumed(self):
+    if torch.cuda.is_available() and torch.cuda.is_initialized():
+      if self.event is None:
+        self.event = torch.cuda.Event()
+      self.event.record()
+
+  def wait_reusable(self):
+    if self.event is not None:
+      self.event.synchronize()
+
+
+_SLOT_ATTR = "_t2r_pin_slot"
+
+
+class _PinnedRing:
+  """Reusable pinned staging: copy each item into the next ring slot.
+
+  `x.pin_memory()` per batch is hipHostMalloc + memcpy EVERY item
+  (multi-ms for image batches); the ring allocates each pinned buffer
+  once and only pays the memcpy afterwards.  Falls back to per-item
+  pinning when an item's structure/shape changes (varlen edge cases).
+  """
+
+  def __init__(self, size: int):
+    self._slots = [_PinSlot() for _ in range(max(2, size))]
+    self._i = 0
+
+  def stage(self, item):
+    slot = self._slots[self._i % len(self._slots)]
+    self._i += 1
+    slot.wait_reusable()
+
+    def stage_tensor(path, x):
+      if not isinstance(x, torch.Tensor) or x.is_cuda:
+        return x
+      buf = slot.buffers.get(path)
+      if buf is None or buf.shape != x.shape or buf.dtype != x.dtype:
+        buf = torch.empty_like(x).pin_memory()
+        slot.buffers[path] = buf
+      buf.copy_(x)
+      return buf
+
+    def stage_struct(prefix, s):
+      if isinstance(s, tuple):
+        return tuple(stage_struct(f"{prefix}/{i}", x)
+                     for i, x in enumerate(s))
+      if isinstance(s, (tsu.TensorSpecStruct, dict)):
+        out = tsu.TensorSpecStruct()
+        for k, v in s.items():
+          out[k] = stage_tensor(f"{prefix}/{k}", v)
+        object.__setattr__(out, _SLOT_ATTR, slot)
+        return out
+      return stage_tensor(prefix, s)
+
+    return stage_struct("", item)
+
+
 class PrefetchIterator:
   """Background-thread prefetch with a bounded queue (prefetch(AUTOTUNE))."""
 
@@ -141,12 +211,13 @@ class PrefetchIterator:
   def __iter__(self):
     q: queue_mod.Queue = queue_mod.Queue(maxsize=self._depth)
     error = []
+    ring = _PinnedRing(self._depth + 3) if self._pin else None
 
     def worker():
       try:
         for item in self._source_fn():
-          if self._pin:
-            item = _pin_struct(item)
+          if ring is not None:
+            item = ring.stage(item)
           q.put(item)
       except BaseException as e:  # propagate to consumer
         error.append(e)
@@ -164,25 +235,12 @@ class PrefetchIterator:
       yield item
 
 
-def _pin_struct(item):
-  def pin(x):
-    if isinstance(x, torch.Tensor) and not x.is_cuda:
-      return x.pin_memory()
-    return x
-  if isinstance(item, tuple):
-    return tuple(_pin_struct(x) for x in item)
-  if isinstance(item, tsu.TensorSpecStruct):
-    out = tsu.TensorSpecStruct()
-    for k, v in item.items():
-      out[k] = pin(v)
-    return out
-  if isinstance(item, dict):
-    return {k: pin(v) for k, v in item.items()}
-  return pin(item)
-
-
 def move_struct_to_device(struct, device, non_blocking=True):
-  """Moves every tensor in a (features, labels) struct to device."""
+  """Moves every tensor in a (features, labels) struct to device.
+
+  When the struct came from the pinned ring, the consumer-side H2D
+  event is recorded here so the producer can safely refill the slot.
+  """
   if struct is None:
     return None
   if isinstance(struct, tuple):
@@ -193,6 +251,9 @@ def move_struct_to_device(struct, device, non_blocking=True):
     for k, v in (struct.items() if hasattr(struct, "items") else []):
       out[k] = v.to(device, non_blocking=non_blocking) \
           if isinstance(v, torch.Tensor) else v
+    slot = getattr(struct, _SLOT_ATTR, None)
+    if slot is not None and device.type == "cuda":
+      slot.mark_consumed()
     return out
   if isinstance(struct, torch.Tensor):
     return struct.to(device, non_blocking=non_blocking)

@@ -243,6 +243,10 @@ class MFMAConv2d(nn.Conv2d):
       if w.dtype != torch.bfloat16:
         w = w.to(torch.bfloat16)
       return _StemConvFunction.apply(x, w)
+    if x.is_cuda and os.environ.get("T2R_LOG_CONV_FALLBACK"):
+      print(f"# conv fallback: x={tuple(x.shape)} {x.dtype} "
+            f"req_grad={x.requires_grad} w={tuple(self.weight.shape)} "
+            f"stride={self.stride} pad={self.padding}", flush=True)
     if x.is_cuda and x.dtype != self.weight.dtype and \
         not torch.is_autocast_enabled():
       # bf16 activations outside autocast: run the fallback in bf16 too.

@@ -287,18 +287,27 @@ wrw4_reduce_kernel(const float* __restrict__ part,
                    w4bf16_t* __restrict__ dw, int RS, long ck,
                    int ngx, int ngy, int group, int C) {
   const long cells = (long)RS * ck;
+  const long gx_stride = (long)ngy * group * ck;
   for (long i = blockIdx.x * 256L + threadIdx.x; i < cells;
        i += (long)gridDim.x * 256) {
     const int rs = (int)(i / ck);
     const long rest = i % ck;
     const int gy = rs / group, g = rs % group;
-    float s = 0.0f;
-    for (int gx = 0; gx < ngx; ++gx) {
-      s += part[(((long)gx * ngy + gy) * group + g) * ck + rest];
+    const float* base = part + ((long)gy * group + g) * ck + rest;
+    // 4 accumulator chains: the single-chain version was latency-bound
+    // (400 WGs x ~150 dependent adds measured 38 us for a 63 MB read).
+    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    int gx = 0;
+    for (; gx + 4 <= ngx; gx += 4) {
+      s0 += base[(gx + 0) * gx_stride];
+      s1 += base[(gx + 1) * gx_stride];
+      s2 += base[(gx + 2) * gx_stride];
+      s3 += base[(gx + 3) * gx_stride];
     }
+    for (; gx < ngx; ++gx) s0 += base[gx * gx_stride];
     const int c = (int)(rest / (ck / C));
     const int k = (int)(rest % (ck / C));
-    dw[((long)k * C + c) * RS + rs] = (w4bf16_t)s;
+    dw[((long)k * C + c) * RS + rs] = (w4bf16_t)(s0 + s1 + s2 + s3);
   }
 }
 
