@@ -110,10 +110,15 @@ class AbstractInputGenerator:
           "set_specification_from_model must be called before "
           "create_dataset_input_fn.")
 
+    # ONE PrefetchIterator per input_fn: its pinned staging ring must
+    # survive re-iteration (the Trainer takes a fresh iterator per
+    # train() segment; rebuilding the ring re-pays hipHostMalloc).
+    prefetcher = pipeline.PrefetchIterator(
+        lambda: self._iterate(mode), depth=prefetch_depth,
+        pin_memory=pin_memory)
+
     def input_fn() -> Iterator:
-      source = lambda: self._iterate(mode)
-      return iter(pipeline.PrefetchIterator(source, depth=prefetch_depth,
-                                            pin_memory=pin_memory))
+      return iter(prefetcher)
 
     return input_fn
 

@@ -231,6 +231,12 @@ class MFMAConv2d(nn.Conv2d):
                      stride=stride, padding=padding, bias=bias)
 
   def forward(self, x):
+    if x.is_cuda and x.dtype == torch.float32 and \
+        torch.is_autocast_enabled():
+      # Autocast would cast inside F.conv2d anyway; casting HERE lets
+      # the dispatch checks see the bf16 tensor (the f32 preprocess
+      # output otherwise sent the 6x6 stem to MIOpen - profiles/).
+      x = x.to(torch.get_autocast_dtype("cuda"))
     if self.bias is None and _supported(x, self.weight, self.stride,
                                         self.padding):
       w = self.weight

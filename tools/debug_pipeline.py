@@ -1,0 +1,109 @@
+"""Instrument the Trainer pipeline path: where do the ms/step go?
+
+  python tools/debug_pipeline.py            # pinned ring (default)
+  T2R_NO_PIN=1 python tools/debug_pipeline.py
+"""
+
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from tensor2robot_amd.data import input_generators
+from tensor2robot_amd.data import pipeline
+from tensor2robot_amd.research.qtopt import t2r_models
+from tensor2robot_amd.specs import tensorspec_utils as tsu
+from tensor2robot_amd.train import train_eval
+from tensor2robot_amd.utils import modes as run_modes
+
+bs = 16
+
+
+class _PoolGenerator(input_generators.AbstractInputGenerator):
+
+  def _iterate(self, mode):
+    g = torch.Generator().manual_seed(9)
+    batches = []
+    for _ in range(3):
+      f = tsu.TensorSpecStruct()
+      f["state/image"] = torch.randint(
+          0, 256, (bs, t2r_models.RAW_HEIGHT, t2r_models.RAW_WIDTH, 3),
+          generator=g, dtype=torch.uint8)
+      off = 0
+      act = torch.rand(bs, t2r_models.ACTION_DIM, generator=g)
+      for name, size in t2r_models.ACTION_COMPONENTS:
+        f["action/" + name] = act[:, off:off + size].clone()
+        off += size
+      l = tsu.TensorSpecStruct()
+      l["grasp_success"] = (torch.rand(bs, 1, generator=g) > 0.5).float()
+      batches.append((f, l))
+    i = 0
+    while True:
+      yield batches[i % len(batches)]
+      i += 1
+
+
+def main():
+  torch.manual_seed(0)
+  model = t2r_models.GraspingModel(device_type="gpu",
+                                   compute_dtype="bfloat16")
+  trainer = train_eval.Trainer(model, model_dir="")
+  gen = _PoolGenerator(batch_size=bs)
+  gen.set_specification_from_model(model, run_modes.TRAIN)
+  deferred = gen.defer_preprocessing()
+  pin = not os.environ.get("T2R_NO_PIN")
+  input_fn = gen.create_dataset_input_fn(run_modes.TRAIN,
+                                         pin_memory=pin)
+  # settle
+  trainer.train(input_fn, 25, preprocess_fn=deferred)
+  torch.cuda.synchronize()
+
+  # instrumented manual loop (same ops as Trainer.train body)
+  from tensor2robot_amd.data import pipeline as pl
+  stage_t = [0.0, 0]
+  orig_stage = pl._PinnedRing.stage
+  def timed_stage(self, item):
+    t0 = time.perf_counter()
+    out = orig_stage(self, item)
+    stage_t[0] += time.perf_counter() - t0
+    stage_t[1] += 1
+    return out
+  pl._PinnedRing.stage = timed_stage
+  iterator = iter(input_fn())
+  t_next = t_move = t_pre = t_step = 0.0
+  n = 30
+  torch.cuda.synchronize()
+  t_all0 = time.perf_counter()
+  for i in range(n):
+    t0 = time.perf_counter()
+    features, labels = next(iterator)
+    t1 = time.perf_counter()
+    features = pipeline.move_struct_to_device(features, trainer.device)
+    labels = pipeline.move_struct_to_device(labels, trainer.device)
+    t2 = time.perf_counter()
+    with trainer._autocast():
+      features, labels = deferred(features, labels)
+    t3 = time.perf_counter()
+    trainer._fast_engine.step(features, labels, trainer.global_step)
+    trainer.global_step += 1
+    t4 = time.perf_counter()
+    t_next += t1 - t0
+    t_move += t2 - t1
+    t_pre += t3 - t2
+    t_step += t4 - t3
+  torch.cuda.synchronize()
+  total = time.perf_counter() - t_all0
+  stage_ms = stage_t[0] / max(stage_t[1], 1) * 1000
+  print(f"pin={pin} nowait={bool(os.environ.get('T2R_RING_NO_WAIT'))} "
+        f"total={total / n * 1000:.3f} ms/step | "
+        f"next={t_next / n * 1000:.3f} move={t_move / n * 1000:.3f} "
+        f"preproc={t_pre / n * 1000:.3f} step={t_step / n * 1000:.3f} "
+        f"stage={stage_ms:.3f}x{stage_t[1]}",
+        flush=True)
+
+
+if __name__ == "__main__":
+  main()
