@@ -168,19 +168,32 @@ class _PinnedRing:
     self._slots = [_PinSlot() for _ in range(max(2, size))]
     self._i = 0
 
+  _DEBUG = bool(__import__("os").environ.get("T2R_RING_DEBUG"))
+
   def stage(self, item):
+    import time as _time
     slot = self._slots[self._i % len(self._slots)]
     self._i += 1
+    t0 = _time.perf_counter() if self._DEBUG else 0.0
     slot.wait_reusable()
+    t1 = _time.perf_counter() if self._DEBUG else 0.0
+    stats = {"alloc": 0.0, "copy": 0.0, "n_alloc": 0}
 
     def stage_tensor(path, x):
       if not isinstance(x, torch.Tensor) or x.is_cuda:
         return x
       buf = slot.buffers.get(path)
       if buf is None or buf.shape != x.shape or buf.dtype != x.dtype:
+        ta = _time.perf_counter() if self._DEBUG else 0.0
         buf = torch.empty_like(x).pin_memory()
+        if self._DEBUG:
+          stats["alloc"] += _time.perf_counter() - ta
+          stats["n_alloc"] += 1
         slot.buffers[path] = buf
+      tc = _time.perf_counter() if self._DEBUG else 0.0
       buf.copy_(x)
+      if self._DEBUG:
+        stats["copy"] += _time.perf_counter() - tc
       return buf
 
     def stage_struct(prefix, s):
@@ -195,7 +208,14 @@ class _PinnedRing:
         return out
       return stage_tensor(prefix, s)
 
-    return stage_struct("", item)
+    out = stage_struct("", item)
+    if self._DEBUG:
+      import time as _t
+      total = _t.perf_counter() - t0
+      print(f"# ring.stage total={total*1e3:.2f} wait={(t1-t0)*1e3:.2f} "
+            f"alloc={stats['alloc']*1e3:.2f}x{stats['n_alloc']} "
+            f"copy={stats['copy']*1e3:.2f}", flush=True)
+    return out
 
 
 class PrefetchIterator:

@@ -33,6 +33,7 @@ input buffers each step (`step(features, labels)` does this).
 from __future__ import annotations
 
 import logging
+import time
 from typing import Callable, Dict, Optional, Tuple
 
 import torch
@@ -191,12 +192,23 @@ class FastStepEngine:
       self._demote()
     self._built = True
 
-  def _capture(self, global_step: int):
+  def _capture(self, global_step: int, attempt: int = 0):
     static_features, static_labels = self._static_structs()
     for i in range(3):  # settle MIOpen algo find before capture
       self._eager_step_inner(static_features, static_labels,
                              global_step)
     torch.cuda.synchronize()
+    # Baseline for the post-capture sanity check: a capture taken while
+    # MIOpen's find is mid-flight bakes find-intermediate (slow)
+    # kernels into the graph permanently — measured 13 ms/step vs 2.8
+    # eager-after-find on the same shapes.  Time eager here, replay
+    # after capture, and recapture once if the graph lost.
+    t0 = time.perf_counter()
+    for _ in range(2):
+      self._eager_step_inner(static_features, static_labels,
+                             global_step)
+    torch.cuda.synchronize()
+    t_eager = time.perf_counter() - t0
 
     if self._lr_schedule is not None:
       self._captured_lr = self._lr_schedule(global_step)
@@ -238,6 +250,34 @@ class FastStepEngine:
         return None
 
       self.opt_graphed = graph_step.GraphedTrainStep(opt_body)
+
+    # Post-capture sanity check (see comment above): replay must not be
+    # slower than eager.  The decision is collective in distributed
+    # mode (the replay includes no collectives, but a recapture does
+    # run unsynced settle steps — every rank must take the same path).
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(2):
+      self.graphed.replay()
+      if self.distributed:
+        dist.all_reduce(self._flat)
+        self.opt_graphed.replay()
+    torch.cuda.synchronize()
+    t_replay = time.perf_counter() - t0
+    slow = t_replay > 1.3 * t_eager + 1e-3
+    if self.distributed:
+      flag = torch.tensor([1.0 if slow else 0.0], device=self.device)
+      dist.all_reduce(flag, op=dist.ReduceOp.MAX)
+      slow = float(flag.item()) > 0.0
+    if slow and attempt == 0:
+      _log.warning(
+          "captured graph slower than eager (%.2f vs %.2f ms/step) — "
+          "find likely mid-flight at capture; recapturing",
+          t_replay / 2 * 1e3, t_eager / 2 * 1e3)
+      self.graphed = None
+      self.opt_graphed = None
+      torch.cuda.synchronize()
+      self._capture(global_step, attempt=1)
 
   def _demote(self):
     self.graphed = None

@@ -10,6 +10,9 @@ import time
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
+import logging
+logging.basicConfig(level=logging.INFO)
+
 import torch
 
 from tensor2robot_amd.data import input_generators
@@ -60,6 +63,7 @@ def main():
   # settle
   trainer.train(input_fn, 25, preprocess_fn=deferred)
   torch.cuda.synchronize()
+  print(f"engine graphed={trainer._fast_engine.is_graphed}", flush=True)
 
   # instrumented manual loop (same ops as Trainer.train body)
   from tensor2robot_amd.data import pipeline as pl
