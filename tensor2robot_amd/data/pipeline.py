@@ -146,10 +146,18 @@ class _PinSlot:
         self.event = torch.cuda.Event()
       self.event.record()
 
-  def wait_reusable(self):
+  def wait_reusable(self, alive=None):
     import os as _os
-    if self.event is not None and not _os.environ.get("T2R_RING_NO_WAIT"):
-      self.event.synchronize()
+    import time as _t
+    if self.event is None or _os.environ.get("T2R_RING_NO_WAIT"):
+      return
+    # Polling (instead of event.synchronize) lets a retiring producer
+    # thread exit promptly at interpreter shutdown instead of blocking
+    # inside the HIP runtime while it tears down.
+    while not self.event.query():
+      if alive is not None and not alive():
+        return
+      _t.sleep(0.0005)
 
 
 _SLOT_ATTR = "_t2r_pin_slot"
@@ -170,12 +178,12 @@ class _PinnedRing:
 
   _DEBUG = bool(__import__("os").environ.get("T2R_RING_DEBUG"))
 
-  def stage(self, item):
+  def stage(self, item, alive=None):
     import time as _time
     slot = self._slots[self._i % len(self._slots)]
     self._i += 1
     t0 = _time.perf_counter() if self._DEBUG else 0.0
-    slot.wait_reusable()
+    slot.wait_reusable(alive)
     t1 = _time.perf_counter() if self._DEBUG else 0.0
     stats = {"alloc": 0.0, "copy": 0.0, "n_alloc": 0}
 
@@ -252,7 +260,9 @@ class PrefetchIterator:
           if self._generation != gen:
             return  # a newer iteration owns the ring now
           if ring is not None:
-            item = ring.stage(item)
+            item = ring.stage(
+                item, alive=lambda: self._generation == gen and
+                threading.main_thread().is_alive())
           while True:
             try:
               q.put(item, timeout=0.5)
