@@ -53,9 +53,14 @@ def _conv_fixed_padding(in_ch: int, out_ch: int, kernel: int,
   identical.  For the even-offset stride-2 case we use torch padding
   (k-1)//2 which yields the same output size; the one-pixel alignment
   difference does not change the architecture contract.
+
+  MFMAConv2d self-dispatches per shape: fused MFMA kernels on the
+  C=K<=64 stride-1 shapes, hand-im2col + rocBLAS GEMM on the ResNet
+  C/K-up-to-512 shapes, torch/MIOpen otherwise.
   """
-  return nn.Conv2d(in_ch, out_ch, kernel, stride=stride,
-                   padding=(kernel - 1) // 2, bias=False)
+  from tensor2robot_amd.ops import conv as mfma_conv
+  return mfma_conv.MFMAConv2d(in_ch, out_ch, kernel, stride=stride,
+                              padding=(kernel - 1) // 2, bias=False)
 
 
 def apply_film(x: torch.Tensor, gamma_beta: Optional[torch.Tensor]

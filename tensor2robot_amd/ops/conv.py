@@ -249,6 +249,15 @@ class MFMAConv2d(nn.Conv2d):
       if w.dtype != torch.bfloat16:
         w = w.to(torch.bfloat16)
       return _StemConvFunction.apply(x, w)
+    if self.bias is None:
+      # ResNet-family shapes (C/K up to 512, stride 1/2, 1x1/3x3):
+      # hand im2col/col2im + rocBLAS GEMM (ops/gemm_conv.py) instead of
+      # MIOpen's helper-kernel storm on these shapes.
+      from tensor2robot_amd.ops import gemm_conv
+      if gemm_conv.supported(x, self.weight, self.stride, self.padding,
+                             self.dilation, self.groups):
+        return gemm_conv.gemm_conv2d(x, self.weight, self.stride,
+                                     self.padding)
     if x.is_cuda and os.environ.get("T2R_LOG_CONV_FALLBACK"):
       print(f"# conv fallback: x={tuple(x.shape)} {x.dtype} "
             f"req_grad={x.requires_grad} w={tuple(self.weight.shape)} "

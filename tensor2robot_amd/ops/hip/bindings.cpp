@@ -37,6 +37,11 @@ at::Tensor conv_s1_wrw3(at::Tensor x, at::Tensor dy, int64_t R, int64_t S,
                         int64_t pad);
 at::Tensor conv_s1_wrw4(at::Tensor x, at::Tensor dy, int64_t R, int64_t S,
                         int64_t pad);
+at::Tensor im2col_nhwc(at::Tensor x, int64_t R, int64_t S, int64_t pad,
+                       int64_t stride);
+at::Tensor col2im_nhwc(at::Tensor dcol, int64_t N, int64_t C, int64_t H,
+                       int64_t W, int64_t R, int64_t S, int64_t pad,
+                       int64_t stride);
 
 at::Tensor conv_stem_nhwc(at::Tensor x, at::Tensor wpk);
 
@@ -73,6 +78,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused space-to-depth [N,3,H,W]->[N,16,H/2,W/2] for the stem");
   m.def("conv_s1_wrw4", &conv_s1_wrw4,
         "MFMA wrw v4: pixel-major x image + ds_read_b64_tr_b16 A-reads");
+  m.def("im2col_nhwc", &im2col_nhwc,
+        "NHWC bf16 im2col gather (GEMM-conv path)");
+  m.def("col2im_nhwc", &col2im_nhwc,
+        "NHWC bf16 col2im gather (GEMM-conv dgrad)");
   m.def("conv_s1_wrw3", &conv_s1_wrw3,
         "MFMA wrw v3: occupancy-first rs-split (256-thr WGs, 3/CU)");
   m.def("conv_s1_wrw2", &conv_s1_wrw2,

@@ -343,3 +343,44 @@ def test_mfma_wrw4_matches_torch(shape):
   err = (dw_t - ref).abs().max().item()
   scale = ref.abs().max().item()
   assert err < 0.01 * max(scale, 1.0), (shape, err, scale)
+
+
+@requires_gpu
+@pytest.mark.parametrize("shape", [
+    # (n, c, h, w, k, r, stride, pad) — ResNet-family GEMM-conv shapes
+    (4, 64, 25, 25, 64, 3, 1, 1),
+    (4, 64, 25, 25, 128, 3, 2, 1),     # downsample
+    (4, 128, 13, 13, 128, 3, 1, 1),
+    (4, 256, 7, 7, 512, 1, 1, 0),      # 1x1 projection
+    (4, 64, 25, 25, 256, 1, 1, 0),     # bottleneck expand
+    (2, 512, 4, 4, 512, 3, 1, 1),      # deep tiny-spatial
+    (2, 64, 14, 14, 128, 1, 2, 0),     # 1x1 stride-2 projection
+])
+def test_gemm_conv_matches_torch(shape):
+  """GEMM-conv (im2col + rocBLAS) numerics vs fp32 reference."""
+  import torch.nn.functional as F
+  from tensor2robot_amd.ops import gemm_conv
+  n, c, h, w, k, r, stride, pad = shape
+  torch.manual_seed(0)
+  x = torch.randn(n, c, h, w, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+  wt = (torch.randn(k, c, r, r, device="cuda") * 0.05).to(torch.bfloat16)
+  wt.requires_grad_(True)
+  y = gemm_conv.gemm_conv2d(x, wt, (stride, stride), (pad, pad))
+  oh = (h + 2 * pad - r) // stride + 1
+  assert y.shape == (n, k, oh, oh)
+  dy = torch.randn_like(y.float()).to(torch.bfloat16)
+  y.backward(dy)
+
+  x32 = x.detach().float().requires_grad_(True)
+  w32 = wt.detach().float().requires_grad_(True)
+  y32 = F.conv2d(x32, w32, stride=stride, padding=pad)
+  y32.backward(dy.float())
+
+  def relerr(a, b):
+    return (a.float() - b).abs().max().item() / max(
+        b.abs().max().item(), 1e-6)
+
+  assert relerr(y, y32) < 0.02, ("y", shape, relerr(y, y32))
+  assert relerr(x.grad, x32.grad) < 0.03, ("dx", shape)
+  assert relerr(wt.grad, w32.grad) < 0.03, ("dw", shape)
