@@ -50,11 +50,12 @@ class _GemmConvFunction(torch.autograd.Function):
       col = x.permute(0, 2, 3, 1).reshape(-1, c)  # NHWC view, no copy
     else:
       col = ext.im2col_nhwc(x, r, s, pad, stride)
-    y_flat = col @ wm                              # [M, K] rocBLAS bf16
-    y = torch.empty((n, k, oh, ow), dtype=y_flat.dtype, device=x.device
+    y = torch.empty((n, k, oh, ow), dtype=torch.bfloat16,
+                    device=x.device
                     ).contiguous(memory_format=torch.channels_last)
-    # channels_last [N,K,OH,OW] storage IS [M,K] row-major.
-    y.permute(0, 2, 3, 1).reshape(-1, k).copy_(y_flat)
+    # channels_last [N,K,OH,OW] storage IS [M,K] row-major: GEMM
+    # writes straight into it, no copy.
+    torch.matmul(col, wm, out=y.permute(0, 2, 3, 1).reshape(-1, k))
     ctx.save_for_backward(x, weight, wm,
                           col if not one_by_one else x)
     ctx.conf = (stride, pad, one_by_one)
