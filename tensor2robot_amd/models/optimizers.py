@@ -98,9 +98,12 @@ def create_adam_optimizer(learning_rate=1e-4, beta1: float = 0.9,
   schedule = _resolve_lr(learning_rate)
 
   def build(params) -> ScheduledOptimizer:
-    use_fused = fused and torch.cuda.is_available()
+    use_cuda = torch.cuda.is_available()
+    # capturable: Adam's step counter lives on-device so the optimizer
+    # step can be recorded into a hipGraph (the Trainer's fast path).
     opt = torch.optim.Adam(params, lr=schedule(0), betas=(beta1, beta2),
-                           eps=epsilon, fused=use_fused)
+                           eps=epsilon, fused=fused and use_cuda,
+                           capturable=use_cuda)
     return ScheduledOptimizer(opt, schedule, clip_gradient_norm)
 
   return build
@@ -141,7 +144,8 @@ def create_rms_prop_optimizer(learning_rate=1e-4, decay: float = 0.9,
 
   def build(params) -> ScheduledOptimizer:
     opt = torch.optim.RMSprop(params, lr=schedule(0), alpha=decay,
-                              momentum=momentum, eps=epsilon)
+                              momentum=momentum, eps=epsilon,
+                              capturable=torch.cuda.is_available())
     return ScheduledOptimizer(opt, schedule, clip_gradient_norm)
 
   return build

@@ -258,6 +258,11 @@ class MFMAConv2d(nn.Conv2d):
                              self.dilation, self.groups):
         return gemm_conv.gemm_conv2d(x, self.weight, self.stride,
                                      self.padding)
+      if gemm_conv.supported_cpad(x, self.weight, self.stride,
+                                  self.padding, self.dilation,
+                                  self.groups):
+        return gemm_conv.gemm_conv2d_cpad(x, self.weight, self.stride,
+                                          self.padding)
     if x.is_cuda and os.environ.get("T2R_LOG_CONV_FALLBACK"):
       print(f"# conv fallback: x={tuple(x.shape)} {x.dtype} "
             f"req_grad={x.requires_grad} w={tuple(self.weight.shape)} "

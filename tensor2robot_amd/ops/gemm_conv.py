@@ -115,3 +115,33 @@ def gemm_conv2d(x, weight, stride, padding) -> torch.Tensor:
   w = weight if weight.dtype == torch.bfloat16 else \
       weight.to(torch.bfloat16)
   return _GemmConvFunction.apply(x, w, stride[0], padding[0])
+
+
+def supported_cpad(x, weight, stride, padding, dilation, groups) -> bool:
+  """Small-C stems (RGB 7x7/2 etc.): zero-pad channels to 8 and run the
+  GEMM path — MIOpen's wrw find on these burns minutes (profiles/)."""
+  import os
+  if os.environ.get("T2R_DISABLE_GEMM_CONV"):
+    return False
+  if not (x.is_cuda and x.dtype == torch.bfloat16):
+    return False
+  if groups != 1 or dilation != (1, 1):
+    return False
+  if stride[0] != stride[1] or stride[0] not in (1, 2):
+    return False
+  if padding[0] != padding[1]:
+    return False
+  k, c, r, s = weight.shape
+  return 0 < c < 8 and k % 8 == 0 and r <= 7 and s <= 7
+
+
+def gemm_conv2d_cpad(x, weight, stride, padding) -> torch.Tensor:
+  n, c, h, w = x.shape
+  pad_c = 8 - c
+  zx = x.new_zeros((n, pad_c, h, w))
+  xp = torch.cat([x, zx], 1).contiguous(
+      memory_format=torch.channels_last)
+  wzero = weight.new_zeros((weight.shape[0], pad_c, weight.shape[2],
+                            weight.shape[3]))
+  wp = torch.cat([weight, wzero], 1).to(torch.bfloat16)
+  return _GemmConvFunction.apply(xp, wp, stride[0], padding[0])

@@ -384,3 +384,25 @@ def test_gemm_conv_matches_torch(shape):
   assert relerr(y, y32) < 0.02, ("y", shape, relerr(y, y32))
   assert relerr(x.grad, x32.grad) < 0.03, ("dx", shape)
   assert relerr(wt.grad, w32.grad) < 0.03, ("dw", shape)
+
+
+@requires_gpu
+def test_gemm_conv_cpad_stem_matches_torch():
+  """RGB 7x7/2 stem via the channel-padded GEMM path."""
+  import torch.nn.functional as F
+  from tensor2robot_amd.ops import gemm_conv
+  torch.manual_seed(0)
+  x = torch.randn(4, 3, 100, 100, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last)
+  wt = (torch.randn(64, 3, 7, 7, device="cuda") * 0.05).to(
+      torch.bfloat16).requires_grad_(True)
+  y = gemm_conv.gemm_conv2d_cpad(x, wt, (2, 2), (3, 3))
+  dy = torch.randn_like(y.float()).to(torch.bfloat16)
+  y.backward(dy)
+  w32 = wt.detach().float().requires_grad_(True)
+  y32 = F.conv2d(x.float(), w32, stride=2, padding=3)
+  y32.backward(dy.float())
+  assert (y.float() - y32).abs().max().item() < \
+      0.02 * y32.abs().max().item()
+  assert (wt.grad.float() - w32.grad).abs().max().item() < \
+      0.03 * max(w32.grad.abs().max().item(), 1.0)
