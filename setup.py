@@ -28,6 +28,7 @@ hip_sources = [
     os.path.join(HIP_DIR, "conv_wrw2.hip"),
     os.path.join(HIP_DIR, "conv_wrw4.hip"),
     os.path.join(HIP_DIR, "im2col.hip"),
+    os.path.join(HIP_DIR, "jpeg_gpu.hip"),
     os.path.join(HIP_DIR, "conv_stem.hip"),
 ]
 

@@ -159,6 +159,16 @@ def decode_jpeg(data: bytes) -> np.ndarray:
   return native.decode_jpeg(data)
 
 
+def native_module():
+  """The built _t2r_native extension (raises if missing)."""
+  native = _load_jpeg_native()
+  if native is None:
+    raise RuntimeError(
+        f"JPEG codec extension not built: {_jpeg_import_error}. "
+        "Run `python setup.py build_ext --inplace`.")
+  return native
+
+
 def decode_image(data: bytes, data_format: Optional[str] = None) -> np.ndarray:
   """Decodes JPEG or PNG bytes (sniffs when data_format is None)."""
   if not data:

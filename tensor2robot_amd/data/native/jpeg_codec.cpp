@@ -14,6 +14,8 @@
 #include <string>
 #include <vector>
 
+#include "jpeg_codec.h"
+
 namespace t2r_jpeg {
 
 // ---------------------------------------------------------------------------
@@ -622,6 +624,166 @@ std::vector<uint8_t> decode(const uint8_t* data, size_t size, int& out_h,
       return img;
     } else {
       // Skip APPn/COM/unknown.
+    }
+    pos = seg_end;
+  }
+  throw std::runtime_error("JPEG decode: no scan found");
+}
+
+// ---------------------------------------------------------------------------
+// Coefficient-level decode (Huffman/entropy only, no dequant/IDCT):
+// feeds the GPU decode path (ops/hip/jpeg_gpu.hip) — the bit-serial
+// entropy scan runs on host threads, every numeric stage (dequant +
+// IDCT + upsample + color convert) runs as HIP kernels.
+// ---------------------------------------------------------------------------
+
+CoeffImage decode_coeffs(const uint8_t* data, size_t size) {
+  CoeffImage out;
+  size_t pos = 0;
+  auto rd_u16 = [&](size_t p) -> int {
+    return (data[p] << 8) | data[p + 1];
+  };
+  if (size < 4 || data[0] != 0xff || data[1] != 0xd8)
+    throw std::runtime_error("JPEG decode: missing SOI");
+  pos = 2;
+  HuffDecTable dc_tabs[4], ac_tabs[4];
+  Component comps[4];
+  int ncomp = 0, height = 0, width = 0, restart_interval = 0;
+  int hmax = 1, vmax = 1;
+
+  while (pos + 4 <= size) {
+    if (data[pos] != 0xff) throw std::runtime_error("JPEG: bad marker");
+    uint8_t marker = data[pos + 1];
+    pos += 2;
+    if (marker == 0xd9) break;
+    if (marker == 0x01 || (marker >= 0xd0 && marker <= 0xd7)) continue;
+    int seglen = rd_u16(pos);
+    size_t seg_end = pos + seglen;
+    if (marker == 0xdb) {
+      size_t p = pos + 2;
+      while (p < seg_end) {
+        int pq = data[p] >> 4, tq_id = data[p] & 15;
+        ++p;
+        for (int i = 0; i < 64; ++i) {
+          int v = pq ? rd_u16(p + 2 * i) : data[p + i];
+          out.qt[tq_id][kZigzag[i]] = (uint16_t)v;
+        }
+        p += pq ? 128 : 64;
+      }
+    } else if (marker == 0xc4) {
+      size_t p = pos + 2;
+      while (p < seg_end) {
+        int cls = data[p] >> 4, id = data[p] & 15;
+        ++p;
+        uint8_t bits[17] = {0};
+        int nvals = 0;
+        for (int i = 1; i <= 16; ++i) {
+          bits[i] = data[p + i - 1];
+          nvals += bits[i];
+        }
+        p += 16;
+        if (cls == 0)
+          build_dec_table(bits, data + p, nvals, dc_tabs[id]);
+        else
+          build_dec_table(bits, data + p, nvals, ac_tabs[id]);
+        p += nvals;
+      }
+    } else if (marker == 0xc0 || marker == 0xc1) {
+      height = rd_u16(pos + 3);
+      width = rd_u16(pos + 5);
+      ncomp = data[pos + 7];
+      if (ncomp > 4) throw std::runtime_error("JPEG: too many components");
+      for (int i = 0; i < ncomp; ++i) {
+        size_t p = pos + 8 + 3 * i;
+        comps[i].id = data[p];
+        comps[i].hs = data[p + 1] >> 4;
+        comps[i].vs = data[p + 1] & 15;
+        comps[i].tq = data[p + 2];
+        hmax = std::max(hmax, comps[i].hs);
+        vmax = std::max(vmax, comps[i].vs);
+      }
+    } else if (marker == 0xc2) {
+      throw std::runtime_error("JPEG: progressive not supported");
+    } else if (marker == 0xdd) {
+      restart_interval = rd_u16(pos + 2);
+    } else if (marker == 0xda) {
+      int ns = data[pos + 2];
+      for (int i = 0; i < ns; ++i) {
+        int cid = data[pos + 3 + 2 * i];
+        int tables = data[pos + 4 + 2 * i];
+        for (int c = 0; c < ncomp; ++c)
+          if (comps[c].id == cid) {
+            comps[c].td = tables >> 4;
+            comps[c].ta = tables & 15;
+          }
+      }
+      pos = seg_end;
+      int mcux = (width + 8 * hmax - 1) / (8 * hmax);
+      int mcuy = (height + 8 * vmax - 1) / (8 * vmax);
+      out.height = height;
+      out.width = width;
+      out.ncomp = ncomp;
+      out.hmax = hmax;
+      out.vmax = vmax;
+      for (int c = 0; c < ncomp; ++c) {
+        out.comps[c].hs = comps[c].hs;
+        out.comps[c].vs = comps[c].vs;
+        out.comps[c].tq = comps[c].tq;
+        out.comps[c].bw = mcux * comps[c].hs;
+        out.comps[c].bh = mcuy * comps[c].vs;
+        out.comps[c].coeffs.assign(
+            (size_t)out.comps[c].bw * out.comps[c].bh * 64, 0);
+        comps[c].dc_pred = 0;
+      }
+      BitReader br(data, size, pos);
+      int mcu_count = 0;
+      for (int my = 0; my < mcuy; ++my) {
+        for (int mx = 0; mx < mcux; ++mx) {
+          if (restart_interval && mcu_count &&
+              mcu_count % restart_interval == 0) {
+            br.reset_to_byte();
+            while (br.pos + 1 < size && data[br.pos] == 0xff &&
+                   data[br.pos + 1] >= 0xd0 && data[br.pos + 1] <= 0xd7) {
+              br.pos += 2;
+              for (int c = 0; c < ncomp; ++c) comps[c].dc_pred = 0;
+            }
+          }
+          ++mcu_count;
+          for (int c = 0; c < ncomp; ++c) {
+            Component& comp = comps[c];
+            auto& oc = out.comps[c];
+            for (int v = 0; v < comp.vs; ++v) {
+              for (int hh = 0; hh < comp.hs; ++hh) {
+                const int by = my * comp.vs + v;
+                const int bx = mx * comp.hs + hh;
+                int16_t* blk =
+                    &oc.coeffs[((size_t)by * oc.bw + bx) * 64];
+                int sym = huff_decode(br, dc_tabs[comp.td]);
+                if (sym < 0) throw std::runtime_error("JPEG: truncated");
+                comp.dc_pred += receive_extend(br, sym);
+                blk[0] = (int16_t)comp.dc_pred;
+                for (int k = 1; k < 64;) {
+                  int rs = huff_decode(br, ac_tabs[comp.ta]);
+                  if (rs < 0)
+                    throw std::runtime_error("JPEG: truncated");
+                  int run = rs >> 4, sbits = rs & 15;
+                  if (sbits == 0) {
+                    if (run != 15) break;
+                    k += 16;
+                    continue;
+                  }
+                  k += run;
+                  if (k > 63)
+                    throw std::runtime_error("JPEG: AC overflow");
+                  blk[kZigzag[k]] = (int16_t)receive_extend(br, sbits);
+                  ++k;
+                }
+              }
+            }
+          }
+        }
+      }
+      return out;
     }
     pos = seg_end;
   }

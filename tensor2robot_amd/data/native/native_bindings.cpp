@@ -7,12 +7,7 @@
 
 namespace py = pybind11;
 
-namespace t2r_jpeg {
-std::vector<uint8_t> encode(const uint8_t* rgb, int h, int w, int channels,
-                            int quality);
-std::vector<uint8_t> decode(const uint8_t* data, size_t size, int& out_h,
-                            int& out_w, int& out_c);
-}  // namespace t2r_jpeg
+#include "jpeg_codec.h"
 
 static py::bytes encode_jpeg(
     py::array_t<uint8_t, py::array::c_style | py::array::forcecast> image,
@@ -44,9 +39,44 @@ static py::array decode_jpeg(py::bytes data) {
   return out;
 }
 
+static py::dict decode_jpeg_coeffs(py::bytes data) {
+  // Huffman/entropy decode only (the GPU decode handoff); the scan is
+  // released from the GIL so batches parallelize across host threads.
+  std::string buf = data;
+  t2r_jpeg::CoeffImage ci;
+  {
+    py::gil_scoped_release release;
+    ci = t2r_jpeg::decode_coeffs((const uint8_t*)buf.data(), buf.size());
+  }
+  py::dict out;
+  out["height"] = ci.height;
+  out["width"] = ci.width;
+  out["ncomp"] = ci.ncomp;
+  out["hmax"] = ci.hmax;
+  out["vmax"] = ci.vmax;
+  py::list comps;
+  for (int c = 0; c < ci.ncomp; ++c) {
+    auto& cp = ci.comps[c];
+    py::dict d;
+    d["hs"] = cp.hs;
+    d["vs"] = cp.vs;
+    py::array_t<int16_t> coeffs({cp.bh, cp.bw, 64});
+    std::memcpy(coeffs.mutable_data(), cp.coeffs.data(),
+                cp.coeffs.size() * sizeof(int16_t));
+    d["coeffs"] = coeffs;
+    py::array_t<uint16_t> q({64});
+    std::memcpy(q.mutable_data(), ci.qt[cp.tq], 64 * sizeof(uint16_t));
+    d["quant"] = q;
+    comps.append(d);
+  }
+  out["comps"] = comps;
+  return out;
+}
+
 PYBIND11_MODULE(_t2r_native, m) {
   m.doc() = "CPU-native codecs: baseline JPEG encode/decode";
   m.def("encode_jpeg", &encode_jpeg, py::arg("image"),
         py::arg("quality") = 90);
   m.def("decode_jpeg", &decode_jpeg, py::arg("data"));
+  m.def("decode_jpeg_coeffs", &decode_jpeg_coeffs, py::arg("data"));
 }

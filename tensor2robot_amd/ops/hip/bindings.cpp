@@ -42,6 +42,13 @@ at::Tensor im2col_nhwc(at::Tensor x, int64_t R, int64_t S, int64_t pad,
 at::Tensor col2im_nhwc(at::Tensor dcol, int64_t N, int64_t C, int64_t H,
                        int64_t W, int64_t R, int64_t S, int64_t pad,
                        int64_t stride);
+at::Tensor jpeg_idct(at::Tensor coeffs, at::Tensor quant, int64_t bh,
+                     int64_t bw);
+at::Tensor jpeg_color(at::Tensor py, at::Tensor pcb, at::Tensor pcr,
+                      int64_t H, int64_t W, int64_t hs_y, int64_t vs_y,
+                      int64_t hs_c, int64_t vs_c, int64_t hmax,
+                      int64_t vmax);
+at::Tensor jpeg_gray(at::Tensor py, int64_t H, int64_t W);
 
 at::Tensor conv_stem_nhwc(at::Tensor x, at::Tensor wpk);
 
@@ -82,6 +89,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "NHWC bf16 im2col gather (GEMM-conv path)");
   m.def("col2im_nhwc", &col2im_nhwc,
         "NHWC bf16 col2im gather (GEMM-conv dgrad)");
+  m.def("jpeg_idct", &jpeg_idct,
+        "JPEG dequant + batched 8x8 IDCT -> f32 component plane");
+  m.def("jpeg_color", &jpeg_color,
+        "JPEG upsample + YCbCr->RGB -> uint8 NHWC");
+  m.def("jpeg_gray", &jpeg_gray, "JPEG grayscale plane -> uint8");
   m.def("conv_s1_wrw3", &conv_s1_wrw3,
         "MFMA wrw v3: occupancy-first rs-split (256-thr WGs, 3/CU)");
   m.def("conv_s1_wrw2", &conv_s1_wrw2,
