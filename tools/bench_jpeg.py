@@ -37,9 +37,10 @@ def main():
   path = os.path.join(tempfile.mkdtemp(), "jpegs.tfrecord")
   with tfrecord.TFRecordWriter(path) as wr:
     for i in range(args.images):
-      img = np.clip(
-          128 + 70 * np.sin(xx / (9.0 + i)) * np.cos(yy / (6.0 + i))
-          + rng.randint(-25, 25, (h, w, 3)), 0, 255).astype(np.uint8)
+      base = (128 + 70 * np.sin(xx / (9.0 + i)) *
+              np.cos(yy / (6.0 + i)))[..., None]
+      img = np.clip(base + rng.randint(-25, 25, (h, w, 3)),
+                    0, 255).astype(np.uint8)
       wr.write(image_codec.encode_jpeg(img, 90))
   records = list(tfrecord.read_records(path))
   print(f"{len(records)} records, avg {np.mean([len(r) for r in records])/1e3:.0f} KB")
