@@ -73,3 +73,25 @@ def test_gpu_jpeg_mixed_quality_groups():
   for i, ref in enumerate(refs1 + refs2):
     diff = np.abs(got[i].astype(int) - ref.astype(int))
     assert diff.max() <= 1, (i, diff.max())
+
+
+@requires_gpu
+def test_parser_gpu_image_decode():
+  """ExampleParser(image_decode_device='cuda') yields CUDA uint8
+  batches decoded by the HIP path."""
+  from tensor2robot_amd.data import example as example_mod
+  from tensor2robot_amd.data import parser as parser_mod
+  from tensor2robot_amd.specs import tensorspec_utils as tsu
+
+  records, refs = _roundtrip_records(3, 48, 64)
+  spec = tsu.TensorSpecStruct()
+  spec["img"] = tsu.ExtendedTensorSpec((48, 64, 3), torch.uint8,
+                                       name="image", data_format="JPEG")
+  recs = [example_mod.encode_example({"image": r}) for r in records]
+  p = parser_mod.ExampleParser(spec, image_decode_device="cuda")
+  out = p(recs)
+  img = out["img"]
+  assert img.is_cuda and img.dtype == torch.uint8
+  got = img.cpu().numpy()
+  for i, ref in enumerate(refs):
+    assert np.abs(got[i].astype(int) - ref.astype(int)).max() <= 1
