@@ -39,6 +39,7 @@ class GraphedTrainStep:
     self._capture()
 
   def _capture(self):
+    import gc
     # Warm up on a side stream so allocator state and autotuned algos
     # settle before capture (the standard graph recipe).
     side = torch.cuda.Stream()
@@ -49,9 +50,19 @@ class GraphedTrainStep:
     torch.cuda.current_stream().wait_stream(side)
     torch.cuda.synchronize()
 
-    self._graph = torch.cuda.CUDAGraph()
-    with torch.cuda.graph(self._graph):
-      self._static_output = self._step_fn()
+    # Quiesce the garbage collector across the capture: a GC cycle
+    # mid-capture destroys dead CUDA objects (tensors, events from
+    # earlier iterations) whose hipFree/hipEventDestroy on a capturing
+    # stream aborts the process (observed: SIGABRT inside
+    # "Garbage-collecting" during another engine's forward).
+    gc.collect()
+    gc.disable()
+    try:
+      self._graph = torch.cuda.CUDAGraph()
+      with torch.cuda.graph(self._graph):
+        self._static_output = self._step_fn()
+    finally:
+      gc.enable()
     _log.info("GraphedTrainStep: captured")
 
   def recapture(self):
