@@ -160,7 +160,13 @@ class MAMLModel(abstract_model.AbstractT2RModel):
     for k, v in _stack_structs(per_task_cond).items():
       predictions["full_inference_output/" + k] = v
     for i, s in enumerate(inner_loss_sums):
-      self.scalar_summary(f"inner_loss_{i}", float(s) / num_tasks)
+      # Keep the tensor: float() here is a device sync per step and
+      # breaks hipGraph capture (scalar summaries materialize at
+      # summary-write time in the Trainer).
+      self.scalar_summary(
+          f"inner_loss_{i}",
+          s / num_tasks if isinstance(s, torch.Tensor)
+          else float(s) / num_tasks)
     predictions = self._select_inference_output(predictions)
     if "condition_output" not in predictions:
       raise ValueError("The required condition_output is not in "
