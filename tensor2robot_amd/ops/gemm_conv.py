@@ -77,7 +77,22 @@ class _GemmConvFunction(torch.autograd.Function):
     else:
       col = col_saved
     if ctx.needs_input_grad[1]:
-      dw_mat = col.t() @ dy_mat                    # [RS*C, K]
+      m = col.shape[0]
+      if m >= 65536:
+        # Tiny-output x huge-contraction GEMM: hipBLASLt runs it
+        # without split-K and strands the chip (measured 56 TF at
+        # M=668k vs 177 TF chunked — tools/probe_gemm_shapes.py).
+        # Chunked bmm + sum restores grid parallelism.
+        chunks = max(2, min(64, m // 8192))
+        mc = m // chunks
+        head = chunks * mc
+        dw_mat = torch.bmm(
+            col[:head].view(chunks, mc, -1).transpose(1, 2),
+            dy_mat[:head].view(chunks, mc, k)).sum(0)
+        if head < m:
+          dw_mat = dw_mat + col[head:].t() @ dy_mat[head:]
+      else:
+        dw_mat = col.t() @ dy_mat                  # [RS*C, K]
       dw = dw_mat.reshape(r, s, c, k).permute(3, 2, 0, 1) \
           .contiguous().to(weight.dtype)
     if ctx.needs_input_grad[0]:
