@@ -44,6 +44,12 @@ from tensor2robot_amd.specs import tensorspec_utils as tsu
 
 _log = logging.getLogger(__name__)
 
+# Set while an engine is settling/benchmarking/capturing: background
+# producers (the pinned-ring pipeline) stay off the GPU runtime so
+# MIOpen/hipBLASLt algorithm timing is clean.
+import threading
+CAPTURE_QUIESCE = threading.Event()
+
 
 def _flatten_tensors(struct) -> Dict[str, torch.Tensor]:
   """TensorSpecStruct/dict of tensors -> flat {path: tensor}."""
@@ -167,6 +173,7 @@ class FastStepEngine:
 
     ok = True
     if self._want_graph:
+      CAPTURE_QUIESCE.set()
       try:
         self._capture(global_step)
       except Exception as e:  # pragma: no cover - runtime-dependent
@@ -174,6 +181,8 @@ class FastStepEngine:
         self.graphed = None
         self.opt_graphed = None
         ok = False
+      finally:
+        CAPTURE_QUIESCE.clear()
     else:
       ok = False
 
