@@ -141,6 +141,9 @@ class _PinSlot:
     self.event = None
 
   def mark_consumed(self):
+    import os as _os
+    if _os.environ.get("T2R_RING_NO_EVENT"):
+      return
     if torch.cuda.is_available() and torch.cuda.is_initialized():
       if self.event is None:
         # blocking=True: the producer's wait sleeps in the driver
@@ -217,7 +220,13 @@ class _PinnedRing:
       tc = _time.perf_counter() if self._DEBUG else 0.0
       buf.copy_(x)
       if self._DEBUG:
-        stats["copy"] += _time.perf_counter() - tc
+        dt = _time.perf_counter() - tc
+        stats["copy"] += dt
+        if x.numel() > 1 << 20:
+          import threading as _th
+          print(f"#   big-copy {x.numel()/1e6:.1f}MB {dt*1e3:.2f}ms "
+                f"pinned={buf.is_pinned()} src_contig={x.is_contiguous()} "
+                f"thread={_th.current_thread().name}", flush=True)
       return buf
 
     def stage_struct(prefix, s):
@@ -309,26 +318,64 @@ class PrefetchIterator:
       yield item
 
 
+_COPY_STREAM = None
+
+
+def _copy_stream():
+  global _COPY_STREAM
+  if _COPY_STREAM is None:
+    _COPY_STREAM = torch.cuda.Stream()
+  return _COPY_STREAM
+
+
 def move_struct_to_device(struct, device, non_blocking=True):
   """Moves every tensor in a (features, labels) struct to device.
 
+  On CUDA the copies ride a dedicated COPY STREAM with one event sync
+  into the compute stream: an H2D enqueued directly on the compute
+  stream pays a compute<->DMA queue handoff per copy (~0.28 ms each on
+  this pool — 10 small feeds cost 2.8 ms of stream time), while the
+  copy-stream route pays one cross-stream event wait total.
+
   When the struct came from the pinned ring, the consumer-side H2D
-  event is recorded here so the producer can safely refill the slot.
+  event is recorded so the producer can safely refill the slot.
   """
   if struct is None:
     return None
   if isinstance(struct, tuple):
     return tuple(move_struct_to_device(s, device, non_blocking)
                  for s in struct)
-  if isinstance(struct, (tsu.TensorSpecStruct, dict)):
-    out = tsu.TensorSpecStruct()
-    for k, v in (struct.items() if hasattr(struct, "items") else []):
-      out[k] = v.to(device, non_blocking=non_blocking) \
-          if isinstance(v, torch.Tensor) else v
-    slot = getattr(struct, _SLOT_ATTR, None)
-    if slot is not None and device.type == "cuda":
-      slot.mark_consumed()
-    return out
   if isinstance(struct, torch.Tensor):
     return struct.to(device, non_blocking=non_blocking)
-  return struct
+  if not isinstance(struct, (tsu.TensorSpecStruct, dict)):
+    return struct
+
+  items = list(struct.items() if hasattr(struct, "items") else [])
+  needs_copy = device.type == "cuda" and any(
+      isinstance(v, torch.Tensor) and not v.is_cuda for _, v in items)
+  out = tsu.TensorSpecStruct()
+  if not needs_copy:
+    for k, v in items:
+      out[k] = v.to(device, non_blocking=non_blocking) \
+          if isinstance(v, torch.Tensor) else v
+    return out
+
+  cur = torch.cuda.current_stream()
+  cs = _copy_stream()
+  cs.wait_stream(cur)  # dst allocations ordered after prior compute
+  with torch.cuda.stream(cs):
+    for k, v in items:
+      if isinstance(v, torch.Tensor):
+        v = v.to(device, non_blocking=True)
+      out[k] = v
+  cur.wait_stream(cs)
+  for v in out.values():
+    if isinstance(v, torch.Tensor) and v.is_cuda:
+      v.record_stream(cur)
+  slot = getattr(struct, _SLOT_ATTR, None)
+  if slot is not None:
+    # Record on the COPY stream: the slot is reusable as soon as its
+    # DMA completes, independent of downstream compute.
+    with torch.cuda.stream(cs):
+      slot.mark_consumed()
+  return out

@@ -65,13 +65,16 @@ def main():
   torch.cuda.synchronize()
   print(f"engine graphed={trainer._fast_engine.is_graphed}", flush=True)
 
+  # GPU-side segment timing via events
+  evs = [[torch.cuda.Event(enable_timing=True) for _ in range(4)]
+         for _ in range(30)]
   # instrumented manual loop (same ops as Trainer.train body)
   from tensor2robot_amd.data import pipeline as pl
   stage_t = [0.0, 0]
   orig_stage = pl._PinnedRing.stage
-  def timed_stage(self, item):
+  def timed_stage(self, item, alive=None):
     t0 = time.perf_counter()
-    out = orig_stage(self, item)
+    out = orig_stage(self, item, alive)
     stage_t[0] += time.perf_counter() - t0
     stage_t[1] += 1
     return out
@@ -85,13 +88,17 @@ def main():
     t0 = time.perf_counter()
     features, labels = next(iterator)
     t1 = time.perf_counter()
+    evs[i][0].record()
     features = pipeline.move_struct_to_device(features, trainer.device)
     labels = pipeline.move_struct_to_device(labels, trainer.device)
+    evs[i][1].record()
     t2 = time.perf_counter()
     with trainer._autocast():
       features, labels = deferred(features, labels)
+    evs[i][2].record()
     t3 = time.perf_counter()
     trainer._fast_engine.step(features, labels, trainer.global_step)
+    evs[i][3].record()
     trainer.global_step += 1
     t4 = time.perf_counter()
     t_next += t1 - t0
@@ -100,6 +107,13 @@ def main():
     t_step += t4 - t3
   torch.cuda.synchronize()
   total = time.perf_counter() - t_all0
+  g_move = sum(evs[i][0].elapsed_time(evs[i][1]) for i in range(5, n)) / (n - 5)
+  g_pre = sum(evs[i][1].elapsed_time(evs[i][2]) for i in range(5, n)) / (n - 5)
+  g_step = sum(evs[i][2].elapsed_time(evs[i][3]) for i in range(5, n)) / (n - 5)
+  g_gap = sum(evs[i - 1][3].elapsed_time(evs[i][0]) for i in range(6, n)) / (n - 6)
+  print(f"GPU segments ms: move={g_move:.3f} preproc={g_pre:.3f} "
+        f"step={g_step:.3f} inter-step gap={g_gap:.3f} "
+        f"graphed={trainer._fast_engine.is_graphed}", flush=True)
   stage_ms = stage_t[0] / max(stage_t[1], 1) * 1000
   print(f"pin={pin} nowait={bool(os.environ.get('T2R_RING_NO_WAIT'))} "
         f"total={total / n * 1000:.3f} ms/step | "
