@@ -102,7 +102,8 @@ class AbstractInputGenerator:
     self._preprocess_fn = preprocess_fn
 
   def create_dataset_input_fn(self, mode, prefetch_depth: int = 4,
-                              pin_memory: bool = False):
+                              pin_memory: bool = False,
+                              h2d_device=None):
     """Returns a zero-arg callable producing the (features, labels) iterator."""
     run_modes.validate(mode)
     if self._feature_spec is None:
@@ -115,7 +116,7 @@ class AbstractInputGenerator:
     # train() segment; rebuilding the ring re-pays hipHostMalloc).
     prefetcher = pipeline.PrefetchIterator(
         lambda: self._iterate(mode), depth=prefetch_depth,
-        pin_memory=pin_memory)
+        pin_memory=pin_memory, h2d_device=h2d_device)
 
     def input_fn() -> Iterator:
       return iter(prefetcher)

@@ -172,6 +172,31 @@ class _PinSlot:
 
 
 _SLOT_ATTR = "_t2r_pin_slot"
+_H2D_EV_ATTR = "_t2r_h2d_ev"
+
+
+def _h2d_struct(item, device):
+  """Pinned struct -> device struct via async copies (call on the copy
+  stream)."""
+  if isinstance(item, tuple):
+    return tuple(_h2d_struct(x, device) for x in item)
+  if isinstance(item, (tsu.TensorSpecStruct, dict)):
+    out = tsu.TensorSpecStruct()
+    for k, v in item.items():
+      out[k] = v.to(device, non_blocking=True) \
+          if isinstance(v, torch.Tensor) else v
+    return out
+  if isinstance(item, torch.Tensor):
+    return item.to(device, non_blocking=True)
+  return item
+
+
+def _attach_h2d_event(item, ev):
+  if isinstance(item, tuple):
+    for x in item:
+      _attach_h2d_event(x, ev)
+  elif isinstance(item, tsu.TensorSpecStruct):
+    object.__setattr__(item, _H2D_EV_ATTR, ev)
 
 
 class _PinnedRing:
@@ -183,9 +208,14 @@ class _PinnedRing:
   pinning when an item's structure/shape changes (varlen edge cases).
   """
 
-  def __init__(self, size: int):
+  def __init__(self, size: int, h2d_device=None):
     self._slots = [_PinSlot() for _ in range(max(2, size))]
     self._i = 0
+    # h2d_device: the producer also issues the H2D on the copy stream
+    # right after staging, so transfers overlap the consumer's compute
+    # (an H2D issued at consumption time serializes against the
+    # previous step and runs ~10x slower under HBM contention).
+    self._h2d = torch.device(h2d_device) if h2d_device else None
 
   _DEBUG = bool(__import__("os").environ.get("T2R_RING_DEBUG"))
 
@@ -242,6 +272,17 @@ class _PinnedRing:
       return stage_tensor(prefix, s)
 
     out = stage_struct("", item)
+    if self._h2d is not None:
+      cs = _copy_stream()
+      with torch.cuda.stream(cs):
+        dev = _h2d_struct(out, self._h2d)
+      ev = torch.cuda.Event(blocking=True)
+      ev.record(cs)
+      # The pinned slot is reusable once its H2D READ completed —
+      # no consumer-side marking needed in this mode.
+      slot.event = ev
+      _attach_h2d_event(dev, ev)
+      out = dev
     if self._DEBUG:
       import time as _t
       total = _t.perf_counter() - t0
@@ -265,11 +306,12 @@ class PrefetchIterator:
   _SENTINEL = object()
 
   def __init__(self, source_fn: Callable[[], Iterator], depth: int = 4,
-               pin_memory: bool = False):
+               pin_memory: bool = False, h2d_device=None):
     self._source_fn = source_fn
     self._depth = depth
     self._pin = pin_memory and torch.cuda.is_available()
-    self._ring = _PinnedRing(depth + 3) if self._pin else None
+    self._ring = _PinnedRing(depth + 3, h2d_device=h2d_device) \
+        if self._pin else None
     self._generation = 0
 
   def __iter__(self):
@@ -355,6 +397,14 @@ def move_struct_to_device(struct, device, non_blocking=True):
       isinstance(v, torch.Tensor) and not v.is_cuda for _, v in items)
   out = tsu.TensorSpecStruct()
   if not needs_copy:
+    ev = getattr(struct, _H2D_EV_ATTR, None)
+    if ev is not None and device.type == "cuda":
+      # Producer-prefetched H2D: one cheap cross-stream wait.
+      cur = torch.cuda.current_stream()
+      cur.wait_event(ev)
+      for _, v in items:
+        if isinstance(v, torch.Tensor) and v.is_cuda:
+          v.record_stream(cur)
     for k, v in items:
       out[k] = v.to(device, non_blocking=non_blocking) \
           if isinstance(v, torch.Tensor) else v

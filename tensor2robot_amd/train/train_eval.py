@@ -530,7 +530,9 @@ def train_eval_model(t2r_model=None,
     # runs the preprocessor on the GPU right after H2D transfer.
     device_preprocess_fn = input_generator_train.defer_preprocessing()
   train_input_fn = input_generator_train.create_dataset_input_fn(
-      TRAIN, pin_memory=trainer.device.type == "cuda")
+      TRAIN, pin_memory=trainer.device.type == "cuda",
+      h2d_device=str(trainer.device)
+      if trainer.device.type == "cuda" else None)
 
   if input_generator_eval is None or not trainer.is_chief:
     result = trainer.train(train_input_fn, max_train_steps,

@@ -413,7 +413,8 @@ def test_train_eval_model_step_parity_with_device_pool():
   gen.set_specification_from_model(model_a, run_modes.TRAIN)
   deferred = gen.defer_preprocessing()
   input_fn_a = gen.create_dataset_input_fn(run_modes.TRAIN,
-                                           pin_memory=True)
+                                           pin_memory=True,
+                                           h2d_device="cuda")
   ms_pipeline = timed_steps(trainer_a, input_fn_a, deferred, 30)
 
   # Path B: device-resident pool (bench.py's shape).

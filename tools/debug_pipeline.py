@@ -58,8 +58,9 @@ def main():
   gen.set_specification_from_model(model, run_modes.TRAIN)
   deferred = gen.defer_preprocessing()
   pin = not os.environ.get("T2R_NO_PIN")
-  input_fn = gen.create_dataset_input_fn(run_modes.TRAIN,
-                                         pin_memory=pin)
+  input_fn = gen.create_dataset_input_fn(
+      run_modes.TRAIN, pin_memory=pin,
+      h2d_device="cuda" if pin else None)
   # settle
   trainer.train(input_fn, 25, preprocess_fn=deferred)
   torch.cuda.synchronize()
