@@ -48,3 +48,23 @@ for _ in range(20):
     s.to(dev, non_blocking=True)
   torch.cuda.synchronize()
 print(f"9 small pinned copies: {(time.perf_counter()-t0)/20*1e3:.3f} ms/round")
+
+# H2D concurrently with busy compute (the real pipeline condition).
+a = torch.randn(8192, 8192, device=dev).to(torch.bfloat16)
+b = torch.randn(8192, 8192, device=dev).to(torch.bfloat16)
+comp = torch.cuda.Stream()
+copy_s = torch.cuda.Stream()
+dst = torch.empty(x.shape, dtype=x.dtype, device=dev)
+torch.cuda.synchronize()
+with torch.cuda.stream(comp):
+  for _ in range(60):
+    a @ b  # keep CUs + HBM busy ~hundreds of ms
+t0 = time.perf_counter()
+n = 20
+with torch.cuda.stream(copy_s):
+  for _ in range(n):
+    dst.copy_(main_pin, non_blocking=True)
+copy_s.synchronize()
+t = (time.perf_counter() - t0) / n
+print(f"H2D under GEMM load: {t*1e3:7.3f} ms {mb/1e3/t:6.1f} GB/s")
+torch.cuda.synchronize()
