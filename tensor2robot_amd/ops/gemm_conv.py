@@ -56,15 +56,16 @@ class _GemmConvFunction(torch.autograd.Function):
     # channels_last [N,K,OH,OW] storage IS [M,K] row-major: GEMM
     # writes straight into it, no copy.
     torch.matmul(col, wm, out=y.permute(0, 2, 3, 1).reshape(-1, k))
-    ctx.save_for_backward(x, weight, wm,
-                          col if not one_by_one else x)
+    # Save x, NOT col: col is R*S*x bytes (multi-GB on big-spatial
+    # ResNet towers) and is a cheap gather to recompute in backward.
+    ctx.save_for_backward(x, weight, wm)
     ctx.conf = (stride, pad, one_by_one)
     return y
 
   @staticmethod
   def backward(ctx, dy):
     ext = ops_mod.require_hip()
-    x, weight, wm, col_saved = ctx.saved_tensors
+    x, weight, wm = ctx.saved_tensors
     stride, pad, one_by_one = ctx.conf
     n, c, h, w = x.shape
     k, _, r, s = weight.shape
@@ -73,9 +74,9 @@ class _GemmConvFunction(torch.autograd.Function):
     dy_mat = dy.permute(0, 2, 3, 1).reshape(-1, k)
     dx = dw = None
     if one_by_one:
-      col = col_saved.permute(0, 2, 3, 1).reshape(-1, c)
+      col = x.permute(0, 2, 3, 1).reshape(-1, c)
     else:
-      col = col_saved
+      col = ext.im2col_nhwc(x, r, s, pad, stride)
     if ctx.needs_input_grad[1]:
       m = col.shape[0]
       if m >= 65536:
