@@ -20,7 +20,11 @@ this module covers what those kernels do not.
 
 from __future__ import annotations
 
+import os as _os
+
 import torch
+
+_DBG = bool(_os.environ.get("T2R_GEMM_DEBUG"))
 
 from tensor2robot_amd import ops as ops_mod
 
@@ -39,6 +43,9 @@ class _GemmConvFunction(torch.autograd.Function):
   def forward(ctx, x, weight, stride, pad):
     ext = ops_mod.require_hip()
     if not x.is_contiguous(memory_format=torch.channels_last):
+      if _DBG:
+        print(f"# gemm fwd x copy {tuple(x.shape)} strides={x.stride()}",
+              flush=True)
       x = x.contiguous(memory_format=torch.channels_last)
     n, c, h, w = x.shape
     k, _, r, s = weight.shape
@@ -69,6 +76,9 @@ class _GemmConvFunction(torch.autograd.Function):
     stride, pad, one_by_one = ctx.conf
     n, c, h, w = x.shape
     k, _, r, s = weight.shape
+    if _DBG and not dy.is_contiguous(memory_format=torch.channels_last):
+      print(f"# gemm bwd dy copy {tuple(dy.shape)} strides={dy.stride()}",
+            flush=True)
     dy = dy.contiguous(memory_format=torch.channels_last) \
         .to(torch.bfloat16)
     dy_mat = dy.permute(0, 2, 3, 1).reshape(-1, k)
@@ -97,11 +107,15 @@ class _GemmConvFunction(torch.autograd.Function):
       dw = dw_mat.reshape(r, s, c, k).permute(3, 2, 0, 1) \
           .contiguous().to(weight.dtype)
     if ctx.needs_input_grad[0]:
-      dcol = dy_mat @ wm.t()                       # [M, RS*C]
       if one_by_one:
+        # GEMM writes straight into the cl dx storage — the earlier
+        # dcol-then-copy_ paid a full extra dx pass per 1x1 conv
+        # (one of the top elementwise slices in the G2V profile).
         dx = torch.empty_like(x)
-        dx.permute(0, 2, 3, 1).reshape(-1, c).copy_(dcol)
+        torch.matmul(dy_mat, wm.t(),
+                     out=dx.permute(0, 2, 3, 1).reshape(-1, c))
       else:
+        dcol = dy_mat @ wm.t()                     # [M, RS*C]
         dx = ext.col2im_nhwc(dcol, n, c, h, w, r, s, pad, stride)
     return dx, dw, None, None
 

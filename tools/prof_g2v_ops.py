@@ -43,11 +43,17 @@ for _ in range(6):
 torch.cuda.synchronize()
 
 from torch.profiler import profile, ProfilerActivity
-with profile(activities=[ProfilerActivity.CUDA],
+with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
              record_shapes=True) as prof:
   for _ in range(3):
     step()
   torch.cuda.synchronize()
 
-print(prof.key_averages(group_by_input_shape=True).table(
-    sort_by="self_cuda_time_total", row_limit=25, max_src_column_width=40))
+avgs = prof.key_averages(group_by_input_shape=True)
+rows = sorted(avgs, key=lambda e: -e.self_device_time_total)
+for e in rows[:60]:
+  if any(e.key.startswith(p) for p in
+         ("aten::copy_", "aten::contiguous", "aten::clone", "aten::to",
+          "aten::_to_copy", "aten::cat", "aten::add")):
+    print(f"{e.self_device_time_total/1e3:9.3f} ms x{e.count:5d} "
+          f"{e.key:20s} {str(e.input_shapes)[:80]}")
