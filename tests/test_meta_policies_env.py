@@ -108,3 +108,68 @@ def test_scheduled_exploration_maml_policy():
   state = np.zeros((64, 64, 3), np.uint8)
   action, _ = policy.sample_action(state)
   assert action.shape == (2,)
+
+
+def test_gym_env_adapters_both_apis():
+  """envs.GymEnvAdapter maps classic-gym and gymnasium step/reset
+  signatures onto the run_env contract (reference run_env.py:50-74)."""
+  from tensor2robot_amd import envs as envs_mod
+
+  class OldGym:
+    def reset(self):
+      return [0.0]
+
+    def step(self, action):
+      return [1.0], 2.0, False, {"k": 1}
+
+  class NewGym:
+    def reset(self):
+      return [0.0], {"info": True}
+
+    def step(self, action):
+      return [1.0], 2.0, False, True, {"k": 2}
+
+  for env_cls, key in ((OldGym, 1), (NewGym, 2)):
+    ad = envs_mod.GymEnvAdapter(env_cls())
+    assert ad.reset() == [0.0]
+    obs, r, done, dbg = ad.step(0)
+    assert obs == [1.0] and r == 2.0 and dbg["k"] == key
+    assert done == (env_cls is NewGym)  # truncated=True in NewGym
+
+  limited = envs_mod.TimeLimitWrapper(
+      envs_mod.GymEnvAdapter(OldGym()), max_episode_steps=3)
+  limited.reset()
+  dones = [limited.step(0)[2] for _ in range(3)]
+  assert dones == [False, False, True]
+
+
+def test_run_env_with_adapter(tmp_path):
+  """The episode runner drives an adapted env end to end."""
+  import numpy as np
+  from tensor2robot_amd import envs as envs_mod
+  from tensor2robot_amd.research.dql_grasping_lib import run_env as re_mod
+
+  class TinyGym:
+    def __init__(self):
+      self.t = 0
+
+    def reset(self):
+      self.t = 0
+      return np.zeros(2, np.float32)
+
+    def step(self, action):
+      self.t += 1
+      return (np.full(2, self.t, np.float32), 1.0, self.t >= 4, {})
+
+  class RandomPolicy:
+    def reset(self):
+      pass
+
+    def sample_action(self, obs, explore_prob):
+      return np.zeros(1, np.float32)
+
+  env = envs_mod.TimeLimitWrapper(envs_mod.GymEnvAdapter(TinyGym()),
+                                  max_episode_steps=10)
+  rewards = re_mod.run_env(env, policy=RandomPolicy(), num_episodes=2,
+                           root_dir=str(tmp_path))
+  assert rewards == [4.0, 4.0]
