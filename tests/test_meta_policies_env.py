@@ -166,7 +166,7 @@ def test_run_env_with_adapter(tmp_path):
       pass
 
     def sample_action(self, obs, explore_prob):
-      return np.zeros(1, np.float32)
+      return np.zeros(1, np.float32), {}
 
   env = envs_mod.TimeLimitWrapper(envs_mod.GymEnvAdapter(TinyGym()),
                                   max_episode_steps=10)
