@@ -371,15 +371,25 @@ at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
       return v == nullptr || v[0] != '8';
     }();
     if (C == 64 && K == 64 && use_ring && small_waves) {
+      static const bool setprio = []() {
+        const char* v = std::getenv("T2R_RING_SETPRIO");
+        return v != nullptr && v[0] == '1';
+      }();
       const int th8 = (OH + 7) / 8;
       const long grid8 = (long)N * th8 * tiles_w;
-      hipLaunchKernelGGL((conv_s1_nhwc_ring_kernel<4, 2, 3, 4>),
-                         dim3(grid8), dim3(256), 0, stream.stream(),
-                         (const cbf16_t*)x.data_ptr(),
-                         (const cbf16_t*)wpk.data_ptr(),
-                         (cbf16_t*)y.data_ptr(),
-                         N, H, W, (int)K, (int)R, (int)S, (int)pad,
-                         OH, OW, th8, tiles_w);
+      auto launch8 = [&](auto kern) {
+        hipLaunchKernelGGL(kern, dim3(grid8), dim3(256), 0,
+                           stream.stream(),
+                           (const cbf16_t*)x.data_ptr(),
+                           (const cbf16_t*)wpk.data_ptr(),
+                           (cbf16_t*)y.data_ptr(),
+                           N, H, W, (int)K, (int)R, (int)S, (int)pad,
+                           OH, OW, th8, tiles_w);
+      };
+      if (setprio)
+        launch8((conv_s1_nhwc_ring_kernel<4, 2, 3, 4, true>));
+      else
+        launch8((conv_s1_nhwc_ring_kernel<4, 2, 3, 4, false>));
     } else if (C == 64 && K == 64)
       use_ring ? (deep_ring ? launch(conv_s1_nhwc_ring_kernel<4, 2, 6>)
                             : launch(conv_s1_nhwc_ring_kernel<4, 2, 3>))
@@ -550,7 +560,8 @@ __device__ __forceinline__ void conv_waitcnt_vm(int count) {
 // a 256-thread WG owns an 8x16 tile -> 2 independent WGs/CU with
 // uncoupled barriers, at the cost of streaming each weight chunk
 // twice per 256 output pixels.
-template <int C16N, int NTILES, int RING_DEPTH, int WAVES>
+template <int C16N, int NTILES, int RING_DEPTH, int WAVES,
+          bool SETPRIO = false>
 __global__ void __launch_bounds__(WAVES * 64, 2)
 conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
                          const cbf16_t* __restrict__ wpk,
@@ -646,6 +657,9 @@ conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
     if (issuer)
       conv_waitcnt_vm(min(RS - 1 - rs, RING_DEPTH - 1) * PPW);
     __builtin_amdgcn_s_barrier();     // chunk rs landed for everyone
+    // T5 (guide): favor the MFMA cluster while issuer waves run their
+    // global_load_lds issue phase (A/B via T2R_RING_SETPRIO).
+    if constexpr (SETPRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int c16 = 0; c16 < C16N; ++c16) {
       cbf16x8 a_frag = *reinterpret_cast<const cbf16x8*>(
@@ -660,6 +674,7 @@ conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
             a_frag, b_frag, acc[nt], 0, 0, 0);
       }
     }
+    if constexpr (SETPRIO) __builtin_amdgcn_s_setprio(0);
     // Slot rs%RING_DEPTH is refilled by the DMA issued at iteration
     // rs+1 (chunk rs+RING_DEPTH): everyone must be done reading
     // before that DMA can be issued.
