@@ -295,7 +295,8 @@ conv_s1_nhwc_big_kernel(const cbf16_t* __restrict__ x,
   }
 }
 
-template <int C16N, int NTILES, int RING_DEPTH, int WAVES = 8>
+template <int C16N, int NTILES, int RING_DEPTH, int WAVES = 8,
+          bool SETPRIO = false>
 __global__ void conv_s1_nhwc_ring_kernel(
     const cbf16_t* __restrict__ x, const cbf16_t* __restrict__ wpk,
     cbf16_t* __restrict__ y, int N, int H, int W, int K, int R, int S,
@@ -377,19 +378,23 @@ at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
       }();
       const int th8 = (OH + 7) / 8;
       const long grid8 = (long)N * th8 * tiles_w;
-      auto launch8 = [&](auto kern) {
-        hipLaunchKernelGGL(kern, dim3(grid8), dim3(256), 0,
-                           stream.stream(),
+      if (setprio) {
+        hipLaunchKernelGGL((conv_s1_nhwc_ring_kernel<4, 2, 3, 4, true>),
+                           dim3(grid8), dim3(256), 0, stream.stream(),
                            (const cbf16_t*)x.data_ptr(),
                            (const cbf16_t*)wpk.data_ptr(),
                            (cbf16_t*)y.data_ptr(),
                            N, H, W, (int)K, (int)R, (int)S, (int)pad,
                            OH, OW, th8, tiles_w);
-      };
-      if (setprio)
-        launch8((conv_s1_nhwc_ring_kernel<4, 2, 3, 4, true>));
-      else
-        launch8((conv_s1_nhwc_ring_kernel<4, 2, 3, 4, false>));
+      } else {
+        hipLaunchKernelGGL((conv_s1_nhwc_ring_kernel<4, 2, 3, 4, false>),
+                           dim3(grid8), dim3(256), 0, stream.stream(),
+                           (const cbf16_t*)x.data_ptr(),
+                           (const cbf16_t*)wpk.data_ptr(),
+                           (cbf16_t*)y.data_ptr(),
+                           N, H, W, (int)K, (int)R, (int)S, (int)pad,
+                           OH, OW, th8, tiles_w);
+      }
     } else if (C == 64 && K == 64)
       use_ring ? (deep_ring ? launch(conv_s1_nhwc_ring_kernel<4, 2, 6>)
                             : launch(conv_s1_nhwc_ring_kernel<4, 2, 3>))
@@ -561,7 +566,7 @@ __device__ __forceinline__ void conv_waitcnt_vm(int count) {
 // uncoupled barriers, at the cost of streaming each weight chunk
 // twice per 256 output pixels.
 template <int C16N, int NTILES, int RING_DEPTH, int WAVES,
-          bool SETPRIO = false>
+          bool SETPRIO>
 __global__ void __launch_bounds__(WAVES * 64, 2)
 conv_s1_nhwc_ring_kernel(const cbf16_t* __restrict__ x,
                          const cbf16_t* __restrict__ wpk,
