@@ -23,6 +23,7 @@ import torch
 from torch import nn
 
 from tensor2robot_amd import gin
+from tensor2robot_amd.specs import tensorspec_utils as tsu
 
 
 @contextlib.contextmanager
@@ -47,6 +48,28 @@ def swap_parameters(module: nn.Module, fast: Dict[str, torch.Tensor]):
   finally:
     for mod, leaf, original in reversed(saved):
       mod._parameters[leaf] = original
+
+
+@contextlib.contextmanager
+def freeze_running_stats(module: nn.Module):
+  """Disable BatchNorm running-stat updates (buffer mutation).
+
+  torch.func transforms forbid in-place mutation of captured tensors;
+  train-mode BN normalizes by batch statistics either way, so under the
+  task-parallel inner loop only the running-average bookkeeping is
+  skipped — the same restriction TF's parallel_for put on stateful ops.
+  """
+  flipped = []
+  for m in module.modules():
+    if isinstance(m, nn.modules.batchnorm._BatchNorm) and \
+        m.track_running_stats:
+      m.track_running_stats = False
+      flipped.append(m)
+  try:
+    yield
+  finally:
+    for m in flipped:
+      m.track_running_stats = True
 
 
 @gin.configurable
@@ -172,3 +195,116 @@ class MAMLInnerLoopGradientDescent:
                                          labels=val_labels, mode=mode,
                                          params=params)
     return [unconditioned, conditioned], inner_outputs, inner_losses
+
+  # ---------------------------------------------------------------------
+  # Task-parallel inner loop: torch.func.vmap over the task dimension.
+  # Reference maml_model.py:229-260 maps `task_learn` over tasks with
+  # parallel_for when use_parallel_for=True; the MI355X equivalent is
+  # vmap(grad(...)) — one fused launch per op across all tasks instead
+  # of num_tasks small launches, which is what actually fills 256 CUs
+  # when per-task sample counts are small.
+  # ---------------------------------------------------------------------
+  def inner_loop_vmapped(
+      self,
+      cond_tensors: Dict[str, torch.Tensor],
+      cond_label_tensors: Dict[str, torch.Tensor],
+      inf_tensors: Dict[str, torch.Tensor],
+      inf_label_tensors: Dict[str, torch.Tensor],
+      static: Dict[str, Dict],
+      inference_network_fn: Callable,
+      model_train_fn: Callable,
+      network: nn.Module,
+      num_steps: int,
+      mode=None,
+      params=None,
+  ):
+    """All tasks adapt simultaneously under torch.func.vmap.
+
+    The `*_tensors` dicts are flat key->tensor maps with a leading task
+    dimension; `static` holds the non-tensor leaves per input role
+    ("cond_f", "cond_l", "inf_f", "inf_l") shared across tasks.
+
+    Semantics match `inner_loop` exactly (same data for every
+    adaptation step, final monitoring forward, conditioned and
+    unconditioned val forwards); returns
+    (uncond_flat, cond_flat, inner_outputs_flat_list, losses) where
+    every tensor has the task dimension back in front and `losses` is
+    [tasks, num_steps+1].
+
+    Constraint: the base network must be buffer-mutation free under
+    vmap (BatchNorm running stats in train mode will raise) — same
+    restriction TF's parallel_for imposed on stateful ops.
+    """
+    from torch.func import grad as func_grad
+    from torch.func import vmap
+
+    params = dict(params or {})
+    params["is_inner_loop"] = True
+    theta = dict(network.named_parameters())
+    adapt_names = [n for n in theta if self._adapts(n)]
+
+    def _mk(tensors: Dict[str, torch.Tensor], role: str):
+      s = tsu.TensorSpecStruct()
+      for k, v in tensors.items():
+        s[k] = v
+      for k, v in static.get(role, {}).items():
+        s[k] = v
+      return s
+
+    def _flat_tensors(outputs):
+      flat = tsu.flatten_spec_structure(outputs)
+      return {k: v for k, v in flat.items()
+              if isinstance(v, torch.Tensor)}
+
+    def _fwd_loss(fast, cf, cl, run_params):
+      full = dict(theta)
+      full.update(fast)
+      with torch.enable_grad(), swap_parameters(network, full):
+        outputs = inference_network_fn(
+            features=_mk(cf, "cond_f"), labels=_mk(cl, "cond_l"),
+            mode=mode, params=run_params)
+        loss = self._extract_train_loss(model_train_fn(
+            features=_mk(cf, "cond_f"), labels=_mk(cl, "cond_l"),
+            inference_outputs=outputs, mode=mode, params=run_params))
+      # grad(has_aux=True) hands back only the aux — carry the loss in it.
+      return loss, (loss, _flat_tensors(outputs))
+
+    def task_learn(cf, cl, inff, infl):
+      fast = {n: theta[n] for n in adapt_names}
+      inner_losses: List[torch.Tensor] = []
+      inner_outs: List[Dict[str, torch.Tensor]] = []
+      for _ in range(num_steps):
+        grads, (loss, outs) = func_grad(
+            lambda fp: _fwd_loss(fp, cf, cl, params), has_aux=True)(fast)
+        new_fast = {}
+        for n in adapt_names:
+          g = grads[n]
+          if not self._use_second_order:
+            g = g.detach()  # first-order MAML (reference :184-185)
+          new_fast[n] = fast[n] - self._lr(n, g.device) * g
+        fast = new_fast
+        inner_losses.append(loss)
+        inner_outs.append(outs)
+      # Monitoring forward on the condition data under adapted weights
+      # (reference :290-306) — loss computed inside the swap.
+      _, (loss_m, outs_m) = _fwd_loss(fast, cf, cl, params)
+      inner_losses.append(loss_m)
+      inner_outs.append(outs_m)
+      # Conditioned val forward under the adapted weights.
+      params_cond = dict(params)
+      params_cond["is_inner_loop"] = False
+      full = dict(theta)
+      full.update(fast)
+      with swap_parameters(network, full):
+        cond = _flat_tensors(inference_network_fn(
+            features=_mk(inff, "inf_f"), labels=_mk(infl, "inf_l"),
+            mode=mode, params=params_cond))
+      # Unconditioned val forward under the original weights (:321-324).
+      uncond = _flat_tensors(inference_network_fn(
+          features=_mk(inff, "inf_f"), labels=_mk(infl, "inf_l"),
+          mode=mode, params=params))
+      return uncond, cond, inner_outs, torch.stack(inner_losses)
+
+    with freeze_running_stats(network):
+      return vmap(task_learn, randomness="different")(
+          cond_tensors, cond_label_tensors, inf_tensors, inf_label_tensors)

@@ -63,11 +63,16 @@ class MAMLModel(abstract_model.AbstractT2RModel):
                use_second_order: bool = True,
                learn_inner_lr: bool = False,
                inner_var_scope: Optional[str] = None,
+               parallel_tasks: bool = False,
                **kwargs):
     super().__init__(**kwargs)
     self._base_model = base_model
     self._maml_preprocessor_cls = preprocessor_cls
     self._num_inner_loop_steps = max(1, num_inner_loop_steps)
+    # Reference maml_model.py use_parallel_for: adapt all tasks at once
+    # (vmap) instead of a Python loop.  Requires a base network whose
+    # forward doesn't mutate buffers (e.g. no train-mode BatchNorm).
+    self._parallel_tasks = parallel_tasks
     self._var_scope = var_scope
     self._inner_learning_rate = inner_learning_rate
     self._use_second_order = use_second_order
@@ -126,6 +131,10 @@ class MAMLModel(abstract_model.AbstractT2RModel):
 
     num_tasks = next(iter(
         tsu.flatten_spec_structure(cond_f).values())).shape[0]
+    if self._parallel_tasks:
+      return self._inference_vmapped(cond_f, cond_l, inf_f,
+                                     unused_inference_labels, num_tasks,
+                                     mode, params)
     per_task_uncond, per_task_cond = [], []
     per_task_inner: List[List] = [
         [] for _ in range(self._num_inner_loop_steps + 1)]
@@ -174,6 +183,57 @@ class MAMLModel(abstract_model.AbstractT2RModel):
     if "inference_output" not in predictions:
       raise ValueError("The required inference_output is not in "
                        f"predictions {list(predictions.keys())}")
+    return predictions
+
+  def _inference_vmapped(self, cond_f, cond_l, inf_f, inf_l, num_tasks,
+                         mode, params):
+    """Task-parallel adaptation path (reference use_parallel_for).
+
+    Splits each input struct into its tensor leaves (vmapped over the
+    task dim) and non-tensor leaves (closed over), runs
+    `inner_loop_vmapped`, and assembles the identical prediction keys
+    the per-task Python loop produces — vmap's output stacking replaces
+    `_stack_structs`.
+    """
+    def _split(struct):
+      flat = tsu.flatten_spec_structure(struct)
+      tensors = {k: v for k, v in flat.items()
+                 if isinstance(v, torch.Tensor)}
+      static = {k: v for k, v in flat.items()
+                if not isinstance(v, torch.Tensor)}
+      return tensors, static
+
+    cf_t, cf_s = _split(cond_f)
+    cl_t, cl_s = _split(cond_l)
+    if_t, if_s = _split(inf_f)
+    il_t, il_s = _split(inf_l)
+    static = {"cond_f": cf_s, "cond_l": cl_s, "inf_f": if_s,
+              "inf_l": il_s}
+    uncond, cond, inner_outs, losses = self._inner_loop.inner_loop_vmapped(
+        cf_t, cl_t, if_t, il_t, static,
+        inference_network_fn=self._base_model.inference_network_fn,
+        model_train_fn=self._base_model.model_train_fn,
+        network=self._base_model.network,
+        num_steps=self._num_inner_loop_steps, mode=mode, params=params)
+
+    predictions = tsu.TensorSpecStruct()
+    for k, v in inner_outs[0].items():
+      predictions["full_condition_output/" + k] = v
+    for pos, outs in enumerate(inner_outs):
+      for k, v in outs.items():
+        predictions[f"full_condition_outputs/output_{pos}/" + k] = v
+    for k, v in uncond.items():
+      predictions["full_inference_output_unconditioned/" + k] = v
+    for k, v in cond.items():
+      predictions["full_inference_output/" + k] = v
+    loss_means = losses.detach().mean(dim=0)  # [num_steps + 1]
+    for i in range(loss_means.shape[0]):
+      self.scalar_summary(f"inner_loss_{i}", loss_means[i])
+    predictions = self._select_inference_output(predictions)
+    for required in ("condition_output", "inference_output"):
+      if required not in predictions:
+        raise ValueError(f"The required {required} is not in "
+                         f"predictions {list(predictions.keys())}")
     return predictions
 
   def _select_inference_output(self, predictions):

@@ -241,6 +241,64 @@ def test_maml_model_specs_and_preprocessor():
       tsu.flatten_spec_structure(out_spec)
 
 
+@pytest.mark.parametrize("second_order", [True, False])
+def test_maml_parallel_tasks_matches_loop(second_order):
+  """vmap path (parallel_tasks) == per-task Python loop, incl. grads."""
+  torch.manual_seed(0)
+  base_loop = mocks.MockT2RModel()
+  model_loop = _MockMAML(base_model=base_loop, device_type="cpu",
+                         compute_dtype="float32", num_inner_loop_steps=2,
+                         use_second_order=second_order)
+  torch.manual_seed(0)
+  base_vmap = mocks.MockT2RModel()
+  model_vmap = _MockMAML(base_model=base_vmap, device_type="cpu",
+                         compute_dtype="float32", num_inner_loop_steps=2,
+                         use_second_order=second_order,
+                         parallel_tasks=True)
+  features, labels = _meta_batch(tasks=3, samples=4, seed=5)
+  # Networks materialize lazily — force both, then share identical
+  # weights so the two paths start from the same theta.
+  _ = model_loop.network
+  _ = model_vmap.network
+  base_vmap.network.load_state_dict(base_loop.network.state_dict())
+  ops_loop = model_loop.model_fn(features, labels, run_modes.TRAIN)
+  ops_vmap = model_vmap.model_fn(features, labels, run_modes.TRAIN)
+  # Same parameters (same seed) -> identical predictions and loss.
+  for k, v in ops_loop.inference_outputs.items():
+    if isinstance(v, torch.Tensor):
+      torch.testing.assert_close(
+          ops_vmap.inference_outputs[k], v, rtol=1e-5, atol=1e-6,
+          msg=lambda m, key=k: f"{key}: {m}")
+  torch.testing.assert_close(ops_vmap.loss, ops_loop.loss,
+                             rtol=1e-5, atol=1e-6)
+  # Outer gradients (incl. second-order term) must agree.
+  ops_loop.loss.backward()
+  ops_vmap.loss.backward()
+  for (n, p_l), (_, p_v) in zip(
+      base_loop.network.named_parameters(),
+      base_vmap.network.named_parameters()):
+    if p_l.grad is None:
+      assert p_v.grad is None or not torch.any(p_v.grad)
+      continue
+    # fp32 accumulation order differs between batched and looped ops.
+    torch.testing.assert_close(p_v.grad, p_l.grad, rtol=1e-3, atol=1e-5,
+                               msg=lambda m, name=n: f"{name}: {m}")
+
+
+def test_maml_parallel_tasks_inner_var_scope():
+  base = mocks.MockT2RModel()
+  model = _MockMAML(base_model=base, device_type="cpu",
+                    compute_dtype="float32", num_inner_loop_steps=1,
+                    inner_var_scope="stack.0", parallel_tasks=True)
+  features, labels = _meta_batch(tasks=2, samples=4, seed=7)
+  ops = model.model_fn(features, labels, run_modes.TRAIN)
+  assert torch.isfinite(ops.loss)
+  cond = ops.inference_outputs["full_inference_output/prediction"]
+  uncond = ops.inference_outputs[
+      "full_inference_output_unconditioned/prediction"]
+  assert not torch.allclose(cond, uncond)
+
+
 def test_maml_model_eval_fn():
   base = mocks.MockT2RModel()
   model = _MockMAML(base_model=base, device_type="cpu",
