@@ -36,11 +36,18 @@ def _pool() -> concurrent.futures.ThreadPoolExecutor:
 
 
 def _huffman_batch(records: Sequence[bytes]) -> List[dict]:
+  import os
   from tensor2robot_amd.data import image_codec
   native = image_codec.native_module()
+  cores = os.cpu_count() or 4
+  # Across-image threads come from the pool; when the batch is smaller
+  # than the core count, spare cores decode restart-marker segments
+  # WITHIN each image (no-op on streams without RST markers).
+  per_image = max(1, cores // max(1, len(records)))
   if len(records) <= 1:
-    return [native.decode_jpeg_coeffs(r) for r in records]
-  return list(_pool().map(native.decode_jpeg_coeffs, records))
+    return [native.decode_jpeg_coeffs(r, per_image) for r in records]
+  return list(_pool().map(
+      lambda r: native.decode_jpeg_coeffs(r, per_image), records))
 
 
 def _geometry_key(ci: dict):

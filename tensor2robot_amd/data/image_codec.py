@@ -138,7 +138,10 @@ def _load_jpeg_native():
   return _jpeg_native
 
 
-def encode_jpeg(image: np.ndarray, quality: int = 90) -> bytes:
+def encode_jpeg(image: np.ndarray, quality: int = 90,
+                restart_interval: int = 0) -> bytes:
+  """restart_interval > 0 emits RSTn markers every N MCUs so the
+  Huffman scan parallelizes at decode time (gpu_jpeg segment decode)."""
   native = _load_jpeg_native()
   if native is None:
     raise RuntimeError(
@@ -147,7 +150,7 @@ def encode_jpeg(image: np.ndarray, quality: int = 90) -> bytes:
   image = np.ascontiguousarray(image, dtype=np.uint8)
   if image.ndim == 2:
     image = image[:, :, None]
-  return native.encode_jpeg(image, quality)
+  return native.encode_jpeg(image, quality, restart_interval)
 
 
 def decode_jpeg(data: bytes) -> np.ndarray:

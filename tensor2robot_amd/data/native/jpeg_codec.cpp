@@ -7,11 +7,13 @@
 // luma sampling, restart markers; encoder emits 4:4:4 YCbCr (or
 // grayscale) with the Annex-K quantization/huffman tables.
 
+#include <algorithm>
 #include <cmath>
 #include <cstdint>
 #include <cstring>
 #include <stdexcept>
 #include <string>
+#include <thread>
 #include <vector>
 
 #include "jpeg_codec.h"
@@ -208,7 +210,7 @@ static void emit_u16(std::vector<uint8_t>& out, uint16_t v) {
 }
 
 std::vector<uint8_t> encode(const uint8_t* rgb, int h, int w, int channels,
-                            int quality) {
+                            int quality, int restart_interval) {
   if (channels != 1 && channels != 3)
     throw std::runtime_error("JPEG encode: channels must be 1 or 3");
   if (quality < 1) quality = 1;
@@ -270,6 +272,12 @@ std::vector<uint8_t> encode(const uint8_t* rgb, int h, int w, int channels,
     for (int i = 1; i <= 16; ++i) out.push_back(s.bits[i]);
     for (int i = 0; i < s.nvals; ++i) out.push_back(s.vals[i]);
   }
+  // DRI — independently decodable entropy segments every N MCUs.
+  if (restart_interval > 0) {
+    emit_marker(out, 0xdd);
+    emit_u16(out, 4);
+    emit_u16(out, restart_interval);
+  }
   // SOS
   emit_marker(out, 0xda);
   emit_u16(out, 6 + 2 * channels);
@@ -292,6 +300,14 @@ std::vector<uint8_t> encode(const uint8_t* rgb, int h, int w, int channels,
   float block[64], coef[64];
   for (int by = 0; by < mcus_y; ++by) {
     for (int bx = 0; bx < mcus_x; ++bx) {
+      int mcu_index = by * mcus_x + bx;
+      if (restart_interval > 0 && mcu_index &&
+          mcu_index % restart_interval == 0) {
+        bw.flush();  // byte-align, pad with 1s
+        emit_marker(out,
+                    0xd0 + ((mcu_index / restart_interval - 1) & 7));
+        pred[0] = pred[1] = pred[2] = 0;
+      }
       for (int ci = 0; ci < channels; ++ci) {
         const int* q = ci == 0 ? qluma : qchroma;
         HuffEncTable& dct = ci == 0 ? dcl : dcc;
@@ -637,7 +653,50 @@ std::vector<uint8_t> decode(const uint8_t* data, size_t size, int& out_h,
 // IDCT + upsample + color convert) runs as HIP kernels.
 // ---------------------------------------------------------------------------
 
-CoeffImage decode_coeffs(const uint8_t* data, size_t size) {
+// Decode MCUs [m0, m1) of the scan into out.comps' coefficient blocks,
+// advancing `br` and the caller's DC predictors.  Disjoint MCU ranges
+// touch disjoint blocks, so restart segments decode concurrently.
+static void decode_coeff_mcus(const HuffDecTable* dc_tabs,
+                              const HuffDecTable* ac_tabs,
+                              const Component* comps, CoeffImage& out,
+                              int ncomp, int mcux, BitReader& br, int m0,
+                              int m1, int* dc_pred) {
+  for (int m = m0; m < m1; ++m) {
+    const int my = m / mcux, mx = m % mcux;
+    for (int c = 0; c < ncomp; ++c) {
+      const Component& comp = comps[c];
+      auto& oc = out.comps[c];
+      for (int v = 0; v < comp.vs; ++v) {
+        for (int hh = 0; hh < comp.hs; ++hh) {
+          const int by = my * comp.vs + v;
+          const int bx = mx * comp.hs + hh;
+          int16_t* blk = &oc.coeffs[((size_t)by * oc.bw + bx) * 64];
+          int sym = huff_decode(br, dc_tabs[comp.td]);
+          if (sym < 0) throw std::runtime_error("JPEG: truncated");
+          dc_pred[c] += receive_extend(br, sym);
+          blk[0] = (int16_t)dc_pred[c];
+          for (int k = 1; k < 64;) {
+            int rs = huff_decode(br, ac_tabs[comp.ta]);
+            if (rs < 0) throw std::runtime_error("JPEG: truncated");
+            int run = rs >> 4, sbits = rs & 15;
+            if (sbits == 0) {
+              if (run != 15) break;  // EOB
+              k += 16;
+              continue;
+            }
+            k += run;
+            if (k > 63) throw std::runtime_error("JPEG: AC overflow");
+            blk[kZigzag[k]] = (int16_t)receive_extend(br, sbits);
+            ++k;
+          }
+        }
+      }
+    }
+  }
+}
+
+CoeffImage decode_coeffs(const uint8_t* data, size_t size,
+                         int num_threads) {
   CoeffImage out;
   size_t pos = 0;
   auto rd_u16 = [&](size_t p) -> int {
@@ -735,53 +794,77 @@ CoeffImage decode_coeffs(const uint8_t* data, size_t size) {
             (size_t)out.comps[c].bw * out.comps[c].bh * 64, 0);
         comps[c].dc_pred = 0;
       }
-      BitReader br(data, size, pos);
-      int mcu_count = 0;
-      for (int my = 0; my < mcuy; ++my) {
-        for (int mx = 0; mx < mcux; ++mx) {
-          if (restart_interval && mcu_count &&
-              mcu_count % restart_interval == 0) {
-            br.reset_to_byte();
-            while (br.pos + 1 < size && data[br.pos] == 0xff &&
-                   data[br.pos + 1] >= 0xd0 && data[br.pos + 1] <= 0xd7) {
-              br.pos += 2;
-              for (int c = 0; c < ncomp; ++c) comps[c].dc_pred = 0;
-            }
+      const int total_mcus = mcux * mcuy;
+      const int expect = restart_interval > 0
+          ? (total_mcus + restart_interval - 1) / restart_interval
+          : 1;
+      if (restart_interval > 0 && num_threads > 1 && expect > 1) {
+        // Restart markers are byte-aligned and unambiguous inside the
+        // entropy stream (0xFF 0x00 is stuffing, 0xFF 0xFF is fill), so
+        // segment starts are found by a plain byte scan — each segment
+        // then Huffman-decodes independently with fresh DC predictors.
+        std::vector<size_t> seg_starts{pos};
+        size_t p = pos;
+        while (p + 1 < size) {
+          if (data[p] != 0xff) { ++p; continue; }
+          uint8_t b = data[p + 1];
+          if (b == 0x00) { p += 2; continue; }       // stuffed byte
+          if (b >= 0xd0 && b <= 0xd7) {              // RSTn
+            seg_starts.push_back(p + 2);
+            p += 2;
+            continue;
           }
-          ++mcu_count;
-          for (int c = 0; c < ncomp; ++c) {
-            Component& comp = comps[c];
-            auto& oc = out.comps[c];
-            for (int v = 0; v < comp.vs; ++v) {
-              for (int hh = 0; hh < comp.hs; ++hh) {
-                const int by = my * comp.vs + v;
-                const int bx = mx * comp.hs + hh;
-                int16_t* blk =
-                    &oc.coeffs[((size_t)by * oc.bw + bx) * 64];
-                int sym = huff_decode(br, dc_tabs[comp.td]);
-                if (sym < 0) throw std::runtime_error("JPEG: truncated");
-                comp.dc_pred += receive_extend(br, sym);
-                blk[0] = (int16_t)comp.dc_pred;
-                for (int k = 1; k < 64;) {
-                  int rs = huff_decode(br, ac_tabs[comp.ta]);
-                  if (rs < 0)
-                    throw std::runtime_error("JPEG: truncated");
-                  int run = rs >> 4, sbits = rs & 15;
-                  if (sbits == 0) {
-                    if (run != 15) break;
-                    k += 16;
-                    continue;
-                  }
-                  k += run;
-                  if (k > 63)
-                    throw std::runtime_error("JPEG: AC overflow");
-                  blk[kZigzag[k]] = (int16_t)receive_extend(br, sbits);
-                  ++k;
-                }
+          if (b == 0xff) { ++p; continue; }          // fill byte
+          break;                                     // EOI / next marker
+        }
+        if ((int)seg_starts.size() == expect) {
+          const int nthreads = std::min(num_threads, expect);
+          std::vector<std::exception_ptr> errs(nthreads);
+          auto work = [&](int t) {
+            try {
+              for (int s = t; s < expect; s += nthreads) {
+                BitReader br(data, size, seg_starts[s]);
+                int preds[4] = {0, 0, 0, 0};
+                decode_coeff_mcus(
+                    dc_tabs, ac_tabs, comps, out, ncomp, mcux, br,
+                    s * restart_interval,
+                    std::min(total_mcus, (s + 1) * restart_interval),
+                    preds);
               }
+            } catch (...) {
+              errs[t] = std::current_exception();
             }
+          };
+          std::vector<std::thread> workers;
+          for (int t = 1; t < nthreads; ++t) workers.emplace_back(work, t);
+          work(0);
+          for (auto& th : workers) th.join();
+          for (auto& e : errs)
+            if (e) std::rethrow_exception(e);
+          return out;
+        }
+        // Marker layout didn't match DRI bookkeeping — decode
+        // sequentially below, which tolerates odd streams.
+      }
+      BitReader br(data, size, pos);
+      int preds[4] = {0, 0, 0, 0};
+      int mcu = 0;
+      while (mcu < total_mcus) {
+        if (restart_interval && mcu) {
+          br.reset_to_byte();
+          while (br.pos + 1 < size && data[br.pos] == 0xff &&
+                 data[br.pos + 1] >= 0xd0 && data[br.pos + 1] <= 0xd7) {
+            br.pos += 2;
+            preds[0] = preds[1] = preds[2] = preds[3] = 0;
           }
         }
+        const int m1 = restart_interval
+            ? std::min(total_mcus,
+                       (mcu / restart_interval + 1) * restart_interval)
+            : total_mcus;
+        decode_coeff_mcus(dc_tabs, ac_tabs, comps, out, ncomp, mcux, br,
+                          mcu, m1, preds);
+        mcu = m1;
       }
       return out;
     }

@@ -7,8 +7,12 @@
 
 namespace t2r_jpeg {
 
+// restart_interval > 0 emits a DRI segment and an RSTn marker every
+// that many MCUs — the entropy stream becomes independently decodable
+// segments, which is what lets decode_coeffs() parallelize the
+// bit-serial Huffman scan inside ONE image.
 std::vector<uint8_t> encode(const uint8_t* rgb, int h, int w, int channels,
-                            int quality);
+                            int quality, int restart_interval = 0);
 std::vector<uint8_t> decode(const uint8_t* data, size_t size, int& out_h,
                             int& out_w, int& out_c);
 
@@ -23,6 +27,9 @@ struct CoeffImage {
     std::vector<int16_t> coeffs;     // [bh][bw][64] natural order
   } comps[4];
 };
-CoeffImage decode_coeffs(const uint8_t* data, size_t size);
+// num_threads > 1 decodes restart-marker segments concurrently (falls
+// back to the sequential scan when the stream has no restart markers).
+CoeffImage decode_coeffs(const uint8_t* data, size_t size,
+                         int num_threads = 1);
 
 }  // namespace t2r_jpeg

@@ -11,16 +11,18 @@ namespace py = pybind11;
 
 static py::bytes encode_jpeg(
     py::array_t<uint8_t, py::array::c_style | py::array::forcecast> image,
-    int quality) {
+    int quality, int restart_interval) {
   if (image.ndim() == 2) {
     auto out = t2r_jpeg::encode(image.data(), image.shape(0),
-                                image.shape(1), 1, quality);
+                                image.shape(1), 1, quality,
+                                restart_interval);
     return py::bytes((const char*)out.data(), out.size());
   }
   if (image.ndim() != 3)
     throw std::runtime_error("encode_jpeg: HxW or HxWxC uint8 expected");
   auto out = t2r_jpeg::encode(image.data(), image.shape(0),
-                              image.shape(1), image.shape(2), quality);
+                              image.shape(1), image.shape(2), quality,
+                              restart_interval);
   return py::bytes((const char*)out.data(), out.size());
 }
 
@@ -39,14 +41,17 @@ static py::array decode_jpeg(py::bytes data) {
   return out;
 }
 
-static py::dict decode_jpeg_coeffs(py::bytes data) {
+static py::dict decode_jpeg_coeffs(py::bytes data, int num_threads) {
   // Huffman/entropy decode only (the GPU decode handoff); the scan is
-  // released from the GIL so batches parallelize across host threads.
+  // released from the GIL so batches parallelize across host threads,
+  // and restart-marker streams additionally parallelize WITHIN one
+  // image (num_threads segments decode concurrently).
   std::string buf = data;
   t2r_jpeg::CoeffImage ci;
   {
     py::gil_scoped_release release;
-    ci = t2r_jpeg::decode_coeffs((const uint8_t*)buf.data(), buf.size());
+    ci = t2r_jpeg::decode_coeffs((const uint8_t*)buf.data(), buf.size(),
+                                 num_threads);
   }
   py::dict out;
   out["height"] = ci.height;
@@ -76,7 +81,8 @@ static py::dict decode_jpeg_coeffs(py::bytes data) {
 PYBIND11_MODULE(_t2r_native, m) {
   m.doc() = "CPU-native codecs: baseline JPEG encode/decode";
   m.def("encode_jpeg", &encode_jpeg, py::arg("image"),
-        py::arg("quality") = 90);
+        py::arg("quality") = 90, py::arg("restart_interval") = 0);
   m.def("decode_jpeg", &decode_jpeg, py::arg("data"));
-  m.def("decode_jpeg_coeffs", &decode_jpeg_coeffs, py::arg("data"));
+  m.def("decode_jpeg_coeffs", &decode_jpeg_coeffs, py::arg("data"),
+        py::arg("num_threads") = 1);
 }

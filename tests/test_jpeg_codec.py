@@ -58,6 +58,52 @@ def test_we_decode_pil_grayscale_and_restart_markers():
   assert np.abs(dec.astype(int) - pil_dec.astype(int)).mean() < 1.5
 
 
+def test_restart_interval_encode_roundtrip_and_pil():
+  """Our encoder's RSTn streams decode identically and conform (PIL)."""
+  img = _gradient(70, 90)  # edge MCUs + restart boundaries interact
+  plain = image_codec.encode_jpeg(img, quality=95)
+  rst = image_codec.encode_jpeg(img, quality=95, restart_interval=5)
+  assert b"\xff\xdd" in rst and b"\xff\xdd" not in plain  # DRI present
+  # Identical pixels either way (same quantization, only DC-pred resets).
+  np.testing.assert_array_equal(image_codec.decode_jpeg(plain),
+                                image_codec.decode_jpeg(rst))
+  pil = np.asarray(PIL_Image.open(io.BytesIO(rst)).convert("RGB"))
+  assert np.abs(pil.astype(int) - img.astype(int)).mean() < 2.0
+
+
+@pytest.mark.parametrize("interval,threads", [(3, 4), (7, 16), (1, 2)])
+def test_parallel_segment_coeff_decode_matches(interval, threads):
+  """Segment-parallel Huffman == sequential scan, coefficient-exact."""
+  native = image_codec.native_module()
+  img = _gradient(56, 72)
+  data = image_codec.encode_jpeg(img, quality=90,
+                                 restart_interval=interval)
+  seq = native.decode_jpeg_coeffs(data, 1)
+  par = native.decode_jpeg_coeffs(data, threads)
+  assert seq["height"] == par["height"] and seq["width"] == par["width"]
+  for cs, cp in zip(seq["comps"], par["comps"]):
+    np.testing.assert_array_equal(cs["coeffs"], cp["coeffs"])
+  # num_threads on a marker-free stream is a no-op.
+  plain = image_codec.encode_jpeg(img, quality=90)
+  a = native.decode_jpeg_coeffs(plain, threads)
+  b = native.decode_jpeg_coeffs(plain, 1)
+  for cs, cp in zip(a["comps"], b["comps"]):
+    np.testing.assert_array_equal(cs["coeffs"], cp["coeffs"])
+
+
+def test_parallel_segment_decode_of_pil_restart_stream():
+  """Foreign (PIL-encoded) restart streams hit the parallel path too."""
+  native = image_codec.native_module()
+  img = _gradient(64, 64)
+  buf = io.BytesIO()
+  PIL_Image.fromarray(img).save(buf, format="JPEG", quality=92,
+                                subsampling=0, restart_marker_rows=1)
+  seq = native.decode_jpeg_coeffs(buf.getvalue(), 1)
+  par = native.decode_jpeg_coeffs(buf.getvalue(), 8)
+  for cs, cp in zip(seq["comps"], par["comps"]):
+    np.testing.assert_array_equal(cs["coeffs"], cp["coeffs"])
+
+
 def test_decode_image_sniffs_jpeg_and_png():
   img = _gradient(32, 32)
   jp = image_codec.encode_jpeg(img, 90)
