@@ -350,3 +350,19 @@ def test_legacy_pkl_assets_roundtrip(tmp_path):
   assert assets.global_step == 7
   loaded = tsu.load_t2r_assets_from_file(out)
   assert "state/image" in loaded.feature_spec
+
+
+def test_convert_pkl_assets_cli(tmp_path):
+  import torch
+  from tensor2robot_amd.bin import convert_pkl_assets
+  spec = tsu.TensorSpecStruct()
+  spec["obs"] = tsu.ExtendedTensorSpec((4,), torch.float32, name="obs")
+  labels = tsu.TensorSpecStruct()
+  labels["act"] = tsu.ExtendedTensorSpec((2,), torch.float32, name="act")
+  assets_dir = tmp_path / "assets.extra"
+  assets_dir.mkdir()
+  tsu.write_input_spec_to_pkl_file(
+      str(assets_dir / tsu.INPUT_SPEC_PKL_FILENAME), spec, labels)
+  out = convert_pkl_assets.main(["--assets_filepath", str(assets_dir)])
+  loaded = tsu.load_t2r_assets_from_file(out)
+  assert "obs" in loaded.feature_spec and "act" in loaded.label_spec
