@@ -434,3 +434,90 @@ def test_train_eval_model_step_parity_with_device_pool():
   # (in-pipeline H2D runs ~6 GB/s where an isolated under-load copy
   # does 26 GB/s) is an open item (profiles/r2_input_pipeline.md).
   assert ms_pipeline <= ms_pool * 4.0 + 1.0, (ms_pipeline, ms_pool)
+
+
+@requires_gpu
+def test_maml_vmap_parallel_tasks_matches_loop_gpu():
+  """Task-parallel (vmap) inner loop == per-task loop on GPU, incl.
+  second-order outer grads (BASELINE config #5 workload)."""
+  from tensor2robot_amd.meta_learning import maml_inner_loop
+  from tensor2robot_amd.specs import tensorspec_utils as tsu
+
+  torch.manual_seed(0)
+  net = torch.nn.Sequential(
+      torch.nn.Linear(6, 64), torch.nn.ReLU(),
+      torch.nn.Linear(64, 64), torch.nn.ReLU(),
+      torch.nn.Linear(64, 2)).cuda()
+
+  class _Base:
+    network = net
+
+    def inference_network_fn(self, features, labels, mode, params=None):
+      return {"prediction": net(features["x"])}
+
+    def model_train_fn(self, features, labels, inference_outputs, mode,
+                       params=None):
+      return torch.nn.functional.mse_loss(
+          inference_outputs["prediction"], labels["y"])
+
+  base = _Base()
+  loop = maml_inner_loop.MAMLInnerLoopGradientDescent(learning_rate=0.05)
+  tasks, samples = 8, 16
+  cf = torch.randn(tasks, samples, 6, device="cuda")
+  cl = torch.randn(tasks, samples, 2, device="cuda")
+  inf = torch.randn(tasks, samples, 6, device="cuda")
+  infl = torch.randn(tasks, samples, 2, device="cuda")
+
+  # Loop path (2 adaptation steps per task).
+  per_task_cond = []
+  losses0 = []
+  for t in range(tasks):
+    f = tsu.TensorSpecStruct(); f["x"] = cf[t]
+    l = tsu.TensorSpecStruct(); l["y"] = cl[t]
+    vf = tsu.TensorSpecStruct(); vf["x"] = inf[t]
+    vl = tsu.TensorSpecStruct(); vl["y"] = infl[t]
+    (uncond, cond), _, inner_losses = loop.inner_loop(
+        [(f, l), (f, l), (vf, vl)], base.inference_network_fn,
+        base.model_train_fn, net, mode="train")
+    per_task_cond.append(cond["prediction"])
+    losses0.append(inner_losses[0])
+  loop_cond = torch.stack(per_task_cond)
+  loop_outer = loop_cond.pow(2).mean()
+  loop_outer.backward()
+  loop_grads = [p.grad.clone() for p in net.parameters()]
+  for p in net.parameters():
+    p.grad = None
+
+  # vmap path.
+  uncond, cond, inner_outs, losses = loop.inner_loop_vmapped(
+      {"x": cf}, {"y": cl}, {"x": inf}, {"y": infl},
+      {"cond_f": {}, "cond_l": {}, "inf_f": {}, "inf_l": {}},
+      base.inference_network_fn, base.model_train_fn, net,
+      num_steps=2, mode="train")
+  torch.testing.assert_close(cond["prediction"], loop_cond,
+                             rtol=1e-4, atol=1e-5)
+  torch.testing.assert_close(losses[:, 0], torch.stack(losses0),
+                             rtol=1e-4, atol=1e-6)
+  vmap_outer = cond["prediction"].pow(2).mean()
+  vmap_outer.backward()
+  for g_loop, p in zip(loop_grads, net.parameters()):
+    torch.testing.assert_close(p.grad, g_loop, rtol=1e-3, atol=1e-5)
+
+
+@requires_gpu
+def test_jpeg_restart_stream_gpu_batch_decode():
+  """Restart-marker streams (segment-parallel Huffman) through the GPU
+  decode path match the plain-stream decode bit-for-bit."""
+  from tensor2robot_amd.data import gpu_jpeg
+  from tensor2robot_amd.data import image_codec
+
+  rng = np.random.default_rng(3)
+  imgs = [(rng.random((96, 128, 3)) * 255).astype(np.uint8)
+          for _ in range(6)]
+  plain = [image_codec.encode_jpeg(im, quality=92) for im in imgs]
+  rst = [image_codec.encode_jpeg(im, quality=92, restart_interval=16)
+         for im in imgs]
+  out_plain = gpu_jpeg.decode_jpeg_batch(plain, device="cuda")
+  out_rst = gpu_jpeg.decode_jpeg_batch(rst, device="cuda")
+  assert out_plain.shape == (6, 96, 128, 3)
+  torch.testing.assert_close(out_rst, out_plain, rtol=0, atol=0)
