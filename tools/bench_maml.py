@@ -95,9 +95,10 @@ def _meta_batch(tasks, samples, state_dim, action_dim, device, seed=0):
 
 def bench(parallel_tasks, args, device):
   torch.manual_seed(0)
+  dev_type = "gpu" if device.type == "cuda" else "cpu"
   base = _StatePolicy(args.state_dim, args.action_dim, args.hidden,
-                      device_type=device.type)
-  model = _BenchMAML(base_model=base, device_type=device.type,
+                      device_type=dev_type)
+  model = _BenchMAML(base_model=base, device_type=dev_type,
                      compute_dtype="float32",
                      num_inner_loop_steps=args.inner_steps,
                      use_second_order=True,
