@@ -24,6 +24,7 @@ hip_sources = [
     os.path.join(HIP_DIR, "spatial_softmax.hip"),
     os.path.join(HIP_DIR, "mfma_probe.hip"),
     os.path.join(HIP_DIR, "conv_s1.hip"),
+    os.path.join(HIP_DIR, "conv_s1_big.hip"),
     os.path.join(HIP_DIR, "conv_wrw.hip"),
     os.path.join(HIP_DIR, "conv_wrw2.hip"),
     os.path.join(HIP_DIR, "conv_wrw4.hip"),

@@ -149,6 +149,72 @@ class _MFMAConvFunction(torch.autograd.Function):
     return dx, dw, None
 
 
+def _bigc_supported(x: torch.Tensor, weight: torch.Tensor, stride,
+                    padding) -> bool:
+  """C-chunked MFMA 3x3 path (conv_s1_big.hip): ResNet-family
+  C=K in {128,256,512} stride-1 3x3s (film_resnet_model.py:100-341,
+  grasp2vec/resnet.py).  Forward and dgrad run the chunk kernel; the
+  weight gradient keeps the im2col+GEMM recipe (gemm_conv.dw_from_col).
+  Opt-out: T2R_DISABLE_MFMA_BIGC=1 falls back to the GEMM-conv path."""
+  if os.environ.get("T2R_DISABLE_MFMA_CONV") or \
+      os.environ.get("T2R_DISABLE_MFMA_BIGC"):
+    return False
+  if not (x.is_cuda and x.dtype == torch.bfloat16):
+    return False
+  k, c, r, s = weight.shape
+  if not (stride == (1, 1) and (r, s) == (3, 3) and
+          padding[0] == padding[1] and padding[0] <= 1):
+    return False
+  # dgrad swaps channel roles, so both must satisfy both constraints
+  # (C-role % 32, K-role % 64) when an input gradient is needed.
+  if x.requires_grad:
+    return c % 64 == 0 and k % 64 == 0 and 96 <= c <= 512 and k <= 512
+  return c % 32 == 0 and k % 64 == 0 and 96 <= c <= 512 and k <= 512
+
+
+class _BigCConvFunction(torch.autograd.Function):
+  """Big-channel 3x3: chunked MFMA fwd/dgrad + GEMM weight gradient."""
+
+  @staticmethod
+  def forward(ctx, x, weight, pad):
+    ext = ops_mod.require_hip()
+    k, c, r, s = weight.shape
+    if not x.is_contiguous(memory_format=torch.channels_last):
+      x = x.contiguous(memory_format=torch.channels_last)
+    if x.requires_grad:
+      wpk, wpk_b = ext.pack_conv_w_pair(weight)
+    else:
+      wpk, wpk_b = ext.pack_conv_w(weight, False), None
+    y = ext.conv_s1_nhwc_cchunk(x, wpk, k, r, s, pad)
+    if wpk_b is None:
+      ctx.save_for_backward(x, weight)
+    else:
+      ctx.save_for_backward(x, weight, wpk_b)
+    ctx.pad = pad
+    return y
+
+  @staticmethod
+  def backward(ctx, dy):
+    ext = ops_mod.require_hip()
+    x, weight = ctx.saved_tensors[:2]
+    wpk_b = ctx.saved_tensors[2] if len(ctx.saved_tensors) > 2 else None
+    k, c, r, s = weight.shape
+    dy = dy.contiguous(memory_format=torch.channels_last) \
+        .to(torch.bfloat16)
+    dx = dw = None
+    if ctx.needs_input_grad[0]:
+      bpad = r - 1 - ctx.pad
+      if wpk_b is None:
+        wpk_b = ext.pack_conv_w(weight, True)
+      dx = ext.conv_s1_nhwc_cchunk(dy, wpk_b, c, r, s, bpad)
+    if ctx.needs_input_grad[1]:
+      from tensor2robot_amd.ops import gemm_conv
+      col = ext.im2col_nhwc(x, r, s, ctx.pad, 1)
+      dy_mat = dy.permute(0, 2, 3, 1).reshape(-1, k)
+      dw = gemm_conv.dw_from_col(col, dy_mat, r, s, c, k, weight.dtype)
+    return dx, dw, None
+
+
 def _space_to_depth_nhwc(x: torch.Tensor, pad_to: int = 16
                          ) -> torch.Tensor:
   """[N,3,H,W] cl -> [N,pad_to,H/2,W/2] cl, c-order (dr, ds, c)."""
@@ -243,6 +309,12 @@ class MFMAConv2d(nn.Conv2d):
       if w.dtype != torch.bfloat16:
         w = w.to(torch.bfloat16)
       return _MFMAConvFunction.apply(x, w, self.padding[0])
+    if self.bias is None and _bigc_supported(x, self.weight, self.stride,
+                                             self.padding):
+      w = self.weight
+      if w.dtype != torch.bfloat16:
+        w = w.to(torch.bfloat16)
+      return _BigCConvFunction.apply(x, w, self.padding[0])
     if self.bias is None and _stem_supported(x, self.weight, self.stride,
                                              self.padding):
       w = self.weight

@@ -36,6 +36,31 @@ def _w_mat(weight: torch.Tensor) -> torch.Tensor:
       .contiguous()
 
 
+def dw_from_col(col, dy_mat, r, s, c, k, out_dtype) -> torch.Tensor:
+  """dw = col^T @ dy_mat -> [K,C,R,S]; chunked for huge contractions.
+
+  hipBLASLt runs tiny-output x huge-contraction GEMMs without split-K
+  and strands the chip (measured 56 TF at M=668k vs 177 TF chunked —
+  tools/probe_gemm_shapes.py); chunked bmm + sum restores grid
+  parallelism.  Shared by the GEMM-conv path and the big-C MFMA conv's
+  weight gradient.
+  """
+  m = col.shape[0]
+  if m >= 65536:
+    chunks = max(2, min(64, m // 8192))
+    mc = m // chunks
+    head = chunks * mc
+    dw_mat = torch.bmm(
+        col[:head].view(chunks, mc, -1).transpose(1, 2),
+        dy_mat[:head].view(chunks, mc, k)).sum(0)
+    if head < m:
+      dw_mat = dw_mat + col[head:].t() @ dy_mat[head:]
+  else:
+    dw_mat = col.t() @ dy_mat                    # [RS*C, K]
+  return dw_mat.reshape(r, s, c, k).permute(3, 2, 0, 1) \
+      .contiguous().to(out_dtype)
+
+
 class _GemmConvFunction(torch.autograd.Function):
   """im2col + rocBLAS GEMM conv (NHWC bf16, stride 1/2)."""
 
@@ -88,24 +113,7 @@ class _GemmConvFunction(torch.autograd.Function):
     else:
       col = ext.im2col_nhwc(x, r, s, pad, stride)
     if ctx.needs_input_grad[1]:
-      m = col.shape[0]
-      if m >= 65536:
-        # Tiny-output x huge-contraction GEMM: hipBLASLt runs it
-        # without split-K and strands the chip (measured 56 TF at
-        # M=668k vs 177 TF chunked — tools/probe_gemm_shapes.py).
-        # Chunked bmm + sum restores grid parallelism.
-        chunks = max(2, min(64, m // 8192))
-        mc = m // chunks
-        head = chunks * mc
-        dw_mat = torch.bmm(
-            col[:head].view(chunks, mc, -1).transpose(1, 2),
-            dy_mat[:head].view(chunks, mc, k)).sum(0)
-        if head < m:
-          dw_mat = dw_mat + col[head:].t() @ dy_mat[head:]
-      else:
-        dw_mat = col.t() @ dy_mat                  # [RS*C, K]
-      dw = dw_mat.reshape(r, s, c, k).permute(3, 2, 0, 1) \
-          .contiguous().to(weight.dtype)
+      dw = dw_from_col(col, dy_mat, r, s, c, k, weight.dtype)
     if ctx.needs_input_grad[0]:
       if one_by_one:
         # GEMM writes straight into the cl dx storage — the earlier

@@ -19,6 +19,8 @@ at::Tensor mfma_probe(at::Tensor A, at::Tensor B);
 
 at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
                         int64_t R, int64_t S, int64_t pad);
+at::Tensor conv_s1_nhwc_cchunk(at::Tensor x, at::Tensor wpk, int64_t K,
+                               int64_t R, int64_t S, int64_t pad);
 
 at::Tensor pack_conv_w(at::Tensor w, bool transpose);
 std::vector<at::Tensor> pack_conv_w_pair(at::Tensor w);
@@ -75,6 +77,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_bn_relu_backward", &fused_bn_relu_backward,
         "Fused BN(+ReLU) backward (NHWC bf16)");
   m.def("mfma_probe", &mfma_probe, "32x32x16 bf16 MFMA layout probe");
+  m.def("conv_s1_nhwc_cchunk", &conv_s1_nhwc_cchunk,
+        "C-chunked MFMA 3x3 stride-1 conv (C%32, K%64, NHWC bf16)");
   m.def("conv_s1_nhwc", &conv_s1_nhwc,
         "MFMA stride-1 NHWC bf16 conv (prepacked weights)");
   m.def("pack_conv_w", &pack_conv_w,

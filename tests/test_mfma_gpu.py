@@ -406,3 +406,49 @@ def test_gemm_conv_cpad_stem_matches_torch():
       0.02 * y32.abs().max().item()
   assert (wt.grad.float() - w32.grad).abs().max().item() < \
       0.03 * max(w32.grad.abs().max().item(), 1.0)
+
+
+@requires_gpu
+@pytest.mark.parametrize("shape", [
+    # n, c, h, w, k, pad   (3x3 stride-1 only)
+    (2, 128, 59, 59, 128, 1),    # G2V/ResNet50 layer2
+    (2, 256, 30, 30, 256, 1),    # layer3
+    (2, 512, 15, 15, 512, 1),    # layer4
+    (2, 128, 25, 25, 128, 1),    # BC-Z ResNet18 @100^2
+    (2, 96, 20, 20, 64, 1),      # C%32 only, K=64
+    (2, 128, 17, 19, 128, 0),    # VALID pad, non-square map
+])
+def test_bigc_conv_matches_torch(shape):
+  """C-chunked MFMA 3x3 (conv_s1_big.hip) vs fp32 reference:
+  forward, dgrad (chunk kernel on flipped pack) and dw (GEMM)."""
+  import torch.nn.functional as F
+  from tensor2robot_amd.ops import conv as conv_mod
+  n, c, h, w, k, pad = shape
+  torch.manual_seed(0)
+  x = torch.randn(n, c, h, w, device="cuda").to(torch.bfloat16) \
+      .contiguous(memory_format=torch.channels_last)
+  need_dx = c % 64 == 0 and k % 64 == 0
+  x.requires_grad_(need_dx)
+  wt = (torch.randn(k, c, 3, 3, device="cuda") * 0.05).to(
+      torch.bfloat16).requires_grad_(True)
+  y = conv_mod._BigCConvFunction.apply(x, wt, pad)
+  oh, ow = h + 2 * pad - 2, w + 2 * pad - 2
+  assert y.shape == (n, k, oh, ow)
+  dy = torch.randn_like(y.float()).to(torch.bfloat16)
+  y.backward(dy)
+
+  x32 = x.detach().float().requires_grad_(True)
+  w32 = wt.detach().float().requires_grad_(True)
+  y32 = F.conv2d(x32, w32, stride=1, padding=pad)
+  y32.backward(dy.float())
+
+  def relerr(a, b):
+    return (a.float() - b).abs().max().item() / max(
+        b.abs().max().item(), 1e-6)
+
+  assert relerr(y, y32) < 0.02, ("y", shape, relerr(y, y32))
+  if need_dx:
+    assert relerr(x.grad, x32.grad) < 0.03, ("dx", shape,
+                                             relerr(x.grad, x32.grad))
+  assert relerr(wt.grad, w32.grad) < 0.03, ("dw", shape,
+                                            relerr(wt.grad, w32.grad))
