@@ -13,9 +13,15 @@ def main():
   db = sqlite3.connect(path)
   cur = db.cursor()
   tabs = [r[0] for r in cur.execute(
-      "SELECT name FROM sqlite_master WHERE type='table'")]
-  sym = [t for t in tabs if "kernel_symbol" in t][0]
-  disp = [t for t in tabs if "kernel_dispatch" in t][0]
+      "SELECT name FROM sqlite_master WHERE type='table' "
+      "OR type='view'")]
+  syms = [t for t in tabs if "kernel_symbol" in t]
+  disps = [t for t in tabs if "kernel_dispatch" in t]
+  if not syms or not disps:
+    print("schema tables/views:", tabs)
+    return
+  sym = syms[0]
+  disp = disps[0]
   q = (f"SELECT s.display_name, COUNT(*), SUM(d.end-d.start)/1e6 "
        f"FROM {disp} d JOIN {sym} s ON d.kernel_id=s.id "
        f"GROUP BY s.display_name ORDER BY 3 DESC LIMIT {n}")
