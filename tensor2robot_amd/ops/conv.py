@@ -117,23 +117,14 @@ class _MFMAConvFunction(torch.autograd.Function):
       sq35 = c == 64 and k == 64 and r == s and r in (3, 5)
       default_v4 = (wrw_mode == "" and c == 64 and k == 64 and
                     r == 5 and s == 5)
-      # Big-spatial 3x3 (e.g. Grasp2Vec 64ch @ 118^2): MIOpen wrw on
-      # these shapes is off the packaged perf DB — its find burns ~6.5 s
-      # of naive-wrw evaluation per process (gpurun_out/r2b G2V trace)
-      # and risks baking a find-intermediate kernel into the hipGraph.
-      # The im2col+chunked-GEMM wrw (the gemm_conv recipe) is find-free.
-      # The QT-Opt 3x3s (<=32^2) stay on MIOpen: in-DB and measured
-      # faster there (profiles/r01).
-      big_spatial_33 = (wrw_mode == "" and r == 3 and s == 3 and
-                        x.shape[2] > 32 and c % 8 == 0 and k % 8 == 0)
-      if big_spatial_33:
-        from tensor2robot_amd.ops import gemm_conv
-        col = ext.im2col_nhwc(x, r, s, ctx.pad, 1)
-        dy_mat = dy.to(torch.bfloat16).permute(0, 2, 3, 1) \
-            .reshape(-1, k)
-        dw = gemm_conv.dw_from_col(col, dy_mat, r, s, c, k,
-                                   weight.dtype)
-      elif default_v4 or (wrw_mode == "4" and sq35):
+      # (A/B note, gpurun_out/r2b_wrw_ab: on the big-spatial G2V 3x3
+      # 64ch @ 118^2, MIOpen wrw steady state is 0.074 ms vs 0.213 for
+      # the find-free im2col+GEMM recipe — 2.9x faster.  Its 3.6-s
+      # first-call find is one-time and absorbed by the untimed warmup,
+      # so MIOpen stays the 3x3 wrw default everywhere; the engine's
+      # replay-vs-eager check guards against find-intermediate kernels
+      # being baked into a graph.)
+      if default_v4 or (wrw_mode == "4" and sq35):
         # v4 emits [K,C,R,S] bf16 directly (permute+cast fused into
         # its reduce kernel).
         dw = ext.conv_s1_wrw4(x, dy.to(torch.bfloat16), r, s, ctx.pad)
