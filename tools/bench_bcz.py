@@ -6,6 +6,11 @@ hipGraph-captured fwd+bwd+opt): on-GPU preprocess (crop 472^2 of
 losses + backward + Adam.
 
   python tools/bench_bcz.py [--steps 50] [--warmup 15] [--batch-size 32]
+
+DP=N (BASELINE config #3, one rank per MI355X over RCCL):
+
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 tools/bench_bcz.py ...
 """
 
 import argparse
@@ -35,7 +40,9 @@ def main():
   p.add_argument("--resnet-size", type=int, default=18)
   p.add_argument("--no-hipgraph", action="store_true")
   args = p.parse_args()
-  assert torch.cuda.is_available()
+  import dist_bench
+  distributed, rank, world_size, device0 = dist_bench.init()
+  use_cuda = torch.cuda.is_available()
 
   model = bcz_model.BCZModel(
       image_size=(100, 100), input_size=(512, 640), num_waypoints=10,
@@ -44,13 +51,15 @@ def main():
           bcz_model.BCZPreprocessor, image_size=(100, 100),
           input_size=(512, 640), crop_size=(472, 472), mock_subtask=True),
       create_optimizer_fn=lambda: optimizers.create_adam_optimizer(1e-4),
-      device_type="gpu", compute_dtype="bfloat16")
+      device_type="gpu" if use_cuda else "cpu",
+      compute_dtype="bfloat16" if use_cuda else "float32")
   trainer = train_eval.Trainer(model, model_dir="",
-                               use_hip_graph=not args.no_hipgraph)
+                               use_hip_graph=use_cuda and
+                               not args.no_hipgraph)
   device = trainer.device
 
   bs = args.batch_size
-  g = torch.Generator().manual_seed(0)
+  g = torch.Generator().manual_seed(1234 + rank)
   pool = []
   for _ in range(2):
     features = tsu.TensorSpecStruct()
@@ -78,32 +87,45 @@ def main():
     trainer.train(lambda: pool_iter, trainer.global_step + n,
                   preprocess_fn=preprocess_fn)
 
-  run_steps(max(args.warmup, 20))
-  torch.cuda.synchronize()
-  # settle probe (find/clock ramp)
-  prev = None
-  for _ in range(10):
-    t0 = time.perf_counter()
-    run_steps(5)
-    torch.cuda.synchronize()
-    win = time.perf_counter() - t0
-    if prev is not None and abs(win - prev) <= 0.05 * prev:
-      break
-    prev = win
-  torch.cuda.synchronize()
+  run_steps(max(args.warmup, 20) if use_cuda else args.warmup)
+  dist_bench.barrier_sync(distributed)
+  if use_cuda:
+    # settle probe (find/clock ramp); collective stop in DP mode
+    prev = None
+    for _ in range(10):
+      t0 = time.perf_counter()
+      run_steps(5)
+      dist_bench.barrier_sync(distributed)
+      win = time.perf_counter() - t0
+      stable = prev is not None and abs(win - prev) <= 0.05 * prev
+      if distributed:
+        import torch.distributed as dist
+        flag = torch.tensor([1.0 if stable else 0.0], device=device)
+        dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+        stable = float(flag.item()) >= 1.0
+      if stable:
+        break
+      prev = win
+  dist_bench.barrier_sync(distributed)
   t0 = time.perf_counter()
   run_steps(args.steps)
-  torch.cuda.synchronize()
-  elapsed = time.perf_counter() - t0
-  print(json.dumps({
-      "metric": "images/sec BC-Z FiLM-ResNet%d train, 100x100 "
-                "(512x640 raw), bs=%d" % (args.resnet_size, bs),
-      "value": round(bs * args.steps / elapsed, 2),
-      "ms_per_step": round(elapsed / args.steps * 1000, 3),
-      "graphed": bool(trainer._fast_engine and
-                      trainer._fast_engine.is_graphed),
-      "dtype": "bf16", "data": "synthetic", "n_gpus": 1,
-  }))
+  dist_bench.barrier_sync(distributed)
+  elapsed = dist_bench.max_over_ranks(time.perf_counter() - t0,
+                                      distributed, device)
+  if rank == 0:
+    print(json.dumps({
+        "metric": "images/sec (whole job) BC-Z FiLM-ResNet%d train, "
+                  "100x100 (512x640 raw), bs=%d/GPU"
+                  % (args.resnet_size, bs),
+        "value": round(bs * args.steps * world_size / elapsed, 2),
+        "ms_per_step": round(elapsed / args.steps * 1000, 3),
+        "graphed": bool(trainer._fast_engine and
+                        trainer._fast_engine.is_graphed),
+        "dtype": "bf16" if use_cuda else "float32",
+        "data": "synthetic", "n_gpus": world_size,
+        "parallelism": "dp%d" % world_size,
+    }))
+  dist_bench.finalize(distributed)
 
 
 if __name__ == "__main__":

@@ -10,6 +10,12 @@ policy adapted per task from condition episodes (reference
 `research/vrgripper/configs/run_train_wtl_statespace_trial.gin`).
 
   python tools/bench_maml.py --tasks 16 --samples 40 --steps 30
+
+DP=N (BASELINE config #5 shape, one rank per MI355X over RCCL;
+gradients all-reduced after the outer backward):
+
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 tools/bench_maml.py ...
 """
 
 import argparse
@@ -93,8 +99,9 @@ def _meta_batch(tasks, samples, state_dim, action_dim, device, seed=0):
   return f, l
 
 
-def bench(parallel_tasks, args, device):
-  torch.manual_seed(0)
+def bench(parallel_tasks, args, device, distributed=False, rank=0,
+          world_size=1):
+  torch.manual_seed(0)  # same init on every rank
   dev_type = "gpu" if device.type == "cuda" else "cpu"
   base = _StatePolicy(args.state_dim, args.action_dim, args.hidden,
                       device_type=dev_type)
@@ -106,13 +113,20 @@ def bench(parallel_tasks, args, device):
   _ = model.network
   model.to_device(device)
   features, labels = _meta_batch(args.tasks, args.samples,
-                                 args.state_dim, args.action_dim, device)
+                                 args.state_dim, args.action_dim, device,
+                                 seed=1234 + rank)
   opt = torch.optim.Adam(model.network.parameters(), lr=1e-3)
 
   def one_step():
     opt.zero_grad(set_to_none=True)
     ops = model.model_fn(features, labels, run_modes.TRAIN)
     ops.loss.backward()
+    if distributed:
+      import torch.distributed as dist
+      for p_ in model.network.parameters():
+        if p_.grad is not None:
+          dist.all_reduce(p_.grad)
+          p_.grad.div_(world_size)
     opt.step()
     return ops.loss
 
@@ -150,18 +164,28 @@ def main():
   p.add_argument("--mode", choices=["both", "loop", "vmap"],
                  default="both")
   args = p.parse_args()
-  device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
+  import dist_bench
+  distributed, rank, world_size, device = dist_bench.init()
   results = []
   if args.mode in ("both", "loop"):
-    results.append(bench(False, args, device))
+    results.append(bench(False, args, device, distributed, rank,
+                         world_size))
   if args.mode in ("both", "vmap"):
-    results.append(bench(True, args, device))
-  for r in results:
-    print(json.dumps(r))
-  if len(results) == 2:
-    print(json.dumps({
-        "speedup_vmap_over_loop": round(
-            results[0]["ms_per_step"] / results[1]["ms_per_step"], 2)}))
+    results.append(bench(True, args, device, distributed, rank,
+                         world_size))
+  if rank == 0:
+    for r in results:
+      r["n_gpus"] = world_size
+      r["parallelism"] = "dp%d" % world_size
+      r["task_steps_per_sec"] = round(
+          r["task_steps_per_sec"] * world_size, 1)
+      print(json.dumps(r))
+    if len(results) == 2:
+      print(json.dumps({
+          "speedup_vmap_over_loop": round(
+              results[0]["ms_per_step"] / results[1]["ms_per_step"],
+              2)}))
+  dist_bench.finalize(distributed)
 
 
 if __name__ == "__main__":
