@@ -25,6 +25,7 @@ hip_sources = [
     os.path.join(HIP_DIR, "mfma_probe.hip"),
     os.path.join(HIP_DIR, "conv_s1.hip"),
     os.path.join(HIP_DIR, "conv_s1_big.hip"),
+    os.path.join(HIP_DIR, "mdn_nll.hip"),
     os.path.join(HIP_DIR, "conv_wrw.hip"),
     os.path.join(HIP_DIR, "conv_wrw2.hip"),
     os.path.join(HIP_DIR, "conv_wrw4.hip"),

@@ -143,12 +143,22 @@ class MDNDecoder(nn.Module):
     self.head = MDNHead(in_dim, action_size, num_alphas=num_mixture,
                         condition_sigmas=condition_sigmas)
     self._mixture: Optional[GaussianMixture] = None
+    self._params: Optional[torch.Tensor] = None
 
   def forward(self, x: torch.Tensor) -> torch.Tensor:
-    self._mixture = self.head.distribution(x)
+    self._params = self.head(x)
+    self._mixture = get_mixture_distribution(
+        self._params, self.head.num_alphas, self.head.sample_size)
     return self._mixture.approximate_mode()
 
   def loss(self, labels: torch.Tensor) -> torch.Tensor:
     if self._mixture is None:
       raise RuntimeError("MDNDecoder.loss called before forward")
+    from tensor2robot_amd.ops import mdn_nll as fused
+    if fused.supported(self._params, self.head.num_alphas):
+      # One HIP kernel per direction instead of the eager op chain
+      # (ops/hip/mdn_nll.hip); identical math incl. softplus + 1e-4.
+      return fused.mdn_nll(self._params, labels,
+                           self.head.num_alphas,
+                           self.head.sample_size).mean()
     return -self._mixture.log_prob(labels).mean()

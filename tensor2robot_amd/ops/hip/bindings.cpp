@@ -21,6 +21,12 @@ at::Tensor conv_s1_nhwc(at::Tensor x, at::Tensor wpk, int64_t K,
                         int64_t R, int64_t S, int64_t pad);
 at::Tensor conv_s1_nhwc_cchunk(at::Tensor x, at::Tensor wpk, int64_t K,
                                int64_t R, int64_t S, int64_t pad);
+std::vector<at::Tensor> mdn_nll_forward(at::Tensor params,
+                                        at::Tensor labels, int64_t A,
+                                        int64_t S);
+at::Tensor mdn_nll_backward(at::Tensor params, at::Tensor labels,
+                            at::Tensor wsave, at::Tensor gout,
+                            int64_t A, int64_t S);
 
 at::Tensor pack_conv_w(at::Tensor w, bool transpose);
 std::vector<at::Tensor> pack_conv_w_pair(at::Tensor w);
@@ -77,6 +83,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_bn_relu_backward", &fused_bn_relu_backward,
         "Fused BN(+ReLU) backward (NHWC bf16)");
   m.def("mfma_probe", &mfma_probe, "32x32x16 bf16 MFMA layout probe");
+  m.def("mdn_nll_forward", &mdn_nll_forward,
+        "fused MDN negative log-likelihood forward");
+  m.def("mdn_nll_backward", &mdn_nll_backward,
+        "fused MDN negative log-likelihood backward");
   m.def("conv_s1_nhwc_cchunk", &conv_s1_nhwc_cchunk,
         "C-chunked MFMA 3x3 stride-1 conv (C%32, K%64, NHWC bf16)");
   m.def("conv_s1_nhwc", &conv_s1_nhwc,
