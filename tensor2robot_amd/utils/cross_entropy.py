@@ -71,3 +71,65 @@ def cross_entropy_optimize(objective_fn, action_size: int,
   mean = np.zeros(action_size, np.float32)
   std = np.full(action_size, initial_std, np.float32)
   return cem.run(objective_fn, mean, std)
+
+
+def cross_entropy_method(sample_fn, objective_fn, update_fn,
+                         initial_params, num_elites: int,
+                         num_iterations: int = 1,
+                         threshold_to_terminate: Optional[float] = None):
+  """Generic CEM maximization (reference `cross_entropy.py:30-107`).
+
+  Sample batches may be lists `[x0..xn]` or dicts of such lists; the
+  elite selection sorts ascending by value and keeps the top
+  `num_elites`, exactly as the reference.  Returns
+  (final_samples, final_values, final_params).
+  """
+  import operator
+  updated_params = initial_params
+  samples, values = None, None
+  for _ in range(num_iterations):
+    samples = sample_fn(**updated_params)
+    values = list(objective_fn(samples))
+    if isinstance(samples, dict):
+      order = [i for i, _ in sorted(enumerate(values),
+                                    key=operator.itemgetter(1))]
+      elite_samples = {
+          k: [v[i] for i in order][-num_elites:]
+          for k, v in samples.items()}
+    else:
+      sorted_samples = [
+          s for s, _ in sorted(zip(samples, values),
+                               key=operator.itemgetter(1))]
+      elite_samples = sorted_samples[-num_elites:]
+    updated_params = update_fn(updated_params, elite_samples)
+    if (threshold_to_terminate is not None and
+        max(values) > threshold_to_terminate):
+      break
+  return samples, values, updated_params
+
+
+def normal_cross_entropy_method(objective_fn, mean, stddev,
+                                num_samples: int, num_elites: int,
+                                num_iterations: int = 1):
+  """CEM with a normal sampler (reference `cross_entropy.py:110-156`).
+
+  Returns (final_mean, final_stddev); the elite refit uses Bessel's
+  correction (ddof=1) like the reference.
+  """
+  size = np.broadcast(mean, stddev).size
+
+  def _sample_fn(mean, stddev):
+    return np.asarray(mean) + np.asarray(stddev) * np.random.randn(
+        num_samples, size)
+
+  def _update_fn(params, elite_samples):
+    del params
+    elite = np.asarray(elite_samples)
+    return {"mean": np.mean(elite, axis=0),
+            "stddev": np.std(elite, axis=0, ddof=1)}
+
+  _, _, final_params = cross_entropy_method(
+      _sample_fn, objective_fn, _update_fn,
+      {"mean": mean, "stddev": stddev}, num_elites,
+      num_iterations=num_iterations)
+  return final_params["mean"], final_params["stddev"]

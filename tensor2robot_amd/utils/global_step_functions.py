@@ -38,3 +38,22 @@ def piecewise_linear(boundaries: Sequence[float],
     return values[-1]
 
   return schedule
+
+
+@gin.configurable
+def exponential_decay(initial_value: float = 0.0001,
+                      decay_steps: int = 10000,
+                      decay_rate: float = 0.9,
+                      staircase: bool = True):
+  """callable(step) -> initial_value * decay_rate^(step/decay_steps)
+  (reference `global_step_functions.py:98-121`; staircase floors the
+  exponent to whole decay intervals like tf.train.exponential_decay)."""
+  assert decay_steps > 0, "decay_steps must be positive"
+
+  def schedule(step) -> float:
+    p = float(step) / float(decay_steps)
+    if staircase:
+      p = float(int(p))
+    return float(initial_value) * float(decay_rate) ** p
+
+  return schedule

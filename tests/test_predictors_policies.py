@@ -347,3 +347,83 @@ def test_tf_example_receiver_matches_numpy_receiver(tmp_path):
   for k in feed_np:
     np.testing.assert_allclose(feed_np[k].numpy(),
                                np.asarray(feed_ex[k]), rtol=1e-6)
+
+
+def test_generic_cross_entropy_method_list_and_dict():
+  """Reference cross_entropy.py:30-107 API: list and dict batches."""
+  np.random.seed(0)
+
+  def sample_fn(mean, stddev):
+    return list(mean + stddev * np.random.randn(64))
+
+  def objective_fn(samples):
+    return [-(s - 3.0) ** 2 for s in samples]
+
+  def update_fn(params, elites):
+    return {"mean": float(np.mean(elites)),
+            "stddev": float(np.std(elites) + 1e-3)}
+
+  samples, values, params = cross_entropy.cross_entropy_method(
+      sample_fn, objective_fn, update_fn,
+      {"mean": 0.0, "stddev": 2.0}, num_elites=8, num_iterations=10)
+  assert abs(params["mean"] - 3.0) < 0.3
+  assert len(samples) == 64 and len(values) == 64
+
+  # Dict-batch form: keys sorted coherently by the same value order.
+  def sample_fn_d(mean, stddev):
+    xs = mean + stddev * np.random.randn(32)
+    return {"x": list(xs), "tag": list(range(32))}
+
+  def objective_fn_d(samples):
+    return [-(x - 1.0) ** 2 for x in samples["x"]]
+
+  def update_fn_d(params, elites):
+    assert set(elites) == {"x", "tag"}
+    assert len(elites["x"]) == 4
+    return {"mean": float(np.mean(elites["x"])),
+            "stddev": float(np.std(elites["x"]) + 1e-3)}
+
+  _, _, params = cross_entropy.cross_entropy_method(
+      sample_fn_d, objective_fn_d, update_fn_d,
+      {"mean": 0.0, "stddev": 2.0}, num_elites=4, num_iterations=10)
+  assert abs(params["mean"] - 1.0) < 0.4
+
+
+def test_generic_cem_early_termination():
+  calls = []
+
+  def sample_fn(mean):
+    calls.append(1)
+    return [mean, mean + 1]
+
+  _, _, _ = cross_entropy.cross_entropy_method(
+      sample_fn, lambda s: [float(x) for x in s],
+      lambda p, e: {"mean": max(e)}, {"mean": 5.0}, num_elites=1,
+      num_iterations=50, threshold_to_terminate=2.0)
+  assert len(calls) == 1  # first batch already exceeds the threshold
+
+
+def test_normal_cross_entropy_method_converges():
+  np.random.seed(1)
+
+  def objective_fn(samples):
+    return [-float(np.sum((s - 2.0) ** 2)) for s in samples]
+
+  mean, stddev = cross_entropy.normal_cross_entropy_method(
+      objective_fn, mean=[0.0, 0.0], stddev=[2.0, 2.0],
+      num_samples=128, num_elites=16, num_iterations=12)
+  assert np.allclose(mean, [2.0, 2.0], atol=0.3)
+  assert np.all(np.asarray(stddev) < 1.0)
+
+
+def test_exponential_decay_schedule():
+  from tensor2robot_amd.utils import global_step_functions as gsf
+  sched = gsf.exponential_decay(initial_value=1.0, decay_steps=100,
+                                decay_rate=0.5, staircase=True)
+  assert sched(0) == 1.0
+  assert sched(99) == 1.0          # staircase: same interval
+  assert sched(100) == 0.5
+  assert sched(250) == 0.25
+  smooth = gsf.exponential_decay(initial_value=1.0, decay_steps=100,
+                                 decay_rate=0.5, staircase=False)
+  assert abs(smooth(50) - 0.5 ** 0.5) < 1e-9
