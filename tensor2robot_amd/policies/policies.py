@@ -56,6 +56,13 @@ class Policy(abc.ABC):
       return self._predictor.global_step
     return -1
 
+  @property
+  def model_path(self):
+    """Path of the restored model (reference policies.py:71-74)."""
+    if self._predictor is not None:
+      return self._predictor.model_path
+    return None
+
 
 @gin.configurable
 class CEMPolicy(Policy):
