@@ -1077,3 +1077,146 @@ def convert_pkl_assets_to_proto_assets(pkl_path: str, assets_path: str,
                      global_step=global_step)
   write_t2r_assets_to_file(assets, assets_path)
   return assets
+
+
+# ---------------------------------------------------------------------------
+# Reference-named API parity (utils/tensorspec_utils.py): thin entry
+# points a reference user would look for, mapped onto the torch-native
+# implementations above.
+# ---------------------------------------------------------------------------
+
+
+def convert_to_tensorspecstruct(inputs) -> TensorSpecStruct:
+  """Any spec/tensor hierarchy -> flat TensorSpecStruct (reference :686)."""
+  if isinstance(inputs, TensorSpecStruct):
+    return inputs
+  return flatten_spec_structure(inputs)
+
+
+def make_placeholders(spec_structure, batch_size=None,
+                      sequence_length=None) -> TensorSpecStruct:
+  """Torch stand-in for TF placeholders (reference :783-813).
+
+  Returns zero tensors shaped per spec: torch has no symbolic batch
+  dim, so batch_size None concretizes to 1; batch_size <= 0 omits the
+  batch dim; > 0 is fixed — the same three cases the reference's
+  placeholder shapes distinguish.
+  """
+  out = TensorSpecStruct()
+  for key, spec in flatten_spec_structure(spec_structure).items():
+    if batch_size is not None and batch_size <= 0:
+      shape = _concrete_shape(spec, None, sequence_length)
+    else:
+      shape = _concrete_shape(spec, batch_size or 1, sequence_length)
+    out[key] = torch.zeros(shape, dtype=spec.dtype)
+  return out
+
+
+def tensorspec_from_tensors(tensors) -> TensorSpecStruct:
+  """Tensor structure -> spec structure with unique names (ref :1043)."""
+  out = TensorSpecStruct()
+  for i, (key, t) in enumerate(
+      flatten_spec_structure(tensors).items()):
+    out[key] = ExtendedTensorSpec.from_tensor(t, name=f"{key}/{i}")
+  return out
+
+
+def add_sequence_length_specs(spec_structure) -> TensorSpecStruct:
+  """Augments with key + '_length' int64 specs for every sequence spec
+  (reference :1280-1289)."""
+  flat = flatten_spec_structure(spec_structure)
+  out = TensorSpecStruct()
+  for key, value in flat.items():
+    out[key] = value
+    if value.is_sequence:
+      out[key + "_length"] = ExtendedTensorSpec(
+          shape=(), dtype=torch.int64, name=(value.name or key) +
+          "_length")
+  return out
+
+
+def is_flat_spec_or_tensors_structure(spec_or_tensors) -> bool:
+  """True iff the structure is already flat {key: leaf} (ref :1430)."""
+  if not isinstance(spec_or_tensors, dict):
+    return False
+  for key, value in spec_or_tensors.items():
+    if isinstance(value, dict):
+      return False
+    if isinstance(key, str) and "/" not in key and isinstance(
+        value, (list, tuple)):
+      return False
+  return True
+
+
+def map_predict_fn_dict(spec_structure, spec_numpy, feed_dict=None,
+                        ignore_batch=False):
+  """Builds/extends a validated feed mapping, refusing overwrites
+  (reference :968-1010)."""
+  flat_np = flatten_spec_structure(spec_numpy)
+  flat_spec = flatten_spec_structure(spec_structure)
+  assert_required(filter_required_flat_tensor_spec(flat_spec))
+  if feed_dict is None:
+    feed_dict = {}
+  for key, value in flat_np.items():
+    if key not in flat_spec:
+      continue
+    if key in feed_dict:
+      raise ValueError(
+          f"We would overwrite existing placeholder mapping {key}.")
+    spec = flat_spec[key]
+    # Specs here carry NO batch (TFModel semantics, reference
+    # ignore_batch docstring) and sequences add a time dim: verify the
+    # trailing dims against the spec shape.
+    sshape = tuple(1 if d is None else d for d in spec.shape)
+    vshape = tuple(np.asarray(value).shape)
+    if sshape and (len(vshape) < len(sshape) or
+                   vshape[len(vshape) - len(sshape):] != sshape):
+      raise ValueError(
+          f"{key}: shape {vshape} does not end with spec {sshape}")
+    feed_dict[key] = value
+  return feed_dict
+
+
+def map_feed_dict_unsafe(feature_placeholders_spec, np_inputs_spec):
+  """Unchecked {key: array} mapping (reference :1012-1041); prefer
+  map_feed_dict."""
+  import logging as _logging
+  _logging.getLogger(__name__).warning(
+      "map_feed_dict_unsafe is deprecated. Please update to "
+      "map_feed_dict.")
+  flat_spec = flatten_spec_structure(feature_placeholders_spec)
+  flat_np = flatten_spec_structure(np_inputs_spec)
+  return {key: flat_np[key] for key in flat_spec if key in flat_np}
+
+
+# Legacy pickle/pbtxt IO under the reference's exact names
+# (reference :1691-1732); the native implementations live above.
+def load_t2r_assets_to_file(filename: str) -> "T2RAssets":
+  return load_t2r_assets_from_file(filename)
+
+
+def write_input_spec_to_file(in_feature_spec, in_label_spec,
+                             filename: str):
+  write_input_spec_to_pkl_file(filename, in_feature_spec, in_label_spec)
+
+
+def load_input_spec_from_file(filename: str):
+  import os as _os
+  if not _os.path.exists(filename):
+    raise ValueError(f"The file {filename} does not exist.")
+  return load_input_spec_from_pkl_file(filename)
+
+
+def write_global_step_to_file(global_step: int, filename: str):
+  import pickle as _pickle
+  with open(filename, "wb") as f:
+    _pickle.dump({"global_step": int(global_step)}, f)
+
+
+def load_global_step_from_file(filename: str) -> int:
+  import os as _os
+  import pickle as _pickle
+  if not _os.path.exists(filename):
+    raise ValueError(f"The file {filename} does not exist.")
+  with open(filename, "rb") as f:
+    return _pickle.load(f)["global_step"]

@@ -366,3 +366,55 @@ def test_convert_pkl_assets_cli(tmp_path):
   out = convert_pkl_assets.main(["--assets_filepath", str(assets_dir)])
   loaded = tsu.load_t2r_assets_from_file(out)
   assert "obs" in loaded.feature_spec and "act" in loaded.label_spec
+
+
+def test_reference_named_api_parity(tmp_path):
+  """Reference utils/tensorspec_utils.py entry-point names resolve and
+  behave (convert/make_placeholders/from_tensors/lengths/feeds/IO)."""
+  import torch
+  spec = tsu.TensorSpecStruct()
+  spec["obs/img"] = tsu.ExtendedTensorSpec((4, 4, 3), torch.uint8,
+                                           name="img")
+  spec["seq"] = tsu.ExtendedTensorSpec((2,), torch.float32, name="s",
+                                       is_sequence=True)
+  flat = tsu.convert_to_tensorspecstruct({"obs": {"img": spec["obs/img"]},
+                                          "seq": spec["seq"]})
+  assert "obs/img" in flat
+
+  ph = tsu.make_placeholders(spec, batch_size=3, sequence_length=5)
+  assert ph["obs/img"].shape == (3, 4, 4, 3)
+  assert ph["seq"].shape == (3, 5, 2)
+  assert tsu.make_placeholders(spec, batch_size=0)["obs/img"].shape == \
+      (4, 4, 3)
+
+  specs_back = tsu.tensorspec_from_tensors(ph)
+  assert tuple(specs_back["obs/img"].shape) == (3, 4, 4, 3)
+
+  with_len = tsu.add_sequence_length_specs(spec)
+  assert "seq_length" in with_len
+  assert with_len["seq_length"].dtype == torch.int64
+
+  assert tsu.is_flat_spec_or_tensors_structure({"a": ph["seq"]})
+  assert not tsu.is_flat_spec_or_tensors_structure({"a": {"b": 1}})
+
+  np_in = tsu.make_random_numpy(spec, batch_size=3, sequence_length=5)
+  fd = tsu.map_predict_fn_dict(spec, np_in)
+  assert set(fd) == {"obs/img", "seq"}
+  try:
+    tsu.map_predict_fn_dict(spec, np_in, feed_dict=fd)
+    assert False, "expected overwrite error"
+  except ValueError:
+    pass
+  unsafe = tsu.map_feed_dict_unsafe(spec, np_in)
+  assert set(unsafe) == {"obs/img", "seq"}
+
+  # IO names.
+  labels = tsu.TensorSpecStruct()
+  labels["y"] = tsu.ExtendedTensorSpec((1,), torch.float32, name="y")
+  pkl = str(tmp_path / "input_specifications.pkl")
+  tsu.write_input_spec_to_file(spec, labels, pkl)
+  f, l = tsu.load_input_spec_from_file(pkl)
+  assert "obs/img" in f and "y" in l
+  gs = str(tmp_path / "global_step.pkl")
+  tsu.write_global_step_to_file(7, gs)
+  assert tsu.load_global_step_from_file(gs) == 7
