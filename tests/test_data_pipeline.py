@@ -11,6 +11,7 @@ from tensor2robot_amd.data import parser as parser_mod
 from tensor2robot_amd.data import pipeline
 from tensor2robot_amd.data import tfrecord
 from tensor2robot_amd.specs import tensorspec_utils as tsu
+from tensor2robot_amd.utils import modes as run_modes
 
 T = tsu.ExtendedTensorSpec
 
@@ -231,3 +232,45 @@ def test_prefetch_iterator_propagates_errors():
   it = pipeline.PrefetchIterator(bad_source, depth=2)
   with pytest.raises(RuntimeError):
     list(it)
+
+
+def test_tfdata_reference_api(tmp_path):
+  """Reference utils/tfdata.py entry points on the native pipeline."""
+  from tensor2robot_amd.data import tfdata
+  rng = np.random.RandomState(0)
+  records = [example_codec.encode_example({
+      "vec": rng.rand(3).astype(np.float32),
+      "label": np.array([i % 2], np.float32),
+  }) for i in range(10)]
+  for shard in range(2):
+    _write_records(tmp_path, records[shard * 5:(shard + 1) * 5],
+                   name=f"data-{shard}.tfrecord")
+  patterns = str(tmp_path / "data-*.tfrecord")
+
+  assert tfdata.get_batch_size(None, 4) == 4
+  assert tfdata.get_batch_size({"batch_size": 8}, 4) == 8
+
+  fmt, lists = tfdata.get_data_format_and_filenames_list(patterns)
+  assert fmt == "tfrecord" and len(lists) == 1 and len(lists[0]) == 2
+
+  fmt, shards, per_shard = tfdata.get_dataset_metadata(patterns)
+  assert (fmt, shards, per_shard) == ("tfrecord", 2, 5)
+
+  feature_spec = tsu.TensorSpecStruct()
+  feature_spec["vec"] = T((3,), torch.float32, name="vec")
+  label_spec = tsu.TensorSpecStruct()
+  label_spec["label"] = T((1,), torch.float32, name="label")
+
+  parsed = list(tfdata.serialized_to_parsed(
+      [records[:4]], feature_spec, label_spec))
+  assert parsed[0][0]["vec"].shape == (4, 3)
+
+  input_fn = tfdata.get_input_fn(feature_spec, label_spec, patterns,
+                                 run_modes.EVAL, batch_size=2)
+  batches = list(input_fn())
+  assert len(batches) == 5  # 10 examples / bs 2, no repeat in EVAL
+  f, l = batches[0]
+  assert f["vec"].shape == (2, 3) and l["label"].shape == (2, 1)
+  # params batch-size override (reference TPU semantics).
+  f, l = next(iter(input_fn({"batch_size": 5})))
+  assert f["vec"].shape == (5, 3)
