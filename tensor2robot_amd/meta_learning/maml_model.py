@@ -274,3 +274,11 @@ class MAMLModel(abstract_model.AbstractT2RModel):
     return self._base_model.model_eval_fn(
         features_flat, labels_flat, inference_flat, train_loss,
         train_outputs, mode, params)
+
+
+def pfor_map_fn(fn, elems):
+  """map_fn over dim 0 via torch.func.vmap (reference maml_model.py:43-69
+  wrapped tf pfor; same contract: slices of `elems` through `fn`,
+  results stacked on dim 0).  fn must be vmap-compatible (no in-place
+  buffer mutation)."""
+  return torch.func.vmap(fn, randomness="different")(elems)

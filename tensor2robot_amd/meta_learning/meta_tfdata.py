@@ -15,7 +15,9 @@ import torch
 
 from tensor2robot_amd.specs import tensorspec_utils as tsu
 
-TrainValPair = collections.namedtuple("TrainValPair", ["train", "val"])
+TrainValPair = collections.namedtuple("TrainValPair",
+                                      ["train", "val", "val_mode"])
+TrainValPair.__new__.__defaults__ = (None,)  # val_mode optional
 
 
 def parallel_read(file_patterns, parse_fn, shuffle_filenames: bool = True,
@@ -126,3 +128,34 @@ def split_train_val(struct, num_train_samples: int) -> TrainValPair:
   train = _map_struct(lambda t: t[:, :num_train_samples], struct)
   val = _map_struct(lambda t: t[:, num_train_samples:], struct)
   return TrainValPair(train, val)
+
+
+def tile_val_mode(pair: TrainValPair) -> TrainValPair:
+  """Tiles pair.val_mode over samples-per-task (reference :154-171).
+
+  Requires num_train_samples_per_task == num_val_samples_per_task,
+  like the reference; returns a new pair (namedtuples are immutable).
+  """
+  train_tensor = next(iter(
+      tsu.flatten_spec_structure(pair.train).values()))
+  val_tensor = next(iter(tsu.flatten_spec_structure(pair.val).values()))
+  n_train, n_val = train_tensor.shape[1], val_tensor.shape[1]
+  if n_train != n_val:
+    raise ValueError("Flattening example and batch dimensions requires "
+                     "num_train_samples and num_val_samples to be the "
+                     "same.")
+  if pair.val_mode is None:
+    raise ValueError("pair.val_mode is not set")
+  return pair._replace(val_mode=pair.val_mode.repeat(n_train, 1))
+
+
+def merge_first_n_dims(structure, n: int):
+  """Merges the first n dims of every tensor (reference :222-238)."""
+  return _map_struct(lambda t: t.reshape(-1, *t.shape[n:]), structure)
+
+
+def expand_batch_dims(structure, batch_sizes):
+  """Unmerges dim 0 into `batch_sizes` (reference :241-257)."""
+  sizes = [int(b) for b in batch_sizes]
+  return _map_struct(lambda t: t.reshape(*sizes, *t.shape[1:]),
+                     structure)
