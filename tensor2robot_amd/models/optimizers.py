@@ -213,3 +213,38 @@ class ExponentialMovingAverage:
 def create_moving_average_optimizer(decay: float = 0.9999):
   """Returns the EMA decay config used by the train loop (reference :133)."""
   return decay
+
+
+@gin.configurable
+def create_gradient_descent_optimizer(
+    learning_rate=1e-4, clip_gradient_norm: float = 0.0):
+  """Plain gradient descent factory (reference optimizers.py:83-100) —
+  SGD without momentum."""
+  return create_sgd_optimizer(learning_rate=learning_rate,
+                              clip_gradient_norm=clip_gradient_norm)
+
+
+@gin.configurable
+def create_swapping_saver(ema, model_dir: str,
+                          keep_checkpoint_every_n_hours: float = 1.0,
+                          max_to_keep: int = 5):
+  """Saver that writes the EMA (averaged) weights (reference
+  optimizers.py:150-160 swapping_saver): every save swaps the averaged
+  parameters in around the snapshot — the MovingAverageOptimizer
+  swapping-saver contract, bound to our Checkpointer."""
+  from tensor2robot_amd.train import checkpointing
+  ckpt = checkpointing.Checkpointer(
+      model_dir, max_to_keep=max_to_keep,
+      keep_checkpoint_every_n_hours=keep_checkpoint_every_n_hours)
+
+  class _SwappingSaver:
+    checkpointer = ckpt
+
+    def save(self, step, network, optimizer=None, extra=None):
+      return ckpt.save(step, network, optimizer=optimizer, ema=ema,
+                       extra=extra)
+
+    def __getattr__(self, name):
+      return getattr(ckpt, name)
+
+  return _SwappingSaver()

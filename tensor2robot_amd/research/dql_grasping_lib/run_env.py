@@ -95,3 +95,10 @@ def run_env(env, policy=None, explore_schedule=None, episode_to_transitions_fn=N
     if summary_writer is not None:
       summary_writer.close()
   return episode_rewards
+
+
+def encode_image_array_as_png_str(image) -> bytes:
+  """uint8 HWC array -> PNG bytes (reference run_env.py helper)."""
+  import numpy as _np
+  from tensor2robot_amd.data import image_codec
+  return image_codec.encode_png(_np.asarray(image, _np.uint8))

@@ -358,3 +358,17 @@ class TD3Hooks(HookBuilder):
         periodic.end(context)
 
     return [_Chain()]
+
+
+# Reference class names (hooks/checkpoint_hooks.py:51,91 and
+# hooks/gin_config_hook_builder.py): the torch-native implementations
+# above under the names a reference user would import.
+CheckpointExportListener = CheckpointExportHook
+LaggedCheckpointListener = LaggedCheckpointExportHook
+
+
+class OperativeGinConfigLoggerHookBuilder(HookBuilder):
+  """Builds GinConfigLoggerHook (reference gin_config_hook_builder.py)."""
+
+  def create_hooks(self, t2r_model, trainer):
+    return [GinConfigLoggerHook()]

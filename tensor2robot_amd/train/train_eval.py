@@ -565,3 +565,59 @@ def predict_from_model(t2r_model=None, input_generator=None,
   input_generator.set_specification_from_model(t2r_model, PREDICT)
   input_fn = input_generator.create_dataset_input_fn(PREDICT)
   return trainer.predict(input_fn, yield_single_examples)
+
+
+# ---------------------------------------------------------------------------
+# Reference-named helpers (utils/train_eval.py:61-95,97-126,687-717).
+# ---------------------------------------------------------------------------
+
+
+def print_spec(tensor_spec):
+  """Logs a spec structure's entries in sorted order (reference :61-70)."""
+  import logging as _logging
+  log = _logging.getLogger(__name__)
+  from tensor2robot_amd.specs import tensorspec_utils as _tsu
+  for key, value in sorted(
+      _tsu.flatten_spec_structure(tensor_spec).items()):
+    log.info("%s: %s", key, value)
+
+
+def print_specification(t2r_model):
+  """Logs the model preprocessor's in-specs (reference :73-95)."""
+  import logging as _logging
+  log = _logging.getLogger(__name__)
+  from tensor2robot_amd.utils import modes as _modes
+  for mode in (_modes.TRAIN, _modes.PREDICT):
+    log.info("Preprocessor in feature specification for mode %s", mode)
+    print_spec(t2r_model.preprocessor.get_in_feature_specification(mode))
+    log.info("Preprocessor in label specification.")
+    print_spec(t2r_model.preprocessor.get_in_label_specification(mode))
+
+
+def provide_input_generator_with_model_information(
+    input_generator_instance, t2r_model, mode):
+  """Fills an input generator with the model's specs + preprocessor
+  (reference :97-126); returns the configured generator."""
+  input_generator_instance.set_specification_from_model(t2r_model, mode)
+  return input_generator_instance
+
+
+def save_copy(src_filename: str, dest_filename: str,
+              overwrite: bool = False, num_retries: int = 3,
+              sleep_time: float = 0.5) -> bool:
+  """Copies a file with retries (reference :687-717); returns success."""
+  import logging as _logging
+  import shutil
+  import time as _time
+  log = _logging.getLogger(__name__)
+  if os.path.exists(dest_filename) and not overwrite:
+    log.info("Not overwriting existing %s", dest_filename)
+    return False
+  for _ in range(num_retries):
+    try:
+      shutil.copyfile(src_filename, dest_filename)
+      return True
+    except OSError as e:
+      log.warning("save_copy failed (%r); retrying", e)
+      _time.sleep(sleep_time)
+  return False

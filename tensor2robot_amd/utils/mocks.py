@@ -129,3 +129,8 @@ class MockInputGenerator(input_generators.GeneratorInputGenerator):
 class MockExportGenerator(
     abstract_export_generator.AbstractExportGenerator):
   pass
+
+
+class MockTF2T2RModel(MockT2RModel):
+  """Reference mocks.py MockTF2T2RModel: the TF2-flavored mock is the
+  same torch model here (no session/graph split to emulate)."""

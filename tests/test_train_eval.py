@@ -10,6 +10,7 @@ from tensor2robot_amd.models import optimizers
 from tensor2robot_amd.train import checkpointing
 from tensor2robot_amd.train import train_eval
 from tensor2robot_amd.utils import mocks
+from tensor2robot_amd.utils import modes as run_modes
 
 
 def _make_model(**kwargs):
@@ -170,3 +171,43 @@ def test_multi_eval_input_generator(tmp_path, monkeypatch):
                 "holdout": "/data/b*.tfrecord"},
       batch_size=2)
   assert gen2._file_patterns == "/data/a*.tfrecord"
+
+
+def test_reference_named_train_eval_helpers(tmp_path):
+  """print_spec/print_specification/provide_input_generator.../save_copy
+  (reference utils/train_eval.py:61-126,687-717)."""
+  from tensor2robot_amd.train import train_eval as te
+  model = mocks.MockT2RModel()
+  te.print_specification(model)  # logs, must not raise
+  gen = mocks.MockInputGenerator(batch_size=4)
+  out = te.provide_input_generator_with_model_information(
+      gen, model, run_modes.TRAIN)
+  assert out is gen
+  f, l = next(iter(gen.create_dataset_input_fn(run_modes.TRAIN)()))
+  assert f["measured_position"].shape[0] == 4
+  src = tmp_path / "src.txt"
+  src.write_text("payload")
+  dst = tmp_path / "dst.txt"
+  assert te.save_copy(str(src), str(dst))
+  assert dst.read_text() == "payload"
+  assert not te.save_copy(str(src), str(dst))  # refuses overwrite
+  assert te.save_copy(str(src), str(dst), overwrite=True)
+
+
+def test_swapping_saver_writes_averaged_weights(tmp_path):
+  """create_swapping_saver stores EMA weights as the canonical state
+  (reference optimizers.py:150-160)."""
+  import torch
+  from tensor2robot_amd.models import optimizers
+  net = torch.nn.Linear(3, 2)
+  ema = optimizers.ExponentialMovingAverage(net, decay=0.5)
+  with torch.no_grad():
+    net.weight.add_(1.0)
+  ema.update()
+  saver = optimizers.create_swapping_saver(ema, str(tmp_path),
+                                           max_to_keep=2)
+  path = saver.save(5, net)
+  state = torch.load(path, weights_only=False)
+  assert torch.allclose(state["model_state"]["weight"],
+                        ema.shadow["weight"])
+  assert not torch.allclose(state["model_state"]["weight"], net.weight)
