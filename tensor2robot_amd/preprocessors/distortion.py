@@ -83,3 +83,23 @@ def preprocess_image(image: torch.Tensor, mode: str,
   if leading is not None:
     image = image.reshape(*leading, *image.shape[1:])
   return image
+
+
+def maybe_distort_and_flip_image_batch(images: torch.Tensor, mode: str):
+  """TRAIN-only photometric distortion + random flips (reference
+  distortion.py:39-53); 4D [B,H,W,C] or 5D [B,T,H,W,C]."""
+  from tensor2robot_amd.utils import modes as run_modes
+  if mode != run_modes.TRAIN:
+    return images
+  squeeze = False
+  if images.dim() == 5:
+    b, t = images.shape[:2]
+    images = images.reshape(b * t, *images.shape[2:])
+    squeeze = (b, t)
+  images = image_transformations.ApplyPhotometricImageDistortions(
+      [images])[0]
+  images = image_transformations.ApplyRandomFlips([images])[0]
+  if squeeze:
+    b, t = squeeze
+    images = images.reshape(b, t, *images.shape[1:])
+  return images

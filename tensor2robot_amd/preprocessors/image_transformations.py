@@ -248,3 +248,39 @@ def mixup(images: torch.Tensor, labels: torch.Tensor, alpha: float = 0.2,
   lam_lab = lam.view(-1, *([1] * (labels.dim() - 1)))
   mixed_labels = lam_lab * labels + (1 - lam_lab) * labels[perm]
   return mixed_images, mixed_labels
+
+
+@gin.configurable
+def ApplyPhotometricImageDistortionsParallel(
+    images: torch.Tensor,
+    random_brightness: bool = False,
+    max_delta_brightness: float = 0.125,
+    random_saturation: bool = False,
+    lower_saturation: float = 0.5,
+    upper_saturation: float = 1.5,
+    random_hue: bool = False,
+    max_delta_hue: float = 0.2,
+    random_contrast: bool = False,
+    lower_contrast: float = 0.5,
+    upper_contrast: float = 1.5,
+    random_noise_level: float = 0.0,
+    random_noise_apply_probability: float = 0.5,
+    custom_distortion_fn=None,
+    generator: Optional[torch.Generator] = None) -> torch.Tensor:
+  """Reference :268-363 entry point: one [B,H,W,3] tensor in/out with
+  per-image independent draws (ApplyPhotometricImageDistortions above
+  already batches that way) + the optional custom_distortion_fn."""
+  out = ApplyPhotometricImageDistortions(
+      [images], random_brightness=random_brightness,
+      max_delta_brightness=max_delta_brightness,
+      random_saturation=random_saturation,
+      lower_saturation=lower_saturation,
+      upper_saturation=upper_saturation, random_hue=random_hue,
+      max_delta_hue=max_delta_hue, random_contrast=random_contrast,
+      lower_contrast=lower_contrast, upper_contrast=upper_contrast,
+      random_noise_levels=random_noise_level,
+      random_noise_apply_probability=random_noise_apply_probability,
+      generator=generator)[0]
+  if custom_distortion_fn is not None:
+    out = torch.clamp(custom_distortion_fn(out), 0.0, 1.0)
+  return out
