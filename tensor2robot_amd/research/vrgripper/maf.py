@@ -144,3 +144,18 @@ class MAFDecoder(nn.Module):
 
   def loss(self, labels) -> torch.Tensor:
     return -self.log_prob(labels["action"]).mean()
+
+
+def init_once(x, name: str) -> torch.Tensor:
+  """Constant-initialized buffer-style tensor (reference maf.py:29-47):
+  the permutation must be drawn ONCE and frozen, never re-sampled per
+  call — here a detached clone registered by the caller."""
+  del name
+  t = torch.as_tensor(np.asarray(x))
+  return t.detach().clone()
+
+
+def maf_bijector(event_size: int, num_flows: int,
+                 hidden_layers) -> MAFBijector:
+  """Chain-of-MAF factory under the reference name (maf.py:50-63)."""
+  return MAFBijector(event_size, num_flows, list(hidden_layers))
