@@ -645,3 +645,15 @@ class BCZModel(abstract_model.AbstractT2RModel):
       metrics.update(get_gripper_accuracy_metrics(
           inference_outputs, features, labels))
     return metrics
+
+
+def xyz_action_trajectory(outputs):
+  """Concats xyz + rotation action streams (reference model.py:621-627)."""
+  if "action/quaternion" in outputs:
+    rotation = outputs["action/quaternion"]
+  elif "action/axis_angle" in outputs:
+    rotation = outputs["action/axis_angle"]
+  else:
+    raise KeyError("outputs carry neither action/quaternion nor "
+                   "action/axis_angle")
+  return torch.cat([outputs["action/xyz"], rotation], dim=-1)
