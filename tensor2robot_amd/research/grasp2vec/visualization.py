@@ -32,3 +32,199 @@ def heatmap_keypoints(heatmaps: torch.Tensor
                       ) -> Tuple[torch.Tensor, torch.Tensor]:
   """Soft arg-max keypoints of [N, 1, H, W] heatmaps -> ([N, 2], map)."""
   return ss_mod.SpatialSoftmax()(heatmaps)
+
+
+# ---------------------------------------------------------------------------
+# Reference visualization surface (visualization.py:31-259).  Torch-
+# native design: every function RETURNS the rendered tensors/statistics
+# instead of writing TF summaries — the caller logs them (the summary
+# sink here is scalar JSONL + TB events; images stay artifacts).
+# ---------------------------------------------------------------------------
+
+
+def plot_labels(labels: torch.Tensor, max_label: int = 1,
+                predictions: torch.Tensor = None,
+                name: str = "") -> torch.Tensor:
+  """Labels (and optionally predictions) as a tiny RGB image
+  (reference :31-52): first 3 batch items, labels in the red channel,
+  predictions (when given) stacked below in green."""
+  del name
+  if max_label > 1:
+    labels = torch.nn.functional.one_hot(
+        labels.long().reshape(-1), max_label).float()
+  labels_image = labels[:3].reshape(1, 3, max_label, 1)
+  empty = torch.zeros_like(labels_image)
+  image = torch.cat([labels_image, empty, empty], dim=-1)
+  if predictions is not None:
+    pred_image = predictions[:3].reshape(1, 3, -1, 1)
+    zero = torch.zeros_like(pred_image)
+    image2 = torch.cat([zero, pred_image, zero], dim=-1)
+    image = torch.cat([image, image2], dim=1)
+  return image
+
+
+def plot_distances(pregrasp: torch.Tensor, goal: torch.Tensor,
+                   postgrasp: torch.Tensor):
+  """Embedding-arithmetic evaluation statistics (reference :55-70):
+  returns the dict of distance/similarity vectors the reference
+  histograms."""
+  goal_normalized = goal / (1e-7 + goal.norm(dim=1, keepdim=True))
+  return {
+      "correct_distances": (pregrasp - (goal + postgrasp)).norm(dim=1),
+      "incorrect_distances": (pregrasp - pregrasp.flip(0)).norm(dim=1),
+      "goal_distances": (goal - goal.flip(0)).norm(dim=1),
+      "pregrasp_sizes": pregrasp.norm(dim=1),
+      "postgrasp_sizes": postgrasp.norm(dim=1),
+      "goal_sizes": goal.norm(dim=1),
+      "goal_cosine_similarity": (goal_normalized[:-1] *
+                                 goal_normalized[1:]).sum(dim=1),
+  }
+
+
+def np_render_keypoints(image, locations, num_images: int = 3,
+                        dot_radius: int = 3):
+  """Soft-argmax locations rasterized over greyed images
+  (reference :107-149).  image [N,H,W,3] float in [0,1]; locations
+  [N,C,2] in [-1,1] (x, y).  Returns uint8 [num_images,H,W,3]."""
+  import colorsys
+  import numpy as np
+  image = np.asarray(image)
+  locations = np.asarray(locations)
+  num_images = int(min(num_images, image.shape[0]))
+  _, h, w, _ = image.shape
+  mx, my = np.meshgrid(np.arange(w), np.arange(h))
+  num_points = locations.shape[1]
+  images = []
+  for i in range(num_images):
+    img = np.tile(np.mean(image[i], axis=2, keepdims=True), [1, 1, 3])
+    img = img / 2.0 + 0.4
+    hues = np.linspace(0, 1, num_points + 1)[:-1]
+    colors = [np.array(colorsys.hsv_to_rgb(hue, 1.0, 0.9))
+              for hue in hues]
+    xs = np.round((locations[i, :, 0] + 1.0) * w / 2.0).astype(int)
+    ys = np.round((locations[i, :, 1] + 1.0) * h / 2.0).astype(int)
+    for x, y, color in zip(xs, ys, colors):
+      dist = np.sqrt((x - mx) ** 2 + (y - my) ** 2)
+      weight = np.maximum(np.minimum(dot_radius - dist, 1.0), 0.0)
+      weight = np.tile(np.expand_dims(weight, 2), [1, 1, 3])
+      img = img * (1 - weight) + weight * color.reshape([1, 1, 3])
+    images.append((img * 255).astype(np.uint8))
+  import numpy as _np
+  return _np.stack(images, 0)
+
+
+def _hsv_to_rgb(hsv: torch.Tensor) -> torch.Tensor:
+  """[..., 3] HSV in [0,1] -> RGB (torch)."""
+  h, s, v = hsv[..., 0], hsv[..., 1], hsv[..., 2]
+  i = torch.floor(h * 6.0)
+  f = h * 6.0 - i
+  p = v * (1.0 - s)
+  q = v * (1.0 - f * s)
+  t = v * (1.0 - (1.0 - f) * s)
+  i = (i % 6).long()
+  choices = torch.stack([
+      torch.stack([v, t, p], dim=-1), torch.stack([q, v, p], dim=-1),
+      torch.stack([p, v, t], dim=-1), torch.stack([p, q, v], dim=-1),
+      torch.stack([t, p, v], dim=-1), torch.stack([v, p, q], dim=-1),
+  ], dim=0)
+  idx = i.unsqueeze(-1).expand(i.shape + (3,)).unsqueeze(0)
+  return choices.gather(0, idx).squeeze(0)
+
+
+def get_softmax_viz(image: torch.Tensor, softmax: torch.Tensor,
+                    nrows: int = None) -> torch.Tensor:
+  """Softmax maps tiled into a grid, superimposed on the greyscale
+  image via HSV (reference :199-238).  image [N,H,W,3],
+  softmax [N,h,w,C] -> [N, 2h*nrows, 2w*ncols, 3]."""
+  import torch.nn.functional as F
+  n, h, w, c = softmax.shape
+  th, tw = h * 2, w * 2
+  if nrows is None:
+    nrows = max(d for d in range(1, int(c ** 0.5) + 1) if c % d == 0)
+  ncols = c // nrows
+  img = softmax / softmax.amax(dim=(1, 2), keepdim=True).clamp_min(1e-12)
+  grey = image.mean(dim=-1, keepdim=True).permute(0, 3, 1, 2)
+  grey = F.interpolate(grey, size=(th, tw), mode="bilinear",
+                       align_corners=False).permute(0, 2, 3, 1)
+  grey = grey.expand(n, th, tw, c).reshape(n, th, tw, c, 1)
+  img = F.interpolate(img.permute(0, 3, 1, 2), size=(th, tw),
+                      mode="bilinear", align_corners=False
+                      ).permute(0, 2, 3, 1).reshape(n, th, tw, c, 1)
+  hsv = torch.cat([img / 2.0 + 0.5, img, grey * 0.7 + 0.3], dim=4)
+  hsv = hsv.reshape(n, th, tw, nrows, ncols, 3)
+  hsv = hsv.permute(0, 3, 1, 4, 2, 5)
+  hsv = hsv.reshape(n, th * nrows, tw * ncols, 3)
+  return _hsv_to_rgb(hsv.clamp(0.0, 1.0))
+
+
+def add_spatial_soft_argmax_viz(image: torch.Tensor,
+                                softmax: torch.Tensor,
+                                locations: torch.Tensor,
+                                max_outputs: int = 3,
+                                num_groups: int = 1,
+                                num_rows: int = 1):
+  """Spatial-softmax visualization bundle (reference :153-196):
+  returns {x, y, softmax_avg, locations_overlay, softmax_grid[s]}."""
+  out = {
+      "x": locations[:, :, 0],
+      "y": locations[:, :, 1],
+      "softmax_avg": softmax.mean(dim=3, keepdim=True),
+      "locations_overlay": torch.from_numpy(np_render_keypoints(
+          image.detach().cpu().numpy(),
+          locations.detach().cpu().numpy(), max_outputs)),
+  }
+  if num_groups > 1:
+    for i, group in enumerate(softmax.chunk(num_groups, dim=3)):
+      out[f"softmax_group_{i}"] = get_softmax_viz(image, group, num_rows)
+  else:
+    out["softmax"] = get_softmax_viz(image, softmax, num_rows)
+  return out
+
+
+# Minimal 5x7 bitmap font (digits + a few glyphs) so put_text needs no
+# cv2; enough for step counters / labels on summary images.
+_FONT5X7 = {
+    "0": ["01110", "10001", "10011", "10101", "11001", "10001", "01110"],
+    "1": ["00100", "01100", "00100", "00100", "00100", "00100", "01110"],
+    "2": ["01110", "10001", "00001", "00010", "00100", "01000", "11111"],
+    "3": ["11110", "00001", "00001", "01110", "00001", "00001", "11110"],
+    "4": ["00010", "00110", "01010", "10010", "11111", "00010", "00010"],
+    "5": ["11111", "10000", "11110", "00001", "00001", "10001", "01110"],
+    "6": ["00110", "01000", "10000", "11110", "10001", "10001", "01110"],
+    "7": ["11111", "00001", "00010", "00100", "01000", "01000", "01000"],
+    "8": ["01110", "10001", "10001", "01110", "10001", "10001", "01110"],
+    "9": ["01110", "10001", "10001", "01111", "00001", "00010", "01100"],
+    ".": ["00000", "00000", "00000", "00000", "00000", "01100", "01100"],
+    "-": ["00000", "00000", "00000", "11111", "00000", "00000", "00000"],
+    ":": ["00000", "01100", "01100", "00000", "01100", "01100", "00000"],
+    " ": ["00000"] * 7,
+}
+
+
+def put_text(imgs, texts, text_size: int = 1, text_pos=(0, 30),
+             text_color=(0.0, 0.0, 1.0)):
+  """Rasterizes text onto a batch of images (reference tf_put_text
+  :241-259, sans the cv2 dependency — a built-in 5x7 bitmap font
+  covering digits/./-/: renders step counters and numeric labels)."""
+  import numpy as np
+  imgs = np.array(imgs, copy=True)
+  color = np.asarray(text_color, dtype=imgs.dtype)
+  x0, y0 = int(text_pos[0]), int(text_pos[1])
+  for i in range(imgs.shape[0]):
+    text = texts[i]
+    if isinstance(text, bytes):
+      text = text.decode("utf-8", "replace")
+    x = x0
+    for ch in str(text):
+      glyph = _FONT5X7.get(ch, _FONT5X7[" "])
+      for r, row in enumerate(glyph):
+        for c, bit in enumerate(row):
+          if bit == "1":
+            ys = y0 + r * text_size
+            xs = x + c * text_size
+            ye = min(ys + text_size, imgs.shape[1])
+            xe = min(xs + text_size, imgs.shape[2])
+            if ys < imgs.shape[1] and xs < imgs.shape[2]:
+              imgs[i, ys:ye, xs:xe, :] = color
+      x += 6 * text_size
+  return imgs

@@ -179,3 +179,10 @@ class GraspingCEMPolicy(policies_mod.CEMPolicy):
           action[..., offset: offset + size], np.float32)
       offset += size
     return feed
+
+
+def pack_features_kuka_e2e(t2r_model, *policy_inputs):
+  """Policy-input packing for the real Kuka E2E env — unimplemented in
+  the reference too (t2r_models.py:46-57 raises NotImplementedError)."""
+  del t2r_model, policy_inputs
+  raise NotImplementedError

@@ -149,3 +149,34 @@ def test_visualization_heatmap():
   assert float(heat.min()) >= 0.0 and float(heat.max()) <= 1.0
   points, _ = visualization.heatmap_keypoints(heat)
   assert points.shape == (2, 2)
+
+
+def test_visualization_reference_surface():
+  """plot_labels/plot_distances/np_render_keypoints/get_softmax_viz/
+  add_spatial_soft_argmax_viz/put_text (reference visualization.py)."""
+  import numpy as np
+  import torch
+  from tensor2robot_amd.research.grasp2vec import visualization as viz
+  img = viz.plot_labels(torch.tensor([0, 1, 3, 2]), max_label=4,
+                        predictions=torch.rand(4, 4))
+  assert img.shape == (1, 6, 4, 3)  # labels row stacked above preds row
+  assert viz.plot_labels(torch.tensor([[1.], [0.], [1.]])).shape == \
+      (1, 3, 1, 3)
+  d = viz.plot_distances(torch.randn(6, 8), torch.randn(6, 8),
+                         torch.randn(6, 8))
+  assert d["correct_distances"].shape == (6,)
+  assert d["goal_cosine_similarity"].shape == (5,)
+  kp = viz.np_render_keypoints(np.random.rand(4, 16, 16, 3),
+                               np.random.uniform(-1, 1, (4, 5, 2)))
+  assert kp.shape == (3, 16, 16, 3) and kp.dtype == np.uint8
+  grid = viz.get_softmax_viz(torch.rand(2, 16, 16, 3),
+                             torch.rand(2, 8, 8, 4))
+  assert grid.shape == (2, 32, 32, 3)
+  assert float(grid.min()) >= 0.0 and float(grid.max()) <= 1.0
+  bundle = viz.add_spatial_soft_argmax_viz(
+      torch.rand(2, 16, 16, 3), torch.rand(2, 8, 8, 4),
+      torch.rand(2, 4, 2) * 2 - 1, num_groups=2)
+  assert {"x", "y", "softmax_avg", "locations_overlay",
+          "softmax_group_0", "softmax_group_1"} <= set(bundle)
+  txt = viz.put_text(np.zeros((1, 40, 60, 3), np.float32), ["12.5"])
+  assert txt.sum() > 0
