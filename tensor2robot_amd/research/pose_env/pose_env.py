@@ -95,3 +95,16 @@ class PoseEnvRandomPolicy:
 
   def SelectAction(self, state, context=None, timestep: int = 0):
     return self._rng.uniform(-1, 1, 2).astype(np.float32)
+
+
+def get_pybullet_urdf_root() -> str:
+  """Reference pose_env.py:25-31 resolves pybullet_data's URDF path;
+  this rebuild replaces PyBullet with the synthetic blob env, so the
+  path is only meaningful when pybullet happens to be installed."""
+  try:
+    import pybullet_data  # type: ignore
+    return pybullet_data.getDataPath()
+  except ImportError as e:
+    raise ImportError(
+        "pybullet is not part of this MI355X image; the synthetic "
+        "PoseEnv (this module) replaces it") from e

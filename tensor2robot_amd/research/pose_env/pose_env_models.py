@@ -181,3 +181,10 @@ def generate_test_tfrecord(path: str, num_records: int = 20,
       }
       writer.write(example_codec.encode_example(features))
   return path
+
+
+# Reference class names (pose_env_models.py:41,183): both reference
+# preprocessors do the same uint8 -> float32 image conversion the
+# shared _PoseEnvPreprocessor implements.
+DefaultPoseEnvContinuousPreprocessor = _PoseEnvPreprocessor
+DefaultPoseEnvRegressionPreprocessor = _PoseEnvPreprocessor
