@@ -394,3 +394,8 @@ def resnet_init_from_checkpoint_fn(checkpoint_path: str,
     return {"missing": missing, "unexpected": unexpected}
 
   return init_fn
+
+
+# Reference name (film_resnet_model.py:77 / grasp2vec/resnet.py:77):
+# strided conv with kernel-size-based explicit padding.
+conv2d_fixed_padding = _conv_fixed_padding

@@ -133,3 +133,8 @@ def normal_cross_entropy_method(objective_fn, mean, stddev,
       {"mean": mean, "stddev": stddev}, num_elites,
       num_iterations=num_iterations)
   return final_params["mean"], final_params["stddev"]
+
+
+# Reference-exact names (utils/cross_entropy.py:30,110).
+CrossEntropyMethodFn = cross_entropy_method
+NormalCrossEntropyMethod = normal_cross_entropy_method
