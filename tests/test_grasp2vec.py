@@ -180,3 +180,46 @@ def test_visualization_reference_surface():
           "softmax_group_0", "softmax_group_1"} <= set(bundle)
   txt = viz.put_text(np.zeros((1, 40, 60, 3), np.float32), ["12.5"])
   assert txt.sum() > 0
+
+
+def test_arithmetic_losses_mask_semantics():
+  """Zeros mask -> 0; ones mask -> mean over batch; mixed mask -> mean
+  over masked rows only (reference losses_test.py:47-126)."""
+  import numpy as np
+  import torch
+  import torch.nn.functional as F
+  g = torch.Generator().manual_seed(0)
+  pre = torch.randn(6, 8, generator=g)
+  goal = torch.randn(6, 8, generator=g)
+  post = torch.randn(6, 8, generator=g)
+  zeros = torch.zeros(6)
+  ones = torch.ones(6)
+  mixed = torch.zeros(6); mixed[0] = 1
+
+  assert float(losses.L2ArithmeticLoss(pre, goal, post, zeros)) == 0.0
+  assert float(losses.CosineArithmeticLoss(pre, goal, post, zeros)) == 0.0
+
+  l2_all = ((pre - goal - post) ** 2).sum(dim=1)
+  torch.testing.assert_close(
+      losses.L2ArithmeticLoss(pre, goal, post, ones), l2_all.mean())
+  torch.testing.assert_close(
+      losses.L2ArithmeticLoss(pre, goal, post, mixed), l2_all[0])
+
+  pa = F.normalize(pre - post, dim=1)
+  pb = F.normalize(goal, dim=1)
+  cos_all = 1.0 - (pa * pb).sum(dim=1)
+  torch.testing.assert_close(
+      losses.CosineArithmeticLoss(pre, goal, post, ones), cos_all.mean())
+  torch.testing.assert_close(
+      losses.CosineArithmeticLoss(pre, goal, post, mixed), cos_all[0])
+
+
+def test_keypoint_accuracy_quadrants():
+  """Perfect quadrant keypoints -> accuracy 1 (reference :143-148)."""
+  import torch
+  kp = torch.tensor([[0.5, -0.5], [-0.5, -0.5], [0.5, 0.5], [-0.5, 0.5]])
+  labels = torch.arange(4)
+  acc, loss = losses.KeypointAccuracy(kp, labels)
+  assert float(acc) == 1.0 and float(loss) > 0.0
+  wrong = losses.KeypointAccuracy(kp.flip(0), labels)[0]
+  assert float(wrong) < 1.0
