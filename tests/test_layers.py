@@ -359,3 +359,12 @@ def test_spatial_softmax_gumbel_training_path():
   p1, _ = mod(feat)
   p2, _ = mod(feat)
   torch.testing.assert_close(p1, p2)
+
+
+def test_malformed_film_enabled_blocks_raises():
+  """5-entry enabled_block_layers on a 4-stage ResNet raises
+  (reference resnet_test.py:57-66)."""
+  net = resnet.ResNet(resnet_size=18, num_classes=10)
+  with pytest.raises(ValueError):
+    resnet.LinearFiLMGenerator(embedding_dim=16, resnet=net,
+                               enabled_block_layers=[True] * 5)
