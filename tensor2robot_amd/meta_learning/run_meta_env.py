@@ -29,6 +29,7 @@ def run_meta_env(env, policy=None, demo_policy_cls=None,
                  explore_schedule=None, episode_to_transitions_fn=None,
                  replay_writer=None, root_dir: Optional[str] = None,
                  task: int = 0, global_step: int = 0,
+                 num_episodes=None,  # accepted, unused (reference :41)
                  num_tasks: int = 10, num_adaptations_per_task: int = 2,
                  num_episodes_per_adaptation: int = 1, num_demos: int = 1,
                  break_after_one_task: bool = False, tag: str = "collect",

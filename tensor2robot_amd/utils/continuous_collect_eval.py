@@ -9,6 +9,7 @@ model appears.  Any side may die and rejoin (poll-based decoupling).
 from __future__ import annotations
 
 import logging
+import os
 import time
 from typing import Callable, Optional
 
@@ -52,13 +53,17 @@ def collect_eval_loop(collect_env=None, eval_env=None, policy_class=None,
       continue
     last_global_step = global_step
     if collect_env is not None:
+      # Reference output layout (:76-77): collected episodes land under
+      # root_dir/policy_collect, eval episodes under root_dir/eval.
       run_agent_fn(collect_env, policy=policy, global_step=global_step,
-                   root_dir=root_dir, num_episodes=num_collect_episodes,
-                   tag="collect")
+                   root_dir=os.path.join(root_dir, "policy_collect")
+                   if root_dir else root_dir,
+                   num_episodes=num_collect_episodes, tag="collect")
     if eval_env is not None:
       run_agent_fn(eval_env, policy=policy, global_step=global_step,
-                   root_dir=root_dir, num_episodes=num_eval_episodes,
-                   tag="eval")
+                   root_dir=os.path.join(root_dir, "eval")
+                   if root_dir else root_dir,
+                   num_episodes=num_eval_episodes, tag="eval")
     loops += 1
     if global_step >= max_steps:
       return global_step
