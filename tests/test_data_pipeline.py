@@ -274,3 +274,31 @@ def test_tfdata_reference_api(tmp_path):
   # params batch-size override (reference TPU semantics).
   f, l = next(iter(input_fn({"batch_size": 5})))
   assert f["vec"].shape == (5, 3)
+
+
+def test_fractional_record_input_generator(tmp_path):
+  """file_fraction keeps the first fraction of shards (reference
+  default_input_generator_test.py:129-141)."""
+  rng = np.random.RandomState(0)
+  for shard in range(4):
+    recs = [example_codec.encode_example({
+        "vec": rng.rand(3).astype(np.float32),
+        "label": np.array([float(shard)], np.float32),
+    }) for _ in range(4)]
+    _write_records(tmp_path, recs, name=f"frac-{shard}.tfrecord")
+  feature_spec = tsu.TensorSpecStruct()
+  feature_spec["vec"] = T((3,), torch.float32, name="vec")
+  label_spec = tsu.TensorSpecStruct()
+  label_spec["label"] = T((1,), torch.float32, name="label")
+
+  gen = input_generators.FractionalRecordInputGenerator(
+      file_fraction=0.5,
+      file_patterns=str(tmp_path / "frac-*.tfrecord"), batch_size=4,
+      shard_by_rank=False)
+  gen.set_feature_specifications(feature_spec, feature_spec)
+  gen.set_label_specifications(label_spec, label_spec)
+  files = gen._resolve_files()
+  assert all(len(v) == 2 for v in files.values())  # 4 shards -> 2
+  f, l = next(iter(gen.create_dataset_input_fn(run_modes.EVAL)()))
+  # Only labels from the first two shards can appear.
+  assert set(np.unique(l["label"].numpy())) <= {0.0, 1.0}
