@@ -500,3 +500,30 @@ def test_full_lifecycle_train_export_collect_retrain(tmp_path):
       model_dir=retrain_dir)
   assert result["global_step"] == 3
   assert np.isfinite(result["loss"])
+
+
+def test_export_warmup_roundtrip_predict_serialized(tmp_path):
+  """Export -> warmup-request records -> predict_serialized: the
+  serialized-example serving contract end to end (reference
+  default_export_generator.py:84-142)."""
+  from tensor2robot_amd.data import tfrecord
+  from tensor2robot_amd.export_generators import (
+      abstract_export_generator)
+  model, export_dir = _train_and_export(tmp_path, steps=10)
+
+  gen = abstract_export_generator.AbstractExportGenerator()
+  gen.set_specification_from_model(model)
+  warmup_path = gen.create_warmup_requests_numpy([1, 3], str(tmp_path))
+  records = list(tfrecord.read_records(warmup_path))
+  assert len(records) == 4  # 1 + 3 single-example records
+
+  predictor = esp.ExportedSavedModelPredictor(export_dir, timeout=5)
+  assert predictor.restore()
+  # Replay the batch-3 warmup group as one serialized predict call.
+  out = predictor.predict_serialized(records[1:])
+  assert out["prediction"].shape == (3, 1)
+  # Serialized path == numpy path on the same (zero) input.
+  ref = predictor.predict(
+      {"measured_position": np.zeros((3, 3), np.float32)})
+  np.testing.assert_allclose(out["prediction"], ref["prediction"],
+                             rtol=1e-5)
