@@ -105,6 +105,50 @@ class MAMLPreprocessorV2(abstract_preprocessor.AbstractPreprocessor):
                                                   mode)
     return unfold(f), (unfold(l) if l is not None else None)
 
+  def create_meta_map_fn(self, num_condition_samples_per_task,
+                         num_inference_samples_per_task):
+    """Batch -> (meta_features, meta_labels) regrouping map
+    (reference preprocessors.py:132-232): the first
+    num_condition_samples_per_task rows of each batch become the
+    condition set, the rest the inference set; every tensor's batch
+    size must equal their sum."""
+    if (num_condition_samples_per_task is None or
+        num_condition_samples_per_task <= 0):
+      raise ValueError(
+          "num_condition_samples_per_task cannot be None and has to be "
+          f"positive but is {num_condition_samples_per_task}.")
+    if (num_inference_samples_per_task is None or
+        num_inference_samples_per_task <= 0):
+      raise ValueError(
+          "num_inference_samples_per_task cannot be None and has to be "
+          f"positive but is {num_inference_samples_per_task}.")
+    ref_batch_size = (num_condition_samples_per_task +
+                      num_inference_samples_per_task)
+
+    def map_fn(features, labels):
+      flat_f = tsu.flatten_spec_structure(features)
+      flat_l = tsu.flatten_spec_structure(labels)
+      for struct in (flat_f, flat_l):
+        for key, t in struct.items():
+          if t.shape[0] != ref_batch_size:
+            raise ValueError(
+                f"{key}: batch size has to be num_condition_samples_"
+                "per_task + num_inference_samples_per_task = "
+                f"{ref_batch_size} but is {t.shape[0]}.")
+      n = num_condition_samples_per_task
+      meta_features = tsu.TensorSpecStruct()
+      for key, t in flat_f.items():
+        meta_features["condition/features/" + key] = t[:n]
+        meta_features["inference/features/" + key] = t[n:]
+      for key, t in flat_l.items():
+        meta_features["condition/labels/" + key] = t[:n]
+      meta_labels = tsu.TensorSpecStruct()
+      for key, t in flat_l.items():
+        meta_labels[key] = t[n:]
+      return meta_features, meta_labels
+
+    return map_fn
+
   def _preprocess_fn(self, features, labels, mode):
     out = tsu.TensorSpecStruct()
     cond_f = features["condition/features"]

@@ -309,3 +309,33 @@ def test_maml_model_eval_fn():
                                 ops.loss, ops.train_outputs,
                                 run_modes.EVAL)
   assert "accuracy" in metrics
+
+
+def test_maml_preprocessor_v2_create_meta_map_fn():
+  """Batch regrouping + validation (reference preprocessors_test.py
+  :130-232)."""
+  prep = meta_prep.MAMLPreprocessorV2(
+      base_preprocessor=mocks.MockT2RModel().preprocessor)
+  for bad in ((None, 1), (1, None), (-1, 1), (1, -1)):
+    with pytest.raises(ValueError):
+      prep.create_meta_map_fn(*bad)
+
+  map_fn = prep.create_meta_map_fn(2, 1)
+  f = tsu.TensorSpecStruct()
+  f["x"] = torch.arange(9.0).reshape(3, 3)
+  l = tsu.TensorSpecStruct()
+  l["y"] = torch.arange(3.0).reshape(3, 1)
+  meta_f, meta_l = map_fn(f, l)
+  torch.testing.assert_close(meta_f["condition/features/x"], f["x"][:2])
+  torch.testing.assert_close(meta_f["inference/features/x"], f["x"][2:])
+  torch.testing.assert_close(meta_f["condition/labels/y"], l["y"][:2])
+  torch.testing.assert_close(meta_l["y"], l["y"][2:])
+
+  # Wrong batch size raises (both too small and too large).
+  for n in (2, 4):
+    bad_f = tsu.TensorSpecStruct()
+    bad_f["x"] = torch.zeros(n, 3)
+    bad_l = tsu.TensorSpecStruct()
+    bad_l["y"] = torch.zeros(n, 1)
+    with pytest.raises(ValueError):
+      map_fn(bad_f, bad_l)
