@@ -52,6 +52,7 @@ ext_modules = [
     Extension(
         name="tensor2robot_amd.ops._t2r_native",
         sources=[os.path.join(NATIVE_DIR, "jpeg_codec.cpp"),
+                 os.path.join(NATIVE_DIR, "example_codec.cpp"),
                  os.path.join(NATIVE_DIR, "native_bindings.cpp")],
         include_dirs=[pybind11.get_include()],
         extra_compile_args=["-O3", "-std=c++17"],

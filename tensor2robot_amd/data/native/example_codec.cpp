@@ -1,0 +1,248 @@
+// Native tf.Example wire-format batch decoder.
+//
+// The reference's Example parsing runs inside TensorFlow's C++ runtime
+// (utils/tfdata.py serialized_to_parsed -> tf.parse_example kernels);
+// this is the equivalent native stage for the MI355X pipeline: the
+// whole batch's proto scan runs WITHOUT the GIL, and only the final
+// {name: list[bytes] | float32 array | int64 array} dicts are built
+// under it.  Semantics mirror data/example.py decode_example exactly
+// (first list field wins, packed or unpacked scalars, two's-complement
+// int64, empty feature -> []).
+
+#include <pybind11/pybind11.h>
+#include <pybind11/numpy.h>
+
+#include <cstdint>
+#include <cstring>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+namespace py = pybind11;
+
+namespace {
+
+struct Slice {
+  const uint8_t* p = nullptr;
+  size_t n = 0;
+};
+
+uint64_t read_varint(const uint8_t* d, size_t n, size_t& pos) {
+  uint64_t result = 0;
+  int shift = 0;
+  while (pos < n) {
+    uint8_t b = d[pos++];
+    result |= (uint64_t)(b & 0x7f) << shift;
+    if (!(b & 0x80)) return result;
+    shift += 7;
+    if (shift >= 64) break;
+  }
+  throw std::runtime_error("Malformed varint");
+}
+
+struct FeatureOut {
+  int kind = 0;  // 0 = empty, 1 = bytes, 2 = float, 3 = int64
+  std::vector<Slice> bytes_items;
+  std::vector<float> floats;
+  std::vector<int64_t> ints;
+};
+
+constexpr int kWtVarint = 0;
+constexpr int kWtI32 = 5;
+constexpr int kWtLen = 2;
+
+void decode_bytes_list(Slice s, FeatureOut& out) {
+  size_t pos = 0;
+  while (pos < s.n) {
+    uint64_t tag = read_varint(s.p, s.n, pos);
+    if ((tag >> 3) != 1 || (tag & 7) != kWtLen)
+      throw std::runtime_error("Malformed BytesList");
+    uint64_t ln = read_varint(s.p, s.n, pos);
+    if (pos + ln > s.n) throw std::runtime_error("Malformed BytesList");
+    out.bytes_items.push_back({s.p + pos, (size_t)ln});
+    pos += ln;
+  }
+}
+
+void decode_float_list(Slice s, FeatureOut& out) {
+  size_t pos = 0;
+  while (pos < s.n) {
+    uint64_t tag = read_varint(s.p, s.n, pos);
+    int field = (int)(tag >> 3), wt = (int)(tag & 7);
+    if (field != 1) throw std::runtime_error("Malformed FloatList");
+    if (wt == kWtLen) {  // packed
+      uint64_t ln = read_varint(s.p, s.n, pos);
+      if (pos + ln > s.n) throw std::runtime_error("Malformed FloatList");
+      size_t cnt = ln / 4;
+      size_t base = out.floats.size();
+      out.floats.resize(base + cnt);
+      memcpy(out.floats.data() + base, s.p + pos, cnt * 4);
+      pos += ln;
+    } else if (wt == kWtI32) {
+      if (pos + 4 > s.n) throw std::runtime_error("Malformed FloatList");
+      float v;
+      memcpy(&v, s.p + pos, 4);
+      out.floats.push_back(v);
+      pos += 4;
+    } else {
+      throw std::runtime_error("Malformed FloatList wire type");
+    }
+  }
+}
+
+void decode_int64_list(Slice s, FeatureOut& out) {
+  size_t pos = 0;
+  while (pos < s.n) {
+    uint64_t tag = read_varint(s.p, s.n, pos);
+    int field = (int)(tag >> 3), wt = (int)(tag & 7);
+    if (field != 1) throw std::runtime_error("Malformed Int64List");
+    if (wt == kWtLen) {  // packed
+      uint64_t ln = read_varint(s.p, s.n, pos);
+      size_t end = pos + ln;
+      if (end > s.n) throw std::runtime_error("Malformed Int64List");
+      while (pos < end)
+        out.ints.push_back((int64_t)read_varint(s.p, s.n, pos));
+    } else if (wt == kWtVarint) {
+      out.ints.push_back((int64_t)read_varint(s.p, s.n, pos));
+    } else {
+      throw std::runtime_error("Malformed Int64List wire type");
+    }
+  }
+}
+
+// Feature message: first of bytes_list(1)/float_list(2)/int64_list(3)
+// wins, exactly like decode_feature's early return.
+FeatureOut decode_feature(Slice s) {
+  FeatureOut out;
+  size_t pos = 0;
+  while (pos < s.n) {
+    uint64_t tag = read_varint(s.p, s.n, pos);
+    int field = (int)(tag >> 3), wt = (int)(tag & 7);
+    if (wt != kWtLen)
+      throw std::runtime_error("Unexpected wire type in Feature");
+    uint64_t ln = read_varint(s.p, s.n, pos);
+    if (pos + ln > s.n) throw std::runtime_error("Malformed Feature");
+    Slice payload{s.p + pos, (size_t)ln};
+    pos += ln;
+    if (field == 1) {
+      out.kind = 1;
+      decode_bytes_list(payload, out);
+      return out;
+    }
+    if (field == 2) {
+      out.kind = 2;
+      decode_float_list(payload, out);
+      return out;
+    }
+    if (field == 3) {
+      out.kind = 3;
+      decode_int64_list(payload, out);
+      return out;
+    }
+  }
+  return out;  // empty feature -> []
+}
+
+using ExampleOut = std::vector<std::pair<Slice, FeatureOut>>;
+
+ExampleOut decode_features_msg(Slice s) {
+  ExampleOut out;
+  size_t pos = 0;
+  while (pos < s.n) {
+    uint64_t tag = read_varint(s.p, s.n, pos);
+    if ((tag >> 3) != 1 || (tag & 7) != kWtLen)
+      throw std::runtime_error("Malformed Features");
+    uint64_t ln = read_varint(s.p, s.n, pos);
+    if (pos + ln > s.n) throw std::runtime_error("Malformed Features");
+    Slice entry{s.p + pos, (size_t)ln};
+    pos += ln;
+    Slice name{};
+    Slice value{};
+    bool has_name = false, has_value = false;
+    size_t epos = 0;
+    while (epos < entry.n) {
+      uint64_t etag = read_varint(entry.p, entry.n, epos);
+      uint64_t eln = read_varint(entry.p, entry.n, epos);
+      if (epos + eln > entry.n)
+        throw std::runtime_error("Malformed Features entry");
+      Slice payload{entry.p + epos, (size_t)eln};
+      epos += eln;
+      if ((etag >> 3) == 1) { name = payload; has_name = true; }
+      else if ((etag >> 3) == 2) { value = payload; has_value = true; }
+    }
+    if (has_name) {
+      FeatureOut f = has_value ? decode_feature(value) : FeatureOut{};
+      out.emplace_back(name, std::move(f));
+    }
+  }
+  return out;
+}
+
+ExampleOut decode_example(Slice s) {
+  size_t pos = 0;
+  while (pos < s.n) {
+    uint64_t tag = read_varint(s.p, s.n, pos);
+    if ((tag & 7) != kWtLen)
+      throw std::runtime_error("Malformed Example");
+    uint64_t ln = read_varint(s.p, s.n, pos);
+    if (pos + ln > s.n) throw std::runtime_error("Malformed Example");
+    Slice payload{s.p + pos, (size_t)ln};
+    pos += ln;
+    if ((tag >> 3) == 1) return decode_features_msg(payload);
+  }
+  return {};
+}
+
+}  // namespace
+
+py::list parse_example_batch(py::sequence records) {
+  // Pin the record buffers under the GIL, scan the whole batch without
+  // it, then materialize Python objects.
+  const size_t n = py::len(records);
+  std::vector<Slice> slices(n);
+  std::vector<py::object> keepalive;
+  keepalive.reserve(n);
+  for (size_t i = 0; i < n; ++i) {
+    py::object rec = records[i];
+    char* buf;
+    Py_ssize_t ln;
+    if (PyBytes_AsStringAndSize(rec.ptr(), &buf, &ln) != 0)
+      throw py::type_error("parse_example_batch expects bytes records");
+    slices[i] = {(const uint8_t*)buf, (size_t)ln};
+    keepalive.push_back(std::move(rec));
+  }
+  std::vector<ExampleOut> parsed(n);
+  {
+    py::gil_scoped_release release;
+    for (size_t i = 0; i < n; ++i) parsed[i] = decode_example(slices[i]);
+  }
+  py::list out;
+  for (size_t i = 0; i < n; ++i) {
+    py::dict d;
+    for (auto& kv : parsed[i]) {
+      py::str name(reinterpret_cast<const char*>(kv.first.p),
+                   kv.first.n);
+      FeatureOut& f = kv.second;
+      if (f.kind == 1) {
+        py::list items;
+        for (auto& b : f.bytes_items)
+          items.append(py::bytes(reinterpret_cast<const char*>(b.p),
+                                 b.n));
+        d[name] = items;
+      } else if (f.kind == 2) {
+        py::array_t<float> arr((py::ssize_t)f.floats.size());
+        memcpy(arr.mutable_data(), f.floats.data(),
+               f.floats.size() * 4);
+        d[name] = arr;
+      } else if (f.kind == 3) {
+        py::array_t<int64_t> arr((py::ssize_t)f.ints.size());
+        memcpy(arr.mutable_data(), f.ints.data(), f.ints.size() * 8);
+        d[name] = arr;
+      } else {
+        d[name] = py::list();
+      }
+    }
+    out.append(d);
+  }
+  return out;
+}

@@ -9,6 +9,8 @@ namespace py = pybind11;
 
 #include "jpeg_codec.h"
 
+py::list parse_example_batch(py::sequence records);
+
 static py::bytes encode_jpeg(
     py::array_t<uint8_t, py::array::c_style | py::array::forcecast> image,
     int quality, int restart_interval) {
@@ -85,4 +87,7 @@ PYBIND11_MODULE(_t2r_native, m) {
   m.def("decode_jpeg", &decode_jpeg, py::arg("data"));
   m.def("decode_jpeg_coeffs", &decode_jpeg_coeffs, py::arg("data"),
         py::arg("num_threads") = 1);
+  m.def("parse_example_batch", &parse_example_batch,
+        py::arg("records"),
+        "batch tf.Example wire decode (GIL-released scan)");
 }

@@ -120,17 +120,28 @@ class ExampleParser:
     self._has_sequence = any(
         s.is_sequence for s in specs.values())
 
+  @staticmethod
+  def _decode_example_batch(records: List[bytes]) -> List[Dict]:
+    """Whole-batch wire decode: the C++ scanner runs GIL-released
+    (data/native/example_codec.cpp) when built, else the python codec."""
+    try:
+      from tensor2robot_amd.ops import _t2r_native
+      return _t2r_native.parse_example_batch(list(records))
+    except ImportError:
+      return [example_codec.decode_example(r) for r in records]
+
   def __call__(self, records: List[bytes]) -> tsu.TensorSpecStruct:
     """records: batch of serialized Example/SequenceExample protos."""
     batch = len(records)
     per_key_values: Dict[str, list] = {k: [] for k in self._specs.keys()}
     lengths: Dict[str, List[int]] = {}
 
-    for raw in records:
-      if self._has_sequence:
-        context, feature_lists = example_codec.decode_sequence_example(raw)
-      else:
-        context, feature_lists = example_codec.decode_example(raw), {}
+    if self._has_sequence:
+      decoded = [example_codec.decode_sequence_example(r)
+                 for r in records]
+    else:
+      decoded = [(c, {}) for c in self._decode_example_batch(records)]
+    for context, feature_lists in decoded:
       for key, spec in self._specs.items():
         name = spec.name or key
         if spec.is_sequence:
