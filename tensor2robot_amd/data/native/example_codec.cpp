@@ -246,3 +246,136 @@ py::list parse_example_batch(py::sequence records) {
   }
   return out;
 }
+
+namespace {
+
+struct SeqOut {
+  ExampleOut context;
+  std::vector<std::pair<Slice, std::vector<FeatureOut>>> feature_lists;
+};
+
+// SequenceExample: context Features (field 1) + FeatureLists (field 2:
+// repeated entries of {name(1), FeatureList(2) = repeated Feature(1)}).
+// Mirrors data/example.py decode_sequence_example.
+SeqOut decode_sequence_example(Slice s) {
+  SeqOut out;
+  size_t pos = 0;
+  while (pos < s.n) {
+    uint64_t tag = read_varint(s.p, s.n, pos);
+    if ((tag & 7) != kWtLen)
+      throw std::runtime_error("Malformed SequenceExample");
+    uint64_t ln = read_varint(s.p, s.n, pos);
+    if (pos + ln > s.n)
+      throw std::runtime_error("Malformed SequenceExample");
+    Slice payload{s.p + pos, (size_t)ln};
+    pos += ln;
+    int field = (int)(tag >> 3);
+    if (field == 1) {
+      out.context = decode_features_msg(payload);
+    } else if (field == 2) {
+      size_t fpos = 0;
+      while (fpos < payload.n) {
+        read_varint(payload.p, payload.n, fpos);  // entry tag
+        uint64_t fln = read_varint(payload.p, payload.n, fpos);
+        if (fpos + fln > payload.n)
+          throw std::runtime_error("Malformed FeatureLists");
+        Slice entry{payload.p + fpos, (size_t)fln};
+        fpos += fln;
+        Slice name{};
+        bool has_name = false;
+        std::vector<FeatureOut> steps;
+        size_t epos = 0;
+        while (epos < entry.n) {
+          uint64_t etag = read_varint(entry.p, entry.n, epos);
+          uint64_t eln = read_varint(entry.p, entry.n, epos);
+          if (epos + eln > entry.n)
+            throw std::runtime_error("Malformed FeatureList entry");
+          Slice inner{entry.p + epos, (size_t)eln};
+          epos += eln;
+          if ((etag >> 3) == 1) {
+            name = inner;
+            has_name = true;
+          } else if ((etag >> 3) == 2) {
+            size_t ipos = 0;
+            while (ipos < inner.n) {
+              read_varint(inner.p, inner.n, ipos);  // Feature tag
+              uint64_t iln = read_varint(inner.p, inner.n, ipos);
+              if (ipos + iln > inner.n)
+                throw std::runtime_error("Malformed FeatureList");
+              steps.push_back(
+                  decode_feature({inner.p + ipos, (size_t)iln}));
+              ipos += iln;
+            }
+          }
+        }
+        if (has_name)
+          out.feature_lists.emplace_back(name, std::move(steps));
+      }
+    }
+  }
+  return out;
+}
+
+py::object feature_to_py(FeatureOut& f) {
+  if (f.kind == 1) {
+    py::list items;
+    for (auto& b : f.bytes_items)
+      items.append(py::bytes(reinterpret_cast<const char*>(b.p), b.n));
+    return items;
+  }
+  if (f.kind == 2) {
+    py::array_t<float> arr((py::ssize_t)f.floats.size());
+    memcpy(arr.mutable_data(), f.floats.data(), f.floats.size() * 4);
+    return arr;
+  }
+  if (f.kind == 3) {
+    py::array_t<int64_t> arr((py::ssize_t)f.ints.size());
+    memcpy(arr.mutable_data(), f.ints.data(), f.ints.size() * 8);
+    return arr;
+  }
+  return py::list();
+}
+
+}  // namespace
+
+py::list parse_sequence_example_batch(py::sequence records) {
+  const size_t n = py::len(records);
+  std::vector<Slice> slices(n);
+  std::vector<py::object> keepalive;
+  keepalive.reserve(n);
+  for (size_t i = 0; i < n; ++i) {
+    py::object rec = records[i];
+    char* buf;
+    Py_ssize_t ln;
+    if (PyBytes_AsStringAndSize(rec.ptr(), &buf, &ln) != 0)
+      throw py::type_error(
+          "parse_sequence_example_batch expects bytes records");
+    slices[i] = {(const uint8_t*)buf, (size_t)ln};
+    keepalive.push_back(std::move(rec));
+  }
+  std::vector<SeqOut> parsed(n);
+  {
+    py::gil_scoped_release release;
+    for (size_t i = 0; i < n; ++i)
+      parsed[i] = decode_sequence_example(slices[i]);
+  }
+  py::list out;
+  for (size_t i = 0; i < n; ++i) {
+    py::dict ctx;
+    for (auto& kv : parsed[i].context) {
+      py::str name(reinterpret_cast<const char*>(kv.first.p),
+                   kv.first.n);
+      ctx[name] = feature_to_py(kv.second);
+    }
+    py::dict fls;
+    for (auto& kv : parsed[i].feature_lists) {
+      py::str name(reinterpret_cast<const char*>(kv.first.p),
+                   kv.first.n);
+      py::list steps;
+      for (auto& f : kv.second) steps.append(feature_to_py(f));
+      fls[name] = steps;
+    }
+    out.append(py::make_tuple(ctx, fls));
+  }
+  return out;
+}

@@ -10,6 +10,7 @@ namespace py = pybind11;
 #include "jpeg_codec.h"
 
 py::list parse_example_batch(py::sequence records);
+py::list parse_sequence_example_batch(py::sequence records);
 
 static py::bytes encode_jpeg(
     py::array_t<uint8_t, py::array::c_style | py::array::forcecast> image,
@@ -90,4 +91,7 @@ PYBIND11_MODULE(_t2r_native, m) {
   m.def("parse_example_batch", &parse_example_batch,
         py::arg("records"),
         "batch tf.Example wire decode (GIL-released scan)");
+  m.def("parse_sequence_example_batch", &parse_sequence_example_batch,
+        py::arg("records"),
+        "batch tf.SequenceExample wire decode (GIL-released scan)");
 }

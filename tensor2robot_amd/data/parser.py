@@ -137,8 +137,12 @@ class ExampleParser:
     lengths: Dict[str, List[int]] = {}
 
     if self._has_sequence:
-      decoded = [example_codec.decode_sequence_example(r)
-                 for r in records]
+      try:
+        from tensor2robot_amd.ops import _t2r_native
+        decoded = _t2r_native.parse_sequence_example_batch(list(records))
+      except ImportError:
+        decoded = [example_codec.decode_sequence_example(r)
+                   for r in records]
     else:
       decoded = [(c, {}) for c in self._decode_example_batch(records)]
     for context, feature_lists in decoded:

@@ -346,3 +346,31 @@ def test_native_parse_example_batch_rejects_garbage():
   from tensor2robot_amd.ops import _t2r_native
   with pytest.raises(RuntimeError):
     _t2r_native.parse_example_batch([b"\xff\xff\xff\xff"])
+
+
+def test_native_parse_sequence_example_batch_matches_python():
+  from tensor2robot_amd.ops import _t2r_native
+  rng = np.random.RandomState(1)
+  records = []
+  for i in range(20):
+    ctx = {"id": np.array([i], np.int64)}
+    fls = {
+        "obs": [rng.randn(4).astype(np.float32) for _ in range(i % 4 + 1)],
+        "img": [[bytes(rng.bytes(10))] for _ in range(i % 3 + 1)],
+    }
+    records.append(example_codec.encode_sequence_example(ctx, fls))
+  records.append(example_codec.encode_sequence_example({}, {}))
+  native = _t2r_native.parse_sequence_example_batch(records)
+  for raw, (nctx, nfls) in zip(records, native):
+    rctx, rfls = example_codec.decode_sequence_example(raw)
+    assert set(nctx) == set(rctx) and set(nfls) == set(rfls)
+    for k in rctx:
+      np.testing.assert_array_equal(np.asarray(nctx[k]),
+                                    np.asarray(rctx[k]))
+    for k in rfls:
+      assert len(nfls[k]) == len(rfls[k])
+      for nstep, rstep in zip(nfls[k], rfls[k]):
+        if isinstance(rstep, list):
+          assert nstep == rstep
+        else:
+          np.testing.assert_array_equal(nstep, rstep)
