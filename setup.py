@@ -55,7 +55,7 @@ ext_modules = [
                  os.path.join(NATIVE_DIR, "example_codec.cpp"),
                  os.path.join(NATIVE_DIR, "native_bindings.cpp")],
         include_dirs=[pybind11.get_include()],
-        extra_compile_args=["-O3", "-std=c++17"],
+        extra_compile_args=["-O3", "-std=c++17", "-msse4.2"],
         language="c++",
     ),
 ]

@@ -3,6 +3,7 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/numpy.h>
 
+#include <string>
 #include <vector>
 
 namespace py = pybind11;
@@ -11,6 +12,7 @@ namespace py = pybind11;
 
 py::list parse_example_batch(py::sequence records);
 py::list parse_sequence_example_batch(py::sequence records);
+py::list read_tfrecord_file(const std::string& path, bool verify_crc);
 
 static py::bytes encode_jpeg(
     py::array_t<uint8_t, py::array::c_style | py::array::forcecast> image,
@@ -94,4 +96,7 @@ PYBIND11_MODULE(_t2r_native, m) {
   m.def("parse_sequence_example_batch", &parse_sequence_example_batch,
         py::arg("records"),
         "batch tf.SequenceExample wire decode (GIL-released scan)");
+  m.def("read_tfrecord_file", &read_tfrecord_file, py::arg("path"),
+        py::arg("verify_crc") = true,
+        "read a whole TFRecord shard (hardware CRC32C verify)");
 }

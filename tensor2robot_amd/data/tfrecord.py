@@ -90,7 +90,23 @@ class TFRecordWriter:
 
 
 def read_records(path: str, verify_crc: bool = False) -> Iterator[bytes]:
-  """Iterates serialized records in one TFRecord file."""
+  """Iterates serialized records in one TFRecord file.
+
+  Shards up to 256 MB read through the native reader
+  (data/native/example_codec.cpp): framing + hardware-CRC32C
+  verification run GIL-released in C++; larger shards stream through
+  the python path below to bound memory."""
+  try:
+    from tensor2robot_amd.ops import _t2r_native
+    native = _t2r_native.read_tfrecord_file
+  except ImportError:
+    native = None
+  if native is not None and os.path.getsize(path) <= 256 * 1024 * 1024:
+    try:
+      yield from native(path, verify_crc)
+    except RuntimeError as e:
+      raise IOError(str(e)) from e
+    return
   with open(path, "rb") as f:
     while True:
       header = f.read(12)
