@@ -154,8 +154,10 @@ def create_rms_prop_optimizer(learning_rate=1e-4, decay: float = 0.9,
 class ExponentialMovingAverage:
   """EMA of model parameters with swap-in/out (swapping-saver semantics).
 
-  The HIP fused multi-tensor EMA update kernel is used on GPU when the
-  extension is built (ops/fused_update); this class is the orchestration.
+  The update runs as two batched multi-tensor kernels
+  (torch._foreach_mul_/add_ — one fused launch each over every shadow
+  tensor) and captures inside hipGraphs; this class is the
+  orchestration around them.
   """
 
   def __init__(self, module: torch.nn.Module, decay: float = 0.9999):
