@@ -46,7 +46,15 @@ def encode_png(image: np.ndarray) -> bytes:
 
 
 def _unfilter(data: np.ndarray, h: int, stride: int, bpp: int) -> np.ndarray:
-  """Reverses PNG scanline filters; returns (h, stride) uint8."""
+  """Reverses PNG scanline filters; returns (h, stride) uint8.
+
+  The sequential Sub/Average/Paeth filters run through the native
+  unfilter (data/native/example_codec.cpp, GIL-released) when the
+  extension is built; this python loop is the reference semantics and
+  the fallback."""
+  native = _load_jpeg_native()
+  if native is not None and hasattr(native, "png_unfilter"):
+    return native.png_unfilter(data.tobytes(), h, stride, bpp)
   out = np.zeros((h, stride), dtype=np.uint8)
   pos = 0
   for y in range(h):

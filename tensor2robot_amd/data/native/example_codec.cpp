@@ -492,3 +492,63 @@ py::list read_tfrecord_file(const std::string& path, bool verify_crc) {
                          r.size()));
   return out;
 }
+
+// PNG scanline unfilter (image_codec._unfilter's hot loop): the
+// Sub/Average/Paeth filters are sequential per pixel and were a
+// per-pixel python loop for foreign (e.g. PIL-written) PNGs; here the
+// whole image unfilters GIL-released.  Filter semantics per the PNG
+// spec, identical to the python reference implementation.
+py::array_t<uint8_t> png_unfilter(py::bytes data, int h, int stride,
+                                  int bpp) {
+  std::string buf = data;
+  const uint8_t* src = (const uint8_t*)buf.data();
+  const size_t need = (size_t)h * (stride + 1);
+  if (buf.size() < need)
+    throw std::runtime_error("png_unfilter: truncated scanline data");
+  py::array_t<uint8_t> out({h, stride});
+  uint8_t* dst = out.mutable_data();
+  {
+    py::gil_scoped_release release;
+    for (int y = 0; y < h; ++y) {
+      const uint8_t ftype = src[(size_t)y * (stride + 1)];
+      const uint8_t* row = src + (size_t)y * (stride + 1) + 1;
+      uint8_t* cur = dst + (size_t)y * stride;
+      const uint8_t* prev = y > 0 ? dst + (size_t)(y - 1) * stride
+                                  : nullptr;
+      switch (ftype) {
+        case 0:
+          memcpy(cur, row, stride);
+          break;
+        case 1:  // Sub
+          for (int x = 0; x < stride; ++x)
+            cur[x] = row[x] + (x >= bpp ? cur[x - bpp] : 0);
+          break;
+        case 2:  // Up
+          for (int x = 0; x < stride; ++x)
+            cur[x] = row[x] + (prev ? prev[x] : 0);
+          break;
+        case 3:  // Average
+          for (int x = 0; x < stride; ++x) {
+            int a = x >= bpp ? cur[x - bpp] : 0;
+            int b = prev ? prev[x] : 0;
+            cur[x] = row[x] + (uint8_t)((a + b) >> 1);
+          }
+          break;
+        case 4:  // Paeth
+          for (int x = 0; x < stride; ++x) {
+            int a = x >= bpp ? cur[x - bpp] : 0;
+            int b = prev ? prev[x] : 0;
+            int c = (prev && x >= bpp) ? prev[x - bpp] : 0;
+            int p = a + b - c;
+            int pa = abs(p - a), pb = abs(p - b), pc = abs(p - c);
+            int pred = (pa <= pb && pa <= pc) ? a : (pb <= pc ? b : c);
+            cur[x] = row[x] + (uint8_t)pred;
+          }
+          break;
+        default:
+          throw std::runtime_error("png_unfilter: unsupported filter");
+      }
+    }
+  }
+  return out;
+}

@@ -13,6 +13,8 @@ namespace py = pybind11;
 py::list parse_example_batch(py::sequence records);
 py::list parse_sequence_example_batch(py::sequence records);
 py::list read_tfrecord_file(const std::string& path, bool verify_crc);
+py::array_t<uint8_t> png_unfilter(py::bytes data, int h, int stride,
+                                  int bpp);
 
 static py::bytes encode_jpeg(
     py::array_t<uint8_t, py::array::c_style | py::array::forcecast> image,
@@ -99,4 +101,7 @@ PYBIND11_MODULE(_t2r_native, m) {
   m.def("read_tfrecord_file", &read_tfrecord_file, py::arg("path"),
         py::arg("verify_crc") = true,
         "read a whole TFRecord shard (hardware CRC32C verify)");
+  m.def("png_unfilter", &png_unfilter, py::arg("data"), py::arg("h"),
+        py::arg("stride"), py::arg("bpp"),
+        "PNG scanline unfilter (GIL-released)");
 }

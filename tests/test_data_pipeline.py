@@ -395,3 +395,35 @@ def test_native_tfrecord_reader_matches_and_verifies(tmp_path):
   open(bad, "wb").write(bytes(blob))
   with pytest.raises(IOError):
     list(tfrecord.read_records(bad, verify_crc=True))
+
+
+def test_png_decode_foreign_filters_exact():
+  """Foreign (PIL-written, Paeth/Sub-filtered) PNGs decode exactly via
+  the native unfilter."""
+  PIL_Image = pytest.importorskip("PIL.Image")
+  import io
+  rng = np.random.RandomState(0)
+  img = (rng.rand(48, 40, 3) * 255).astype(np.uint8)
+  buf = io.BytesIO()
+  PIL_Image.fromarray(img).save(buf, format="PNG")
+  np.testing.assert_array_equal(image_codec.decode_png(buf.getvalue()),
+                                img)
+  # And the pure-python fallback agrees with the native path.
+  from tensor2robot_amd.data.image_codec import _PNG_SIG
+  import tensor2robot_amd.data.image_codec as ic
+  native = ic._load_jpeg_native()
+  if native is not None:
+    h, stride, bpp = 4, 9, 3
+    raw = rng.randint(0, 256, h * (stride + 1), dtype=np.uint8)
+    raw[::stride + 1] = rng.randint(0, 5, h)  # valid filter types
+    nat = native.png_unfilter(raw.tobytes(), h, stride, bpp)
+    # temporarily disable native to exercise the python loop
+    saved = ic._jpeg_native
+    ic._jpeg_native = None
+    ic._jpeg_import_error = ImportError("disabled for test")
+    try:
+      ref = ic._unfilter(raw, h, stride, bpp)
+    finally:
+      ic._jpeg_native = saved
+      ic._jpeg_import_error = None
+    np.testing.assert_array_equal(nat, ref)
