@@ -19,25 +19,29 @@ py::array_t<uint8_t> png_unfilter(py::bytes data, int h, int stride,
 static py::bytes encode_jpeg(
     py::array_t<uint8_t, py::array::c_style | py::array::forcecast> image,
     int quality, int restart_interval) {
-  if (image.ndim() == 2) {
-    auto out = t2r_jpeg::encode(image.data(), image.shape(0),
-                                image.shape(1), 1, quality,
-                                restart_interval);
-    return py::bytes((const char*)out.data(), out.size());
-  }
-  if (image.ndim() != 3)
+  const int ndim = image.ndim();
+  if (ndim != 2 && ndim != 3)
     throw std::runtime_error("encode_jpeg: HxW or HxWxC uint8 expected");
-  auto out = t2r_jpeg::encode(image.data(), image.shape(0),
-                              image.shape(1), image.shape(2), quality,
-                              restart_interval);
+  const int h = image.shape(0), w = image.shape(1);
+  const int c = ndim == 2 ? 1 : image.shape(2);
+  const uint8_t* data = image.data();
+  std::vector<uint8_t> out;
+  {
+    py::gil_scoped_release release;
+    out = t2r_jpeg::encode(data, h, w, c, quality, restart_interval);
+  }
   return py::bytes((const char*)out.data(), out.size());
 }
 
 static py::array decode_jpeg(py::bytes data) {
   std::string buf = data;
   int h = 0, w = 0, c = 0;
-  auto img = t2r_jpeg::decode((const uint8_t*)buf.data(), buf.size(), h, w,
-                              c);
+  std::vector<uint8_t> img;
+  {
+    py::gil_scoped_release release;
+    img = t2r_jpeg::decode((const uint8_t*)buf.data(), buf.size(), h, w,
+                           c);
+  }
   if (c == 1) {
     py::array_t<uint8_t> out({h, w});
     std::memcpy(out.mutable_data(), img.data(), img.size());
