@@ -427,3 +427,30 @@ def test_png_decode_foreign_filters_exact():
       ic._jpeg_native = saved
       ic._jpeg_import_error = None
     np.testing.assert_array_equal(nat, ref)
+
+
+def test_native_decoders_survive_fuzz():
+  """Malformed wire data raises (never crashes) in the native decoders
+  — same error surface as the python codec (incl. UnicodeDecodeError
+  for invalid UTF-8 names)."""
+  from tensor2robot_amd.ops import _t2r_native as native
+  rng = np.random.RandomState(7)
+  ok = (RuntimeError, TypeError, ValueError, UnicodeDecodeError)
+  base = example_codec.encode_example(
+      {"a": np.arange(6, dtype=np.float32), "b": [b"hello"]})
+  for _ in range(500):
+    blob = bytearray(base)
+    for _ in range(rng.randint(1, 4)):
+      blob[rng.randint(len(blob))] = rng.randint(256)
+    for fn in (native.parse_example_batch,
+               native.parse_sequence_example_batch):
+      try:
+        fn([bytes(blob)])
+      except ok:
+        pass
+  for _ in range(300):
+    raw = bytes(rng.bytes(rng.randint(0, 60)))
+    try:
+      native.parse_example_batch([raw])
+    except ok:
+      pass
