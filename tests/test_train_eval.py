@@ -211,3 +211,25 @@ def test_swapping_saver_writes_averaged_weights(tmp_path):
   assert torch.allclose(state["model_state"]["weight"],
                         ema.shadow["weight"])
   assert not torch.allclose(state["model_state"]["weight"], net.weight)
+
+
+def test_training_is_seed_deterministic(tmp_path):
+  """Same seeds -> bit-identical training trajectory on CPU (the
+  torch-native analog of the reference's graph-level determinism)."""
+  def run(seed, d):
+    torch.manual_seed(seed)
+    np.random.seed(seed)
+    model = mocks.MockT2RModel(
+        create_optimizer_fn=lambda: optimizers.create_adam_optimizer(
+            1e-2))
+    gen = mocks.MockInputGenerator(batch_size=8, seed=3)
+    return train_eval.train_eval_model(
+        t2r_model=model, input_generator_train=gen,
+        input_generator_eval=None, max_train_steps=15,
+        model_dir=str(d))["loss"]
+
+  a = run(11, tmp_path / "a")
+  b = run(11, tmp_path / "b")
+  c = run(12, tmp_path / "c")
+  assert a == b
+  assert c != a
