@@ -140,6 +140,7 @@ std::vector<at::Tensor> mdn_nll_forward(at::Tensor params,
   TORCH_CHECK(labels_f.numel() == M * S, "mdn_nll: labels shape");
   auto nll = at::empty({M}, params.options().dtype(at::kFloat));
   auto wsave = at::empty({M, A}, params.options().dtype(at::kFloat));
+  if (M == 0) return {nll, wsave};
   const int grid = (int)((M + 255) / 256);
   auto stream = at::cuda::getCurrentCUDAStream();
   if (params.scalar_type() == at::kBFloat16)
@@ -170,6 +171,7 @@ at::Tensor mdn_nll_backward(at::Tensor params, at::Tensor labels,
   const long P = A + 2 * A * S;
   const long M = params.numel() / P;
   auto dparams = at::empty_like(params);
+  if (M == 0) return dparams;
   const int grid = (int)((M + 255) / 256);
   auto stream = at::cuda::getCurrentCUDAStream();
   if (params.scalar_type() == at::kBFloat16)

@@ -339,3 +339,35 @@ def test_maml_preprocessor_v2_create_meta_map_fn():
     bad_l["y"] = torch.zeros(n, 1)
     with pytest.raises(ValueError):
       map_fn(bad_f, bad_l)
+
+
+def test_maml_parallel_tasks_learned_inner_lr():
+  """learn_inner_lr composes with the vmap path: lrs receive outer
+  gradients and the update matches the loop path."""
+  torch.manual_seed(0)
+  base_l = mocks.MockT2RModel()
+  m_l = _MockMAML(base_model=base_l, device_type="cpu",
+                  compute_dtype="float32", num_inner_loop_steps=1,
+                  learn_inner_lr=True)
+  torch.manual_seed(0)
+  base_v = mocks.MockT2RModel()
+  m_v = _MockMAML(base_model=base_v, device_type="cpu",
+                  compute_dtype="float32", num_inner_loop_steps=1,
+                  learn_inner_lr=True, parallel_tasks=True)
+  _ = m_l.network
+  _ = m_v.network
+  m_v.network.load_state_dict(m_l.network.state_dict())
+  features, labels = _meta_batch(tasks=2, samples=4, seed=9)
+  ops_l = m_l.model_fn(features, labels, run_modes.TRAIN)
+  ops_v = m_v.model_fn(features, labels, run_modes.TRAIN)
+  torch.testing.assert_close(ops_v.loss, ops_l.loss, rtol=1e-5,
+                             atol=1e-6)
+  ops_l.loss.backward()
+  ops_v.loss.backward()
+  lrs_l = dict(m_l.network["inner_lrs"].named_parameters())
+  lrs_v = dict(m_v.network["inner_lrs"].named_parameters())
+  assert lrs_l and set(lrs_l) == set(lrs_v)
+  for k in lrs_l:
+    assert lrs_v[k].grad is not None
+    torch.testing.assert_close(lrs_v[k].grad, lrs_l[k].grad,
+                               rtol=1e-3, atol=1e-6)
