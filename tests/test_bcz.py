@@ -185,3 +185,54 @@ def test_preprocessor_mixup_blends_images_and_future_labels():
       run_modes.EVAL)
   torch.testing.assert_close(l2["future/xyz"],
                              torch.tensor([[0.0], [1.0]]))
+
+
+@pytest.mark.parametrize(
+    "residual_xyz,residual_angle,angle_format,residual_gripper", [
+        (True, False, "quaternion", False),
+        (False, False, "quaternion", False),
+        (True, True, "axis_angle", False),
+        (False, False, "axis_angle", False),
+    ])
+def test_bcz_pose_component_configurations(tmp_path, residual_xyz,
+                                           residual_angle, angle_format,
+                                           residual_gripper):
+  """Action-component configurations train (reference
+  model_test.py:72-95 test_pose_components)."""
+  angle_size = 3 if angle_format == "axis_angle" else 4
+  action_components = [
+      ("xyz", 3, residual_xyz, 100.0),
+      (angle_format, angle_size, residual_angle, 10.0),
+      ("target_close", 1, residual_gripper, 1.0),
+  ]
+  model = _small_model(action_components=action_components,
+                       state_components=[])
+  gen = input_generators.DefaultRandomInputGenerator(batch_size=2,
+                                                     seed=3)
+  result = train_eval.train_eval_model(
+      t2r_model=model, input_generator_train=gen,
+      input_generator_eval=None, max_train_steps=2,
+      model_dir=str(tmp_path))
+  assert result["global_step"] == 2
+  assert np.isfinite(result["loss"])
+
+
+def test_bcz_all_components(tmp_path):
+  """Every component family at once, incl. arm_joints (reference
+  model_test.py:54-70 test_all_components)."""
+  action_components = [
+      ("xyz", 3, True, 100.0),
+      ("quaternion", 4, False, 10.0),
+      ("axis_angle", 3, True, 10.0),
+      ("arm_joints", 7, True, 1.0),
+      ("target_close", 1, False, 1.0),
+  ]
+  model = _small_model(action_components=action_components)
+  gen = input_generators.DefaultRandomInputGenerator(batch_size=2,
+                                                     seed=4)
+  result = train_eval.train_eval_model(
+      t2r_model=model, input_generator_train=gen,
+      input_generator_eval=None, max_train_steps=2,
+      model_dir=str(tmp_path))
+  assert result["global_step"] == 2
+  assert np.isfinite(result["loss"])
