@@ -72,3 +72,16 @@ def test_train_eval_gin_helper(tmp_path):
                      "configs", "run_train_reg_model.gin")
   result = tet.test_train_eval_gin(str(tmp_path), cfg)
   assert result["global_step"] == 1
+
+
+def test_random_train_qtopt_e2e_model(tmp_path):
+  """The reference's exact QT-Opt fixture test (t2r_models_test.py:41):
+  random-train + random-predict the E2E grasping model by name."""
+  from tensor2robot_amd.research.qtopt import t2r_models
+  name = "Grasping44E2EOpenCloseTerminateGripperStatusHeightToBottom"
+  fixture = t2r_test_fixture.T2RModelFixture()
+  result = fixture.random_train(t2r_models, name,
+                                model_dir=str(tmp_path))
+  assert np.isfinite(result["loss"])
+  prediction = fixture.random_predict(t2r_models, name)
+  assert prediction is not None
