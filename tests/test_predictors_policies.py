@@ -527,3 +527,18 @@ def test_export_warmup_roundtrip_predict_serialized(tmp_path):
       {"measured_position": np.zeros((3, 3), np.float32)})
   np.testing.assert_allclose(out["prediction"], ref["prediction"],
                              rtol=1e-5)
+
+
+def test_saved_model_v2_predictor_names(tmp_path):
+  """The reference's three SavedModel predictor flavors resolve and
+  serve here (saved_model_v2_predictor_test.py parity — one servable
+  format, three names)."""
+  from tensor2robot_amd.predictors import saved_model_v2_predictor as v2
+  _, export_dir = _train_and_export(tmp_path, steps=10)
+  for cls in (v2.SavedModelPredictorBase, v2.SavedModelTF1Predictor,
+              v2.SavedModelTF2Predictor):
+    predictor = cls(export_dir, timeout=5)
+    assert predictor.restore()
+    out = predictor.predict(
+        {"measured_position": np.ones((2, 3), np.float32)})
+    assert out["prediction"].shape == (2, 1)
