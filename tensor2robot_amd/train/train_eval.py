@@ -308,8 +308,12 @@ class Trainer:
       # the job-level metric is sum(totals)/sum(counts) over ranks
       # (reference semantics: one evaluator sees all data).
       keys = sorted(totals)
+      # RCCL (backend "nccl") only reduces device tensors — a CPU
+      # tensor here works on gloo but errors on GPU jobs.
       vec = torch.tensor([totals[k] for k in keys] + [float(count)],
-                         dtype=torch.float64)
+                         dtype=torch.float64,
+                         device=self.device
+                         if self.device.type == "cuda" else "cpu")
       torch.distributed.all_reduce(vec)
       totals = {k: float(vec[i]) for i, k in enumerate(keys)}
       count = int(vec[-1])
