@@ -177,10 +177,12 @@ class DataParallelEngine:
 
 
 def all_reduce_scalar(value: float, average: bool = True) -> float:
-  """Scalar metric reduction across the DP group."""
+  """Scalar metric reduction across the DP group.  The tensor lives on
+  the GPU when the backend is nccl/RCCL (which rejects CPU tensors)."""
   if not dist.is_available() or not dist.is_initialized():
     return value
-  t = torch.tensor([value], dtype=torch.float64)
+  device = "cuda" if dist.get_backend() == "nccl" and       torch.cuda.is_available() else "cpu"
+  t = torch.tensor([value], dtype=torch.float64, device=device)
   dist.all_reduce(t)
   if average:
     t /= dist.get_world_size()
